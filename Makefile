@@ -1,0 +1,46 @@
+# edgehip — MI355X-native FUSE object-storage data path.
+# Host C++ builds with g++; HIP device code builds with hipcc for gfx950 ONLY
+# (no CUDA shims, no multi-arch fat binaries). A GPU-less box builds everything
+# (hipcc cross-compiles) and runs the host engine; engine=gpu fails loudly
+# there by design.
+#
+#   make             -> bin/edgehip (CLI) + lib/libedgehip.so (ctypes API)
+#   make test-bins   -> standalone GPU kernel-test binaries (run via gpurun)
+
+ROCM     ?= /opt/rocm
+HIPCC    ?= $(ROCM)/bin/hipcc
+CXX      ?= g++
+ARCH     ?= gfx950
+CXXFLAGS ?= -O2 -g -std=c++17 -Wall -Wextra -fPIC -pthread
+HIPFLAGS ?= -O3 -std=c++17 --offload-arch=$(ARCH) -fPIC
+LDFLAGS  ?= -L$(ROCM)/lib -lamdhip64 -lrccl -Wl,-rpath,$(ROCM)/lib -pthread
+
+HOST_SRCS := src/obs.cc src/bufpool.cc src/conf.cc src/manifest.cc src/blake3.cc src/lz4.cc \
+             src/gf256.cc src/store.cc src/obj.cc src/vfs.cc src/fuse.cc \
+             src/engine/engine.cc
+HIP_SRCS  := src/engine/gpu_engine.hip
+HOST_OBJS := $(HOST_SRCS:.cc=.o)
+HIP_OBJS  := $(HIP_SRCS:.hip=.o)
+OBJS      := $(HOST_OBJS) $(HIP_OBJS)
+
+.PHONY: all clean test-bins
+
+all: bin/edgehip lib/libedgehip.so
+
+bin/edgehip: $(OBJS) src/main.o
+	@mkdir -p bin
+	$(CXX) $(CXXFLAGS) -o $@ $^ $(LDFLAGS)
+
+lib/libedgehip.so: $(OBJS) src/capi.o
+	@mkdir -p lib
+	$(CXX) $(CXXFLAGS) -shared -o $@ $^ $(LDFLAGS)
+
+%.o: %.cc
+	$(CXX) $(CXXFLAGS) -c -o $@ $<
+
+%.o: %.hip
+	$(HIPCC) $(HIPFLAGS) -c -o $@ $<
+
+clean:
+	rm -f $(OBJS) src/main.o src/capi.o
+	rm -rf bin lib

@@ -1,0 +1,121 @@
+// C API for the Python test/bench harness (ctypes). The product data path is
+// the native CLI + library; Python is glue only.
+#include <memory>
+#include <new>
+
+#include "blake3.h"
+#include "conf.h"
+#include "engine/engine.h"
+#include "gf256.h"
+#include "lz4.h"
+#include "manifest.h"
+#include "obj.h"
+#include "obs.h"
+
+using namespace edgehip;
+
+extern "C" {
+
+void eh_blake3(const void* data, size_t len, uint8_t* out32) { blake3_hash(data, len, out32); }
+
+uint32_t eh_crc32c(const void* data, size_t len) { return crc32c(data, len); }
+
+size_t eh_lz4_compress_block(const uint8_t* src, size_t n, uint8_t* dst, size_t cap) {
+    return lz4_compress_block(src, n, dst, cap);
+}
+size_t eh_lz4_decompress_block(const uint8_t* src, size_t n, uint8_t* dst, size_t cap) {
+    size_t r = lz4_decompress_block(src, n, dst, cap);
+    return r == SIZE_MAX ? (size_t)-1 : r;
+}
+// framed chunk compress: returns framed size, 0 if incompressible, -1 on error
+int64_t eh_lz4_chunk_compress(const uint8_t* src, size_t n, uint8_t* dst, size_t cap) {
+    std::vector<uint8_t> out;
+    if (!lz4_chunk_compress(src, n, &out)) return 0;
+    if (out.size() > cap) return -1;
+    std::memcpy(dst, out.data(), out.size());
+    return (int64_t)out.size();
+}
+int eh_lz4_chunk_decompress(const uint8_t* src, size_t n, uint8_t* dst, size_t logical) {
+    return (int)lz4_chunk_decompress(src, n, dst, logical);
+}
+
+int eh_rs_encode(int k, int m, const uint8_t* const* data, uint8_t* const* parity, size_t len) {
+    rs_encode(k, m, data, parity, len);
+    return 0;
+}
+int eh_rs_reconstruct(int k, int m, uint8_t* const* shards, const uint8_t* present, size_t len) {
+    bool pres[64];
+    for (int i = 0; i < k + m; i++) pres[i] = present[i] != 0;
+    return (int)rs_reconstruct(k, m, shards, pres, len);
+}
+void eh_rs_parity_matrix(int k, int m, uint8_t* out) {
+    auto M = rs_parity_matrix(k, m);
+    std::memcpy(out, M.data(), M.size());
+}
+
+// ---- manifest ----
+int64_t eh_manifest_roundtrip(const uint8_t* in, size_t len, uint8_t* out, size_t cap) {
+    Manifest m;
+    if (Manifest::decode(in, len, &m) != St::ok) return -1;
+    std::vector<uint8_t> enc;
+    m.encode(&enc);
+    if (enc.size() > cap) return -2;
+    std::memcpy(out, enc.data(), enc.size());
+    return (int64_t)enc.size();
+}
+
+// ---- engine/object harness ----
+struct EhEngine {
+    Conf conf;
+    std::unique_ptr<Engine> eng;
+};
+
+void* eh_engine_create(const char* opts) {
+    auto* e = new (std::nothrow) EhEngine;
+    if (!e) return nullptr;
+    if (opts && *opts && e->conf.apply_opts(opts) != St::ok) {
+        delete e;
+        return nullptr;
+    }
+    try {
+        e->eng = make_engine(e->conf);
+    } catch (const std::exception& ex) {
+        LOGE("eh_engine_create: %s", ex.what());
+        delete e;
+        return nullptr;
+    }
+    return e;
+}
+void eh_engine_destroy(void* e) { delete (EhEngine*)e; }
+uint64_t eh_engine_stored_bytes(void* e) { return ((EhEngine*)e)->eng->stored_bytes(); }
+uint64_t eh_engine_stored_chunks(void* e) { return ((EhEngine*)e)->eng->stored_chunks(); }
+int eh_engine_is_gpu(void* e) { return ((EhEngine*)e)->eng->is_gpu() ? 1 : 0; }
+
+void* eh_object_create(void* e) {
+    auto* ee = (EhEngine*)e;
+    return new Object(ee->eng.get(), &ee->conf);
+}
+void eh_object_destroy(void* o) { delete (Object*)o; }
+int eh_object_write(void* o, uint64_t off, const void* data, uint64_t len) {
+    return (int)((Object*)o)->write(off, (const uint8_t*)data, len);
+}
+int64_t eh_object_read(void* o, uint64_t off, void* out, uint64_t len) {
+    uint64_t got = 0;
+    St s = ((Object*)o)->read(off, len, (uint8_t*)out, &got);
+    return s == St::ok ? (int64_t)got : (int64_t)st_errno(s);
+}
+int eh_object_flush(void* o) { return (int)((Object*)o)->flush(); }
+int eh_object_truncate(void* o, uint64_t size) { return (int)((Object*)o)->truncate(size); }
+uint64_t eh_object_size(void* o) { return ((Object*)o)->size(); }
+int64_t eh_object_commit(void* o, uint8_t* manifest_out, size_t cap) {
+    std::vector<uint8_t> bytes;
+    St s = ((Object*)o)->commit(&bytes);
+    if (s != St::ok) return (int64_t)st_errno(s);
+    if (manifest_out) {
+        if (bytes.size() > cap) return -1000;
+        std::memcpy(manifest_out, bytes.data(), bytes.size());
+    }
+    return (int64_t)bytes.size();
+}
+
+}  // extern "C"

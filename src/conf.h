@@ -1,0 +1,61 @@
+// Config / flag system.
+// Precedence: CLI -o key=val  >  env EDGEHIP_KEY  >  defaults.
+// Every option is documented in docs/options.md; judged-config presets live
+// in conf/presets/*.conf (simple key=val lines, one per option).
+#pragma once
+
+#include <string>
+
+#include "common.h"
+
+namespace edgehip {
+
+enum class FpAlgo : uint8_t { none = 0, blake3 = 1 };
+enum class Compress : uint8_t { none = 0, lz4 = 1 };
+enum class EngineKind : uint8_t { host = 0, gpu = 1 };
+
+struct Conf {
+    // data layout
+    uint64_t chunk_size = 1 * MiB;     // fixed 1 MiB default
+    FpAlgo fp = FpAlgo::blake3;        // §9 Q3
+    Compress compress = Compress::lz4;
+    bool dedup = true;                 // fingerprint-index dedup on ingest
+    bool verify_read = false;          // re-hash chunks on read (expensive)
+    bool crc = true;                   // crc32c per stored chunk (verify-lite)
+    uint64_t prealloc = 0;             // pre-fault this many pool bytes at init
+
+    // erasure coding (config 4): stripe = rs_k data chunks + rs_m parity
+    int rs_k = 0;                      // 0 = replication off / no EC
+    int rs_m = 0;
+
+    // engine
+    EngineKind engine = EngineKind::host;  // host for config 1; gpu otherwise
+    int ngpus = 1;                         // devices used by this mount
+    int gpu_first = 0;                     // first device index
+    uint64_t arena_bytes = 16 * GiB;       // HBM arena per GPU (chunk store+cache)
+    uint64_t pinned_bytes = 512 * MiB;     // pinned staging ring per GPU
+    uint64_t spill_bytes = 64 * GiB;       // host spill tier cap (0 = off)
+    std::string spill_dir;                 // backing dir for spill (empty = RAM)
+
+    // fuse frontend
+    int fuse_threads = 8;
+    uint32_t max_write = 1 * MiB;
+    uint32_t max_readahead = 16 * MiB;  // deep readahead pipelines seq reads
+    bool writeback = true;             // FUSE writeback cache (ack from page cache)
+    bool direct_io = false;            // force FOPEN_DIRECT_IO on opens
+    bool kernel_cache = true;          // keep_cache on open
+
+    // misc
+    int log_level = 2;                 // info
+    std::string fault;                 // fault injection: "site:prob[,site:prob]"
+
+    // Parse "-o key=val[,key=val...]" fragments and env. Returns St::inval on
+    // unknown key or bad value (strict: typos in bench configs must not pass).
+    St apply_kv(const std::string& key, const std::string& val);
+    St apply_opts(const std::string& opts);   // comma-separated k=v list
+    St apply_env();                           // EDGEHIP_* variables
+    St load_file(const std::string& path);    // key=val lines, # comments
+    std::string dump() const;                 // render effective config
+};
+
+}  // namespace edgehip

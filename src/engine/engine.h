@@ -1,0 +1,63 @@
+// The data-plane engine interface: the object layer hands
+// whole chunks down; the engine fingerprints, compresses, dedups, stores and
+// retrieves them. Two implementations:
+//   HostEngine (engine.cc)    — CPU: blake3.cc/lz4.cc + HostStore. Used by the
+//                               judged config 1 (plumbing, no GPU) and tests.
+//   GpuEngine  (gpu_engine.cc)— MI355X: HIP kernels (src/kernels/*.hip), HBM
+//                               arena chunk store, pinned staging, per-GPU
+//                               streams. The product data path (configs 2-5).
+// The FUSE worker threads call these synchronously per batch; GpuEngine
+// pipelines H2D / kernels / D2H on separate HIP streams inside a call.
+#pragma once
+
+#include <memory>
+
+#include "../bufpool.h"
+#include "../common.h"
+#include "../conf.h"
+#include "../manifest.h"
+
+namespace edgehip {
+
+struct PutChunk {
+    const uint8_t* data;  // host pointer, chunk bytes (uncompressed)
+    uint32_t len;
+    Blob* donate = nullptr;  // optional: pool buffer holding `data`; the engine
+                             // ALWAYS consumes it (stores or recycles) when set
+};
+
+struct GetChunk {
+    const ChunkRef* ref;
+    uint8_t* out;       // host destination
+    uint32_t off;       // offset within the UNCOMPRESSED chunk
+    uint32_t len;       // bytes to produce
+};
+
+class Engine {
+  public:
+    virtual ~Engine() = default;
+
+    // Ingest n chunks; fills refs[i] (fp, lens, flags, crc). Dedup (if enabled)
+    // bumps refcounts instead of re-storing.
+    virtual St put_chunks(const PutChunk* put, ChunkRef* refs, int n) = 0;
+
+    // Produce uncompressed chunk byte ranges into host buffers.
+    virtual St get_chunks(const GetChunk* get, int n) = 0;
+
+    // Drop one reference (unlink / overwrite).
+    virtual St release_chunk(const ChunkRef& ref) = 0;
+
+    virtual uint64_t stored_bytes() const = 0;
+    virtual uint64_t stored_chunks() const = 0;
+    // True when this engine runs on the GPU data path (used by tests to assert
+    // the native path is live on a GPU box).
+    virtual bool is_gpu() const = 0;
+};
+
+std::unique_ptr<Engine> make_host_engine(const Conf& conf);
+// Throws std::runtime_error if HIP init fails (no silent CPU fallback: the GPU
+// data path must fail loudly when the extension/device is missing).
+std::unique_ptr<Engine> make_gpu_engine(const Conf& conf);
+std::unique_ptr<Engine> make_engine(const Conf& conf);  // dispatch on conf.engine
+
+}  // namespace edgehip

@@ -1,0 +1,194 @@
+// LZ4 block format, written from the public format specification:
+// token (hi=literal len, lo=matchlen-4, 15 = extend with 0xFF runs),
+// literals, 2-byte LE offset, matchlen extension. Last sequence is
+// literals-only; last 5 bytes are literals; a match never starts within the
+// final 12 bytes. Verified by round-trip + golden tests (tests/unit/test_lz4.py
+// cross-checks against an independent pure-Python decoder).
+#include "lz4.h"
+
+namespace edgehip {
+
+namespace {
+constexpr size_t MINMATCH = 4;
+constexpr size_t LAST_LITERALS = 5;
+constexpr size_t MFLIMIT = 12;
+constexpr int HASH_LOG = 13;
+
+inline uint32_t read32(const uint8_t* p) { uint32_t v; std::memcpy(&v, p, 4); return v; }
+inline uint32_t hash4(uint32_t v) { return (v * 2654435761u) >> (32 - HASH_LOG); }
+}  // namespace
+
+size_t lz4_compress_block(const uint8_t* src, size_t n, uint8_t* dst, size_t max_out) {
+    if (n == 0) return 0;
+    uint16_t table[1 << HASH_LOG];
+    std::memset(table, 0, sizeof(table));
+    const uint8_t* ip = src;
+    const uint8_t* anchor = src;
+    const uint8_t* iend = src + n;
+    const uint8_t* mflimit = n >= MFLIMIT ? iend - MFLIMIT : src;
+    uint8_t* op = dst;
+    uint8_t* oend = dst + max_out;
+
+    auto emit = [&](const uint8_t* lit, size_t lit_len, size_t off, size_t mlen) -> bool {
+        // token + literal length
+        size_t need = 1 + lit_len / 255 + 1 + lit_len + (mlen ? 2 + (mlen - MINMATCH) / 255 + 1 : 0);
+        if (op + need > oend) return false;
+        uint8_t* tok = op++;
+        size_t ll = lit_len;
+        if (ll >= 15) {
+            *tok = 15 << 4;
+            ll -= 15;
+            while (ll >= 255) { *op++ = 255; ll -= 255; }
+            *op++ = (uint8_t)ll;
+        } else {
+            *tok = (uint8_t)(ll << 4);
+        }
+        std::memcpy(op, lit, lit_len);
+        op += lit_len;
+        if (mlen) {
+            *op++ = (uint8_t)off;
+            *op++ = (uint8_t)(off >> 8);
+            size_t ml = mlen - MINMATCH;
+            if (ml >= 15) {
+                *tok |= 15;
+                ml -= 15;
+                while (ml >= 255) { *op++ = 255; ml -= 255; }
+                *op++ = (uint8_t)ml;
+            } else {
+                *tok |= (uint8_t)ml;
+            }
+        }
+        return true;
+    };
+
+    if (n >= MFLIMIT + MINMATCH) {
+        table[hash4(read32(ip))] = 0;
+        ip++;
+        while (ip <= mflimit) {
+            uint32_t h = hash4(read32(ip));
+            const uint8_t* match = src + table[h];
+            table[h] = (uint16_t)(ip - src);
+            if (ip - match < 65536 && match < ip && read32(match) == read32(ip)) {
+                // extend backward
+                while (ip > anchor && match > src && ip[-1] == match[-1]) { ip--; match--; }
+                // extend forward
+                size_t mlen = MINMATCH;
+                while (ip + mlen < iend - LAST_LITERALS && ip[mlen] == match[mlen]) mlen++;
+                if (mlen < MINMATCH) { ip++; continue; }
+                if (!emit(anchor, (size_t)(ip - anchor), (size_t)(ip - match), mlen)) return 0;
+                ip += mlen;
+                anchor = ip;
+                if (ip > mflimit) break;
+                table[hash4(read32(ip - 2))] = (uint16_t)(ip - 2 - src);
+            } else {
+                ip++;
+            }
+        }
+    }
+    if (!emit(anchor, (size_t)(iend - anchor), 0, 0)) return 0;
+    return (size_t)(op - dst);
+}
+
+size_t lz4_decompress_block(const uint8_t* src, size_t n, uint8_t* dst, size_t dst_cap) {
+    const uint8_t* ip = src;
+    const uint8_t* iend = src + n;
+    uint8_t* op = dst;
+    uint8_t* oend = dst + dst_cap;
+    while (ip < iend) {
+        uint8_t tok = *ip++;
+        size_t lit = tok >> 4;
+        if (lit == 15) {
+            uint8_t b;
+            do {
+                if (ip >= iend) return SIZE_MAX;
+                b = *ip++;
+                lit += b;
+            } while (b == 255);
+        }
+        if (ip + lit > iend || op + lit > oend) return SIZE_MAX;
+        std::memcpy(op, ip, lit);
+        ip += lit;
+        op += lit;
+        if (ip == iend) break;  // last sequence: literals only
+        if (ip + 2 > iend) return SIZE_MAX;
+        size_t off = ip[0] | (size_t)ip[1] << 8;
+        ip += 2;
+        if (off == 0 || off > (size_t)(op - dst)) return SIZE_MAX;
+        size_t mlen = (tok & 15) + MINMATCH;
+        if ((tok & 15) == 15) {
+            uint8_t b;
+            do {
+                if (ip >= iend) return SIZE_MAX;
+                b = *ip++;
+                mlen += b;
+            } while (b == 255);
+        }
+        if (op + mlen > oend) return SIZE_MAX;
+        const uint8_t* m = op - off;
+        for (size_t i = 0; i < mlen; i++) op[i] = m[i];  // byte copy: overlaps allowed
+        op += mlen;
+    }
+    return (size_t)(op - dst);
+}
+
+bool lz4_chunk_compress(const uint8_t* src, size_t n, std::vector<uint8_t>* out) {
+    size_t nblocks = div_up(n, kLz4Block);
+    if (nblocks == 0) return false;
+    size_t header = 8 + nblocks * 8;
+    out->assign(header, 0);
+    uint32_t* h32 = (uint32_t*)out->data();
+    h32[0] = kLz4Magic;
+    h32[1] = (uint32_t)nblocks;
+    std::vector<uint8_t> tmp;
+    for (size_t b = 0; b < nblocks; b++) {
+        size_t off = b * kLz4Block;
+        size_t blen = std::min(kLz4Block, n - off);
+        tmp.resize(lz4_bound(blen));
+        // worth compressing only if it actually shrinks
+        size_t c = lz4_compress_block(src + off, blen, tmp.data(), blen > 64 ? blen - 32 : blen);
+        size_t payload_off = out->size() - header;
+        uint32_t len_flags;
+        if (c == 0 || c >= blen) {
+            len_flags = (uint32_t)blen | 0x80000000u;  // raw
+            out->insert(out->end(), src + off, src + off + blen);
+        } else {
+            len_flags = (uint32_t)c;
+            out->insert(out->end(), tmp.begin(), tmp.begin() + c);
+        }
+        while ((out->size() - header) % 4) out->push_back(0);  // 4-byte align next block
+        h32 = (uint32_t*)out->data();  // insert may have reallocated
+        h32[2 + 2 * b] = (uint32_t)payload_off;
+        h32[3 + 2 * b] = len_flags;
+    }
+    return out->size() < n;
+}
+
+St lz4_chunk_decompress(const uint8_t* src, size_t n, uint8_t* dst, size_t logical) {
+    if (n < 8) return St::bad_format;
+    const uint32_t* h32 = (const uint32_t*)src;
+    if (h32[0] != kLz4Magic) return St::bad_format;
+    size_t nblocks = h32[1];
+    size_t header = 8 + nblocks * 8;
+    if (n < header || nblocks != div_up(logical, kLz4Block)) return St::bad_format;
+    const uint8_t* payload = src + header;
+    size_t payload_len = n - header;
+    for (size_t b = 0; b < nblocks; b++) {
+        size_t off = h32[2 + 2 * b];
+        uint32_t lf = h32[3 + 2 * b];
+        size_t blen = lf & 0x7fffffffu;
+        bool raw = lf & 0x80000000u;
+        size_t dst_off = b * kLz4Block;
+        size_t want = std::min(kLz4Block, logical - dst_off);
+        if (off + blen > payload_len) return St::bad_format;
+        if (raw) {
+            if (blen != want) return St::bad_format;
+            std::memcpy(dst + dst_off, payload + off, blen);
+        } else {
+            size_t got = lz4_decompress_block(payload + off, blen, dst + dst_off, want);
+            if (got != want) return St::bad_format;
+        }
+    }
+    return St::ok;
+}
+
+}  // namespace edgehip

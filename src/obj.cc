@@ -1,0 +1,275 @@
+#include "obj.h"
+
+#include "obs.h"
+
+namespace edgehip {
+
+// Full chunks accumulated per object before one batched engine call; batching
+// is what gives the GPU engine kernel-launch-amortizing work.
+static constexpr int kPutBatch = 8;
+
+Object::~Object() = default;
+
+void Object::drop_ref(const ChunkRef& r) {
+    if (!r.fp.is_zero()) eng_->release_chunk(r);
+}
+
+St Object::load_chunk_for_rmw(uint64_t idx, Dirty* d) {
+    const uint32_t cs = (uint32_t)conf_->chunk_size;
+    if (!d->data.p) d->data = BufPool::inst().get(cs);
+    d->data.len = cs;
+    if (idx < mf_.chunks.size() && !mf_.chunks[idx].fp.is_zero()) {
+        const ChunkRef& r = mf_.chunks[idx];
+        GetChunk g{&r, d->data.p, 0, r.logical_len};
+        St s = eng_->get_chunks(&g, 1);
+        if (s != St::ok) return s;
+        if (r.logical_len < cs)
+            std::memset(d->data.p + r.logical_len, 0, cs - r.logical_len);
+        d->len = std::max(d->len, r.logical_len);
+    } else {
+        std::memset(d->data.p, 0, cs);
+    }
+    d->loaded = true;
+    return St::ok;
+}
+
+St Object::write(uint64_t off, const uint8_t* data, uint64_t len) {
+    std::lock_guard<std::mutex> g(mu_);
+    const uint64_t cs = conf_->chunk_size;
+    uint64_t pos = off;
+    const uint8_t* p = data;
+    uint64_t end = off + len;
+    uint64_t last_idx = len ? (end - 1) / cs : off / cs;
+    while (pos < end) {
+        uint64_t idx = pos / cs;
+        uint32_t o = (uint32_t)(pos - idx * cs);
+        uint32_t n = (uint32_t)std::min<uint64_t>(cs - o, end - pos);
+        auto it = dirty_.find(idx);
+        if (it == dirty_.end()) {
+            Dirty d;
+            bool full_overwrite = (o == 0 && n == cs);
+            bool committed = idx < mf_.chunks.size() && !mf_.chunks[idx].fp.is_zero();
+            if (!full_overwrite && committed) {
+                St s = load_chunk_for_rmw(idx, &d);
+                if (s != St::ok) return s;
+            } else {
+                d.data = BufPool::inst().get((uint32_t)cs);
+                d.data.len = (uint32_t)cs;
+                // full overwrites need no init; partial writes into a fresh
+                // chunk must read back zeros around the written range
+                if (!full_overwrite) std::memset(d.data.p, 0, cs);
+                d.loaded = !committed;  // nothing to merge for fresh chunks
+            }
+            it = dirty_.emplace(idx, std::move(d)).first;
+        } else if (!it->second.loaded && !(o == 0 && n == cs)) {
+            // sparse write into a chunk that has committed content underneath
+            bool committed = idx < mf_.chunks.size() && !mf_.chunks[idx].fp.is_zero();
+            if (committed) {
+                // merge committed bytes under the already-written ones:
+                Dirty merged;
+                St s = load_chunk_for_rmw(idx, &merged);
+                if (s != St::ok) return s;
+                // overlay previously written prefix? Dirty entries without
+                // loaded start zeroed and are only written ranges; we cannot
+                // distinguish written-zero from hole — so a not-loaded entry
+                // with committed data must be loaded BEFORE its first partial
+                // write. Entries reach here only via full_overwrite creation,
+                // where the whole chunk is new data; merging is then wrong.
+                // full_overwrite entries have len==cs: keep them as-is.
+                if (it->second.len < cs) {
+                    std::memcpy(merged.data.p, it->second.data.p, it->second.len);
+                    // (written range is always a prefix for non-loaded entries:
+                    // they were created by a full overwrite => len==cs; guarded)
+                }
+                if (it->second.len == cs) merged = std::move(it->second);
+                merged.loaded = true;
+                it->second = std::move(merged);
+            } else {
+                it->second.loaded = true;
+            }
+        }
+        Dirty& d = it->second;
+        std::memcpy(d.data.p + o, p, n);
+        d.len = std::max<uint32_t>(d.len, o + n);
+        pos += n;
+        p += n;
+    }
+    size_ = std::max(size_, end);
+    metrics().write_ops++;
+    metrics().write_bytes += len;
+
+    // flush completed chunks (all but the one still growing at the tail)
+    int full = 0;
+    for (auto& kv : dirty_)
+        if (kv.second.len == cs && kv.first != last_idx) full++;
+    if (full >= kPutBatch) {
+        std::vector<uint64_t> idxs;
+        std::vector<PutChunk> puts;
+        for (auto& kv : dirty_) {
+            if (kv.second.len == cs && kv.first != last_idx) {
+                idxs.push_back(kv.first);
+                puts.push_back({kv.second.data.p, kv.second.len, &kv.second.data});
+            }
+        }
+        std::vector<ChunkRef> refs(puts.size());
+        St s = eng_->put_chunks(puts.data(), refs.data(), (int)puts.size());
+        if (s != St::ok) return s;
+        if (mf_.chunks.size() <= idxs.back()) mf_.chunks.resize(idxs.back() + 1);
+        for (size_t i = 0; i < idxs.size(); i++) {
+            drop_ref(mf_.chunks[idxs[i]]);
+            mf_.chunks[idxs[i]] = refs[i];
+            dirty_.erase(idxs[i]);
+        }
+    }
+    return St::ok;
+}
+
+St Object::read(uint64_t off, uint64_t len, uint8_t* out, uint64_t* got) {
+    std::lock_guard<std::mutex> g(mu_);
+    if (off >= size_) { *got = 0; return St::ok; }
+    len = std::min(len, size_ - off);
+    *got = len;
+    const uint64_t cs = conf_->chunk_size;
+    uint64_t pos = off, end = off + len;
+    uint8_t* o = out;
+    std::vector<GetChunk> gets;
+    while (pos < end) {
+        uint64_t idx = pos / cs;
+        uint32_t co = (uint32_t)(pos - idx * cs);
+        uint32_t n = (uint32_t)std::min<uint64_t>(cs - co, end - pos);
+        auto it = dirty_.find(idx);
+        if (it != dirty_.end()) {
+            Dirty& d = it->second;
+            bool committed = idx < mf_.chunks.size() && !mf_.chunks[idx].fp.is_zero();
+            if (!d.loaded && committed && co + n > d.len) {
+                // need bytes under the written prefix — merge committed content
+                Dirty merged;
+                St s = load_chunk_for_rmw(idx, &merged);
+                if (s != St::ok) return s;
+                std::memcpy(merged.data.p, d.data.p, d.len);
+                merged.len = std::max(merged.len, d.len);
+                d = std::move(merged);
+            }
+            uint32_t avail = d.len > co ? std::min(n, d.len - co) : 0;
+            if (avail) std::memcpy(o, d.data.p + co, avail);
+            if (avail < n) std::memset(o + avail, 0, n - avail);
+        } else if (idx < mf_.chunks.size() && !mf_.chunks[idx].fp.is_zero()) {
+            const ChunkRef& r = mf_.chunks[idx];
+            uint32_t avail = r.logical_len > co ? std::min(n, r.logical_len - co) : 0;
+            if (avail) gets.push_back({&r, o, co, avail});
+            if (avail < n) std::memset(o + avail, 0, n - avail);
+        } else {
+            std::memset(o, 0, n);  // hole
+        }
+        pos += n;
+        o += n;
+    }
+    if (!gets.empty()) {
+        St s = eng_->get_chunks(gets.data(), (int)gets.size());
+        if (s != St::ok) return s;
+    }
+    metrics().read_ops++;
+    metrics().read_bytes += len;
+    return St::ok;
+}
+
+St Object::flush_locked() {
+    if (dirty_.empty()) return St::ok;
+    std::vector<uint64_t> idxs;
+    std::vector<PutChunk> puts;
+    const uint64_t cs = conf_->chunk_size;
+    for (auto& kv : dirty_) {
+        Dirty& d = kv.second;
+        // clamp the tail chunk to the object size
+        uint64_t logical_end = std::min<uint64_t>(size_ - kv.first * cs, cs);
+        uint32_t len = (uint32_t)std::min<uint64_t>(d.len, logical_end);
+        bool committed = kv.first < mf_.chunks.size() && !mf_.chunks[kv.first].fp.is_zero();
+        if (!d.loaded && committed && len < logical_end) {
+            Dirty merged;
+            St s = load_chunk_for_rmw(kv.first, &merged);
+            if (s != St::ok) return s;
+            std::memcpy(merged.data.p, d.data.p, d.len);
+            merged.len = std::max(merged.len, d.len);
+            d = std::move(merged);
+            len = (uint32_t)std::min<uint64_t>(std::max<uint32_t>(d.len, (uint32_t)logical_end), logical_end);
+        }
+        if (len == 0) continue;
+        idxs.push_back(kv.first);
+        puts.push_back({d.data.p, len, &d.data});
+    }
+    if (!puts.empty()) {
+        std::vector<ChunkRef> refs(puts.size());
+        St s = eng_->put_chunks(puts.data(), refs.data(), (int)puts.size());
+        if (s != St::ok) return s;
+        uint64_t maxidx = idxs.back();
+        if (mf_.chunks.size() <= maxidx) mf_.chunks.resize(maxidx + 1);
+        for (size_t i = 0; i < idxs.size(); i++) {
+            drop_ref(mf_.chunks[idxs[i]]);
+            mf_.chunks[idxs[i]] = refs[i];
+        }
+    }
+    dirty_.clear();
+    mf_.logical_size = size_;
+    return St::ok;
+}
+
+St Object::flush() {
+    std::lock_guard<std::mutex> g(mu_);
+    return flush_locked();
+}
+
+St Object::truncate(uint64_t size) {
+    std::lock_guard<std::mutex> g(mu_);
+    const uint64_t cs = conf_->chunk_size;
+    if (size < size_) {
+        uint64_t nchunks = div_up(size, cs);
+        for (auto it = dirty_.begin(); it != dirty_.end();) {
+            if (it->first >= nchunks) it = dirty_.erase(it);
+            else ++it;
+        }
+        for (uint64_t i = nchunks; i < mf_.chunks.size(); i++) drop_ref(mf_.chunks[i]);
+        if (mf_.chunks.size() > nchunks) mf_.chunks.resize(nchunks);
+        // shorten the (new) last chunk if it straddles the cut
+        if (size % cs != 0 && nchunks > 0) {
+            uint64_t idx = nchunks - 1;
+            uint32_t keep = (uint32_t)(size - idx * cs);
+            auto it = dirty_.find(idx);
+            if (it != dirty_.end()) {
+                it->second.len = std::min(it->second.len, keep);
+                if (it->second.data.len >= keep)
+                    std::memset(it->second.data.p + keep, 0, it->second.data.len - keep);
+            } else if (idx < mf_.chunks.size() && !mf_.chunks[idx].fp.is_zero() &&
+                       mf_.chunks[idx].logical_len > keep) {
+                Dirty d;
+                St s = load_chunk_for_rmw(idx, &d);
+                if (s != St::ok) return s;
+                std::memset(d.data.p + keep, 0, d.data.len - keep);
+                d.len = keep;
+                drop_ref(mf_.chunks[idx]);
+                mf_.chunks[idx] = ChunkRef{};
+                dirty_[idx] = std::move(d);
+            }
+        }
+    }
+    size_ = size;
+    return St::ok;
+}
+
+St Object::commit(std::vector<uint8_t>* manifest_bytes) {
+    std::lock_guard<std::mutex> g(mu_);
+    St s = flush_locked();
+    if (s != St::ok) return s;
+    mf_.logical_size = size_;
+    if (manifest_bytes) mf_.encode(manifest_bytes);
+    return St::ok;
+}
+
+St Object::set_manifest(const Manifest& m) {
+    std::lock_guard<std::mutex> g(mu_);
+    if (!dirty_.empty()) return St::inval;
+    mf_ = m;
+    size_ = m.logical_size;
+    return St::ok;
+}
+
+}  // namespace edgehip

@@ -1,0 +1,57 @@
+// Object layer: offset→chunk mapping, write buffering with
+// batched engine puts, read planning, commit (manifest build).
+// One Object per file inode; FUSE worker threads serialize on obj mutex (the
+// engine below parallelizes across objects and within batches).
+#pragma once
+
+#include <map>
+#include <memory>
+#include <mutex>
+
+#include "common.h"
+#include "engine/engine.h"
+#include "manifest.h"
+
+namespace edgehip {
+
+class Object {
+  public:
+    Object(Engine* eng, const Conf* conf) : eng_(eng), conf_(conf) {
+        mf_.chunk_size = conf->chunk_size;
+        mf_.fp_algo = (uint8_t)conf->fp;
+        mf_.compress = (uint8_t)conf->compress;
+    }
+    ~Object();
+
+    St read(uint64_t off, uint64_t len, uint8_t* out, uint64_t* got);
+    St write(uint64_t off, const uint8_t* data, uint64_t len);
+    St flush();                    // drain write buffers into the engine
+    St truncate(uint64_t size);
+    uint64_t size() const { return size_; }
+
+    // commit(): flush + rebuild the manifest bytes (the durable form).
+    St commit(std::vector<uint8_t>* manifest_bytes = nullptr);
+
+    const Manifest& manifest() const { return mf_; }
+    St set_manifest(const Manifest& m);   // adopt (remount / clone)
+
+  private:
+    struct Dirty {
+        Blob data;            // pool buffer, chunk_size capacity
+        uint32_t len = 0;     // valid prefix length
+        bool loaded = false;  // true once committed content was merged
+    };
+
+    St flush_locked();
+    St load_chunk_for_rmw(uint64_t idx, Dirty* d);
+    void drop_ref(const ChunkRef& r);
+
+    Engine* eng_;
+    const Conf* conf_;
+    std::mutex mu_;
+    Manifest mf_;                    // committed chunk refs (index = chunk idx)
+    std::map<uint64_t, Dirty> dirty_;
+    uint64_t size_ = 0;              // logical size incl. dirty extension
+};
+
+}  // namespace edgehip

@@ -1,0 +1,30 @@
+// Observability: counters + stats rendering.
+// Counters are plain relaxed atomics — the hot path pays one uncontended
+// fetch_add per op; histograms can come later if profiles show need.
+#pragma once
+
+#include <atomic>
+#include <string>
+
+#include "common.h"
+
+namespace edgehip {
+
+struct Metrics {
+    std::atomic<uint64_t> fuse_reqs{0};
+    std::atomic<uint64_t> read_ops{0}, read_bytes{0};
+    std::atomic<uint64_t> write_ops{0}, write_bytes{0};
+    std::atomic<uint64_t> chunks_put{0}, chunks_get{0};
+    std::atomic<uint64_t> dedup_hits{0}, dedup_bytes_saved{0};
+    std::atomic<uint64_t> lz4_in_bytes{0}, lz4_out_bytes{0};
+    std::atomic<uint64_t> rs_encoded_stripes{0}, rs_reconstructs{0};
+    std::atomic<uint64_t> cache_hits{0}, cache_misses{0};
+    std::atomic<uint64_t> h2d_bytes{0}, d2h_bytes{0};
+    std::atomic<uint64_t> spilled_chunks{0}, spill_bytes{0};
+
+    std::string render() const;  // text for the in-mount .edgehip/stats file
+};
+
+Metrics& metrics();
+
+}  // namespace edgehip

@@ -1,0 +1,50 @@
+import os
+import shutil
+import subprocess
+import sys
+from pathlib import Path
+
+import pytest
+
+REPO = Path(__file__).resolve().parent.parent
+sys.path.insert(0, str(REPO))
+
+
+def pytest_configure(config):
+    config.addinivalue_line("markers", "gpu: needs an MI355X (run via gpurun)")
+    config.addinivalue_line("markers", "fuse: needs /dev/fuse + CAP_SYS_ADMIN")
+
+
+def pytest_collection_modifyitems(config, items):
+    have_fuse = os.path.exists("/dev/fuse") and os.geteuid() == 0
+    skip_fuse = pytest.mark.skip(reason="no /dev/fuse or not root")
+    for item in items:
+        if "fuse" in item.keywords and not have_fuse:
+            item.add_marker(skip_fuse)
+
+
+@pytest.fixture(scope="session", autouse=True)
+def build_native():
+    """Ensure the native library + CLI are built before tests run."""
+    lib = REPO / "lib" / "libedgehip.so"
+    cli = REPO / "bin" / "edgehip"
+    if not lib.exists() or not cli.exists():
+        subprocess.run(["make", "-j8"], cwd=REPO, check=True,
+                       stdout=subprocess.DEVNULL)
+    return lib
+
+
+@pytest.fixture()
+def tmp_mount(tmp_path):
+    mnt = tmp_path / "mnt"
+    mnt.mkdir()
+    yield mnt
+    # safety: never leave a stale mount behind
+    subprocess.run(["umount", "-l", str(mnt)], check=False,
+                   stderr=subprocess.DEVNULL)
+
+
+def rand_bytes(n, seed=0):
+    import numpy as np
+    rng = np.random.default_rng(seed)
+    return rng.integers(0, 256, size=n, dtype=np.uint8).tobytes()

@@ -1,0 +1,115 @@
+"""Integration: real FUSE mount on the host engine (judged config 1 shape).
+Needs /dev/fuse + root (the 'fuse' marker skips otherwise)."""
+
+import os
+import subprocess
+
+import pytest
+
+from edgehip.mount import Mount
+from tests.conftest import rand_bytes
+
+pytestmark = pytest.mark.fuse
+
+HOSTOPTS = "engine=host,fp=none,compress=none,dedup=0,log=3"
+CONTENTOPTS = "engine=host,fp=blake3,compress=lz4,dedup=1,log=3"
+
+
+def test_mount_basic_posix(tmp_mount):
+    with Mount(tmp_mount, HOSTOPTS):
+        p = tmp_mount / "hello.txt"
+        p.write_text("hello edgehip\n")
+        assert p.read_text() == "hello edgehip\n"
+        assert p.stat().st_size == 14
+        (tmp_mount / "d").mkdir()
+        (tmp_mount / "d" / "x").write_bytes(b"x" * 100)
+        assert sorted(os.listdir(tmp_mount)) == ["d", "hello.txt"]
+        os.rename(tmp_mount / "hello.txt", tmp_mount / "d" / "hello2.txt")
+        assert sorted(os.listdir(tmp_mount / "d")) == ["hello2.txt", "x"]
+        (tmp_mount / "d" / "x").unlink()
+        (tmp_mount / "d" / "hello2.txt").unlink()
+        os.rmdir(tmp_mount / "d")
+        assert os.listdir(tmp_mount) == []
+    assert not os.path.ismount(tmp_mount)
+
+
+def test_large_file_roundtrip(tmp_mount):
+    with Mount(tmp_mount, HOSTOPTS):
+        data = rand_bytes(32 << 20, seed=1)
+        p = tmp_mount / "big.bin"
+        p.write_bytes(data)
+        assert p.stat().st_size == len(data)
+        assert p.read_bytes() == data
+
+
+def test_random_4k_reads(tmp_mount):
+    with Mount(tmp_mount, CONTENTOPTS):
+        data = rand_bytes(8 << 20, seed=2)
+        p = tmp_mount / "obj.bin"
+        p.write_bytes(data)
+        import random
+        rng = random.Random(3)
+        with open(p, "rb") as f:
+            for _ in range(200):
+                off = rng.randrange(0, len(data) - 4096)
+                f.seek(off)
+                assert f.read(4096) == data[off:off + 4096]
+
+
+def test_append_and_rewrite(tmp_mount):
+    with Mount(tmp_mount, CONTENTOPTS):
+        p = tmp_mount / "log.txt"
+        with open(p, "ab") as f:
+            for i in range(100):
+                f.write(f"line {i}\n".encode())
+        lines = p.read_bytes().decode().splitlines()
+        assert lines[0] == "line 0" and lines[99] == "line 99"
+        # in-place rewrite
+        with open(p, "r+b") as f:
+            f.seek(7)
+            f.write(b"XXXX")
+        assert p.read_bytes()[7:11] == b"XXXX"
+
+
+def test_truncate_via_mount(tmp_mount):
+    with Mount(tmp_mount, CONTENTOPTS):
+        p = tmp_mount / "t.bin"
+        data = rand_bytes(3 << 20, seed=4)
+        p.write_bytes(data)
+        os.truncate(p, 1 << 20)
+        assert p.stat().st_size == 1 << 20
+        assert p.read_bytes() == data[:1 << 20]
+
+
+def test_dedup_through_mount(tmp_mount):
+    with Mount(tmp_mount, CONTENTOPTS):
+        data = rand_bytes(4 << 20, seed=5)
+        for i in range(3):
+            (tmp_mount / f"copy{i}.bin").write_bytes(data)
+        for i in range(3):
+            assert (tmp_mount / f"copy{i}.bin").read_bytes() == data
+
+
+def test_many_small_files(tmp_mount):
+    with Mount(tmp_mount, CONTENTOPTS):
+        for i in range(200):
+            (tmp_mount / f"f{i:03}.txt").write_bytes(b"v" * i)
+        names = sorted(os.listdir(tmp_mount))
+        assert len(names) == 200
+        assert (tmp_mount / "f199.txt").stat().st_size == 199
+
+
+def test_cp_and_cmp_binary(tmp_mount):
+    with Mount(tmp_mount, CONTENTOPTS):
+        src = tmp_mount.parent / "src.bin"
+        src.write_bytes(rand_bytes(16 << 20, seed=6))
+        subprocess.run(["cp", str(src), str(tmp_mount / "dst.bin")], check=True)
+        rc = subprocess.run(["cmp", str(src), str(tmp_mount / "dst.bin")]).returncode
+        assert rc == 0
+
+
+def test_statfs(tmp_mount):
+    with Mount(tmp_mount, HOSTOPTS):
+        st = os.statvfs(tmp_mount)
+        assert st.f_bsize == 4096
+        assert st.f_blocks > 0
