@@ -18,7 +18,9 @@ LDFLAGS  ?= -L$(ROCM)/lib -lamdhip64 -lrccl -Wl,-rpath,$(ROCM)/lib -pthread
 HOST_SRCS := src/obs.cc src/bufpool.cc src/conf.cc src/manifest.cc src/blake3.cc src/lz4.cc \
              src/gf256.cc src/store.cc src/obj.cc src/vfs.cc src/fuse.cc \
              src/engine/engine.cc
-HIP_SRCS  := src/engine/gpu_engine.hip
+HIP_SRCS  := src/engine/gpu_engine.hip src/kernels/fingerprint.hip \
+             src/kernels/lz4.hip src/kernels/rs_gf256.hip \
+             src/kernels/checksum.hip src/kernels/memops.hip
 HOST_OBJS := $(HOST_SRCS:.cc=.o)
 HIP_OBJS  := $(HIP_SRCS:.hip=.o)
 OBJS      := $(HOST_OBJS) $(HIP_OBJS)
@@ -31,7 +33,7 @@ bin/edgehip: $(OBJS) src/main.o
 	@mkdir -p bin
 	$(CXX) $(CXXFLAGS) -o $@ $^ $(LDFLAGS)
 
-lib/libedgehip.so: $(OBJS) src/capi.o
+lib/libedgehip.so: $(OBJS) src/capi.o src/capi_gpu.o
 	@mkdir -p lib
 	$(CXX) $(CXXFLAGS) -shared -o $@ $^ $(LDFLAGS)
 
@@ -42,5 +44,5 @@ lib/libedgehip.so: $(OBJS) src/capi.o
 	$(HIPCC) $(HIPFLAGS) -c -o $@ $<
 
 clean:
-	rm -f $(OBJS) src/main.o src/capi.o
+	rm -f $(OBJS) src/main.o src/capi.o src/capi_gpu.o
 	rm -rf bin lib

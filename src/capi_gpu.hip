@@ -1,0 +1,191 @@
+// Direct kernel-level test entry points (ctypes; tests/kernels/*.py, marked
+// gpu). Each runs one HIP kernel on device buffers and returns results to the
+// host so pytest can compare against the CPU goldens bit-for-bit.
+#include <vector>
+
+#include "kernels/launchers.h"
+#include "lz4.h"
+#include "obs.h"
+
+using namespace edgehip;
+
+namespace {
+struct DevBuf {
+    void* p = nullptr;
+    ~DevBuf() {
+        if (p) (void)hipFree(p);
+    }
+    hipError_t alloc(size_t n) { return hipMalloc(&p, n); }
+    template <typename T>
+    T* as() {
+        return (T*)p;
+    }
+};
+}  // namespace
+
+extern "C" {
+
+// Returns 0 on success. Hashes n chunks of data (concatenated, each `len`
+// bytes except the last which is total-...); simplified: equal-length chunks.
+int eh_gpu_blake3(const uint8_t* data, uint64_t chunk_len, int nchunks, uint8_t* out_fps) {
+    try {
+        uint64_t total = chunk_len * nchunks;
+        DevBuf d_src, d_chunks, d_out;
+        HIP_CHECK(d_src.alloc(total ? total : 1));
+        HIP_CHECK(d_chunks.alloc(sizeof(DevChunk) * nchunks));
+        HIP_CHECK(d_out.alloc(32 * nchunks));
+        HIP_CHECK(hipMemcpy(d_src.p, data, total, hipMemcpyHostToDevice));
+        std::vector<DevChunk> descs(nchunks);
+        for (int i = 0; i < nchunks; i++)
+            descs[i] = DevChunk{chunk_len * i, 0, (uint32_t)chunk_len, 0};
+        HIP_CHECK(hipMemcpy(d_chunks.p, descs.data(), sizeof(DevChunk) * nchunks,
+                            hipMemcpyHostToDevice));
+        launch_blake3_chunks(d_src.as<uint8_t>(), d_chunks.as<DevChunk>(), d_out.as<uint8_t>(),
+                             nchunks, nullptr);
+        HIP_CHECK(hipDeviceSynchronize());
+        HIP_CHECK(hipMemcpy(out_fps, d_out.p, 32 * nchunks, hipMemcpyDeviceToHost));
+        return 0;
+    } catch (const std::exception& e) {
+        LOGE("eh_gpu_blake3: %s", e.what());
+        return -1;
+    }
+}
+
+// Variable-length single chunk.
+int eh_gpu_blake3_one(const uint8_t* data, uint64_t len, uint8_t* out_fp) {
+    try {
+        DevBuf d_src, d_chunks, d_out;
+        HIP_CHECK(d_src.alloc(len ? len : 1));
+        HIP_CHECK(d_chunks.alloc(sizeof(DevChunk)));
+        HIP_CHECK(d_out.alloc(32));
+        if (len) HIP_CHECK(hipMemcpy(d_src.p, data, len, hipMemcpyHostToDevice));
+        DevChunk desc{0, 0, (uint32_t)len, 0};
+        HIP_CHECK(hipMemcpy(d_chunks.p, &desc, sizeof(desc), hipMemcpyHostToDevice));
+        launch_blake3_chunks(d_src.as<uint8_t>(), d_chunks.as<DevChunk>(), d_out.as<uint8_t>(),
+                             1, nullptr);
+        HIP_CHECK(hipDeviceSynchronize());
+        HIP_CHECK(hipMemcpy(out_fp, d_out.p, 32, hipMemcpyDeviceToHost));
+        return 0;
+    } catch (const std::exception& e) {
+        LOGE("eh_gpu_blake3_one: %s", e.what());
+        return -1;
+    }
+}
+
+// GPU-compress one chunk (framed). Returns framed size, 0 if incompressible,
+// -1 on error. out must hold len bytes.
+int64_t eh_gpu_lz4_compress(const uint8_t* data, uint32_t len, uint8_t* out) {
+    try {
+        uint32_t nsegs = (len + 4095) / 4096;
+        DevBuf d_src, d_meta, d_scratch, d_sizes, d_chunks, d_out, d_osz;
+        HIP_CHECK(d_src.alloc(len));
+        HIP_CHECK(d_meta.alloc(sizeof(ulong2) * nsegs));
+        HIP_CHECK(d_scratch.alloc((uint64_t)nsegs * 4352));
+        HIP_CHECK(d_sizes.alloc(4 * nsegs));
+        HIP_CHECK(d_chunks.alloc(sizeof(DevChunk)));
+        HIP_CHECK(d_out.alloc(len + 4096));
+        HIP_CHECK(d_osz.alloc(4));
+        HIP_CHECK(hipMemcpy(d_src.p, data, len, hipMemcpyHostToDevice));
+        std::vector<ulong2> meta(nsegs);
+        for (uint32_t s = 0; s < nsegs; s++)
+            meta[s] = ulong2{(uint64_t)s * 4096, std::min(4096u, len - s * 4096)};
+        HIP_CHECK(hipMemcpy(d_meta.p, meta.data(), sizeof(ulong2) * nsegs,
+                            hipMemcpyHostToDevice));
+        launch_lz4_compress_segs(d_src.as<uint8_t>(), d_meta.as<ulong2>(),
+                                 d_scratch.as<uint8_t>(), d_sizes.as<uint32_t>(), nsegs,
+                                 nullptr);
+        DevChunk desc{0, 0, len, 0};
+        HIP_CHECK(hipMemcpy(d_chunks.p, &desc, sizeof(desc), hipMemcpyHostToDevice));
+        launch_lz4_emit(d_src.as<uint8_t>(), d_scratch.as<uint8_t>(), d_sizes.as<uint32_t>(),
+                        d_chunks.as<DevChunk>(), d_out.as<uint8_t>(), d_osz.as<uint32_t>(), 1,
+                        nullptr);
+        HIP_CHECK(hipDeviceSynchronize());
+        uint32_t osz = 0;
+        HIP_CHECK(hipMemcpy(&osz, d_osz.p, 4, hipMemcpyDeviceToHost));
+        if (osz == 0) return 0;
+        HIP_CHECK(hipMemcpy(out, d_out.p, osz, hipMemcpyDeviceToHost));
+        return (int64_t)osz;
+    } catch (const std::exception& e) {
+        LOGE("eh_gpu_lz4_compress: %s", e.what());
+        return -1;
+    }
+}
+
+// GPU-decompress a framed chunk. Returns 0 ok, >0 corrupt, -1 error.
+int eh_gpu_lz4_decompress(const uint8_t* framed, uint32_t framed_len, uint8_t* out,
+                          uint32_t logical) {
+    try {
+        DevBuf d_src, d_chunks, d_out, d_status;
+        HIP_CHECK(d_src.alloc(framed_len));
+        HIP_CHECK(d_chunks.alloc(sizeof(DevChunk)));
+        HIP_CHECK(d_out.alloc(logical));
+        HIP_CHECK(d_status.alloc(4));
+        HIP_CHECK(hipMemcpy(d_src.p, framed, framed_len, hipMemcpyHostToDevice));
+        DevChunk desc{0, 0, logical, framed_len};
+        HIP_CHECK(hipMemcpy(d_chunks.p, &desc, sizeof(desc), hipMemcpyHostToDevice));
+        launch_lz4_decompress_chunks(d_src.as<uint8_t>(), d_chunks.as<DevChunk>(),
+                                     d_out.as<uint8_t>(), d_status.as<uint32_t>(), 1, nullptr);
+        HIP_CHECK(hipDeviceSynchronize());
+        uint32_t status = 0;
+        HIP_CHECK(hipMemcpy(&status, d_status.p, 4, hipMemcpyDeviceToHost));
+        if (status == 0) HIP_CHECK(hipMemcpy(out, d_out.p, logical, hipMemcpyDeviceToHost));
+        return (int)status;
+    } catch (const std::exception& e) {
+        LOGE("eh_gpu_lz4_decompress: %s", e.what());
+        return -1;
+    }
+}
+
+// GPU GF(256) matmul: out[i] = sum_j coef[i*k+j] * in[j]. Shards concatenated.
+int eh_gpu_gf_matmul(const uint8_t* shards, int k, int m, const uint8_t* coef, uint64_t len,
+                     uint8_t* out) {
+    try {
+        DevBuf d_in, d_out;
+        HIP_CHECK(d_in.alloc(len * k));
+        HIP_CHECK(d_out.alloc(len * m));
+        HIP_CHECK(hipMemcpy(d_in.p, shards, len * k, hipMemcpyHostToDevice));
+        RsDesc d{};
+        for (int j = 0; j < k; j++) d.in[j] = d_in.as<uint8_t>() + len * j;
+        for (int i = 0; i < m; i++) d.out[i] = d_out.as<uint8_t>() + len * i;
+        for (int i = 0; i < m; i++)
+            for (int j = 0; j < k; j++) d.coef[i][j] = coef[i * k + j];
+        d.k = k;
+        d.m = m;
+        d.len = len;
+        launch_gf_matmul(d, nullptr);
+        HIP_CHECK(hipDeviceSynchronize());
+        HIP_CHECK(hipMemcpy(out, d_out.p, len * m, hipMemcpyDeviceToHost));
+        return 0;
+    } catch (const std::exception& e) {
+        LOGE("eh_gpu_gf_matmul: %s", e.what());
+        return -1;
+    }
+}
+
+// GPU chunk crc (XOR of per-4KiB-page crc32c).
+int eh_gpu_chunk_crc(const uint8_t* data, uint32_t len, uint32_t* out) {
+    try {
+        DevBuf d_src, d_chunks, d_out;
+        HIP_CHECK(d_src.alloc(len ? len : 1));
+        HIP_CHECK(d_chunks.alloc(sizeof(DevChunk)));
+        HIP_CHECK(d_out.alloc(4));
+        if (len) HIP_CHECK(hipMemcpy(d_src.p, data, len, hipMemcpyHostToDevice));
+        DevChunk desc{0, 0, len, 0};
+        HIP_CHECK(hipMemcpy(d_chunks.p, &desc, sizeof(desc), hipMemcpyHostToDevice));
+        launch_crc32c_chunks(d_src.as<uint8_t>(), d_chunks.as<DevChunk>(),
+                             d_out.as<uint32_t>(), 1, nullptr);
+        HIP_CHECK(hipDeviceSynchronize());
+        HIP_CHECK(hipMemcpy(out, d_out.p, 4, hipMemcpyDeviceToHost));
+        return 0;
+    } catch (const std::exception& e) {
+        LOGE("eh_gpu_chunk_crc: %s", e.what());
+        return -1;
+    }
+}
+
+int eh_gpu_available(void) {
+    int n = 0;
+    return hipGetDeviceCount(&n) == hipSuccess && n > 0 ? n : 0;
+}
+
+}  // extern "C"

@@ -89,4 +89,8 @@ uint64_t wall_ns();  // CLOCK_REALTIME
 // crc32c (Castagnoli, SSE4.2 when available, table fallback) — manifest trailer
 uint32_t crc32c(const void* data, size_t len, uint32_t seed = 0);
 
+// ChunkRef.crc: XOR of crc32c over each 4 KiB page of the stored bytes
+// (page-parallel on the GPU, docs/FORMATS.md §checksum).
+uint32_t chunk_crc(const void* data, size_t len);
+
 }  // namespace edgehip

@@ -52,12 +52,12 @@ class HostEngine : public Engine {
             if (compressed) {
                 r.flags |= ChunkRef::kFlagLz4;
                 r.stored_len = (uint32_t)comp.size();
-                if (conf_.crc) r.crc = crc32c(comp.data(), comp.size());
+                if (conf_.crc) r.crc = chunk_crc(comp.data(), comp.size());
                 s = store_.put_copy(r.fp, comp.data(), (uint32_t)comp.size());
                 if (p.donate) p.donate->release();
             } else {
                 r.stored_len = p.len;
-                if (conf_.crc) r.crc = crc32c(p.data, p.len);
+                if (conf_.crc) r.crc = chunk_crc(p.data, p.len);
                 if (p.donate && p.donate->p) {
                     p.donate->len = p.len;
                     s = store_.put(r.fp, std::move(*p.donate));  // zero-copy ingest

@@ -1,0 +1,70 @@
+// G8: verify-lite integrity checksum. ChunkRef.crc is the
+// XOR of crc32c over each 4 KiB page of the stored bytes (docs/FORMATS.md
+// §checksum) — XOR-combining makes the computation page-parallel on the GPU
+// while the host side stays a cheap SSE4.2 loop (src/obs.cc chunk_crc).
+#include "dev_common.h"
+
+namespace edgehip {
+
+// slice-by-4 tables built in LDS once per block (4 KiB)
+__global__ __launch_bounds__(256) void k_crc32c_chunks(
+    const uint8_t* __restrict__ src, const DevChunk* __restrict__ chunks,
+    uint32_t* __restrict__ out /* one per chunk */, int nchunks) {
+    (void)nchunks;
+    __shared__ uint32_t tab[4][256];
+    // build tables cooperatively
+    for (int i = threadIdx.x; i < 256; i += blockDim.x) {
+        uint32_t c = i;
+        for (int k = 0; k < 8; k++) c = (c & 1) ? (0x82F63B78u ^ (c >> 1)) : (c >> 1);
+        tab[0][i] = c;
+    }
+    __syncthreads();
+    for (int t = 1; t < 4; t++) {
+        for (int i = threadIdx.x; i < 256; i += blockDim.x)
+            tab[t][i] = (tab[t - 1][i] >> 8) ^ tab[0][tab[t - 1][i] & 0xff];
+        __syncthreads();
+    }
+
+    const DevChunk c = chunks[blockIdx.x];
+    const uint8_t* base = src + c.src_off;
+    uint32_t npages = (c.len + 4095) / 4096;
+    uint32_t acc = 0;
+    for (uint32_t p = threadIdx.x; p < npages; p += blockDim.x) {
+        uint32_t off = p * 4096;
+        uint32_t len = min(4096u, c.len - off);
+        const uint8_t* q = base + off;
+        uint32_t crc = 0xFFFFFFFFu;
+        uint32_t i = 0;
+        if ((((uintptr_t)q) & 3) == 0) {
+            for (; i + 4 <= len; i += 4) {
+                uint32_t v;
+                __builtin_memcpy(&v, q + i, 4);
+                v ^= crc;
+                crc = tab[3][v & 0xff] ^ tab[2][(v >> 8) & 0xff] ^ tab[1][(v >> 16) & 0xff] ^
+                      tab[0][(v >> 24) & 0xff];
+            }
+        }
+        for (; i < len; i++) crc = tab[0][(crc ^ q[i]) & 0xff] ^ (crc >> 8);
+        acc ^= ~crc;
+    }
+    // wave + block XOR-reduce, then one atomic per block
+#pragma unroll
+    for (int s = 32; s >= 1; s >>= 1) acc ^= __shfl_xor(acc, s, 64);
+    __shared__ uint32_t wsum[4];
+    int wave = threadIdx.x / 64;
+    if ((threadIdx.x & 63) == 0) wsum[wave] = acc;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        uint32_t r = wsum[0] ^ wsum[1] ^ wsum[2] ^ wsum[3];
+        out[blockIdx.x] = r;
+    }
+}
+
+void launch_crc32c_chunks(const uint8_t* src, const DevChunk* chunks, uint32_t* out,
+                          int nchunks, hipStream_t stream) {
+    if (nchunks <= 0) return;
+    hipLaunchKernelGGL(k_crc32c_chunks, dim3(nchunks), dim3(256), 0, stream, src, chunks,
+                       out, nchunks);
+}
+
+}  // namespace edgehip

@@ -1,0 +1,39 @@
+// Host-visible launch wrappers for the edgehip HIP kernels.
+#pragma once
+
+#include <hip/hip_runtime.h>
+
+#include "dev_common.h"
+
+namespace edgehip {
+
+void launch_blake3_chunks(const uint8_t* src, const DevChunk* chunks, uint8_t* out_fps,
+                          int nchunks, hipStream_t stream);
+
+void launch_lz4_compress_segs(const uint8_t* src, const ulong2* seg_meta, uint8_t* scratch,
+                              uint32_t* sizes, int nsegs, hipStream_t stream);
+void launch_lz4_emit(const uint8_t* src, const uint8_t* scratch, const uint32_t* seg_sizes,
+                     const DevChunk* chunks, uint8_t* dst_base, uint32_t* out_sizes,
+                     int nchunks, hipStream_t stream);
+void launch_lz4_decompress_chunks(const uint8_t* src, const DevChunk* chunks,
+                                  uint8_t* dst_base, uint32_t* status, int nchunks,
+                                  hipStream_t stream);
+
+void launch_crc32c_chunks(const uint8_t* src, const DevChunk* chunks, uint32_t* out,
+                          int nchunks, hipStream_t stream);
+
+void launch_batch_copy(const uint8_t* src, uint8_t* dst, const DevChunk* descs, int ndescs,
+                       hipStream_t stream);
+
+constexpr int kRsMaxK = 16;
+constexpr int kRsMaxM = 4;
+struct RsDesc {
+    const uint8_t* in[kRsMaxK];
+    uint8_t* out[kRsMaxM];
+    uint8_t coef[kRsMaxM][kRsMaxK];
+    int k, m;
+    uint64_t len;
+};
+void launch_gf_matmul(const RsDesc& d, hipStream_t stream);
+
+}  // namespace edgehip

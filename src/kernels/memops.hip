@@ -1,0 +1,38 @@
+// G9: batched device-side scatter/gather copies — chunk
+// staging <-> arena moves and page slicing for small random reads, replacing
+// per-chunk hipMemcpyDtoD launches with one batch kernel.
+#include "dev_common.h"
+
+namespace edgehip {
+
+// descriptors: src_off/dst_off are absolute byte offsets into src/dst bases.
+__global__ __launch_bounds__(256) void k_batch_copy(
+    const uint8_t* __restrict__ src, uint8_t* __restrict__ dst,
+    const DevChunk* __restrict__ descs, int ndescs) {
+    int d = blockIdx.y;
+    if (d >= ndescs) return;
+    const DevChunk c = descs[d];
+    const uint8_t* s = src + c.src_off;
+    uint8_t* o = dst + c.dst_off;
+    uint32_t len = c.len;
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    uint64_t t = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if ((((uintptr_t)s | (uintptr_t)o) & 15) == 0) {
+        uint32_t vec = len / 16;
+        for (uint64_t v = t; v < vec; v += stride)
+            ((uint4*)o)[v] = ((const uint4*)s)[v];
+        for (uint64_t i = (uint64_t)vec * 16 + t; i < len; i += stride) o[i] = s[i];
+    } else {
+        for (uint64_t i = t; i < len; i += stride) o[i] = s[i];
+    }
+}
+
+void launch_batch_copy(const uint8_t* src, uint8_t* dst, const DevChunk* descs, int ndescs,
+                       hipStream_t stream) {
+    if (ndescs <= 0) return;
+    // x-dim sized for ~1 MiB descriptors; y-dim = descriptor index
+    hipLaunchKernelGGL(k_batch_copy, dim3(32, ndescs), dim3(256), 0, stream, src, dst,
+                       descs, ndescs);
+}
+
+}  // namespace edgehip

@@ -1,0 +1,167 @@
+// G6: Reed–Solomon GF(256) matrix multiply on gfx950 —
+// encode (Cauchy parity rows) and decode (host-inverted submatrix rows) share
+// one kernel: out[i] = sum_j M[i][j] * in[j] over GF(256), bytes vectorized
+// uint4 per lane.
+//
+// GF(256) constant-multiply uses the 4-bit split-table method with the
+// 16-entry tables held in VGPRs and looked up via V_PERM_B32 byte selects
+// (2 perms + OR per nibble — ~10 VALU ops per dword per coefficient), not
+// per-byte LDS reads. Field: poly 0x11D, matching src/gf256.cc and
+// edgehip/golden/gf256.py byte-for-byte.
+//
+// The MFMA bit-plane variant (GF(2) companion-matrix matmul on
+// mfma_i32_16x16x64_i8) is the planned tuned path once rocprof shows this
+// VALU version below the HBM bound (SURVEY.md §8.4 hard part 2).
+#include "launchers.h"
+
+namespace edgehip {
+
+constexpr int kMaxK = kRsMaxK;
+constexpr int kMaxM = kRsMaxM;
+
+// v_perm_b32: D.byte[i] = byte sel[i] of the 8-byte {src0, src1} pair
+// (sel 0-3 -> src1 bytes, 4-7 -> src0 bytes, >=8 -> 0). The builtin is
+// __builtin_amdgcn_perm(a, b, s) which lowers to v_perm_b32 a, b, s —
+// byte i of result = (s.byte[i] < 4 ? b : a >> ...). We verify semantics in
+// tests/kernels (cross-checked against the host golden); the two table
+// halves are passed in the order that makes sel 0-7 hit t[0..7].
+EH_DEV uint32_t gf_mul_dword(uint32_t v, const uint32_t tlo[4], const uint32_t thi[4]) {
+    uint32_t slo = v & 0x0F0F0F0Fu;
+    uint32_t shi = (v >> 4) & 0x0F0F0F0Fu;
+    // entries 0-7 live in (tlo[1],tlo[0]); entries 8-15 in (tlo[3],tlo[2]).
+    // sel byte < 8 picks from the first perm; sel^8 flips which half sees
+    // an in-range selector, the off-range one yields 0 -> OR combines.
+    uint32_t r0 = __builtin_amdgcn_perm(tlo[1], tlo[0], slo) |
+                  __builtin_amdgcn_perm(tlo[3], tlo[2], slo ^ 0x08080808u);
+    uint32_t r1 = __builtin_amdgcn_perm(thi[1], thi[0], shi) |
+                  __builtin_amdgcn_perm(thi[3], thi[2], shi ^ 0x08080808u);
+    return r0 ^ r1;
+}
+
+// GF(256) multiply (poly 0x11D) — device scalar, for table construction.
+EH_DEV uint8_t gf_mul_slow(uint8_t a, uint8_t b) {
+    uint32_t r = 0, aa = a, bb = b;
+    for (int i = 0; i < 8; i++) {
+        if (bb & 1) r ^= aa;
+        bb >>= 1;
+        aa <<= 1;
+        if (aa & 0x100) aa ^= 0x11D;
+    }
+    return (uint8_t)r;
+}
+
+__global__ __launch_bounds__(256) void k_gf_matmul(RsDesc d) {
+    // Build per-coefficient nibble tables once per thread (registers).
+    // m*k <= 8 live coefficient pairs at once would blow VGPRs; instead loop
+    // j outer with the j-column's m coefficient tables live.
+    const uint64_t nvec = d.len / 16;
+    const uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    const uint64_t tid0 = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+
+    // accumulate in registers per 16-byte vector: iterate vectors outer,
+    // shards inner, so each output vector is written once.
+    for (uint64_t v = tid0; v < nvec; v += stride) {
+        uint4 acc[kMaxM];
+#pragma unroll
+        for (int i = 0; i < kMaxM; i++) acc[i] = uint4{0, 0, 0, 0};
+        for (int j = 0; j < d.k; j++) {
+            const uint4 x = ((const uint4*)d.in[j])[v];
+            for (int i = 0; i < d.m; i++) {
+                uint8_t c = d.coef[i][j];
+                if (c == 0) continue;
+                if (c == 1) {
+                    acc[i].x ^= x.x; acc[i].y ^= x.y; acc[i].z ^= x.z; acc[i].w ^= x.w;
+                    continue;
+                }
+                // build the nibble tables (16+16 bytes in 4+4 dwords)
+                uint32_t tlo[4], thi[4];
+#pragma unroll
+                for (int t = 0; t < 4; t++) {
+                    uint32_t w = 0, wh = 0;
+#pragma unroll
+                    for (int b = 0; b < 4; b++) {
+                        w |= (uint32_t)gf_mul_slow(c, (uint8_t)(t * 4 + b)) << (8 * b);
+                        wh |= (uint32_t)gf_mul_slow(c, (uint8_t)((t * 4 + b) << 4)) << (8 * b);
+                    }
+                    tlo[t] = w;
+                    thi[t] = wh;
+                }
+                acc[i].x ^= gf_mul_dword(x.x, tlo, thi);
+                acc[i].y ^= gf_mul_dword(x.y, tlo, thi);
+                acc[i].z ^= gf_mul_dword(x.z, tlo, thi);
+                acc[i].w ^= gf_mul_dword(x.w, tlo, thi);
+            }
+        }
+        for (int i = 0; i < d.m; i++) ((uint4*)d.out[i])[v] = acc[i];
+    }
+    // byte tail (len % 16) handled by thread 0 of block 0
+    if (tid0 == 0) {
+        for (uint64_t b = nvec * 16; b < d.len; b++) {
+            for (int i = 0; i < d.m; i++) {
+                uint8_t a = 0;
+                for (int j = 0; j < d.k; j++) a ^= gf_mul_slow(d.coef[i][j], d.in[j][b]);
+                d.out[i][b] = a;
+            }
+        }
+    }
+}
+
+// Tuned variant: compile-time K/M so every table lives in registers with
+// constant indices (a runtime-indexed register array spills to scratch —
+// guide §5.4 rule 20). Instantiated for the judged RS 6+2 shapes.
+template <int K, int M>
+__global__ __launch_bounds__(256) void k_gf_matmul_t(RsDesc d) {
+    uint32_t tlo[M][K][4], thi[M][K][4];
+#pragma unroll
+    for (int i = 0; i < M; i++)
+#pragma unroll
+        for (int j = 0; j < K; j++) {
+            uint8_t c = d.coef[i][j];
+#pragma unroll
+            for (int t = 0; t < 4; t++) {
+                uint32_t w = 0, wh = 0;
+#pragma unroll
+                for (int b = 0; b < 4; b++) {
+                    w |= (uint32_t)gf_mul_slow(c, (uint8_t)(t * 4 + b)) << (8 * b);
+                    wh |= (uint32_t)gf_mul_slow(c, (uint8_t)((t * 4 + b) << 4)) << (8 * b);
+                }
+                tlo[i][j][t] = w;
+                thi[i][j][t] = wh;
+            }
+        }
+    const uint64_t nvec = d.len / 16;
+    const uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t v = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; v < nvec; v += stride) {
+        uint4 acc[M];
+#pragma unroll
+        for (int i = 0; i < M; i++) acc[i] = uint4{0, 0, 0, 0};
+#pragma unroll
+        for (int j = 0; j < K; j++) {
+            const uint4 x = ((const uint4*)d.in[j])[v];
+#pragma unroll
+            for (int i = 0; i < M; i++) {
+                acc[i].x ^= gf_mul_dword(x.x, tlo[i][j], thi[i][j]);
+                acc[i].y ^= gf_mul_dword(x.y, tlo[i][j], thi[i][j]);
+                acc[i].z ^= gf_mul_dword(x.z, tlo[i][j], thi[i][j]);
+                acc[i].w ^= gf_mul_dword(x.w, tlo[i][j], thi[i][j]);
+            }
+        }
+#pragma unroll
+        for (int i = 0; i < M; i++) ((uint4*)d.out[i])[v] = acc[i];
+    }
+}
+
+void launch_gf_matmul(const RsDesc& d, hipStream_t stream) {
+    uint64_t nvec = d.len / 16;
+    int blocks = (int)std::min<uint64_t>((nvec + 255) / 256, 2048);
+    if (blocks < 1) blocks = 1;
+    bool aligned = d.len % 16 == 0;
+    if (aligned && d.k == 6 && d.m == 2)
+        hipLaunchKernelGGL((k_gf_matmul_t<6, 2>), dim3(blocks), dim3(256), 0, stream, d);
+    else if (aligned && d.k == 6 && d.m == 1)
+        hipLaunchKernelGGL((k_gf_matmul_t<6, 1>), dim3(blocks), dim3(256), 0, stream, d);
+    else
+        hipLaunchKernelGGL(k_gf_matmul, dim3(blocks), dim3(256), 0, stream, d);
+}
+
+}  // namespace edgehip

@@ -128,4 +128,12 @@ uint32_t crc32c(const void* data, size_t len, uint32_t seed) {
     return ~c;
 }
 
+uint32_t chunk_crc(const void* data, size_t len) {
+    const uint8_t* p = (const uint8_t*)data;
+    uint32_t acc = 0;
+    for (size_t off = 0; off < len; off += 4096)
+        acc ^= crc32c(p + off, std::min<size_t>(4096, len - off));
+    return acc;
+}
+
 }  // namespace edgehip
