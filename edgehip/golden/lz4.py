@@ -55,14 +55,27 @@ def decompress_chunk(framed: bytes, logical: int) -> bytes:
     out = bytearray()
     for b in range(nblocks):
         off, len_flags = struct.unpack_from("<II", framed, 8 + 8 * b)
-        blen = len_flags & 0x7FFFFFFF
+        blen = len_flags & 0x3FFFFFFF
         raw = bool(len_flags & 0x80000000)
+        seg = bool(len_flags & 0x40000000)
         want = min(LZ4_BLOCK, logical - b * LZ4_BLOCK)
         blob = payload[off:off + blen]
         if raw:
             if blen != want:
                 raise ValueError("raw block length mismatch")
             out += blob
+        elif seg:
+            # GPU segmented: u16 length table, segments packed = one stream;
+            # also verify per-segment decode (the parallel property)
+            ns = (want + 4095) // 4096
+            lens = struct.unpack_from(f"<{ns}H", blob, 0)
+            pos = ns * 2
+            for s, ln in enumerate(lens):
+                swant = min(4096, want - s * 4096)
+                out += decompress_block(blob[pos:pos + ln], swant)
+                pos += ln
+            if pos != blen:
+                raise ValueError("segment table inconsistent")
         else:
             out += decompress_block(blob, want)
     if len(out) != logical:

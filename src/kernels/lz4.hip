@@ -1,14 +1,13 @@
 // G4/G5: LZ4 block-format compress/decompress on gfx950.
 //
 // Parallel layout: every 4 KiB segment of a chunk is compressed independently
-// by ONE thread (greedy hash-table match finder, table in LDS). A
-// concatenation of per-segment LZ4 streams is itself a valid LZ4 stream for
-// the concatenated output (each segment ends in a literals-only sequence and
-// matches never cross segment starts), so the result decodes with any
-// spec-conformant LZ4 decoder. The chunk frame (docs/FORMATS.md §LZ4)
-// records per-64 KiB-block {off, len_flags}; GPU-written blocks set the
-// SEGMENTED bit (bit 30) and prefix the payload with 16 u16 segment lengths
-// so decompression runs one thread per segment.
+// by ONE thread (greedy hash-table match finder, table in LDS), each segment
+// a self-contained LZ4 block-format stream. The chunk frame
+// (docs/FORMATS.md §LZ4) records per-64 KiB-block {off, len_flags};
+// GPU-written blocks set the SEGMENTED bit (bit 30) and prefix the payload
+// with a u16 segment-length table so decompression runs one thread per
+// segment (decoders MUST decode per segment: a segment's terminal
+// literals-only token would parse as a match token mid-stream).
 //
 // Judged configs use random bytes: the per-thread loop short-circuits
 // incompressible data (hash misses dominate -> emits one big literal run),
@@ -157,12 +156,12 @@ __global__ __launch_bounds__(256) void k_lz4_emit(
         uint32_t blk_len = min(kBlk, len - b * kBlk);
         uint32_t s0 = b * kSegPerBlk;
         uint32_t ns = (blk_len + kSeg - 1) / kSeg;
-        uint32_t total = ((ns * 2 + 3) & ~3u);
+        uint32_t total = ns * 2;  // u16 segment-length table (tight)
         bool ok = true;
         for (uint32_t s = 0; s < ns; s++) {
             uint32_t sz = seg_sizes[first_seg + s0 + s];
             if (sz == 0) ok = false;
-            total += (sz + 3) & ~3u;
+            total += sz;
         }
         if (!ok || total >= blk_len) {
             blk_raw[b] = 1;
@@ -215,20 +214,17 @@ __global__ __launch_bounds__(256) void k_lz4_emit(
                 ((uint4*)bp)[v] = ((const uint4*)s)[v];
             for (uint32_t i = vec * 16 + tid; i < blk_len; i += blockDim.x) bp[i] = s[i];
         } else {
-            uint32_t tbl_bytes = (ns * 2 + 3) & ~3u;
-            if (tid < (int)ns) {
-                ((uint16_t*)bp)[tid] = (uint16_t)seg_sizes[first_seg + b * kSegPerBlk + tid];
-                if (ns * 2 < tbl_bytes && tid == 0)
-                    for (uint32_t i = ns * 2; i < tbl_bytes; i++) bp[i] = 0;
-            }
+            if (tid < (int)ns)
+                __builtin_memcpy(bp + 2 * tid,
+                                 (const uint8_t*)&seg_sizes[first_seg + b * kSegPerBlk + tid], 2);
             __syncthreads();
-            // segment payload copies: thread groups of 16 per segment
-            uint32_t so = tbl_bytes;
+            // segment payload copies, packed back to back
+            uint32_t so = ns * 2;
             for (uint32_t s = 0; s < ns; s++) {
                 uint32_t sz = seg_sizes[first_seg + b * kSegPerBlk + s];
                 const uint8_t* sp = scratch + (uint64_t)(first_seg + b * kSegPerBlk + s) * kSegOutStride;
                 for (uint32_t i = tid; i < sz; i += blockDim.x) bp[so + i] = sp[i];
-                so += (sz + 3) & ~3u;
+                so += sz;
             }
             __syncthreads();
         }
@@ -318,13 +314,15 @@ __global__ __launch_bounds__(256) void k_lz4_decompress_chunks(
             uint32_t ns = (blk_len + kSeg - 1) / kSeg;
             if (!raw && seg && s < ns) {
                 const uint8_t* bp = payload + h32[2 + 2 * b];
-                const uint16_t* seg_len = (const uint16_t*)bp;
-                uint32_t tbl_bytes = (ns * 2 + 3) & ~3u;
-                uint32_t so = tbl_bytes;
-                for (uint32_t k = 0; k < s; k++) so += (seg_len[k] + 3) & ~3u;
+                uint32_t so = ns * 2;
+                uint16_t slen = 0;
+                for (uint32_t k = 0; k <= s; k++) {
+                    __builtin_memcpy(&slen, bp + 2 * k, 2);  // table may be unaligned
+                    if (k < s) so += slen;
+                }
                 uint8_t* bo = out + b * kBlk;
                 uint32_t want = min(kSeg, blk_len - s * kSeg);
-                uint32_t got = lz4_decompress_seq(bp + so, seg_len[s], bo + s * kSeg, want);
+                uint32_t got = lz4_decompress_seq(bp + so, slen, bo + s * kSeg, want);
                 if (got != want) status[c] = 1;
             }
         }

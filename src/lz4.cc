@@ -175,14 +175,35 @@ St lz4_chunk_decompress(const uint8_t* src, size_t n, uint8_t* dst, size_t logic
     for (size_t b = 0; b < nblocks; b++) {
         size_t off = h32[2 + 2 * b];
         uint32_t lf = h32[3 + 2 * b];
-        size_t blen = lf & 0x7fffffffu;
+        size_t blen = lf & 0x3fffffffu;
         bool raw = lf & 0x80000000u;
+        bool seg = lf & 0x40000000u;
         size_t dst_off = b * kLz4Block;
         size_t want = std::min(kLz4Block, logical - dst_off);
         if (off + blen > payload_len) return St::bad_format;
         if (raw) {
             if (blen != want) return St::bad_format;
             std::memcpy(dst + dst_off, payload + off, blen);
+        } else if (seg) {
+            // GPU-written: u16 segment-length table, then per-4 KiB-segment
+            // LZ4 streams packed back to back. Each segment MUST be decoded
+            // with its own stream (a segment's terminal literals-only token
+            // would read as a match token mid-stream).
+            size_t ns = div_up(want, kLz4Seg);
+            if (blen < ns * 2) return St::bad_format;
+            const uint8_t* tbl = payload + off;
+            size_t pos = ns * 2;
+            for (size_t s = 0; s < ns; s++) {
+                uint16_t slen;
+                std::memcpy(&slen, tbl + 2 * s, 2);
+                size_t swant = std::min(kLz4Seg, want - s * kLz4Seg);
+                if (pos + slen > blen) return St::bad_format;
+                size_t got = lz4_decompress_block(tbl + pos, slen,
+                                                  dst + dst_off + s * kLz4Seg, swant);
+                if (got != swant) return St::bad_format;
+                pos += slen;
+            }
+            if (pos != blen) return St::bad_format;
         } else {
             size_t got = lz4_decompress_block(payload + off, blen, dst + dst_off, want);
             if (got != want) return St::bad_format;

@@ -34,6 +34,7 @@ inline size_t lz4_bound(size_t n) { return n + n / 255 + 16; }
 // Block logical size is kLz4Block (64 KiB) except the final block.
 constexpr uint32_t kLz4Magic = 0x5A4C4845;
 constexpr size_t kLz4Block = 64 * KiB;
+constexpr size_t kLz4Seg = 4 * KiB;  // GPU parallel-compression segment
 
 // Compress a whole chunk with framing. Returns true and sets *out_len if the
 // framed result is smaller than n; false => caller stores raw (no framing).
