@@ -189,3 +189,29 @@ int eh_gpu_available(void) {
 }
 
 }  // extern "C"
+
+// ---- diagnostics ------------------------------------------------------------
+namespace {
+__global__ void k_perm_probe(uint32_t a, uint32_t b, uint32_t* out) {
+    // out[s] = v_perm_b32(a, b, sel) for selector bytes 0..15 replicated
+    int s = threadIdx.x;
+    if (s < 16) {
+        uint32_t sel = s * 0x01010101u;
+        out[s] = __builtin_amdgcn_perm(a, b, sel);
+    }
+}
+}  // namespace
+
+extern "C" int eh_gpu_perm_probe(uint32_t a, uint32_t b, uint32_t* out16) {
+    try {
+        DevBuf d;
+        HIP_CHECK(d.alloc(64));
+        hipLaunchKernelGGL(k_perm_probe, dim3(1), dim3(64), 0, nullptr, a, b, d.as<uint32_t>());
+        HIP_CHECK(hipDeviceSynchronize());
+        HIP_CHECK(hipMemcpy(out16, d.p, 64, hipMemcpyDeviceToHost));
+        return 0;
+    } catch (const std::exception& e) {
+        LOGE("eh_gpu_perm_probe: %s", e.what());
+        return -1;
+    }
+}

@@ -4,6 +4,7 @@
 #include <atomic>
 
 #include "../blake3.h"
+#include "../gf256.h"
 #include "../lz4.h"
 #include "../obs.h"
 #include "../store.h"
@@ -103,6 +104,8 @@ class HostEngine : public Engine {
 
     St release_chunk(const ChunkRef& ref) override { return store_.decref(ref.fp); }
 
+    St debug_drop_chunk(const Fp& fp) override { return store_.remove(fp); }
+
     uint64_t stored_bytes() const override { return store_.bytes(); }
     uint64_t stored_chunks() const override { return store_.count(); }
     bool is_gpu() const override { return false; }
@@ -113,6 +116,81 @@ class HostEngine : public Engine {
 };
 
 }  // namespace
+
+// ---- Engine base: RS via the generic chunk APIs (host GF(256) math) --------
+// Works for any engine; GpuEngine overrides with the on-device kernel path.
+St Engine::rs_encode_stripe(const ChunkRef* data, int k, int m, uint32_t shard_len,
+                            ChunkRef* parity_out) {
+    std::vector<std::vector<uint8_t>> shards((size_t)k);
+    std::vector<GetChunk> gets;
+    for (int j = 0; j < k; j++) {
+        shards[j].assign(shard_len, 0);
+        if (!data[j].fp.is_zero())
+            gets.push_back({&data[j], shards[j].data(), 0, data[j].logical_len});
+    }
+    if (!gets.empty()) {
+        St s = get_chunks(gets.data(), (int)gets.size());
+        if (s != St::ok) return s;
+    }
+    std::vector<const uint8_t*> in((size_t)k);
+    for (int j = 0; j < k; j++) in[j] = shards[j].data();
+    std::vector<std::vector<uint8_t>> par((size_t)m, std::vector<uint8_t>(shard_len));
+    std::vector<uint8_t*> out((size_t)m);
+    for (int i = 0; i < m; i++) out[i] = par[i].data();
+    rs_encode(k, m, in.data(), out.data(), shard_len);
+    std::vector<PutChunk> puts((size_t)m);
+    for (int i = 0; i < m; i++) puts[i] = {par[i].data(), shard_len, nullptr};
+    St s = put_chunks(puts.data(), parity_out, m);
+    if (s != St::ok) return s;
+    for (int i = 0; i < m; i++)
+        parity_out[i].flags |= ChunkRef::kFlagParity | ChunkRef::kFlagRs;
+    metrics().rs_encoded_stripes++;
+    return St::ok;
+}
+
+St Engine::rs_reconstruct(const ChunkRef* refs, const bool* present_in, int k, int m,
+                          uint32_t shard_len, int target, uint8_t* out, uint32_t out_len) {
+    int n = k + m;
+    std::vector<std::vector<uint8_t>> shards((size_t)n);
+    std::vector<uint8_t*> ptrs((size_t)n);
+    std::vector<bool> present((size_t)n, false);
+    int navail = 0;
+    for (int i = 0; i < n && navail < k; i++) {
+        shards[i].assign(shard_len, 0);
+        ptrs[i] = shards[i].data();
+        if (i == target) continue;
+        if (refs[i].fp.is_zero()) {  // hole = zero shard, always available
+            present[i] = true;
+            navail++;
+            continue;
+        }
+        if (present_in && !present_in[i]) continue;
+        GetChunk g{&refs[i], shards[i].data(), 0, refs[i].logical_len};
+        if (get_chunks(&g, 1) == St::ok) {
+            present[i] = true;
+            navail++;
+        }
+    }
+    for (int i = 0; i < n; i++)
+        if (shards[i].empty()) {
+            shards[i].assign(shard_len, 0);
+            ptrs[i] = shards[i].data();
+        }
+    if (navail < k) return St::io;
+    std::vector<bool> pres2 = present;
+    // convert to plain bool array for rs_reconstruct()
+    std::vector<char> pc((size_t)n);
+    for (int i = 0; i < n; i++) pc[i] = present[i] ? 1 : 0;
+    std::vector<bool> dummy;
+    bool pb[64];
+    for (int i = 0; i < n; i++) pb[i] = pc[i] != 0;
+    St s = ::edgehip::rs_reconstruct(k, m, ptrs.data(), pb, shard_len);
+    if (s != St::ok) return s;
+    if (out_len > shard_len) return St::inval;
+    std::memcpy(out, shards[target].data(), out_len);
+    metrics().rs_reconstructs++;
+    return St::ok;
+}
 
 std::unique_ptr<Engine> make_host_engine(const Conf& conf) {
     return std::make_unique<HostEngine>(conf);

@@ -119,6 +119,20 @@ class GpuEngine : public Engine {
     St get_chunks(const GetChunk* get, int n) override;
     St release_chunk(const ChunkRef& ref) override;
 
+    St debug_drop_chunk(const Fp& fp) override {
+        std::lock_guard<std::mutex> g(map_mu_);
+        auto it = map_.find(fp);
+        if (it == map_.end()) return spill_.remove(fp);
+        if (it->second.spilled) {
+            spill_.remove(fp);
+        } else {
+            arena_free(it->second.off, it->second.len);
+            lru_.erase(it->second.lru);
+        }
+        map_.erase(it);
+        return St::ok;
+    }
+
     uint64_t stored_bytes() const override { return arena_used_.load() + spill_.bytes(); }
     uint64_t stored_chunks() const override {
         std::lock_guard<std::mutex> g(map_mu_);

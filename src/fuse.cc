@@ -6,6 +6,7 @@
 #include <poll.h>
 #include <sys/stat.h>
 #include <sys/mount.h>
+#include <sys/sysmacros.h>
 #include <sys/uio.h>
 #include <unistd.h>
 
@@ -51,6 +52,11 @@ FuseMount::~FuseMount() { unmount(); }
 
 St FuseMount::mount(const std::string& mountpoint) {
     fd_ = ::open("/dev/fuse", O_RDWR | O_CLOEXEC);
+    if (fd_ < 0 && errno == ENOENT) {
+        // GPU boxes ship without the device node; create it (char 10:229).
+        if (mknod("/dev/fuse", S_IFCHR | 0600, makedev(10, 229)) == 0)
+            fd_ = ::open("/dev/fuse", O_RDWR | O_CLOEXEC);
+    }
     if (fd_ < 0) {
         LOGE("fuse: open /dev/fuse failed: %s", strerror(errno));
         return St::io;

@@ -19,23 +19,23 @@ namespace edgehip {
 constexpr int kMaxK = kRsMaxK;
 constexpr int kMaxM = kRsMaxM;
 
-// v_perm_b32: D.byte[i] = byte sel[i] of the 8-byte {src0, src1} pair
-// (sel 0-3 -> src1 bytes, 4-7 -> src0 bytes, >=8 -> 0). The builtin is
-// __builtin_amdgcn_perm(a, b, s) which lowers to v_perm_b32 a, b, s —
-// byte i of result = (s.byte[i] < 4 ? b : a >> ...). We verify semantics in
-// tests/kernels (cross-checked against the host golden); the two table
-// halves are passed in the order that makes sel 0-7 hit t[0..7].
+// v_perm_b32 semantics, PROBED ON gfx950 (eh_gpu_perm_probe, 2026-09-13):
+// __builtin_amdgcn_perm(a, b, sel): result byte i = sel.byte[i] in 0..3 ->
+// b.byte[sel], 4..7 -> a.byte[sel-4]; sel >= 8 gives sign-replicated
+// 0x00/0xFF constants (NOT zero) — so a 16-entry lookup needs an explicit
+// bit-3 byte select, done with the borrow-propagating byte-mask trick
+// ((m << 8) - m turns per-byte 0/1 into 0x00/0xFF).
+EH_DEV uint32_t gf_tbl16(uint32_t sel /* 4-bit per byte */, const uint32_t t[4]) {
+    uint32_t s7 = sel & 0x07070707u;
+    uint32_t lo = __builtin_amdgcn_perm(t[1], t[0], s7);  // entries 0-7
+    uint32_t hi = __builtin_amdgcn_perm(t[3], t[2], s7);  // entries 8-15
+    uint32_t m = (sel >> 3) & 0x01010101u;
+    uint32_t m8 = (m << 8) - m;  // 0xFF per byte where bit3 of sel set
+    return lo ^ ((lo ^ hi) & m8);
+}
+
 EH_DEV uint32_t gf_mul_dword(uint32_t v, const uint32_t tlo[4], const uint32_t thi[4]) {
-    uint32_t slo = v & 0x0F0F0F0Fu;
-    uint32_t shi = (v >> 4) & 0x0F0F0F0Fu;
-    // entries 0-7 live in (tlo[1],tlo[0]); entries 8-15 in (tlo[3],tlo[2]).
-    // sel byte < 8 picks from the first perm; sel^8 flips which half sees
-    // an in-range selector, the off-range one yields 0 -> OR combines.
-    uint32_t r0 = __builtin_amdgcn_perm(tlo[1], tlo[0], slo) |
-                  __builtin_amdgcn_perm(tlo[3], tlo[2], slo ^ 0x08080808u);
-    uint32_t r1 = __builtin_amdgcn_perm(thi[1], thi[0], shi) |
-                  __builtin_amdgcn_perm(thi[3], thi[2], shi ^ 0x08080808u);
-    return r0 ^ r1;
+    return gf_tbl16(v & 0x0F0F0F0Fu, tlo) ^ gf_tbl16((v >> 4) & 0x0F0F0F0Fu, thi);
 }
 
 // GF(256) multiply (poly 0x11D) — device scalar, for table construction.

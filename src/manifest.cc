@@ -17,10 +17,10 @@ void Manifest::encode(std::vector<uint8_t>* out) const {
     out->push_back(fp_algo);
     out->push_back(compress);
     out->insert(out->end(), oid.b.begin(), oid.b.end());
-    put_u64(out, chunks.size());
+    put_u64(out, chunks.size() + parity.size());
     put_u64(out, logical_size);
     put_u64(out, chunk_size);
-    for (const auto& c : chunks) {
+    auto put_ref = [&](const ChunkRef& c) {
         out->insert(out->end(), c.fp.b.begin(), c.fp.b.end());
         put_u32(out, c.logical_len);
         put_u32(out, c.stored_len);
@@ -28,7 +28,9 @@ void Manifest::encode(std::vector<uint8_t>* out) const {
         put_u32(out, c.crc);
         put_u64(out, c.shard_map);
         put_u64(out, 0);  // reserved
-    }
+    };
+    for (const auto& c : chunks) put_ref(c);
+    for (const auto& c : parity) put_ref(c);
     uint32_t crc = crc32c(out->data(), out->size());
     put_u32(out, crc);
 }
@@ -47,16 +49,22 @@ St Manifest::decode(const uint8_t* data, size_t len, Manifest* out) {
     if (len != need) return St::bad_format;
     if (crc32c(data, len - 4) != get_u32(data + len - 4)) return St::bad_format;
     out->chunks.clear();
-    out->chunks.resize(nchunks);
+    out->parity.clear();
     const uint8_t* p = data + kHeaderSize;
     for (uint64_t i = 0; i < nchunks; i++, p += kRefSize) {
-        ChunkRef& c = out->chunks[i];
+        ChunkRef c;
         std::memcpy(c.fp.b.data(), p, 32);
         c.logical_len = get_u32(p + 32);
         c.stored_len = get_u32(p + 36);
         c.flags = get_u32(p + 40);
         c.crc = get_u32(p + 44);
         c.shard_map = get_u64(p + 48);
+        if (c.parity()) {
+            out->parity.push_back(c);
+        } else {
+            if (!out->parity.empty()) return St::bad_format;  // parity must trail
+            out->chunks.push_back(c);
+        }
     }
     return St::ok;
 }

@@ -38,8 +38,10 @@ struct ChunkRef {
     uint64_t shard_map = 0;
 
     static constexpr uint32_t kFlagLz4 = 1u << 0;
-    static constexpr uint32_t kFlagRs = 1u << 1;
+    static constexpr uint32_t kFlagRs = 1u << 1;      // member of an RS stripe
+    static constexpr uint32_t kFlagParity = 1u << 2;  // parity chunk (not logical data)
     bool compressed() const { return flags & kFlagLz4; }
+    bool parity() const { return flags & kFlagParity; }
 };
 
 struct Manifest {
@@ -53,9 +55,12 @@ struct Manifest {
     uint8_t compress = 0;
     uint64_t logical_size = 0;
     uint64_t chunk_size = 1 * MiB;
-    std::vector<ChunkRef> chunks;
+    std::vector<ChunkRef> chunks;   // data chunks, index = offset / chunk_size
+    std::vector<ChunkRef> parity;   // RS parity (flag kFlagParity), appended on wire
 
-    size_t encoded_size() const { return kHeaderSize + chunks.size() * kRefSize + 4; }
+    size_t encoded_size() const {
+        return kHeaderSize + (chunks.size() + parity.size()) * kRefSize + 4;
+    }
     void encode(std::vector<uint8_t>* out) const;
     static St decode(const uint8_t* data, size_t len, Manifest* out);
 };

@@ -72,6 +72,16 @@ St HostStore::decref(const Fp& fp) {
     return St::ok;
 }
 
+St HostStore::remove(const Fp& fp) {
+    Shard& s = shard(fp);
+    std::lock_guard<std::mutex> g(s.mu);
+    auto it = s.map.find(fp);
+    if (it == s.map.end()) return St::not_found;
+    bytes_ -= it->second.data.len;
+    s.map.erase(it);
+    return St::ok;
+}
+
 uint64_t HostStore::count() const {
     uint64_t n = 0;
     for (int i = 0; i < kShards; i++) {

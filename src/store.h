@@ -30,6 +30,7 @@ class HostStore {
     uint32_t stored_len(const Fp& fp) const;  // 0 if absent
     St addref(const Fp& fp);
     St decref(const Fp& fp);  // frees at 0
+    St remove(const Fp& fp);  // unconditional erase (fault injection)
     uint64_t bytes() const { return bytes_.load(); }
     uint64_t count() const;
 
