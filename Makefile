@@ -11,8 +11,8 @@ ROCM     ?= /opt/rocm
 HIPCC    ?= $(ROCM)/bin/hipcc
 CXX      ?= g++
 ARCH     ?= gfx950
-CXXFLAGS ?= -O2 -g -std=c++17 -Wall -Wextra -fPIC -pthread
-HIPFLAGS ?= -O3 -std=c++17 --offload-arch=$(ARCH) -fPIC
+CXXFLAGS ?= -O2 -g -std=c++17 -Wall -Wextra -fPIC -pthread -MMD -MP
+HIPFLAGS ?= -O3 -std=c++17 --offload-arch=$(ARCH) -fPIC -MMD -MP
 LDFLAGS  ?= -L$(ROCM)/lib -lamdhip64 -lrccl -Wl,-rpath,$(ROCM)/lib -pthread
 
 HOST_SRCS := src/obs.cc src/bufpool.cc src/conf.cc src/manifest.cc src/blake3.cc src/lz4.cc \
@@ -45,4 +45,7 @@ lib/libedgehip.so: $(OBJS) src/capi.o src/capi_gpu.o
 
 clean:
 	rm -f $(OBJS) src/main.o src/capi.o src/capi_gpu.o
+	rm -f $(OBJS:.o=.d) src/main.d src/capi.d src/capi_gpu.d
 	rm -rf bin lib
+
+-include $(OBJS:.o=.d) src/main.d src/capi.d src/capi_gpu.d

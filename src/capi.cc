@@ -119,3 +119,9 @@ int64_t eh_object_commit(void* o, uint8_t* manifest_out, size_t cap) {
 }
 
 }  // extern "C"
+
+extern "C" int eh_engine_drop_chunk(void* e, const uint8_t* fp32) {
+    Fp f;
+    std::memcpy(f.b.data(), fp32, 32);
+    return (int)((EhEngine*)e)->eng->debug_drop_chunk(f);
+}

@@ -14,6 +14,86 @@ void Object::drop_ref(const ChunkRef& r) {
     if (!r.fp.is_zero()) eng_->release_chunk(r);
 }
 
+void Object::mark_stripe_dirty(uint64_t chunk_idx) {
+    if (conf_->rs_k > 0) rs_dirty_.insert(chunk_idx / conf_->rs_k);
+}
+
+// Refresh parity for dirty stripes whose data chunks are all committed.
+// Stripe s covers chunks [s*k, s*k+k); chunks beyond the object are holes
+// (zero shards). Parity refs live at mf_.parity[s*m + i].
+St Object::rs_update_locked() {
+    const int k = conf_->rs_k, m = conf_->rs_m;
+    if (k <= 0 || m <= 0) return St::ok;
+    const uint32_t cs = (uint32_t)conf_->chunk_size;
+    for (auto it = rs_dirty_.begin(); it != rs_dirty_.end();) {
+        uint64_t s = *it;
+        // any still-dirty data chunk in the stripe? leave for the next flush
+        bool pending = false;
+        for (int j = 0; j < k && !pending; j++)
+            pending = dirty_.count(s * k + j) != 0;
+        if (pending) {
+            ++it;
+            continue;
+        }
+        std::vector<ChunkRef> data((size_t)k);
+        for (int j = 0; j < k; j++) {
+            uint64_t idx = s * k + j;
+            if (idx < mf_.chunks.size()) {
+                data[j] = mf_.chunks[idx];
+                data[j].flags |= ChunkRef::kFlagRs;
+            }
+        }
+        std::vector<ChunkRef> par((size_t)m);
+        St st = eng_->rs_encode_stripe(data.data(), k, m, cs, par.data());
+        if (st != St::ok) return st;
+        if (mf_.parity.size() < (s + 1) * m) mf_.parity.resize((s + 1) * m);
+        for (int i = 0; i < m; i++) {
+            drop_ref(mf_.parity[s * m + i]);
+            par[i].shard_map = (uint64_t)s | ((uint64_t)i << 48);
+            mf_.parity[s * m + i] = par[i];
+        }
+        // data refs carry the RS flag from now on
+        for (int j = 0; j < k; j++) {
+            uint64_t idx = s * k + j;
+            if (idx < mf_.chunks.size() && !mf_.chunks[idx].fp.is_zero())
+                mf_.chunks[idx].flags |= ChunkRef::kFlagRs;
+        }
+        it = rs_dirty_.erase(it);
+    }
+    return St::ok;
+}
+
+// Reconstruct chunk `idx` from its stripe (k-1 data + m parity survivors),
+// write the uncompressed bytes into chunk_buf (chunk_size bytes) and repair
+// the store by re-ingesting them (content addressing gives back the same fp).
+St Object::rs_reconstruct_chunk_locked(uint64_t idx, uint8_t* chunk_buf) {
+    const int k = conf_->rs_k, m = conf_->rs_m;
+    if (k <= 0 || m <= 0) return St::not_found;
+    const uint32_t cs = (uint32_t)conf_->chunk_size;
+    uint64_t s = idx / k;
+    if (mf_.parity.size() < (s + 1) * (uint64_t)m) return St::not_found;  // no parity yet
+    std::vector<ChunkRef> refs((size_t)(k + m));
+    for (int j = 0; j < k; j++) {
+        uint64_t c = s * k + j;
+        if (c < mf_.chunks.size()) refs[j] = mf_.chunks[c];
+    }
+    for (int i = 0; i < m; i++) refs[k + i] = mf_.parity[s * m + i];
+    int target = (int)(idx - s * k);
+    St st = eng_->rs_reconstruct(refs.data(), nullptr, k, m, cs, target, chunk_buf, cs);
+    if (st != St::ok) return st;
+    LOGW("obj: reconstructed chunk %llu via RS %d+%d", (unsigned long long)idx, k, m);
+    // repair: re-ingest the chunk (same content -> same fingerprint)
+    const ChunkRef& old = mf_.chunks[idx];
+    PutChunk put{chunk_buf, old.logical_len, nullptr};
+    ChunkRef fixed;
+    st = eng_->put_chunks(&put, &fixed, 1);
+    if (st == St::ok) {
+        fixed.flags |= ChunkRef::kFlagRs;
+        mf_.chunks[idx] = fixed;
+    }
+    return St::ok;
+}
+
 St Object::load_chunk_for_rmw(uint64_t idx, Dirty* d) {
     const uint32_t cs = (uint32_t)conf_->chunk_size;
     if (!d->data.p) d->data = BufPool::inst().get(cs);
@@ -119,6 +199,7 @@ St Object::write(uint64_t off, const uint8_t* data, uint64_t len) {
             drop_ref(mf_.chunks[idxs[i]]);
             mf_.chunks[idxs[i]] = refs[i];
             dirty_.erase(idxs[i]);
+            mark_stripe_dirty(idxs[i]);
         }
     }
     return St::ok;
@@ -166,6 +247,21 @@ St Object::read(uint64_t off, uint64_t len, uint8_t* out, uint64_t* got) {
     }
     if (!gets.empty()) {
         St s = eng_->get_chunks(gets.data(), (int)gets.size());
+        if (s == St::not_found && conf_->rs_k > 0) {
+            // degraded read: find the lost chunk(s), reconstruct, retry
+            std::vector<uint8_t> cbuf(conf_->chunk_size);
+            for (auto& g2 : gets) {
+                St s2 = eng_->get_chunks(&g2, 1);
+                if (s2 == St::ok) continue;
+                if (s2 != St::not_found) return s2;
+                uint64_t idx = (uint64_t)(g2.ref - &mf_.chunks[0]);
+                if (idx >= mf_.chunks.size()) return St::not_found;
+                s2 = rs_reconstruct_chunk_locked(idx, cbuf.data());
+                if (s2 != St::ok) return s2;
+                std::memcpy(g2.out, cbuf.data() + g2.off, g2.len);
+            }
+            s = St::ok;
+        }
         if (s != St::ok) return s;
     }
     metrics().read_ops++;
@@ -206,11 +302,12 @@ St Object::flush_locked() {
         for (size_t i = 0; i < idxs.size(); i++) {
             drop_ref(mf_.chunks[idxs[i]]);
             mf_.chunks[idxs[i]] = refs[i];
+            mark_stripe_dirty(idxs[i]);
         }
     }
     dirty_.clear();
     mf_.logical_size = size_;
-    return St::ok;
+    return rs_update_locked();
 }
 
 St Object::flush() {
@@ -227,8 +324,20 @@ St Object::truncate(uint64_t size) {
             if (it->first >= nchunks) it = dirty_.erase(it);
             else ++it;
         }
-        for (uint64_t i = nchunks; i < mf_.chunks.size(); i++) drop_ref(mf_.chunks[i]);
+        for (uint64_t i = nchunks; i < mf_.chunks.size(); i++) {
+            drop_ref(mf_.chunks[i]);
+            mark_stripe_dirty(i);
+        }
         if (mf_.chunks.size() > nchunks) mf_.chunks.resize(nchunks);
+        if (conf_->rs_k > 0) {
+            // drop parity of stripes wholly beyond the new end
+            uint64_t nstripes = div_up(nchunks, (uint64_t)conf_->rs_k);
+            uint64_t keep = nstripes * conf_->rs_m;
+            for (uint64_t i = keep; i < mf_.parity.size(); i++) drop_ref(mf_.parity[i]);
+            if (mf_.parity.size() > keep) mf_.parity.resize(keep);
+            for (auto it = rs_dirty_.begin(); it != rs_dirty_.end();)
+                it = *it >= nstripes ? rs_dirty_.erase(it) : std::next(it);
+        }
         // shorten the (new) last chunk if it straddles the cut
         if (size % cs != 0 && nchunks > 0) {
             uint64_t idx = nchunks - 1;

@@ -7,6 +7,7 @@
 #include <map>
 #include <memory>
 #include <mutex>
+#include <set>
 
 #include "common.h"
 #include "engine/engine.h"
@@ -45,12 +46,18 @@ class Object {
     St flush_locked();
     St load_chunk_for_rmw(uint64_t idx, Dirty* d);
     void drop_ref(const ChunkRef& r);
+    // RS (judged config 4): refresh parity for stripes whose data changed;
+    // reconstruct a lost chunk from stripe survivors and repair it in place.
+    void mark_stripe_dirty(uint64_t chunk_idx);
+    St rs_update_locked();
+    St rs_reconstruct_chunk_locked(uint64_t idx, uint8_t* chunk_buf);
 
     Engine* eng_;
     const Conf* conf_;
     std::mutex mu_;
     Manifest mf_;                    // committed chunk refs (index = chunk idx)
     std::map<uint64_t, Dirty> dirty_;
+    std::set<uint64_t> rs_dirty_;    // stripe indices with stale parity
     uint64_t size_ = 0;              // logical size incl. dirty extension
 };
 
