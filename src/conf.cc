@@ -102,6 +102,7 @@ St Conf::apply_kv(const std::string& k, const std::string& v) {
         g_log_level = (LogLevel)log_level;
         return St::ok;
     }
+    if (k == "journal") { journal = v; return St::ok; }
     if (k == "fault") { fault = v; return St::ok; }
     // standard mount options we accept and ignore (mount(8) may add them)
     if (k == "rw" || k == "ro" || k == "nosuid" || k == "nodev" || k == "noexec" ||

@@ -8,6 +8,8 @@
 
 #include "common.h"
 
+namespace edgehip { class Journal; }
+
 namespace edgehip {
 
 enum class FpAlgo : uint8_t { none = 0, blake3 = 1 };
@@ -44,6 +46,11 @@ struct Conf {
     bool writeback = true;             // FUSE writeback cache (ack from page cache)
     bool direct_io = false;            // force FOPEN_DIRECT_IO on opens
     bool kernel_cache = true;          // keep_cache on open
+
+    // durability: journal dir for namespace + manifests +
+    // chunk mirror; empty = volatile (in-memory/HBM only)
+    std::string journal;
+    Journal* journal_ptr = nullptr;    // runtime wiring (not a parsed option)
 
     // misc
     int log_level = 2;                 // info

@@ -7,6 +7,7 @@
 #include "../gf256.h"
 #include "../lz4.h"
 #include "../obs.h"
+#include "../journal.h"
 #include "../store.h"
 
 namespace edgehip {
@@ -68,6 +69,8 @@ class HostEngine : public Engine {
             }
             metrics().lz4_out_bytes += r.stored_len;
             if (s != St::ok && s != St::exists) return s;
+            if (conf_.journal_ptr && s == St::ok)
+                conf_.journal_ptr->put_chunk(r.fp, p.data, p.len);  // logical bytes
             metrics().chunks_put++;
         }
         return St::ok;
@@ -80,13 +83,21 @@ class HostEngine : public Engine {
             const ChunkRef& r = *g.ref;
             if ((uint64_t)g.off + g.len > r.logical_len) return St::inval;
             metrics().chunks_get++;
-            if (!r.compressed()) {
+            uint32_t stored = store_.stored_len(r.fp);
+            if (stored == 0) {
+                // miss: try the journal chunk mirror (remount path)
+                if (!load_from_journal(r)) return St::not_found;
+                stored = store_.stored_len(r.fp);
+                if (stored == 0) return St::not_found;
+            }
+            // the stored copy may differ from the ref (re-ingest after
+            // remount); compression is decided by stored != logical
+            if (stored == r.logical_len) {
                 St s = store_.get(r.fp, g.out, g.off, g.len);
                 if (s != St::ok) return s;
             } else {
-                // decompress only the 64 KiB blocks covering [off, off+len)
-                tmp.resize(r.stored_len);
-                St s = store_.get(r.fp, tmp.data(), 0, r.stored_len);
+                tmp.resize(stored);
+                St s = store_.get(r.fp, tmp.data(), 0, stored);
                 if (s != St::ok) return s;
                 if (g.off == 0 && g.len == r.logical_len) {
                     s = lz4_chunk_decompress(tmp.data(), tmp.size(), g.out, r.logical_len);
@@ -102,7 +113,25 @@ class HostEngine : public Engine {
         return St::ok;
     }
 
-    St release_chunk(const ChunkRef& ref) override { return store_.decref(ref.fp); }
+    // Remount: re-ingest a chunk's logical bytes from the journal mirror
+    // (content addressing reproduces the same fingerprint).
+    bool load_from_journal(const ChunkRef& r) {
+        if (!conf_.journal_ptr) return false;
+        std::vector<uint8_t> bytes;
+        if (conf_.journal_ptr->get_chunk(r.fp, &bytes) != St::ok) return false;
+        if (bytes.size() != r.logical_len) return false;
+        PutChunk p{bytes.data(), (uint32_t)bytes.size(), nullptr};
+        ChunkRef nr;
+        if (put_chunks(&p, &nr, 1) != St::ok) return false;
+        return nr.fp == r.fp;
+    }
+
+    St release_chunk(const ChunkRef& ref) override {
+        St s = store_.decref(ref.fp);
+        if (conf_.journal_ptr && s == St::ok && !store_.has(ref.fp))
+            conf_.journal_ptr->del_chunk(ref.fp);
+        return s;
+    }
 
     St debug_drop_chunk(const Fp& fp) override { return store_.remove(fp); }
 

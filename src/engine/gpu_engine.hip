@@ -17,6 +17,7 @@
 #include <vector>
 
 #include "../kernels/launchers.h"
+#include "../journal.h"
 #include "../lz4.h"
 #include "../obs.h"
 #include "../store.h"
@@ -314,6 +315,8 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
                 metrics().chunks_put++;
             }
         }
+        if (!dup && conf_.journal_ptr)
+            conf_.journal_ptr->put_chunk(r.fp, put[i].data, put[i].len);
         if (put[i].donate) put[i].donate->release();
     }
 
@@ -397,6 +400,21 @@ St GpuEngine::get_chunks(const GetChunk* get, int n) {
         bool spilled = false;
         uint64_t aoff = 0;
         uint32_t stored = 0;
+        bool found;
+        {
+            std::lock_guard<std::mutex> lk(map_mu_);
+            found = map_.count(r.fp) != 0;
+        }
+        if (!found && conf_.journal_ptr) {
+            // remount: re-ingest the chunk's logical bytes from the journal
+            std::vector<uint8_t> bytes;
+            if (conf_.journal_ptr->get_chunk(r.fp, &bytes) == St::ok &&
+                bytes.size() == r.logical_len) {
+                PutChunk p{bytes.data(), (uint32_t)bytes.size(), nullptr};
+                ChunkRef nr;
+                (void)put_chunks(&p, &nr, 1);
+            }
+        }
         {
             std::lock_guard<std::mutex> lk(map_mu_);
             auto it = map_.find(r.fp);
@@ -510,6 +528,7 @@ St GpuEngine::release_chunk(const ChunkRef& ref) {
             lru_.erase(it->second.lru);
         }
         map_.erase(it);
+        if (conf_.journal_ptr) conf_.journal_ptr->del_chunk(ref.fp);
     }
     return St::ok;
 }

@@ -406,7 +406,7 @@ void FuseMount::handle(const uint8_t* req, size_t len, std::vector<uint8_t>* rep
         }
         case FUSE_FSYNC: {
             auto* f = (const fuse_fsync_in*)arg;
-            reply_st(vfs_->flush(f->fh));
+            reply_st(vfs_->fsync(f->fh));
             return;
         }
         case FUSE_OPENDIR: {

@@ -1,0 +1,59 @@
+// Durability journal (SURVEY.md §3.1 H10, §6.4 — the storage-framework
+// checkpoint/resume): an append-only, crc-protected log of namespace
+// operations and committed manifests, plus a chunk mirror directory.
+// `edgehip mount` with journal=<dir> replays the log to rebuild the
+// filesystem; chunk bytes are mirrored to <dir>/chunks/<fp-hex> on commit
+// (crash discards only un-fsynced data — the POSIX contract).
+//
+// Record format (little-endian):
+//   u32 magic 'EHJ1' | u8 type | u8 pad[3] | u32 body_len | body | u32 crc32c
+// Types: 1 MKDIR {path}  2 FILE {u16 path_len, path, manifest}  3 UNLINK
+//        {path}  4 RENAME {u16 old_len, old, new}  5 RMDIR {path}
+#pragma once
+
+#include <mutex>
+#include <string>
+#include <vector>
+
+#include "common.h"
+
+namespace edgehip {
+
+class Journal {
+  public:
+    ~Journal();
+    // Opens (creates) <dir>/journal.log and <dir>/chunks/. Returns ok.
+    St open(const std::string& dir);
+    bool enabled() const { return fd_ >= 0; }
+    const std::string& dir() const { return dir_; }
+
+    St mkdir_rec(const std::string& path);
+    St rmdir_rec(const std::string& path);
+    St unlink_rec(const std::string& path);
+    St rename_rec(const std::string& oldp, const std::string& newp);
+    St file_rec(const std::string& path, const std::vector<uint8_t>& manifest);
+    St sync();  // fdatasync the log
+
+    // Chunk mirror: store/load/remove stored chunk bytes by fingerprint.
+    St put_chunk(const Fp& fp, const uint8_t* data, uint32_t len);
+    St get_chunk(const Fp& fp, std::vector<uint8_t>* out) const;
+    St del_chunk(const Fp& fp);
+
+    struct Replay {
+        uint8_t type;
+        std::string path, path2;
+        std::vector<uint8_t> manifest;
+    };
+    // Read every valid record (stops at the first torn/corrupt one).
+    St replay(std::vector<Replay>* out) const;
+
+  private:
+    St append(uint8_t type, const std::vector<uint8_t>& body);
+    std::string chunk_path(const Fp& fp) const;
+
+    std::string dir_;
+    int fd_ = -1;
+    std::mutex mu_;
+};
+
+}  // namespace edgehip

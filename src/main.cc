@@ -4,6 +4,7 @@
 #include <malloc.h>
 
 #include "conf.h"
+#include "journal.h"
 #include "engine/engine.h"
 #include "fuse.h"
 #include "obs.h"
@@ -52,6 +53,11 @@ int main(int argc, char** argv) {
         }
         g_log_level = (LogLevel)conf.log_level;
         LOGI("conf: %s", conf.dump().c_str());
+        Journal journal;
+        if (!conf.journal.empty()) {
+            if (journal.open(conf.journal) != St::ok) return 1;
+            conf.journal_ptr = &journal;
+        }
         std::unique_ptr<Engine> eng;
         try {
             eng = make_engine(conf);
@@ -59,7 +65,11 @@ int main(int argc, char** argv) {
             LOGE("engine init failed: %s", e.what());
             return 1;
         }
-        Vfs vfs(eng.get(), &conf);
+        Vfs vfs(eng.get(), &conf, conf.journal_ptr);
+        if (vfs.restore() != St::ok) {
+            LOGE("journal replay failed");
+            return 1;
+        }
         FuseMount fm(&vfs, &conf);
         g_mount = &fm;
         signal(SIGINT, on_signal);

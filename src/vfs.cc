@@ -6,7 +6,8 @@
 
 namespace edgehip {
 
-Vfs::Vfs(Engine* eng, const Conf* conf) : eng_(eng), conf_(conf) {
+Vfs::Vfs(Engine* eng, const Conf* conf, Journal* journal)
+    : eng_(eng), conf_(conf), journal_(journal) {
     auto root = std::make_unique<Inode>();
     root->ino = 1;
     root->mode = S_IFDIR | 0755;
@@ -20,6 +21,172 @@ Vfs::Vfs(Engine* eng, const Conf* conf) : eng_(eng), conf_(conf) {
 Inode* Vfs::get(uint64_t ino) {
     auto it = inodes_.find(ino);
     return it == inodes_.end() ? nullptr : it->second.get();
+}
+
+Inode* Vfs::get_const(uint64_t ino) const {
+    auto it = inodes_.find(ino);
+    return it == inodes_.end() ? nullptr : it->second.get();
+}
+
+std::string Vfs::path_locked(uint64_t ino) const {
+    std::vector<const std::string*> parts;
+    const Inode* in = get_const(ino);
+    int guard = 0;
+    while (in && in->ino != 1 && guard++ < 256) {
+        parts.push_back(&in->name);
+        in = get_const(in->parent);
+    }
+    std::string p;
+    for (auto it = parts.rbegin(); it != parts.rend(); ++it) {
+        p.push_back('/');
+        p += **it;
+    }
+    return p.empty() ? "/" : p;
+}
+
+St Vfs::resolve_parent_locked(const std::string& path, uint64_t* parent, std::string* leaf) {
+    uint64_t cur = 1;
+    size_t pos = 1;  // skip leading '/'
+    if (path.empty() || path[0] != '/') return St::inval;
+    while (true) {
+        size_t slash = path.find('/', pos);
+        std::string comp = path.substr(pos, slash == std::string::npos ? std::string::npos
+                                                                       : slash - pos);
+        if (slash == std::string::npos) {
+            *parent = cur;
+            *leaf = comp;
+            return St::ok;
+        }
+        Inode* p = get(cur);
+        if (!p) return St::not_found;
+        auto it = p->children.find(comp);
+        if (it == p->children.end()) return St::not_found;
+        cur = it->second;
+        pos = slash + 1;
+    }
+}
+
+St Vfs::restore() {
+    if (!journal_ || !journal_->enabled()) return St::ok;
+    std::vector<Journal::Replay> recs;
+    St s = journal_->replay(&recs);
+    if (s != St::ok) return s;
+    std::lock_guard<std::mutex> g(mu_);
+    for (auto& r : recs) {
+        uint64_t parent;
+        std::string leaf;
+        switch (r.type) {
+            case 1: {  // mkdir
+                if (resolve_parent_locked(r.path, &parent, &leaf) != St::ok) break;
+                Inode* p = get(parent);
+                if (!p || p->children.count(leaf)) break;
+                auto in = std::make_unique<Inode>();
+                in->ino = alloc_ino();
+                in->mode = S_IFDIR | 0755;
+                in->nlink = 2;
+                in->nlookup = 0;
+                in->parent = parent;
+                in->name = leaf;
+                in->atime = in->mtime = in->ctime = (int64_t)wall_ns();
+                p->children[leaf] = in->ino;
+                p->nlink++;
+                inodes_.emplace(in->ino, std::move(in));
+                break;
+            }
+            case 2: {  // file + manifest
+                if (resolve_parent_locked(r.path, &parent, &leaf) != St::ok) break;
+                Inode* p = get(parent);
+                if (!p) break;
+                Manifest m;
+                if (Manifest::decode(r.manifest.data(), r.manifest.size(), &m) != St::ok)
+                    break;
+                auto it = p->children.find(leaf);
+                Inode* f;
+                if (it == p->children.end()) {
+                    auto in = std::make_unique<Inode>();
+                    in->ino = alloc_ino();
+                    in->mode = S_IFREG | 0644;
+                    in->nlookup = 0;
+                    in->parent = parent;
+                    in->name = leaf;
+                    in->atime = in->mtime = in->ctime = (int64_t)wall_ns();
+                    in->obj = std::make_shared<Object>(eng_, conf_);
+                    f = in.get();
+                    p->children[leaf] = in->ino;
+                    inodes_.emplace(f->ino, std::move(in));
+                } else {
+                    f = get(it->second);
+                    if (!f || !f->obj) break;
+                    f->obj = std::make_shared<Object>(eng_, conf_);  // latest wins
+                }
+                f->obj->set_manifest(m);
+                break;
+            }
+            case 3: {  // unlink
+                if (resolve_parent_locked(r.path, &parent, &leaf) != St::ok) break;
+                Inode* p = get(parent);
+                if (!p) break;
+                auto it = p->children.find(leaf);
+                if (it == p->children.end()) break;
+                inodes_.erase(it->second);
+                p->children.erase(it);
+                break;
+            }
+            case 5: {  // rmdir
+                if (resolve_parent_locked(r.path, &parent, &leaf) != St::ok) break;
+                Inode* p = get(parent);
+                if (!p) break;
+                auto it = p->children.find(leaf);
+                if (it == p->children.end()) break;
+                inodes_.erase(it->second);
+                p->children.erase(it);
+                p->nlink--;
+                break;
+            }
+            case 4: {  // rename
+                uint64_t np;
+                std::string nleaf;
+                if (resolve_parent_locked(r.path, &parent, &leaf) != St::ok) break;
+                if (resolve_parent_locked(r.path2, &np, &nleaf) != St::ok) break;
+                Inode* p = get(parent);
+                Inode* q = get(np);
+                if (!p || !q) break;
+                auto it = p->children.find(leaf);
+                if (it == p->children.end()) break;
+                uint64_t ino = it->second;
+                auto dst = q->children.find(nleaf);
+                if (dst != q->children.end()) {
+                    inodes_.erase(dst->second);
+                    q->children.erase(dst);
+                }
+                p->children.erase(it);
+                q->children[nleaf] = ino;
+                Inode* moving = get(ino);
+                if (moving) {
+                    moving->parent = np;
+                    moving->name = nleaf;
+                }
+                break;
+            }
+        }
+    }
+    LOGI("vfs: journal replayed: %zu records, %zu inodes", recs.size(), inodes_.size());
+    return St::ok;
+}
+
+St Vfs::commit_and_journal(uint64_t ino, const std::shared_ptr<Object>& obj) {
+    std::vector<uint8_t> bytes;
+    St s = obj->commit(journal_ && journal_->enabled() ? &bytes : nullptr);
+    if (s != St::ok) return s;
+    if (journal_ && journal_->enabled()) {
+        std::string p;
+        {
+            std::lock_guard<std::mutex> g(mu_);
+            p = path_locked(ino);
+        }
+        journal_->file_rec(p, bytes);
+    }
+    return St::ok;
 }
 
 St Vfs::lookup(uint64_t parent, const std::string& name, Inode** out) {
@@ -72,6 +239,7 @@ St Vfs::create(uint64_t parent, const std::string& name, uint32_t mode, uint32_t
     in->gid = gid;
     in->nlookup = 1;
     in->parent = parent;
+    in->name = name;
     in->atime = in->mtime = in->ctime = (int64_t)wall_ns();
     in->obj = std::make_shared<Object>(eng_, conf_);
     Inode* raw = in.get();
@@ -97,12 +265,14 @@ St Vfs::mkdir(uint64_t parent, const std::string& name, uint32_t mode, uint32_t 
     in->nlink = 2;
     in->nlookup = 1;
     in->parent = parent;
+    in->name = name;
     in->atime = in->mtime = in->ctime = (int64_t)wall_ns();
     Inode* raw = in.get();
     p->children[name] = in->ino;
     p->nlink++;
     p->mtime = p->ctime = in->ctime;
     inodes_.emplace(raw->ino, std::move(in));
+    if (journal_ && journal_->enabled()) journal_->mkdir_rec(path_locked(raw->ino));
     *out = raw;
     return St::ok;
 }
@@ -124,6 +294,7 @@ St Vfs::unlink(uint64_t parent, const std::string& name) {
     if (!c) return St::not_found;
     if (c->is_dir()) return St::is_dir;
     LOGD("vfs: unlink %s (ino %lu)", name.c_str(), (unsigned long)c->ino);
+    if (journal_ && journal_->enabled()) journal_->unlink_rec(path_locked(c->ino));
     c->unlinked = true;
     c->nlink--;
     p->children.erase(it);
@@ -142,6 +313,7 @@ St Vfs::rmdir(uint64_t parent, const std::string& name) {
     if (!c) return St::not_found;
     if (!c->is_dir()) return St::not_dir;
     if (!c->children.empty()) return St::not_empty;
+    if (journal_ && journal_->enabled()) journal_->rmdir_rec(path_locked(c->ino));
     c->unlinked = true;
     p->children.erase(it);
     p->nlink--;
@@ -160,6 +332,12 @@ St Vfs::rename(uint64_t parent, const std::string& name, uint64_t newparent,
     if (it == p->children.end()) return St::not_found;
     uint64_t ino = it->second;
     LOGD("vfs: rename %s -> %s", name.c_str(), newname.c_str());
+    if (journal_ && journal_->enabled()) {
+        std::string oldp = path_locked(ino);
+        std::string newp = path_locked(newparent);
+        if (newp == "/") newp.clear();
+        journal_->rename_rec(oldp, newp + "/" + newname);
+    }
     auto dst = np->children.find(newname);
     if (dst != np->children.end()) {
         if (noreplace) return St::exists;
@@ -175,10 +353,13 @@ St Vfs::rename(uint64_t parent, const std::string& name, uint64_t newparent,
     Inode* moving = get(ino);
     p->children.erase(it);
     np->children[newname] = ino;
-    if (moving && moving->is_dir() && parent != newparent) {
-        p->nlink--;
-        np->nlink++;
+    if (moving) {
+        moving->name = newname;
         moving->parent = newparent;
+        if (moving->is_dir() && parent != newparent) {
+            p->nlink--;
+            np->nlink++;
+        }
     }
     int64_t t = (int64_t)wall_ns();
     p->mtime = p->ctime = t;
@@ -220,15 +401,17 @@ Inode* Vfs::handle_inode(uint64_t fh) {
 
 St Vfs::release(uint64_t fh) {
     std::shared_ptr<Object> obj;
+    uint64_t ino = 0;
     {
         std::lock_guard<std::mutex> g(mu_);
         auto it = handles_.find(fh);
         if (it == handles_.end()) return St::inval;
-        Inode* in = get(it->second);
-        if (in && in->obj) obj = in->obj;
+        ino = it->second;
+        Inode* in = get(ino);
+        if (in && in->obj && !in->unlinked) obj = in->obj;
         handles_.erase(it);
     }
-    if (obj) return obj->commit();
+    if (obj) return commit_and_journal(ino, obj);
     return St::ok;
 }
 
@@ -262,14 +445,23 @@ St Vfs::write(uint64_t fh, uint64_t off, const uint8_t* data, uint64_t len) {
 
 St Vfs::flush(uint64_t fh) {
     std::shared_ptr<Object> obj;
+    uint64_t ino = 0;
     {
         std::lock_guard<std::mutex> g(mu_);
         auto it = handles_.find(fh);
         if (it == handles_.end()) return St::inval;
-        Inode* in = get(it->second);
-        if (in && in->obj) obj = in->obj;
+        ino = it->second;
+        Inode* in = get(ino);
+        if (in && in->obj && !in->unlinked) obj = in->obj;
     }
-    if (obj) return obj->commit();
+    if (obj) return commit_and_journal(ino, obj);
+    return St::ok;
+}
+
+St Vfs::fsync(uint64_t fh) {
+    St s = flush(fh);
+    if (s != St::ok) return s;
+    if (journal_ && journal_->enabled()) return journal_->sync();
     return St::ok;
 }
 

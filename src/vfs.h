@@ -15,6 +15,7 @@
 #include "common.h"
 #include "conf.h"
 #include "engine/engine.h"
+#include "journal.h"
 #include "obj.h"
 
 namespace edgehip {
@@ -29,6 +30,7 @@ struct Inode {
     std::map<std::string, uint64_t> children; // dirs only
     std::shared_ptr<Object> obj;              // files only
     uint64_t parent = 0;
+    std::string name;                         // dentry name (journal paths)
     bool unlinked = false;
 
     bool is_dir() const { return S_ISDIR(mode); }
@@ -43,7 +45,10 @@ struct DirEntry {
 
 class Vfs {
   public:
-    Vfs(Engine* eng, const Conf* conf);
+    Vfs(Engine* eng, const Conf* conf, Journal* journal = nullptr);
+
+    // Remount: rebuild the namespace + manifests from the journal.
+    St restore();
 
     // All methods return St; outputs via pointers. ino arguments are FUSE
     // nodeids (root = 1).
@@ -68,9 +73,11 @@ class Vfs {
     St read(uint64_t fh, uint64_t off, uint64_t len, uint8_t* out, uint64_t* got);
     St write(uint64_t fh, uint64_t off, const uint8_t* data, uint64_t len);
     St flush(uint64_t fh);
+    St fsync(uint64_t fh);  // flush + journal fdatasync
 
     void forget(uint64_t ino, uint64_t n);
     Inode* get(uint64_t ino);
+    Inode* get_const(uint64_t ino) const;
     void ref(Inode* in) { in->nlookup++; }
 
     uint64_t inode_count() const { return inodes_.size(); }
@@ -80,9 +87,14 @@ class Vfs {
   private:
     uint64_t alloc_ino() { return next_ino_++; }
     void maybe_free(uint64_t ino);
+    std::string path_locked(uint64_t ino) const;
+    St commit_and_journal(uint64_t ino, const std::shared_ptr<Object>& obj);
+    St resolve_parent_locked(const std::string& path, uint64_t* parent,
+                             std::string* leaf);
 
     Engine* eng_;
     const Conf* conf_;
+    Journal* journal_;
     std::mutex mu_;
     std::unordered_map<uint64_t, std::unique_ptr<Inode>> inodes_;
     std::unordered_map<uint64_t, uint64_t> handles_;  // fh -> ino
