@@ -189,8 +189,13 @@ class Obj:
     def __exit__(self, *a):
         self.close()
 
-    def write(self, off: int, data: bytes):
-        rc = self.lib._so.eh_object_write(self.h, off, data, len(data))
+    def write(self, off: int, data):
+        if isinstance(data, memoryview):
+            n = data.nbytes
+            buf = (ctypes.c_ubyte * n).from_buffer(data.cast("B"))
+            rc = self.lib._so.eh_object_write(self.h, off, buf, n)
+        else:
+            rc = self.lib._so.eh_object_write(self.h, off, data, len(data))
         if rc != 0:
             raise OSError(-rc, "object write failed")
 

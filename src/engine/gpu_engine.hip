@@ -1,14 +1,20 @@
-// GpuEngine — the MI355X data plane (SURVEY.md §3.1 H7/H8/H9/H10).
+// GpuEngine — the MI355X data plane (SURVEY.md §3.1 H7/H8/H9/H10/H11).
 //
-// Chunks live in a per-GPU HBM arena (the chunk store IS device memory —
-// 288 GB HBM3E per GPU); ingest pipelines H2D staging, BLAKE3 fingerprints,
-// LZ4 compression and crc32c on the GPU (src/kernels/*.hip), then batch-copies
-// stored bytes into arena slots. Reads decompress on-device and stage D2H
-// through pinned buffers. Cold chunks spill to a host RAM tier when the arena
-// fills (config 4's 1 TiB object exceeds one GPU's HBM).
+// Chunks live in per-GPU HBM arenas (the chunk store IS device memory —
+// 288 GB HBM3E per GPU). With ngpus > 1 the engine is the single-node
+// "storage cluster": chunks place across devices by fingerprint (EP-style
+// sharding), ingest pipelines run on a rotating device and
+// ship stored bytes to the owner arena over xGMI (hipMemcpyPeerAsync = the
+// C1/C2 p2p stripe path; each GPU has 7 direct links — SURVEY.md §3.3
+// prefers fan-out p2p over ring collectives for bulk data), and RS parity
+// is computed on-device from peer-gathered shards (G6 kernel).
 //
-// Fails LOUDLY when no GPU is present — no silent CPU fallback on the GPU
-// path (a GPU test passing on eager fallback is a failure, not a feature).
+// Ingest pipeline per batch: H2D staging -> BLAKE3 (G3) -> LZ4 (G4) ->
+// crc32c (G8) -> batch copy (G9) / peer copy into the owner arena. Reads
+// decompress on the owner device (G5) and stage D2H through pinned buffers.
+// Cold chunks spill to a host RAM tier (config 4's 1 TiB > HBM).
+//
+// Fails LOUDLY when no GPU is present — no silent CPU fallback.
 #include <algorithm>
 #include <list>
 #include <mutex>
@@ -16,8 +22,9 @@
 #include <unordered_map>
 #include <vector>
 
-#include "../kernels/launchers.h"
+#include "../gf256.h"
 #include "../journal.h"
+#include "../kernels/launchers.h"
 #include "../lz4.h"
 #include "../obs.h"
 #include "../store.h"
@@ -34,23 +41,73 @@ constexpr uint32_t kSegStride = 4352;
 constexpr uint32_t kMaxSegs = kMaxBatch * 256;  // 1 MiB chunks -> 256 segs each
 
 // pinned metadata buffer layout (256 KiB)
-constexpr uint64_t kOffChunks = 0;                 // DevChunk[kMaxBatch]
-constexpr uint64_t kOffChunks2 = 4 * KiB;          // DevChunk[kMaxBatch]
-constexpr uint64_t kOffSegMeta = 8 * KiB;          // ulong2[kMaxSegs] = 128 KiB
-constexpr uint64_t kOffFps = 140 * KiB;            // 32 B * kMaxBatch
-constexpr uint64_t kOffSizes = 144 * KiB;          // u32 * kMaxBatch
-constexpr uint64_t kOffCrcs = 148 * KiB;           // u32 * kMaxBatch
-constexpr uint64_t kOffCopy = 152 * KiB;           // DevChunk[kMaxBatch]
-constexpr uint64_t kOffStatus = 156 * KiB;         // u32 * kMaxBatch
+constexpr uint64_t kOffChunks = 0;
+constexpr uint64_t kOffChunks2 = 4 * KiB;
+constexpr uint64_t kOffSegMeta = 8 * KiB;   // 128 KiB
+constexpr uint64_t kOffFps = 140 * KiB;
+constexpr uint64_t kOffSizes = 144 * KiB;
+constexpr uint64_t kOffCrcs = 148 * KiB;
+constexpr uint64_t kOffCopy = 152 * KiB;
+constexpr uint64_t kOffStatus = 156 * KiB;
 constexpr uint64_t kPinMetaSz = 256 * KiB;
 
 struct ArenaLoc {
     uint64_t off = 0;
-    uint32_t len = 0;  // stored bytes
+    uint32_t len = 0;   // stored bytes
     uint32_t crc = 0;
     uint32_t refs = 1;
+    int16_t dev = 0;    // owning device slot
     bool spilled = false;
+    bool compressed = false;
     std::list<Fp>::iterator lru;  // valid when !spilled
+};
+
+// Per-device context: arena + allocator + streams + staging.
+struct DevCtx {
+    int hip_dev = 0;
+    hipStream_t s_pipe{}, s_read{};
+    uint8_t* arena = nullptr;
+    uint64_t arena_sz = 0;
+    std::mutex alloc_mu;
+    std::vector<std::vector<uint64_t>> free_cls;
+    uint64_t bump = 0;
+    std::atomic<uint64_t> used{0};
+
+    uint8_t *d_stage = nullptr, *d_comp = nullptr, *d_rdstage = nullptr,
+            *d_segscratch = nullptr, *d_fps = nullptr, *d_rs = nullptr;
+    DevChunk *d_chunks = nullptr, *d_copy = nullptr;
+    ulong2* d_segmeta = nullptr;
+    uint32_t *d_segsizes = nullptr, *d_sizes = nullptr, *d_crcs = nullptr;
+    uint8_t *pin_in = nullptr, *pin_out = nullptr, *pin_meta = nullptr;
+    uint64_t stage_sz = 0;
+    std::mutex pipe_mu;  // staging buffers + streams
+    std::mutex rs_mu;    // d_rs stripe buffer
+
+    bool arena_alloc(uint32_t bytes, uint64_t* off) {
+        int cls = (int)((bytes + kGranule - 1) / kGranule) - 1;
+        std::lock_guard<std::mutex> g(alloc_mu);
+        if (cls < (int)free_cls.size() && !free_cls[cls].empty()) {
+            *off = free_cls[cls].back();
+            free_cls[cls].pop_back();
+            used += (uint64_t)(cls + 1) * kGranule;
+            return true;
+        }
+        uint64_t need = (uint64_t)(cls + 1) * kGranule;
+        if (bump + need <= arena_sz) {
+            *off = bump;
+            bump += need;
+            used += need;
+            return true;
+        }
+        return false;
+    }
+    void arena_free(uint64_t off, uint32_t bytes) {
+        int cls = (int)((bytes + kGranule - 1) / kGranule) - 1;
+        std::lock_guard<std::mutex> g(alloc_mu);
+        if ((int)free_cls.size() <= cls) free_cls.resize(cls + 1);
+        free_cls[cls].push_back(off);
+        used -= (uint64_t)(cls + 1) * kGranule;
+    }
 };
 
 class GpuEngine : public Engine {
@@ -58,55 +115,71 @@ class GpuEngine : public Engine {
     explicit GpuEngine(const Conf& conf) : conf_(conf), spill_(conf.spill_bytes) {
         if (conf.chunk_size > 1 * MiB)
             throw std::runtime_error("GpuEngine: chunk_size > 1 MiB not supported");
-        int ndev = 0;
-        hipError_t e = hipGetDeviceCount(&ndev);
-        if (e != hipSuccess || ndev <= conf.gpu_first)
+        int nvis = 0;
+        hipError_t e = hipGetDeviceCount(&nvis);
+        if (e != hipSuccess || nvis <= conf.gpu_first)
             throw std::runtime_error("GpuEngine: no MI355X visible (hipGetDeviceCount: " +
                                      std::string(hipGetErrorString(e)) + ")");
-        dev_ = conf.gpu_first;
-        HIP_CHECK(hipSetDevice(dev_));
-        HIP_CHECK(hipStreamCreateWithFlags(&s_pipe_, hipStreamNonBlocking));
-        HIP_CHECK(hipStreamCreateWithFlags(&s_read_, hipStreamNonBlocking));
-
-        arena_sz_ = conf.arena_bytes;
-        HIP_CHECK(hipMalloc(&arena_, arena_sz_));
-
+        int ndev = std::min(conf.ngpus, nvis - conf.gpu_first);
+        if (ndev < conf.ngpus)
+            LOGW("GpuEngine: requested %d GPUs, only %d visible — using %d", conf.ngpus,
+                 nvis - conf.gpu_first, ndev);
+        devs_ = std::vector<DevCtx>(ndev);
         const uint64_t cs = conf.chunk_size;
-        stage_sz_ = (uint64_t)kMaxBatch * cs;
-        HIP_CHECK(hipMalloc(&d_stage_, stage_sz_));
-        HIP_CHECK(hipMalloc(&d_comp_, stage_sz_));
-        HIP_CHECK(hipMalloc(&d_rdstage_, stage_sz_));
-        HIP_CHECK(hipMalloc(&d_segscratch_, (uint64_t)kMaxSegs * kSegStride));
-        HIP_CHECK(hipMalloc(&d_chunks_, sizeof(DevChunk) * kMaxBatch * 2));
-        HIP_CHECK(hipMalloc(&d_segmeta_, sizeof(ulong2) * kMaxSegs));
-        HIP_CHECK(hipMalloc(&d_segsizes_, sizeof(uint32_t) * kMaxSegs));
-        HIP_CHECK(hipMalloc(&d_sizes_, sizeof(uint32_t) * kMaxBatch));
-        HIP_CHECK(hipMalloc(&d_fps_, 32 * kMaxBatch));
-        HIP_CHECK(hipMalloc(&d_crcs_, sizeof(uint32_t) * kMaxBatch));
-        HIP_CHECK(hipMalloc(&d_copy_, sizeof(DevChunk) * kMaxBatch));
-
-        HIP_CHECK(hipHostMalloc(&pin_in_, stage_sz_, hipHostMallocDefault));
-        HIP_CHECK(hipHostMalloc(&pin_out_, stage_sz_, hipHostMallocDefault));
-        HIP_CHECK(hipHostMalloc(&pin_meta_, kPinMetaSz, hipHostMallocDefault));
-
-        LOGI("GpuEngine: dev %d arena %.1f GiB, staging %.0f MiB x3, spill cap %.1f GiB",
-             dev_, arena_sz_ / (double)GiB, stage_sz_ / (double)MiB,
+        for (int d = 0; d < ndev; d++) {
+            DevCtx& D = devs_[d];
+            D.hip_dev = conf.gpu_first + d;
+            HIP_CHECK(hipSetDevice(D.hip_dev));
+            // xGMI p2p: enable access to every other device (7 links/GPU)
+            for (int o = 0; o < ndev; o++) {
+                if (o == d) continue;
+                hipError_t pe = hipDeviceEnablePeerAccess(conf.gpu_first + o, 0);
+                if (pe != hipSuccess && pe != hipErrorPeerAccessAlreadyEnabled)
+                    LOGW("GpuEngine: peer access %d->%d: %s", D.hip_dev,
+                         conf.gpu_first + o, hipGetErrorString(pe));
+            }
+            HIP_CHECK(hipStreamCreateWithFlags(&D.s_pipe, hipStreamNonBlocking));
+            HIP_CHECK(hipStreamCreateWithFlags(&D.s_read, hipStreamNonBlocking));
+            D.arena_sz = conf.arena_bytes;
+            HIP_CHECK(hipMalloc(&D.arena, D.arena_sz));
+            D.stage_sz = (uint64_t)kMaxBatch * cs;
+            HIP_CHECK(hipMalloc(&D.d_stage, D.stage_sz));
+            HIP_CHECK(hipMalloc(&D.d_comp, D.stage_sz));
+            HIP_CHECK(hipMalloc(&D.d_rdstage, D.stage_sz));
+            HIP_CHECK(hipMalloc(&D.d_rs, (uint64_t)(kRsMaxK + kRsMaxM) * cs));
+            HIP_CHECK(hipMalloc(&D.d_segscratch, (uint64_t)kMaxSegs * kSegStride));
+            HIP_CHECK(hipMalloc(&D.d_chunks, sizeof(DevChunk) * kMaxBatch * 2));
+            HIP_CHECK(hipMalloc(&D.d_segmeta, sizeof(ulong2) * kMaxSegs));
+            HIP_CHECK(hipMalloc(&D.d_segsizes, sizeof(uint32_t) * kMaxSegs));
+            HIP_CHECK(hipMalloc(&D.d_sizes, sizeof(uint32_t) * kMaxBatch));
+            HIP_CHECK(hipMalloc(&D.d_fps, 32 * kMaxBatch));
+            HIP_CHECK(hipMalloc(&D.d_crcs, sizeof(uint32_t) * kMaxBatch));
+            HIP_CHECK(hipMalloc(&D.d_copy, sizeof(DevChunk) * kMaxBatch));
+            HIP_CHECK(hipHostMalloc(&D.pin_in, D.stage_sz, hipHostMallocDefault));
+            HIP_CHECK(hipHostMalloc(&D.pin_out, D.stage_sz, hipHostMallocDefault));
+            HIP_CHECK(hipHostMalloc(&D.pin_meta, kPinMetaSz, hipHostMallocDefault));
+        }
+        LOGI("GpuEngine: %d dev (first %d), arena %.1f GiB/GPU, spill cap %.1f GiB",
+             ndev, conf.gpu_first, conf.arena_bytes / (double)GiB,
              conf.spill_bytes / (double)GiB);
     }
 
     ~GpuEngine() override {
-        (void)hipSetDevice(dev_);
-        (void)hipDeviceSynchronize();
-        for (void* p : {(void*)arena_, (void*)d_stage_, (void*)d_comp_, (void*)d_rdstage_,
-                        (void*)d_segscratch_, (void*)d_chunks_, (void*)d_segmeta_,
-                        (void*)d_segsizes_, (void*)d_sizes_, (void*)d_fps_, (void*)d_crcs_,
-                        (void*)d_copy_})
-            (void)hipFree(p);
-        (void)hipHostFree(pin_in_);
-        (void)hipHostFree(pin_out_);
-        (void)hipHostFree(pin_meta_);
-        (void)hipStreamDestroy(s_pipe_);
-        (void)hipStreamDestroy(s_read_);
+        for (auto& D : devs_) {
+            (void)hipSetDevice(D.hip_dev);
+            (void)hipDeviceSynchronize();
+            for (void* p : {(void*)D.arena, (void*)D.d_stage, (void*)D.d_comp,
+                            (void*)D.d_rdstage, (void*)D.d_rs, (void*)D.d_segscratch,
+                            (void*)D.d_chunks, (void*)D.d_segmeta, (void*)D.d_segsizes,
+                            (void*)D.d_sizes, (void*)D.d_fps, (void*)D.d_crcs,
+                            (void*)D.d_copy})
+                (void)hipFree(p);
+            (void)hipHostFree(D.pin_in);
+            (void)hipHostFree(D.pin_out);
+            (void)hipHostFree(D.pin_meta);
+            (void)hipStreamDestroy(D.s_pipe);
+            (void)hipStreamDestroy(D.s_read);
+        }
     }
 
     St put_chunks(const PutChunk* put, ChunkRef* refs, int n) override {
@@ -119,6 +192,8 @@ class GpuEngine : public Engine {
 
     St get_chunks(const GetChunk* get, int n) override;
     St release_chunk(const ChunkRef& ref) override;
+    St rs_encode_stripe(const ChunkRef* data, int k, int m, uint32_t shard_len,
+                        ChunkRef* parity_out) override;
 
     St debug_drop_chunk(const Fp& fp) override {
         std::lock_guard<std::mutex> g(map_mu_);
@@ -127,14 +202,18 @@ class GpuEngine : public Engine {
         if (it->second.spilled) {
             spill_.remove(fp);
         } else {
-            arena_free(it->second.off, it->second.len);
+            devs_[it->second.dev].arena_free(it->second.off, it->second.len);
             lru_.erase(it->second.lru);
         }
         map_.erase(it);
         return St::ok;
     }
 
-    uint64_t stored_bytes() const override { return arena_used_.load() + spill_.bytes(); }
+    uint64_t stored_bytes() const override {
+        uint64_t b = spill_.bytes();
+        for (auto& D : devs_) b += D.used.load();
+        return b;
+    }
     uint64_t stored_chunks() const override {
         std::lock_guard<std::mutex> g(map_mu_);
         return map_.size();
@@ -142,65 +221,21 @@ class GpuEngine : public Engine {
     bool is_gpu() const override { return true; }
 
   private:
-    int size_class(uint32_t bytes) const {
-        return (int)((bytes + kGranule - 1) / kGranule) - 1;
-    }
+    // chunk placement: fingerprint byte spreads chunks across devices
+    // (EP-style sharding; dedup-consistent because fp is content-derived)
+    int owner_of(const Fp& fp) const { return fp.b[4] % (int)devs_.size(); }
 
-    bool arena_alloc(uint32_t bytes, uint64_t* off) {
-        int cls = size_class(bytes);
-        std::lock_guard<std::mutex> g(alloc_mu_);
-        if (cls < (int)free_.size() && !free_[cls].empty()) {
-            *off = free_[cls].back();
-            free_[cls].pop_back();
-            arena_used_ += (uint64_t)(cls + 1) * kGranule;
-            return true;
-        }
-        uint64_t need = (uint64_t)(cls + 1) * kGranule;
-        if (bump_ + need <= arena_sz_) {
-            *off = bump_;
-            bump_ += need;
-            arena_used_ += need;
-            return true;
-        }
-        return false;
-    }
-
-    void arena_free(uint64_t off, uint32_t bytes) {
-        int cls = size_class(bytes);
-        std::lock_guard<std::mutex> g(alloc_mu_);
-        if ((int)free_.size() <= cls) free_.resize(cls + 1);
-        free_[cls].push_back(off);
-        arena_used_ -= (uint64_t)(cls + 1) * kGranule;
-    }
-
-    St evict_for(uint32_t need_bytes);  // map_mu_ held
+    St evict_for(int dev, uint32_t need_bytes);  // map_mu_ held
     St flush_batch(const PutChunk* put, ChunkRef* refs, int n);
+    St gather_chunk_to_dev(const ChunkRef& ref, int ddev, uint8_t* dst);
 
     Conf conf_;
-    int dev_ = 0;
-    hipStream_t s_pipe_{}, s_read_{};
-
-    uint8_t* arena_ = nullptr;
-    uint64_t arena_sz_ = 0;
-    std::mutex alloc_mu_;
-    std::vector<std::vector<uint64_t>> free_;
-    uint64_t bump_ = 0;
-    std::atomic<uint64_t> arena_used_{0};
-
-    uint8_t *d_stage_ = nullptr, *d_comp_ = nullptr, *d_rdstage_ = nullptr,
-            *d_segscratch_ = nullptr, *d_fps_ = nullptr;
-    DevChunk *d_chunks_ = nullptr, *d_copy_ = nullptr;
-    ulong2* d_segmeta_ = nullptr;
-    uint32_t *d_segsizes_ = nullptr, *d_sizes_ = nullptr, *d_crcs_ = nullptr;
-    uint64_t stage_sz_ = 0;
-
-    uint8_t *pin_in_ = nullptr, *pin_out_ = nullptr, *pin_meta_ = nullptr;
+    std::vector<DevCtx> devs_;
+    std::atomic<uint32_t> rr_{0};  // ingest/compute device rotation
 
     mutable std::mutex map_mu_;
     std::unordered_map<Fp, ArenaLoc, FpHash> map_;
     std::list<Fp> lru_;  // front = hottest
-
-    std::mutex pipe_mu_;  // staging buffers + streams
 
     HostStore spill_;
 };
@@ -209,16 +244,19 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
     const uint32_t cs = (uint32_t)conf_.chunk_size;
     const bool do_lz4 = conf_.compress == Compress::lz4;
     const bool do_fp = conf_.fp == FpAlgo::blake3;
+    const int ingest_dev = (int)(rr_.fetch_add(1) % devs_.size());
+    DevCtx& D = devs_[ingest_dev];
 
-    std::lock_guard<std::mutex> pg(pipe_mu_);
+    std::lock_guard<std::mutex> pg(D.pipe_mu);
+    HIP_CHECK(hipSetDevice(D.hip_dev));
 
-    auto* h_chunks = (DevChunk*)(pin_meta_ + kOffChunks);
-    auto* h_chunks2 = (DevChunk*)(pin_meta_ + kOffChunks2);
-    auto* h_segmeta = (ulong2*)(pin_meta_ + kOffSegMeta);
+    auto* h_chunks = (DevChunk*)(D.pin_meta + kOffChunks);
+    auto* h_chunks2 = (DevChunk*)(D.pin_meta + kOffChunks2);
+    auto* h_segmeta = (ulong2*)(D.pin_meta + kOffSegMeta);
     int nsegs = 0;
     uint64_t off = 0;
     for (int i = 0; i < n; i++) {
-        std::memcpy(pin_in_ + off, put[i].data, put[i].len);
+        std::memcpy(D.pin_in + off, put[i].data, put[i].len);
         h_chunks[i] = DevChunk{off, off, put[i].len, (uint32_t)nsegs};
         h_chunks2[i] = DevChunk{off, (uint64_t)i * cs, put[i].len, (uint32_t)nsegs};
         uint32_t ns = (put[i].len + kSegSz - 1) / kSegSz;
@@ -228,38 +266,39 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
         nsegs += ns;
         off += round_up(put[i].len, 64);
     }
-    HIP_CHECK(hipMemcpyAsync(d_stage_, pin_in_, off, hipMemcpyHostToDevice, s_pipe_));
-    HIP_CHECK(hipMemcpyAsync(d_chunks_, h_chunks, sizeof(DevChunk) * n, hipMemcpyHostToDevice,
-                             s_pipe_));
+    HIP_CHECK(hipMemcpyAsync(D.d_stage, D.pin_in, off, hipMemcpyHostToDevice, D.s_pipe));
+    HIP_CHECK(hipMemcpyAsync(D.d_chunks, h_chunks, sizeof(DevChunk) * n,
+                             hipMemcpyHostToDevice, D.s_pipe));
     metrics().h2d_bytes += off;
 
-    if (do_fp) launch_blake3_chunks(d_stage_, d_chunks_, d_fps_, n, s_pipe_);
+    if (do_fp) launch_blake3_chunks(D.d_stage, D.d_chunks, D.d_fps, n, D.s_pipe);
     if (do_lz4) {
-        HIP_CHECK(hipMemcpyAsync(d_segmeta_, h_segmeta, sizeof(ulong2) * nsegs,
-                                 hipMemcpyHostToDevice, s_pipe_));
-        launch_lz4_compress_segs(d_stage_, d_segmeta_, d_segscratch_, d_segsizes_, nsegs,
-                                 s_pipe_);
-        HIP_CHECK(hipMemcpyAsync(d_chunks_ + kMaxBatch, h_chunks2, sizeof(DevChunk) * n,
-                                 hipMemcpyHostToDevice, s_pipe_));
-        launch_lz4_emit(d_stage_, d_segscratch_, d_segsizes_, d_chunks_ + kMaxBatch, d_comp_,
-                        d_sizes_, n, s_pipe_);
+        HIP_CHECK(hipMemcpyAsync(D.d_segmeta, h_segmeta, sizeof(ulong2) * nsegs,
+                                 hipMemcpyHostToDevice, D.s_pipe));
+        launch_lz4_compress_segs(D.d_stage, D.d_segmeta, D.d_segscratch, D.d_segsizes, nsegs,
+                                 D.s_pipe);
+        HIP_CHECK(hipMemcpyAsync(D.d_chunks + kMaxBatch, h_chunks2, sizeof(DevChunk) * n,
+                                 hipMemcpyHostToDevice, D.s_pipe));
+        launch_lz4_emit(D.d_stage, D.d_segscratch, D.d_segsizes, D.d_chunks + kMaxBatch,
+                        D.d_comp, D.d_sizes, n, D.s_pipe);
     }
 
-    auto* h_fps = (uint8_t*)(pin_meta_ + kOffFps);
-    auto* h_sizes = (uint32_t*)(pin_meta_ + kOffSizes);
-    auto* h_crcs = (uint32_t*)(pin_meta_ + kOffCrcs);
+    auto* h_fps = (uint8_t*)(D.pin_meta + kOffFps);
+    auto* h_sizes = (uint32_t*)(D.pin_meta + kOffSizes);
+    auto* h_crcs = (uint32_t*)(D.pin_meta + kOffCrcs);
     if (do_fp)
-        HIP_CHECK(hipMemcpyAsync(h_fps, d_fps_, 32 * n, hipMemcpyDeviceToHost, s_pipe_));
+        HIP_CHECK(hipMemcpyAsync(h_fps, D.d_fps, 32 * n, hipMemcpyDeviceToHost, D.s_pipe));
     if (do_lz4)
-        HIP_CHECK(hipMemcpyAsync(h_sizes, d_sizes_, 4 * n, hipMemcpyDeviceToHost, s_pipe_));
-    HIP_CHECK(hipStreamSynchronize(s_pipe_));
+        HIP_CHECK(hipMemcpyAsync(h_sizes, D.d_sizes, 4 * n, hipMemcpyDeviceToHost, D.s_pipe));
+    HIP_CHECK(hipStreamSynchronize(D.s_pipe));
 
     struct Pend {
         int i;
-        uint64_t src_off;
+        uint64_t src_off;   // in D.d_stage (raw) or D.d_comp (framed)
         bool from_comp;
         uint32_t stored;
         uint64_t arena_off;
+        int dev;            // owner device
     };
     std::vector<Pend> pend;
     for (int i = 0; i < n; i++) {
@@ -291,27 +330,30 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
                     it->second.refs++;
                     r.stored_len = it->second.len;
                     r.crc = it->second.crc;
-                    r.flags = (it->second.len != r.logical_len) ? ChunkRef::kFlagLz4 : 0;
+                    r.flags = it->second.compressed ? ChunkRef::kFlagLz4 : 0;
                     metrics().dedup_hits++;
                     metrics().dedup_bytes_saved += put[i].len;
                     dup = true;
                 }
             }
             if (!dup) {
+                int odev = owner_of(r.fp);
                 uint64_t aoff;
-                if (!arena_alloc(stored, &aoff)) {
-                    St s = evict_for(stored);
+                if (!devs_[odev].arena_alloc(stored, &aoff)) {
+                    St s = evict_for(odev, stored);
                     if (s != St::ok) return s;
-                    if (!arena_alloc(stored, &aoff)) return St::no_space;
+                    if (!devs_[odev].arena_alloc(stored, &aoff)) return St::no_space;
                 }
                 ArenaLoc loc;
                 loc.off = aoff;
                 loc.len = stored;
+                loc.dev = (int16_t)odev;
+                loc.compressed = compressed;
                 lru_.push_front(r.fp);
                 loc.lru = lru_.begin();
                 map_.emplace(r.fp, loc);
                 pend.push_back({i, compressed ? (uint64_t)i * cs : h_chunks[i].src_off,
-                                compressed, stored, aoff});
+                                compressed, stored, aoff, odev});
                 metrics().chunks_put++;
             }
         }
@@ -321,61 +363,83 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
     }
 
     if (!pend.empty()) {
-        auto* h_copy = (DevChunk*)(pin_meta_ + kOffCopy);
+        // crc over stored bytes first (from the ingest device's staging copy)
+        if (conf_.crc) {
+            auto* h_cd = (DevChunk*)(D.pin_meta + kOffCopy);
+            for (int pass = 0; pass < 2; pass++) {
+                int m = 0;
+                std::vector<int> idxmap;
+                for (size_t j = 0; j < pend.size(); j++) {
+                    if ((int)pend[j].from_comp != pass) continue;
+                    h_cd[m] = DevChunk{pend[j].src_off, 0, pend[j].stored, 0};
+                    idxmap.push_back((int)j);
+                    m++;
+                }
+                if (!m) continue;
+                HIP_CHECK(hipMemcpyAsync(D.d_copy, h_cd, sizeof(DevChunk) * m,
+                                         hipMemcpyHostToDevice, D.s_pipe));
+                launch_crc32c_chunks(pass ? D.d_comp : D.d_stage, D.d_copy, D.d_crcs, m,
+                                     D.s_pipe);
+                HIP_CHECK(hipMemcpyAsync(h_crcs, D.d_crcs, 4 * m, hipMemcpyDeviceToHost,
+                                         D.s_pipe));
+                HIP_CHECK(hipStreamSynchronize(D.s_pipe));
+                std::lock_guard<std::mutex> g(map_mu_);
+                for (int x = 0; x < m; x++) {
+                    int j = idxmap[x];
+                    refs[pend[j].i].crc = h_crcs[x];
+                    auto it = map_.find(refs[pend[j].i].fp);
+                    if (it != map_.end()) it->second.crc = h_crcs[x];
+                }
+            }
+        }
+        // local-owner chunks: batch copy kernel; remote owners: xGMI peer copy
+        auto* h_copy = (DevChunk*)(D.pin_meta + kOffCopy);
         for (int pass = 0; pass < 2; pass++) {
             int m = 0;
             for (auto& p : pend) {
-                if ((int)p.from_comp != pass) continue;
+                if ((int)p.from_comp != pass || p.dev != ingest_dev) continue;
                 h_copy[m++] = DevChunk{p.src_off, p.arena_off, p.stored, 0};
             }
             if (!m) continue;
-            HIP_CHECK(hipMemcpyAsync(d_copy_, h_copy, sizeof(DevChunk) * m,
-                                     hipMemcpyHostToDevice, s_pipe_));
-            launch_batch_copy(pass ? d_comp_ : d_stage_, arena_, d_copy_, m, s_pipe_);
-            HIP_CHECK(hipStreamSynchronize(s_pipe_));
+            HIP_CHECK(hipMemcpyAsync(D.d_copy, h_copy, sizeof(DevChunk) * m,
+                                     hipMemcpyHostToDevice, D.s_pipe));
+            launch_batch_copy(pass ? D.d_comp : D.d_stage, D.arena, D.d_copy, m, D.s_pipe);
         }
-        if (conf_.crc) {
-            for (size_t j = 0; j < pend.size(); j++)
-                h_copy[j] = DevChunk{pend[j].arena_off, 0, pend[j].stored, 0};
-            HIP_CHECK(hipMemcpyAsync(d_copy_, h_copy, sizeof(DevChunk) * pend.size(),
-                                     hipMemcpyHostToDevice, s_pipe_));
-            launch_crc32c_chunks(arena_, d_copy_, d_crcs_, (int)pend.size(), s_pipe_);
-            HIP_CHECK(hipMemcpyAsync(h_crcs, d_crcs_, 4 * pend.size(), hipMemcpyDeviceToHost,
-                                     s_pipe_));
-            HIP_CHECK(hipStreamSynchronize(s_pipe_));
-            std::lock_guard<std::mutex> g(map_mu_);
-            for (size_t j = 0; j < pend.size(); j++) {
-                refs[pend[j].i].crc = h_crcs[j];
-                auto it = map_.find(refs[pend[j].i].fp);
-                if (it != map_.end()) it->second.crc = h_crcs[j];
-            }
+        for (auto& p : pend) {
+            if (p.dev == ingest_dev) continue;
+            DevCtx& O = devs_[p.dev];
+            HIP_CHECK(hipMemcpyPeerAsync(O.arena + p.arena_off, O.hip_dev,
+                                         (p.from_comp ? D.d_comp : D.d_stage) + p.src_off,
+                                         D.hip_dev, p.stored, D.s_pipe));
         }
+        HIP_CHECK(hipStreamSynchronize(D.s_pipe));
     }
     return St::ok;
 }
 
-St GpuEngine::evict_for(uint32_t need_bytes) {
+St GpuEngine::evict_for(int dev, uint32_t need_bytes) {
     if (conf_.spill_bytes == 0) return St::no_space;
     uint64_t freed = 0;
-    while (freed < (uint64_t)need_bytes * 4 && !lru_.empty()) {
-        Fp victim = lru_.back();
+    auto it2 = lru_.end();
+    while (freed < (uint64_t)need_bytes * 4 && it2 != lru_.begin()) {
+        --it2;
+        Fp victim = *it2;
         auto it = map_.find(victim);
-        if (it == map_.end() || it->second.spilled) {
-            lru_.pop_back();
-            continue;
-        }
+        if (it == map_.end() || it->second.spilled || it->second.dev != dev) continue;
         ArenaLoc& loc = it->second;
+        DevCtx& O = devs_[dev];
         Blob b = BufPool::inst().get(loc.len);
         b.len = loc.len;
-        HIP_CHECK(hipMemcpyAsync(b.p, arena_ + loc.off, loc.len, hipMemcpyDeviceToHost,
-                                 s_read_));
-        HIP_CHECK(hipStreamSynchronize(s_read_));
+        HIP_CHECK(hipSetDevice(O.hip_dev));
+        HIP_CHECK(hipMemcpyAsync(b.p, O.arena + loc.off, loc.len, hipMemcpyDeviceToHost,
+                                 O.s_read));
+        HIP_CHECK(hipStreamSynchronize(O.s_read));
         St s = spill_.put(victim, std::move(b));
         if (s != St::ok && s != St::exists) return s;
-        arena_free(loc.off, loc.len);
+        O.arena_free(loc.off, loc.len);
         freed += round_up(loc.len, kGranule);
         loc.spilled = true;
-        lru_.pop_back();
+        it2 = lru_.erase(loc.lru);  // returns next; loop steps back from it
         metrics().spilled_chunks++;
         metrics().spill_bytes += loc.len;
     }
@@ -389,24 +453,21 @@ St GpuEngine::get_chunks(const GetChunk* get, int n) {
         uint64_t arena_off;
         uint32_t stored;
         bool compressed;
+        int dev;
     };
-    std::vector<DevReq> dev;
-    dev.reserve(n);
+    std::vector<DevReq> dev_reqs;
+    dev_reqs.reserve(n);
     for (int i = 0; i < n; i++) {
         const GetChunk& g = get[i];
         const ChunkRef& r = *g.ref;
         if ((uint64_t)g.off + g.len > r.logical_len) return St::inval;
         metrics().chunks_get++;
-        bool spilled = false;
-        uint64_t aoff = 0;
-        uint32_t stored = 0;
         bool found;
         {
             std::lock_guard<std::mutex> lk(map_mu_);
             found = map_.count(r.fp) != 0;
         }
         if (!found && conf_.journal_ptr) {
-            // remount: re-ingest the chunk's logical bytes from the journal
             std::vector<uint8_t> bytes;
             if (conf_.journal_ptr->get_chunk(r.fp, &bytes) == St::ok &&
                 bytes.size() == r.logical_len) {
@@ -415,6 +476,11 @@ St GpuEngine::get_chunks(const GetChunk* get, int n) {
                 (void)put_chunks(&p, &nr, 1);
             }
         }
+        bool spilled = false;
+        uint64_t aoff = 0;
+        uint32_t stored = 0;
+        bool comp = false;
+        int odev = 0;
         {
             std::lock_guard<std::mutex> lk(map_mu_);
             auto it = map_.find(r.fp);
@@ -423,6 +489,8 @@ St GpuEngine::get_chunks(const GetChunk* get, int n) {
             spilled = loc.spilled;
             aoff = loc.off;
             stored = loc.len;
+            comp = loc.compressed;
+            odev = loc.dev;
             if (!spilled) {
                 metrics().cache_hits++;
                 lru_.erase(loc.lru);
@@ -433,7 +501,7 @@ St GpuEngine::get_chunks(const GetChunk* get, int n) {
             }
         }
         if (spilled) {
-            if (!r.compressed()) {
+            if (!comp) {
                 St s = spill_.get(r.fp, g.out, g.off, g.len);
                 if (s != St::ok) return s;
             } else {
@@ -445,74 +513,197 @@ St GpuEngine::get_chunks(const GetChunk* get, int n) {
                 std::memcpy(g.out, full.data() + g.off, g.len);
             }
         } else {
-            dev.push_back({&g, aoff, stored, r.compressed()});
+            dev_reqs.push_back({&g, aoff, stored, comp, odev});
         }
     }
-    if (dev.empty()) return St::ok;
+    if (dev_reqs.empty()) return St::ok;
 
-    std::lock_guard<std::mutex> pg(pipe_mu_);
-    auto* h_dc = (DevChunk*)(pin_meta_ + kOffChunks);
-    auto* h_status = (uint32_t*)(pin_meta_ + kOffStatus);
+    for (size_t d = 0; d < devs_.size(); d++) {
+        DevCtx& D = devs_[d];
+        std::vector<DevReq*> mine;
+        for (auto& q : dev_reqs)
+            if (q.dev == (int)d) mine.push_back(&q);
+        if (mine.empty()) continue;
+        std::lock_guard<std::mutex> pg(D.pipe_mu);
+        HIP_CHECK(hipSetDevice(D.hip_dev));
+        auto* h_dc = (DevChunk*)(D.pin_meta + kOffChunks);
+        auto* h_status = (uint32_t*)(D.pin_meta + kOffStatus);
 
-    // sub-batches: <= kMaxBatch compressed chunks (d_rdstage_ slots) and
-    // pinned-out bytes <= stage_sz_
-    size_t j0 = 0;
-    while (j0 < dev.size()) {
-        int m = 0;  // compressed chunks in this sub-batch
-        uint64_t outb = 0;
-        size_t j1 = j0;
-        while (j1 < dev.size()) {
-            const DevReq& q = dev[j1];
-            int nm = m + (q.compressed ? 1 : 0);
-            uint64_t no = outb + round_up(q.g->len, 64);
-            if (nm > kMaxBatch || no > stage_sz_) break;
-            m = nm;
-            outb = no;
-            j1++;
-        }
-        // decompress pass
-        int mi = 0;
-        for (size_t j = j0; j < j1; j++) {
-            if (!dev[j].compressed) continue;
-            h_dc[mi] = DevChunk{dev[j].arena_off, (uint64_t)mi * cs, dev[j].g->ref->logical_len,
-                                dev[j].stored};
-            mi++;
-        }
-        if (mi) {
-            HIP_CHECK(hipMemcpyAsync(d_chunks_, h_dc, sizeof(DevChunk) * mi,
-                                     hipMemcpyHostToDevice, s_read_));
-            launch_lz4_decompress_chunks(arena_, d_chunks_, d_rdstage_, d_sizes_, mi, s_read_);
-            HIP_CHECK(hipMemcpyAsync(h_status, d_sizes_, 4 * mi, hipMemcpyDeviceToHost,
-                                     s_read_));
-        }
-        // D2H requested ranges
-        uint64_t pout = 0;
-        mi = 0;
-        for (size_t j = j0; j < j1; j++) {
-            const GetChunk& g = *dev[j].g;
-            const uint8_t* src;
-            if (dev[j].compressed) {
-                src = d_rdstage_ + (uint64_t)mi * cs + g.off;
-                mi++;
-            } else {
-                src = arena_ + dev[j].arena_off + g.off;
+        size_t j0 = 0;
+        while (j0 < mine.size()) {
+            int m = 0;
+            uint64_t outb = 0;
+            size_t j1 = j0;
+            while (j1 < mine.size()) {
+                const DevReq& q = *mine[j1];
+                int nm = m + (q.compressed ? 1 : 0);
+                uint64_t no = outb + round_up(q.g->len, 64);
+                if (nm > kMaxBatch || no > D.stage_sz) break;
+                m = nm;
+                outb = no;
+                j1++;
             }
-            HIP_CHECK(hipMemcpyAsync(pin_out_ + pout, src, g.len, hipMemcpyDeviceToHost,
-                                     s_read_));
-            pout += round_up(g.len, 64);
+            int mi = 0;
+            for (size_t j = j0; j < j1; j++) {
+                if (!mine[j]->compressed) continue;
+                h_dc[mi] = DevChunk{mine[j]->arena_off, (uint64_t)mi * cs,
+                                    mine[j]->g->ref->logical_len, mine[j]->stored};
+                mi++;
+            }
+            if (mi) {
+                HIP_CHECK(hipMemcpyAsync(D.d_chunks, h_dc, sizeof(DevChunk) * mi,
+                                         hipMemcpyHostToDevice, D.s_read));
+                launch_lz4_decompress_chunks(D.arena, D.d_chunks, D.d_rdstage, D.d_sizes, mi,
+                                             D.s_read);
+                HIP_CHECK(hipMemcpyAsync(h_status, D.d_sizes, 4 * mi, hipMemcpyDeviceToHost,
+                                         D.s_read));
+            }
+            uint64_t pout = 0;
+            mi = 0;
+            for (size_t j = j0; j < j1; j++) {
+                const GetChunk& g = *mine[j]->g;
+                const uint8_t* src;
+                if (mine[j]->compressed) {
+                    src = D.d_rdstage + (uint64_t)mi * cs + g.off;
+                    mi++;
+                } else {
+                    src = D.arena + mine[j]->arena_off + g.off;
+                }
+                HIP_CHECK(hipMemcpyAsync(D.pin_out + pout, src, g.len, hipMemcpyDeviceToHost,
+                                         D.s_read));
+                pout += round_up(g.len, 64);
+            }
+            HIP_CHECK(hipStreamSynchronize(D.s_read));
+            metrics().d2h_bytes += pout;
+            for (int x = 0; x < mi; x++)
+                if (h_status[x]) return St::bad_format;
+            pout = 0;
+            for (size_t j = j0; j < j1; j++) {
+                const GetChunk& g = *mine[j]->g;
+                std::memcpy(g.out, D.pin_out + pout, g.len);
+                pout += round_up(g.len, 64);
+            }
+            j0 = j1;
         }
-        HIP_CHECK(hipStreamSynchronize(s_read_));
-        metrics().d2h_bytes += pout;
-        for (int k = 0; k < mi; k++)
-            if (h_status[k]) return St::bad_format;
-        pout = 0;
-        for (size_t j = j0; j < j1; j++) {
-            const GetChunk& g = *dev[j].g;
-            std::memcpy(g.out, pin_out_ + pout, g.len);
-            pout += round_up(g.len, 64);
-        }
-        j0 = j1;
     }
+    return St::ok;
+}
+
+// Gather the UNCOMPRESSED bytes of one stored chunk into dst on device `ddev`
+// (zero-fill holes, on-owner decompress, xGMI peer copy). Part of the RS
+// stripe path (C2 analog).
+St GpuEngine::gather_chunk_to_dev(const ChunkRef& ref, int ddev, uint8_t* dst) {
+    const uint32_t cs = (uint32_t)conf_.chunk_size;
+    DevCtx& T = devs_[ddev];
+    HIP_CHECK(hipSetDevice(T.hip_dev));
+    if (ref.fp.is_zero()) {  // hole = zero shard
+        HIP_CHECK(hipMemsetAsync(dst, 0, cs, T.s_pipe));
+        return St::ok;
+    }
+    uint64_t aoff = 0;
+    uint32_t stored = 0;
+    bool comp = false, spilled = false;
+    int odev = 0;
+    {
+        std::lock_guard<std::mutex> lk(map_mu_);
+        auto it = map_.find(ref.fp);
+        if (it == map_.end()) return St::not_found;
+        spilled = it->second.spilled;
+        aoff = it->second.off;
+        stored = it->second.len;
+        comp = it->second.compressed;
+        odev = it->second.dev;
+    }
+    if (spilled) {
+        std::vector<uint8_t> full(ref.logical_len);
+        GetChunk g{&ref, full.data(), 0, ref.logical_len};
+        St s = get_chunks(&g, 1);
+        if (s != St::ok) return s;
+        HIP_CHECK(hipSetDevice(T.hip_dev));
+        HIP_CHECK(hipMemcpyAsync(dst, full.data(), ref.logical_len, hipMemcpyHostToDevice,
+                                 T.s_pipe));
+    } else if (!comp) {
+        DevCtx& O = devs_[odev];
+        if (odev == ddev) {
+            HIP_CHECK(hipMemcpyAsync(dst, O.arena + aoff, stored, hipMemcpyDeviceToDevice,
+                                     T.s_pipe));
+        } else {
+            HIP_CHECK(hipMemcpyPeerAsync(dst, T.hip_dev, O.arena + aoff, O.hip_dev, stored,
+                                         T.s_pipe));
+        }
+    } else {
+        DevCtx& O = devs_[odev];
+        std::lock_guard<std::mutex> og(O.pipe_mu);
+        HIP_CHECK(hipSetDevice(O.hip_dev));
+        auto* h_dc = (DevChunk*)(O.pin_meta + kOffChunks);
+        auto* h_status = (uint32_t*)(O.pin_meta + kOffStatus);
+        h_dc[0] = DevChunk{aoff, 0, ref.logical_len, stored};
+        HIP_CHECK(hipMemcpyAsync(O.d_chunks, h_dc, sizeof(DevChunk), hipMemcpyHostToDevice,
+                                 O.s_read));
+        launch_lz4_decompress_chunks(O.arena, O.d_chunks, O.d_rdstage, O.d_sizes, 1, O.s_read);
+        HIP_CHECK(hipMemcpyAsync(h_status, O.d_sizes, 4, hipMemcpyDeviceToHost, O.s_read));
+        HIP_CHECK(hipStreamSynchronize(O.s_read));
+        if (h_status[0]) return St::bad_format;
+        if (odev == ddev) {
+            HIP_CHECK(hipMemcpyAsync(dst, O.d_rdstage, ref.logical_len,
+                                     hipMemcpyDeviceToDevice, O.s_read));
+        } else {
+            HIP_CHECK(hipMemcpyPeerAsync(dst, T.hip_dev, O.d_rdstage, O.hip_dev,
+                                         ref.logical_len, O.s_read));
+        }
+        HIP_CHECK(hipStreamSynchronize(O.s_read));
+    }
+    if (ref.logical_len < cs) {
+        HIP_CHECK(hipSetDevice(T.hip_dev));
+        HIP_CHECK(hipMemsetAsync(dst + ref.logical_len, 0, cs - ref.logical_len, T.s_pipe));
+    }
+    return St::ok;
+}
+
+// On-device RS encode (C1/C2 + G6): gather the stripe's uncompressed shards
+// onto one device over xGMI, run the GF(256) kernel, ingest parity chunks.
+St GpuEngine::rs_encode_stripe(const ChunkRef* data, int k, int m, uint32_t shard_len,
+                               ChunkRef* parity_out) {
+    if (shard_len != conf_.chunk_size || k > kRsMaxK || m > kRsMaxM)
+        return Engine::rs_encode_stripe(data, k, m, shard_len, parity_out);
+    const int cdev = (int)(rr_.fetch_add(1) % devs_.size());
+    DevCtx& C = devs_[cdev];
+    std::lock_guard<std::mutex> rg(C.rs_mu);
+    const uint32_t cs = (uint32_t)conf_.chunk_size;
+    for (int j = 0; j < k; j++) {
+        St s = gather_chunk_to_dev(data[j], cdev, C.d_rs + (uint64_t)j * cs);
+        if (s != St::ok) return s;
+    }
+    HIP_CHECK(hipSetDevice(C.hip_dev));
+    HIP_CHECK(hipStreamSynchronize(C.s_pipe));
+    RsDesc d{};
+    auto M = rs_parity_matrix(k, m);
+    for (int j = 0; j < k; j++) d.in[j] = C.d_rs + (uint64_t)j * cs;
+    for (int i = 0; i < m; i++) d.out[i] = C.d_rs + (uint64_t)(k + i) * cs;
+    for (int i = 0; i < m; i++)
+        for (int j = 0; j < k; j++) d.coef[i][j] = M[(size_t)i * k + j];
+    d.k = k;
+    d.m = m;
+    d.len = cs;
+    launch_gf_matmul(d, C.s_pipe);
+    HIP_CHECK(hipStreamSynchronize(C.s_pipe));
+    // D2H parity, then ingest through the normal pipeline (hash/compress/crc)
+    std::vector<uint8_t> host((size_t)m * cs);
+    {
+        std::lock_guard<std::mutex> pg(C.pipe_mu);
+        for (int i = 0; i < m; i++)
+            HIP_CHECK(hipMemcpyAsync(host.data() + (size_t)i * cs,
+                                     C.d_rs + (uint64_t)(k + i) * cs, cs,
+                                     hipMemcpyDeviceToHost, C.s_read));
+        HIP_CHECK(hipStreamSynchronize(C.s_read));
+    }
+    std::vector<PutChunk> puts((size_t)m);
+    for (int i = 0; i < m; i++) puts[i] = {host.data() + (size_t)i * cs, cs, nullptr};
+    St s = put_chunks(puts.data(), parity_out, m);
+    if (s != St::ok) return s;
+    for (int i = 0; i < m; i++)
+        parity_out[i].flags |= ChunkRef::kFlagParity | ChunkRef::kFlagRs;
+    metrics().rs_encoded_stripes++;
     return St::ok;
 }
 
@@ -524,7 +715,7 @@ St GpuEngine::release_chunk(const ChunkRef& ref) {
         if (it->second.spilled) {
             spill_.decref(ref.fp);
         } else {
-            arena_free(it->second.off, it->second.len);
+            devs_[it->second.dev].arena_free(it->second.off, it->second.len);
             lru_.erase(it->second.lru);
         }
         map_.erase(it);
