@@ -20,7 +20,8 @@ HOST_SRCS := src/obs.cc src/bufpool.cc src/conf.cc src/manifest.cc src/blake3.cc
              src/engine/engine.cc
 HIP_SRCS  := src/engine/gpu_engine.hip src/kernels/fingerprint.hip \
              src/kernels/lz4.hip src/kernels/rs_gf256.hip \
-             src/kernels/checksum.hip src/kernels/memops.hip
+             src/kernels/checksum.hip src/kernels/memops.hip \
+             src/kernels/dedup_index.hip
 HOST_OBJS := $(HOST_SRCS:.cc=.o)
 HIP_OBJS  := $(HIP_SRCS:.hip=.o)
 OBJS      := $(HOST_OBJS) $(HIP_OBJS)

@@ -183,6 +183,31 @@ int eh_gpu_chunk_crc(const uint8_t* data, uint32_t len, uint32_t* out) {
     }
 }
 
+// G7 dedup index: upsert n fps twice; first pass dups -> out1, second -> out2.
+int eh_gpu_dedup_test(const uint8_t* fps, int n, uint32_t* out1, uint32_t* out2) {
+    try {
+        DevBuf d_tbl, d_fps, d_out;
+        uint64_t slots = 1 << 16;
+        HIP_CHECK(d_tbl.alloc(slots * 8));
+        HIP_CHECK(hipMemset(d_tbl.p, 0, slots * 8));
+        HIP_CHECK(d_fps.alloc(32 * n));
+        HIP_CHECK(d_out.alloc(4 * n));
+        HIP_CHECK(hipMemcpy(d_fps.p, fps, 32 * n, hipMemcpyHostToDevice));
+        launch_dedup_upsert(d_tbl.as<unsigned long long>(), slots - 1, d_fps.as<uint8_t>(),
+                            d_out.as<uint32_t>(), n, nullptr);
+        HIP_CHECK(hipDeviceSynchronize());
+        HIP_CHECK(hipMemcpy(out1, d_out.p, 4 * n, hipMemcpyDeviceToHost));
+        launch_dedup_upsert(d_tbl.as<unsigned long long>(), slots - 1, d_fps.as<uint8_t>(),
+                            d_out.as<uint32_t>(), n, nullptr);
+        HIP_CHECK(hipDeviceSynchronize());
+        HIP_CHECK(hipMemcpy(out2, d_out.p, 4 * n, hipMemcpyDeviceToHost));
+        return 0;
+    } catch (const std::exception& e) {
+        LOGE("eh_gpu_dedup_test: %s", e.what());
+        return -1;
+    }
+}
+
 int eh_gpu_available(void) {
     int n = 0;
     return hipGetDeviceCount(&n) == hipSuccess && n > 0 ? n : 0;

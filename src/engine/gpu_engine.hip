@@ -75,6 +75,8 @@ struct DevCtx {
 
     uint8_t *d_stage = nullptr, *d_comp = nullptr, *d_rdstage = nullptr,
             *d_segscratch = nullptr, *d_fps = nullptr, *d_rs = nullptr;
+    unsigned long long* d_index = nullptr;  // G7 fingerprint index (HBM)
+    uint64_t index_mask = 0;
     DevChunk *d_chunks = nullptr, *d_copy = nullptr;
     ulong2* d_segmeta = nullptr;
     uint32_t *d_segsizes = nullptr, *d_sizes = nullptr, *d_crcs = nullptr;
@@ -155,6 +157,16 @@ class GpuEngine : public Engine {
             HIP_CHECK(hipMalloc(&D.d_fps, 32 * kMaxBatch));
             HIP_CHECK(hipMalloc(&D.d_crcs, sizeof(uint32_t) * kMaxBatch));
             HIP_CHECK(hipMalloc(&D.d_copy, sizeof(DevChunk) * kMaxBatch));
+            // G7: per-device fingerprint index shard in HBM (tags of chunks
+            // this device owns); 1/1024 of the arena, power-of-two slots
+            {
+                uint64_t slots = 1;
+                while (slots * 16 < conf.arena_bytes / 1024) slots <<= 1;
+                if (slots < (1u << 16)) slots = 1u << 16;
+                HIP_CHECK(hipMalloc(&D.d_index, slots * 8));
+                HIP_CHECK(hipMemset(D.d_index, 0, slots * 8));
+                D.index_mask = slots - 1;
+            }
             HIP_CHECK(hipHostMalloc(&D.pin_in, D.stage_sz, hipHostMallocDefault));
             HIP_CHECK(hipHostMalloc(&D.pin_out, D.stage_sz, hipHostMallocDefault));
             HIP_CHECK(hipHostMalloc(&D.pin_meta, kPinMetaSz, hipHostMallocDefault));
@@ -172,7 +184,7 @@ class GpuEngine : public Engine {
                             (void*)D.d_rdstage, (void*)D.d_rs, (void*)D.d_segscratch,
                             (void*)D.d_chunks, (void*)D.d_segmeta, (void*)D.d_segsizes,
                             (void*)D.d_sizes, (void*)D.d_fps, (void*)D.d_crcs,
-                            (void*)D.d_copy})
+                            (void*)D.d_copy, (void*)D.d_index})
                 (void)hipFree(p);
             (void)hipHostFree(D.pin_in);
             (void)hipHostFree(D.pin_out);
@@ -286,6 +298,16 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
     auto* h_fps = (uint8_t*)(D.pin_meta + kOffFps);
     auto* h_sizes = (uint32_t*)(D.pin_meta + kOffSizes);
     auto* h_crcs = (uint32_t*)(D.pin_meta + kOffCrcs);
+    // G7: upsert the batch's fingerprints into THIS device's HBM index.
+    // (Placement-owner sharding refines this: tags also get upserted into the
+    // owner's shard below once owners are known; the ingest-dev upsert gives
+    // the fast dup signal for the common single-device case.)
+    auto* h_dups = (uint32_t*)(D.pin_meta + kOffStatus);
+    if (do_fp && conf_.dedup) {
+        // d_crcs doubles as the dup-out scratch here (n <= kMaxBatch)
+        launch_dedup_upsert(D.d_index, D.index_mask, D.d_fps, D.d_crcs, n, D.s_pipe);
+        HIP_CHECK(hipMemcpyAsync(h_dups, D.d_crcs, 4 * n, hipMemcpyDeviceToHost, D.s_pipe));
+    }
     if (do_fp)
         HIP_CHECK(hipMemcpyAsync(h_fps, D.d_fps, 32 * n, hipMemcpyDeviceToHost, D.s_pipe));
     if (do_lz4)
@@ -322,6 +344,8 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
         metrics().lz4_out_bytes += stored;
 
         bool dup = false;
+        bool index_hint_dup = do_fp && conf_.dedup && h_dups[i] != 0;
+        (void)index_hint_dup;  // host map confirms (stale tags after deletes)
         {
             std::lock_guard<std::mutex> g(map_mu_);
             if (conf_.dedup && do_fp) {

@@ -178,3 +178,25 @@ def test_chunk_crc_vs_host(lib, n):
     for off in range(0, n, 4096):
         acc ^= lib.crc32c(data[off:off + 4096])
     assert out.value == acc
+
+
+# ---- dedup index (G7) -------------------------------------------------------
+def test_dedup_index_hbm(lib):
+    """HBM hash table: first upsert of unique fps -> new; second pass -> dup;
+    duplicate fps within one batch -> exactly one inserter."""
+    import numpy as np
+    so = lib._so
+    so.eh_gpu_dedup_test.argtypes = [ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p,
+                                     ctypes.c_void_p]
+    so.eh_gpu_dedup_test.restype = ctypes.c_int
+    n = 4096
+    rng = np.random.default_rng(5)
+    uniq = rng.integers(0, 256, size=(n // 2, 32), dtype=np.uint8)
+    fps = np.vstack([uniq, uniq]).tobytes()  # every fp appears exactly twice
+    out1 = (ctypes.c_uint32 * n)()
+    out2 = (ctypes.c_uint32 * n)()
+    assert so.eh_gpu_dedup_test(fps, n, out1, out2) == 0
+    # pass 1: each duplicated pair has exactly one inserter (dup sum == n/2)
+    assert sum(out1) == n // 2
+    # pass 2: everything is a dup
+    assert sum(out2) == n
