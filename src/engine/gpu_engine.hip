@@ -49,6 +49,8 @@ constexpr uint64_t kOffSizes = 144 * KiB;
 constexpr uint64_t kOffCrcs = 148 * KiB;
 constexpr uint64_t kOffCopy = 152 * KiB;
 constexpr uint64_t kOffStatus = 156 * KiB;
+constexpr uint64_t kOffCrcDesc = 160 * KiB;   // DevChunk[2*kMaxBatch]
+constexpr uint64_t kOffCopyDesc = 168 * KiB;  // DevChunk[2*kMaxBatch]
 constexpr uint64_t kPinMetaSz = 256 * KiB;
 
 struct ArenaLoc {
@@ -155,8 +157,8 @@ class GpuEngine : public Engine {
             HIP_CHECK(hipMalloc(&D.d_segsizes, sizeof(uint32_t) * kMaxSegs));
             HIP_CHECK(hipMalloc(&D.d_sizes, sizeof(uint32_t) * kMaxBatch));
             HIP_CHECK(hipMalloc(&D.d_fps, 32 * kMaxBatch));
-            HIP_CHECK(hipMalloc(&D.d_crcs, sizeof(uint32_t) * kMaxBatch));
-            HIP_CHECK(hipMalloc(&D.d_copy, sizeof(DevChunk) * kMaxBatch));
+            HIP_CHECK(hipMalloc(&D.d_crcs, sizeof(uint32_t) * kMaxBatch * 2));
+            HIP_CHECK(hipMalloc(&D.d_copy, sizeof(DevChunk) * kMaxBatch * 2));
             // G7: per-device fingerprint index shard in HBM (tags of chunks
             // this device owns); 1/1024 of the arena, power-of-two slots
             {
@@ -387,10 +389,14 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
     }
 
     if (!pend.empty()) {
-        // crc over stored bytes first (from the ingest device's staging copy)
+        // crc over stored bytes (from the ingest device's staging copy)
+        std::vector<int> crc_maps[2];
         if (conf_.crc) {
-            auto* h_cd = (DevChunk*)(D.pin_meta + kOffCopy);
             for (int pass = 0; pass < 2; pass++) {
+                auto* h_cd = (DevChunk*)(D.pin_meta + kOffCrcDesc) + pass * kMaxBatch;
+                DevChunk* d_cd = D.d_copy + pass * kMaxBatch;
+                uint32_t* d_out = D.d_crcs + 0;  // reused per pass via offsets below
+                (void)d_out;
                 int m = 0;
                 std::vector<int> idxmap;
                 for (size_t j = 0; j < pend.size(); j++) {
@@ -400,34 +406,30 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
                     m++;
                 }
                 if (!m) continue;
-                HIP_CHECK(hipMemcpyAsync(D.d_copy, h_cd, sizeof(DevChunk) * m,
+                HIP_CHECK(hipMemcpyAsync(d_cd, h_cd, sizeof(DevChunk) * m,
                                          hipMemcpyHostToDevice, D.s_pipe));
-                launch_crc32c_chunks(pass ? D.d_comp : D.d_stage, D.d_copy, D.d_crcs, m,
-                                     D.s_pipe);
-                HIP_CHECK(hipMemcpyAsync(h_crcs, D.d_crcs, 4 * m, hipMemcpyDeviceToHost,
-                                         D.s_pipe));
-                HIP_CHECK(hipStreamSynchronize(D.s_pipe));
-                std::lock_guard<std::mutex> g(map_mu_);
-                for (int x = 0; x < m; x++) {
-                    int j = idxmap[x];
-                    refs[pend[j].i].crc = h_crcs[x];
-                    auto it = map_.find(refs[pend[j].i].fp);
-                    if (it != map_.end()) it->second.crc = h_crcs[x];
-                }
+                launch_crc32c_chunks(pass ? D.d_comp : D.d_stage, d_cd,
+                                     D.d_crcs + pass * kMaxBatch, m, D.s_pipe);
+                HIP_CHECK(hipMemcpyAsync(h_crcs + pass * kMaxBatch,
+                                         D.d_crcs + pass * kMaxBatch, 4 * m,
+                                         hipMemcpyDeviceToHost, D.s_pipe));
+                crc_maps[pass] = idxmap;
             }
         }
         // local-owner chunks: batch copy kernel; remote owners: xGMI peer copy
-        auto* h_copy = (DevChunk*)(D.pin_meta + kOffCopy);
+        // (independent of the crc kernels — one sync covers everything)
         for (int pass = 0; pass < 2; pass++) {
+            auto* h_copy = (DevChunk*)(D.pin_meta + kOffCopyDesc) + pass * kMaxBatch;
+            DevChunk* d_cp = D.d_copy + (2 + pass) * kMaxBatch;
             int m = 0;
             for (auto& p : pend) {
                 if ((int)p.from_comp != pass || p.dev != ingest_dev) continue;
                 h_copy[m++] = DevChunk{p.src_off, p.arena_off, p.stored, 0};
             }
             if (!m) continue;
-            HIP_CHECK(hipMemcpyAsync(D.d_copy, h_copy, sizeof(DevChunk) * m,
+            HIP_CHECK(hipMemcpyAsync(d_cp, h_copy, sizeof(DevChunk) * m,
                                      hipMemcpyHostToDevice, D.s_pipe));
-            launch_batch_copy(pass ? D.d_comp : D.d_stage, D.arena, D.d_copy, m, D.s_pipe);
+            launch_batch_copy(pass ? D.d_comp : D.d_stage, D.arena, d_cp, m, D.s_pipe);
         }
         for (auto& p : pend) {
             if (p.dev == ingest_dev) continue;
@@ -437,6 +439,18 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
                                          D.hip_dev, p.stored, D.s_pipe));
         }
         HIP_CHECK(hipStreamSynchronize(D.s_pipe));
+        if (conf_.crc) {
+            std::lock_guard<std::mutex> g(map_mu_);
+            for (int pass = 0; pass < 2; pass++) {
+                for (size_t x = 0; x < crc_maps[pass].size(); x++) {
+                    int j = crc_maps[pass][x];
+                    uint32_t c = h_crcs[pass * kMaxBatch + x];
+                    refs[pend[j].i].crc = c;
+                    auto it = map_.find(refs[pend[j].i].fp);
+                    if (it != map_.end()) it->second.crc = c;
+                }
+            }
+        }
     }
     return St::ok;
 }

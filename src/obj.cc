@@ -6,7 +6,7 @@ namespace edgehip {
 
 // Full chunks accumulated per object before one batched engine call; batching
 // is what gives the GPU engine kernel-launch-amortizing work.
-static constexpr int kPutBatch = 8;
+static constexpr int kPutBatch = 32;
 
 Object::~Object() = default;
 
