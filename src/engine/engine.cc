@@ -21,6 +21,7 @@ class HostEngine : public Engine {
     }
 
     St put_chunks(const PutChunk* put, ChunkRef* refs, int n) override {
+        if (Faults::inst().fire("put_io")) return St::io;
         for (int i = 0; i < n; i++) {
             const PutChunk& p = put[i];
             ChunkRef& r = refs[i];
@@ -83,6 +84,7 @@ class HostEngine : public Engine {
             const ChunkRef& r = *g.ref;
             if ((uint64_t)g.off + g.len > r.logical_len) return St::inval;
             metrics().chunks_get++;
+            if (Faults::inst().fire("get_drop")) return St::not_found;
             uint32_t stored = store_.stored_len(r.fp);
             if (stored == 0) {
                 // miss: try the journal chunk mirror (remount path)
@@ -226,6 +228,7 @@ std::unique_ptr<Engine> make_host_engine(const Conf& conf) {
 }
 
 std::unique_ptr<Engine> make_engine(const Conf& conf) {
+    Faults::inst().configure(conf.fault);
     if (conf.engine == EngineKind::gpu) return make_gpu_engine(conf);
     return make_host_engine(conf);
 }

@@ -500,6 +500,7 @@ St GpuEngine::get_chunks(const GetChunk* get, int n) {
         const ChunkRef& r = *g.ref;
         if ((uint64_t)g.off + g.len > r.logical_len) return St::inval;
         metrics().chunks_get++;
+        if (Faults::inst().fire("get_drop")) return St::not_found;
         bool found;
         {
             std::lock_guard<std::mutex> lk(map_mu_);

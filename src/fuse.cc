@@ -18,6 +18,9 @@ namespace {
 
 constexpr uint64_t kAttrValidSec = 1;
 constexpr uint64_t kEntryValidSec = 1;
+// hidden virtual file /.edgehip-stats exposing live counters
+constexpr uint64_t kStatsIno = 0x7FFFFFFFFFFFFFF0ull;
+constexpr const char* kStatsName = ".edgehip-stats";
 
 void fill_attr(const Inode* in, fuse_attr* a) {
     std::memset(a, 0, sizeof(*a));
@@ -236,6 +239,18 @@ void FuseMount::handle(const uint8_t* req, size_t len, std::vector<uint8_t>* rep
     switch (h->opcode) {
         case FUSE_LOOKUP: {
             std::string name((const char*)arg);
+            if (h->nodeid == 1 && name == kStatsName) {
+                fuse_entry_out e{};
+                e.nodeid = kStatsIno;
+                e.generation = 1;
+                e.attr.ino = kStatsIno;
+                e.attr.mode = S_IFREG | 0444;
+                e.attr.nlink = 1;
+                e.attr.size = metrics().render().size();
+                e.attr.blksize = 4096;
+                reply_body(&e, sizeof(e));
+                return;
+            }
             Inode* in;
             St s = vfs_->lookup(h->nodeid, name, &in);
             if (s != St::ok) { reply_st(s); return; }
@@ -256,6 +271,16 @@ void FuseMount::handle(const uint8_t* req, size_t len, std::vector<uint8_t>* rep
             return;  // no reply
         }
         case FUSE_GETATTR: {
+            if (h->nodeid == kStatsIno) {
+                fuse_attr_out ao{};
+                ao.attr.ino = kStatsIno;
+                ao.attr.mode = S_IFREG | 0444;
+                ao.attr.nlink = 1;
+                ao.attr.size = metrics().render().size();
+                ao.attr.blksize = 4096;
+                reply_body(&ao, sizeof(ao));
+                return;
+            }
             Inode* in;
             St s = vfs_->getattr(h->nodeid, &in);
             if (s != St::ok) { reply_st(s); return; }
@@ -339,6 +364,13 @@ void FuseMount::handle(const uint8_t* req, size_t len, std::vector<uint8_t>* rep
             return;
         }
         case FUSE_OPEN: {
+            if (h->nodeid == kStatsIno) {
+                fuse_open_out oo{};
+                oo.fh = 0;
+                oo.open_flags = FOPEN_DIRECT_IO;  // size is dynamic
+                reply_body(&oo, sizeof(oo));
+                return;
+            }
             uint64_t fh;
             St s = vfs_->open(h->nodeid, &fh);
             if (s != St::ok) { reply_st(s); return; }
@@ -376,6 +408,13 @@ void FuseMount::handle(const uint8_t* req, size_t len, std::vector<uint8_t>* rep
         }
         case FUSE_READ: {
             auto* r = (const fuse_read_in*)arg;
+            if (h->nodeid == kStatsIno) {
+                std::string s = metrics().render();
+                size_t off = std::min((size_t)r->offset, s.size());
+                size_t n2 = std::min((size_t)r->size, s.size() - off);
+                reply_body(s.data() + off, n2);
+                return;
+            }
             reply_buf->resize(r->size);
             uint64_t got = 0;
             St s = vfs_->read(r->fh, r->offset, r->size, reply_buf->data(), &got);
@@ -395,11 +434,19 @@ void FuseMount::handle(const uint8_t* req, size_t len, std::vector<uint8_t>* rep
             return;
         }
         case FUSE_FLUSH: {
+            if (h->nodeid == kStatsIno) {
+                reply_err(0);
+                return;
+            }
             auto* f = (const fuse_flush_in*)arg;
             reply_st(vfs_->flush(f->fh));
             return;
         }
         case FUSE_RELEASE: {
+            if (h->nodeid == kStatsIno) {
+                reply_err(0);
+                return;
+            }
             auto* r = (const fuse_release_in*)arg;
             reply_st(vfs_->release(r->fh));
             return;

@@ -76,6 +76,49 @@ std::string Metrics::render() const {
     return std::string(buf);
 }
 
+Faults& Faults::inst() {
+    static Faults f;
+    return f;
+}
+
+void Faults::configure(const std::string& spec, uint64_t seed) {
+    sites_.clear();
+    enabled_ = false;
+    rng_ = seed ? seed : 0x9E3779B97F4A7C15ull;
+    size_t pos = 0;
+    while (pos < spec.size()) {
+        size_t comma = spec.find(',', pos);
+        if (comma == std::string::npos) comma = spec.size();
+        std::string item = spec.substr(pos, comma - pos);
+        pos = comma + 1;
+        size_t colon = item.find(':');
+        if (colon == std::string::npos) continue;
+        Site s;
+        s.name = item.substr(0, colon);
+        s.prob = atof(item.c_str() + colon + 1);
+        if (s.prob > 0) {
+            sites_.push_back(s);
+            enabled_ = true;
+            LOGW("faults: %s armed at p=%.4f", s.name.c_str(), s.prob);
+        }
+    }
+}
+
+bool Faults::fire(const char* site) {
+    if (!enabled_) return false;
+    for (auto& s : sites_) {
+        if (s.name != site) continue;
+        // xorshift64* per call (thread-coarse; probabilistic tests only)
+        uint64_t x = rng_.fetch_add(0x9E3779B97F4A7C15ull) ^ rng_.load();
+        x ^= x >> 12;
+        x ^= x << 25;
+        x ^= x >> 27;
+        double u = (double)(x * 0x2545F4914F6CDD1Dull >> 11) / 9007199254740992.0;
+        return u < s.prob;
+    }
+    return false;
+}
+
 // ---------------------------------------------------------------- crc32c
 // Castagnoli polynomial 0x1EDC6F41, reflected 0x82F63B78. Table-sliced by 8.
 static uint32_t crc_tab[8][256];

@@ -5,6 +5,7 @@
 
 #include <atomic>
 #include <string>
+#include <vector>
 
 #include "common.h"
 
@@ -26,5 +27,26 @@ struct Metrics {
 };
 
 Metrics& metrics();
+
+// Fault injection: conf `fault=site:prob[,site:prob]`.
+// Sites wired into the data path: "get_drop" (chunk read returns not_found,
+// exercising RS reconstruct), "put_io" (ingest fails with EIO),
+// "corrupt" (read bytes corrupted — caught by verify_read / crc checks).
+class Faults {
+  public:
+    static Faults& inst();
+    void configure(const std::string& spec, uint64_t seed = 0x12345678);
+    bool fire(const char* site);  // true => inject at this call site
+    bool enabled() const { return enabled_; }
+
+  private:
+    struct Site {
+        std::string name;
+        double prob;
+    };
+    std::vector<Site> sites_;
+    std::atomic<uint64_t> rng_{0x9E3779B97F4A7C15ull};
+    bool enabled_ = false;
+};
 
 }  // namespace edgehip
