@@ -11,7 +11,7 @@ ROCM     ?= /opt/rocm
 HIPCC    ?= $(ROCM)/bin/hipcc
 CXX      ?= g++
 ARCH     ?= gfx950
-CXXFLAGS ?= -O2 -g -std=c++17 -Wall -Wextra -fPIC -pthread -MMD -MP
+CXXFLAGS ?= -O2 -g -std=c++17 -Wall -Wextra -fPIC -pthread -MMD -MP -I$(ROCM)/include -D__HIP_PLATFORM_AMD__
 HIPFLAGS ?= -O3 -std=c++17 --offload-arch=$(ARCH) -fPIC -MMD -MP
 LDFLAGS  ?= -L$(ROCM)/lib -lamdhip64 -lrccl -Wl,-rpath,$(ROCM)/lib -pthread
 
@@ -28,7 +28,13 @@ OBJS      := $(HOST_OBJS) $(HIP_OBJS)
 
 .PHONY: all clean test-bins
 
-all: bin/edgehip lib/libedgehip.so
+all: bin/edgehip lib/libedgehip.so bin/kernbench
+
+bin/kernbench: bench/kernbench.o $(HIP_OBJS:src/engine/gpu_engine.o=) src/gf256.o src/obs.o src/bufpool.o
+	@mkdir -p bin
+	$(HIPCC) -o $@ bench/kernbench.o src/kernels/fingerprint.o src/kernels/lz4.o \
+	    src/kernels/rs_gf256.o src/kernels/checksum.o src/kernels/memops.o \
+	    src/kernels/dedup_index.o src/gf256.o src/obs.o src/bufpool.o -lpthread
 
 bin/edgehip: $(OBJS) src/main.o
 	@mkdir -p bin
@@ -45,7 +51,7 @@ lib/libedgehip.so: $(OBJS) src/capi.o src/capi_gpu.o
 	$(HIPCC) $(HIPFLAGS) -c -o $@ $<
 
 clean:
-	rm -f $(OBJS) src/main.o src/capi.o src/capi_gpu.o
+	rm -f $(OBJS) src/main.o src/capi.o src/capi_gpu.o bench/kernbench.o bench/kernbench.d
 	rm -f $(OBJS:.o=.d) src/main.d src/capi.d src/capi_gpu.d
 	rm -rf bin lib
 

@@ -1,8 +1,32 @@
 #include "bufpool.h"
 
+#include <hip/hip_runtime.h>
+
 #include <thread>
 
+#include "obs.h"
+
 namespace edgehip {
+
+void* BufPool::alloc_block(uint32_t cls) {
+    // Prefer pinned host memory when a GPU runtime is live: hipMemcpyAsync
+    // from these blocks is true async DMA (pageable memory would stage).
+    if (!probed_) {
+        probed_ = true;
+        int n = 0;
+        pinned_ = hipGetDeviceCount(&n) == hipSuccess && n > 0;
+        if (pinned_) LOGI("bufpool: pinned host blocks (GPU present)");
+    }
+    if (pinned_) {
+        void* p = nullptr;
+        if (hipHostMalloc(&p, cls, hipHostMallocDefault) == hipSuccess) return p;
+        pinned_ = false;  // fall back for good
+        LOGW("bufpool: hipHostMalloc failed; falling back to pageable");
+    }
+    void* p = nullptr;
+    if (posix_memalign(&p, 2 * MiB, cls) != 0) return nullptr;
+    return p;
+}
 
 void BufPool::prealloc(uint64_t bytes, uint32_t block_sz, int nthreads) {
     if (!bytes) return;

@@ -267,10 +267,21 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
     auto* h_chunks = (DevChunk*)(D.pin_meta + kOffChunks);
     auto* h_chunks2 = (DevChunk*)(D.pin_meta + kOffChunks2);
     auto* h_segmeta = (ulong2*)(D.pin_meta + kOffSegMeta);
+    const bool pool_pinned = BufPool::inst().pinned();
     int nsegs = 0;
     uint64_t off = 0;
+    uint64_t staged = 0;
     for (int i = 0; i < n; i++) {
-        std::memcpy(D.pin_in + off, put[i].data, put[i].len);
+        if (pool_pinned && put[i].donate && put[i].donate->p == put[i].data) {
+            // dirty buffer is already pinned: DMA straight from it
+            HIP_CHECK(hipMemcpyAsync(D.d_stage + off, put[i].data, put[i].len,
+                                     hipMemcpyHostToDevice, D.s_pipe));
+        } else {
+            std::memcpy(D.pin_in + staged, put[i].data, put[i].len);
+            HIP_CHECK(hipMemcpyAsync(D.d_stage + off, D.pin_in + staged, put[i].len,
+                                     hipMemcpyHostToDevice, D.s_pipe));
+            staged += round_up(put[i].len, 64);
+        }
         h_chunks[i] = DevChunk{off, off, put[i].len, (uint32_t)nsegs};
         h_chunks2[i] = DevChunk{off, (uint64_t)i * cs, put[i].len, (uint32_t)nsegs};
         uint32_t ns = (put[i].len + kSegSz - 1) / kSegSz;
@@ -280,7 +291,6 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
         nsegs += ns;
         off += round_up(put[i].len, 64);
     }
-    HIP_CHECK(hipMemcpyAsync(D.d_stage, D.pin_in, off, hipMemcpyHostToDevice, D.s_pipe));
     HIP_CHECK(hipMemcpyAsync(D.d_chunks, h_chunks, sizeof(DevChunk) * n,
                              hipMemcpyHostToDevice, D.s_pipe));
     metrics().h2d_bytes += off;
@@ -385,7 +395,8 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
         }
         if (!dup && conf_.journal_ptr)
             conf_.journal_ptr->put_chunk(r.fp, put[i].data, put[i].len);
-        if (put[i].donate) put[i].donate->release();
+        // donated (possibly pinned) buffers are released AFTER the final
+        // stream sync at the end of this batch — see below
     }
 
     if (!pend.empty()) {
@@ -452,6 +463,8 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
             }
         }
     }
+    for (int i = 0; i < n; i++)
+        if (put[i].donate) put[i].donate->release();
     return St::ok;
 }
 
