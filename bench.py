@@ -57,29 +57,44 @@ class ObjectIo:
     def close(self):
         self.eng.close()
 
-    def run_step(self, step_idx, gib, rand_buf, read_buf):
-        # 8 MiB library-call granularity: the FUSE path gets the same batching
-        # from 1 MiB kernel requests via the object layer's write buffer and
-        # 16 MiB kernel readahead; this drives the identical engine batches.
-        o = self.eng.object()
+    def run_step(self, step_idx, gib, rand_buf, read_buf, jobs=4):
+        # fio-numjobs analog: `jobs` threads, each streaming its own object
+        # (ctypes releases the GIL, so host memcpys and engine batches overlap
+        # across jobs exactly like concurrent FUSE worker threads would).
+        import threading
         unit = 8 * MIB
-        n = gib * 1024 // 8
+        per_job = max(1, gib * 1024 // 8 // jobs)
         src = memoryview(rand_buf)
-        try:
-            for i in range(n):
-                off = (i % 128) * unit
-                o.write(i * unit, src[off:off + unit])
-            o.flush()
-            total = 0
-            for i in range(n):
-                got = o.read(i * unit, unit)
-                total += len(got)
-            if total != n * unit:
-                raise RuntimeError("short read")
-        finally:
-            o.truncate(0)
-            o.close()
-        return 2 * n * unit
+        errs = []
+
+        def job(j):
+            try:
+                o = self.eng.object()
+                try:
+                    for i in range(per_job):
+                        off = ((i * jobs + j) % 128) * unit
+                        o.write(i * unit, src[off:off + unit])
+                    o.flush()
+                    total = 0
+                    for i in range(per_job):
+                        got = o.read(i * unit, unit)
+                        total += len(got)
+                    if total != per_job * unit:
+                        raise RuntimeError("short read")
+                finally:
+                    o.truncate(0)
+                    o.close()
+            except Exception as e:  # pragma: no cover
+                errs.append(e)
+
+        ts = [threading.Thread(target=job, args=(j,)) for j in range(jobs)]
+        for t in ts:
+            t.start()
+        for t in ts:
+            t.join()
+        if errs:
+            raise errs[0]
+        return 2 * per_job * unit * jobs
 
 
 def fuse_usable():
@@ -95,32 +110,50 @@ def fuse_usable():
         return False
 
 
-def run_step(mnt, step_idx, gib, rand_buf, read_buf):
-    """One step: seq-write `gib` GiB (1 MiB buffered writes + fsync), then
-    seq-read it back with O_DIRECT 1 MiB reads. Returns bytes moved."""
-    path = os.path.join(mnt, f"bench_{step_idx}.bin")
-    nmb = gib * 1024
+def run_step(mnt, step_idx, gib, rand_buf, read_buf, jobs=4):
+    """One step: `jobs` fio-style workers each seq-write gib/jobs GiB (1 MiB
+    buffered writes + fsync) then seq-read it back with O_DIRECT 1 MiB reads
+    (fio numjobs=N seq write + seq read). Returns bytes moved."""
+    import mmap as mmap_mod
+    import threading
     src = memoryview(rand_buf)
-    fd = os.open(path, os.O_WRONLY | os.O_CREAT | os.O_TRUNC, 0o644)
-    try:
-        for i in range(nmb):
-            off = (i % 1024) * MIB
-            os.write(fd, src[off:off + MIB])
-        os.fsync(fd)
-    finally:
-        os.close(fd)
-    fd = os.open(path, os.O_RDONLY | os.O_DIRECT)
-    try:
-        total = 0
-        while total < nmb * MIB:
-            n = os.readv(fd, [read_buf])
-            if n <= 0:
-                raise RuntimeError(f"short read at {total}")
-            total += n
-    finally:
-        os.close(fd)
-    os.unlink(path)  # release chunks so the arena doesn't grow across steps
-    return 2 * nmb * MIB
+    per_job = max(1, gib * 1024 // jobs)
+    errs = []
+
+    def job(j):
+        try:
+            path = os.path.join(mnt, f"bench_{step_idx}_{j}.bin")
+            fd = os.open(path, os.O_WRONLY | os.O_CREAT | os.O_TRUNC, 0o644)
+            try:
+                for i in range(per_job):
+                    off = ((i * jobs + j) % 1024) * MIB
+                    os.write(fd, src[off:off + MIB])
+                os.fsync(fd)
+            finally:
+                os.close(fd)
+            rbuf = mmap_mod.mmap(-1, MIB)
+            fd = os.open(path, os.O_RDONLY | os.O_DIRECT)
+            try:
+                total = 0
+                while total < per_job * MIB:
+                    n = os.readv(fd, [rbuf])
+                    if n <= 0:
+                        raise RuntimeError(f"short read at {total}")
+                    total += n
+            finally:
+                os.close(fd)
+            os.unlink(path)
+        except Exception as e:  # pragma: no cover
+            errs.append(e)
+
+    ts = [threading.Thread(target=job, args=(j,)) for j in range(jobs)]
+    for t in ts:
+        t.start()
+    for t in ts:
+        t.join()
+    if errs:
+        raise errs[0]
+    return 2 * per_job * MIB * jobs
 
 
 def main():
@@ -132,6 +165,7 @@ def main():
                     help="GiB written (and read back) per step per rank")
     ap.add_argument("--engine", default=None, choices=[None, "gpu", "host"],
                     help="override engine (default: gpu if a GPU is visible)")
+    ap.add_argument("--jobs", type=int, default=4, help="parallel I/O jobs (fio numjobs)")
     ap.add_argument("--mnt-base", default="/tmp/edgehip_bench")
     args = ap.parse_args()
 
@@ -186,24 +220,27 @@ def main():
     if use_fuse:
         with Mount(mnt, opts):
             for w in range(args.warmup):
-                run_step(mnt, 1000 + w, max(1, args.gib_per_step // 2), rng_buf, read_buf)
+                run_step(mnt, 1000 + w, max(1, args.gib_per_step // 2), rng_buf, read_buf,
+                         jobs=args.jobs)
             barrier()
             t0 = time.perf_counter()
             moved = 0
             for s in range(args.steps):
-                moved += run_step(mnt, s, args.gib_per_step, rng_buf, read_buf)
+                moved += run_step(mnt, s, args.gib_per_step, rng_buf, read_buf,
+                                  jobs=args.jobs)
             barrier()
             t1 = time.perf_counter()
     else:
         io = ObjectIo(opts)
         try:
             for w in range(args.warmup):
-                io.run_step(1000 + w, max(1, args.gib_per_step // 2), rng_buf, read_buf)
+                io.run_step(1000 + w, max(1, args.gib_per_step // 2), rng_buf, read_buf,
+                            jobs=args.jobs)
             barrier()
             t0 = time.perf_counter()
             moved = 0
             for s in range(args.steps):
-                moved += io.run_step(s, args.gib_per_step, rng_buf, read_buf)
+                moved += io.run_step(s, args.gib_per_step, rng_buf, read_buf, jobs=args.jobs)
             barrier()
             t1 = time.perf_counter()
         finally:

@@ -1,0 +1,67 @@
+// Distribution layer: RCCL collectives over xGMI
+// for metadata, complementing the engine's fan-out p2p for bulk chunk data.
+//
+// Design split per SURVEY §3.3: xGMI is 7 point-to-point links per GPU, so a
+// ring collective is bound by ONE link (~153 GB/s) while fan-out p2p reaches
+// ~1 TB/s — bulk shard traffic therefore moves via hipMemcpyPeerAsync in the
+// engine (C1/C2), and RCCL carries the small, latency-tolerant metadata:
+//   C4  publish_manifests : ncclAllGather of fixed-size ChunkRef records at
+//                           commit — every GPU learns the new chunk->owner map
+//                           (feeds future GPU-side read service + repair).
+//   C5  dedup deltas      : ncclBroadcast of new fingerprint tags so every
+//                           device's HBM index shard can pre-warm lookups.
+//   C6  epoch barrier     : 4-byte ncclAllReduce as the commit barrier.
+// C3 (parity reduce-scatter): RCCL has no GF(2) XOR op, so sharded-parity
+// accumulation uses p2p partials + local GF combine in the engine
+// (SURVEY §6.8); revisit via bit-plane ncclSum only if profiles show C3 hot.
+//
+// One process, one communicator spanning the node's devices (ncclCommInitAll;
+// "nccl" on ROCm IS RCCL). Multi-process mode (one rank per GPU — the
+// driver's scaling bench) bootstraps via ncclUniqueId exchange.
+#pragma once
+
+#include <cstdint>
+#include <vector>
+
+#include "../common.h"
+#include "../manifest.h"
+
+namespace edgehip {
+
+class DistCtx {
+  public:
+    ~DistCtx();
+
+    // Single-process mode over hip devices [first, first+n).
+    St init_local(int ngpus, int first_dev);
+    bool active() const { return ndev_ > 1; }
+    int ndev() const { return ndev_; }
+
+    // C4: every device contributes its batch of new ChunkRef records; all
+    // devices receive the concatenation into their HBM manifest table.
+    // records are padded to `slots` per rank (zero-filled).
+    St allgather_manifests(int dev_slot, const ChunkRef* recs, int n, int slots);
+
+    // C5: broadcast new fingerprint tags from `root_slot` to all devices'
+    // index-warm buffers.
+    St broadcast_tags(int root_slot, const uint64_t* tags, int n);
+
+    // C6: commit barrier — tiny all-reduce; returns the max epoch seen.
+    St epoch_barrier(uint64_t epoch, uint64_t* max_epoch);
+
+    // device-resident manifest table (per device slot): records land here
+    void* manifest_table(int dev_slot) const;
+    uint64_t manifest_count(int dev_slot) const;
+
+  private:
+    int ndev_ = 0;
+    int first_dev_ = 0;
+    std::vector<void*> comms_;        // ncclComm_t per device
+    std::vector<void*> streams_;      // hipStream_t per device
+    std::vector<void*> d_tables_;     // device manifest tables
+    std::vector<void*> d_scratch_;    // per-device allgather scratch
+    std::vector<uint64_t> counts_;
+    uint64_t table_cap_ = 0;
+};
+
+}  // namespace edgehip

@@ -64,6 +64,15 @@ class Engine {
                               uint32_t shard_len, int target, uint8_t* out,
                               uint32_t out_len);
 
+    // C4/C6: publish freshly committed manifest records to
+    // every device (RCCL all-gather over xGMI) + commit epoch barrier.
+    // No-op on single-device / host engines.
+    virtual St publish_manifest(const ChunkRef* recs, int n) {
+        (void)recs;
+        (void)n;
+        return St::ok;
+    }
+
     // Test hook: drop a chunk's stored bytes, simulating shard loss
     // (tests/faults; the read path must then reconstruct via RS).
     virtual St debug_drop_chunk(const Fp& fp) = 0;

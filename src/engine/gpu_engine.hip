@@ -27,6 +27,7 @@
 #include "../kernels/launchers.h"
 #include "../lz4.h"
 #include "../obs.h"
+#include "../dist/dist.h"
 #include "../store.h"
 #include "engine.h"
 
@@ -173,6 +174,10 @@ class GpuEngine : public Engine {
             HIP_CHECK(hipHostMalloc(&D.pin_out, D.stage_sz, hipHostMallocDefault));
             HIP_CHECK(hipHostMalloc(&D.pin_meta, kPinMetaSz, hipHostMallocDefault));
         }
+        if (ndev > 1) {
+            if (dist_.init_local(ndev, conf.gpu_first) != St::ok)
+                throw std::runtime_error("GpuEngine: RCCL communicator init failed");
+        }
         LOGI("GpuEngine: %d dev (first %d), arena %.1f GiB/GPU, spill cap %.1f GiB",
              ndev, conf.gpu_first, conf.arena_bytes / (double)GiB,
              conf.spill_bytes / (double)GiB);
@@ -208,6 +213,20 @@ class GpuEngine : public Engine {
     St release_chunk(const ChunkRef& ref) override;
     St rs_encode_stripe(const ChunkRef* data, int k, int m, uint32_t shard_len,
                         ChunkRef* parity_out) override;
+
+    St publish_manifest(const ChunkRef* recs, int n) override {
+        if (!dist_.active()) return St::ok;
+        // contribute from slot 0 (single writer process); C5 pre-warm: also
+        // broadcast the new fingerprint tags; C6 epoch barrier
+        St s = dist_.allgather_manifests(0, recs, n, 64);
+        if (s != St::ok) return s;
+        std::vector<uint64_t> tags((size_t)n);
+        for (int i = 0; i < n; i++) std::memcpy(&tags[i], recs[i].fp.b.data(), 8);
+        s = dist_.broadcast_tags(0, tags.data(), n);
+        if (s != St::ok) return s;
+        uint64_t e = epoch_.fetch_add(1) + 1, me = 0;
+        return dist_.epoch_barrier(e, &me);
+    }
 
     St debug_drop_chunk(const Fp& fp) override {
         std::lock_guard<std::mutex> g(map_mu_);
@@ -246,6 +265,8 @@ class GpuEngine : public Engine {
     Conf conf_;
     std::vector<DevCtx> devs_;
     std::atomic<uint32_t> rr_{0};  // ingest/compute device rotation
+    DistCtx dist_;
+    std::atomic<uint64_t> epoch_{0};
 
     mutable std::mutex map_mu_;
     std::unordered_map<Fp, ArenaLoc, FpHash> map_;

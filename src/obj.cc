@@ -200,6 +200,7 @@ St Object::write(uint64_t off, const uint8_t* data, uint64_t len) {
             mf_.chunks[idxs[i]] = refs[i];
             dirty_.erase(idxs[i]);
             mark_stripe_dirty(idxs[i]);
+            pending_publish_.push_back(refs[i]);
         }
     }
     return St::ok;
@@ -303,6 +304,7 @@ St Object::flush_locked() {
             drop_ref(mf_.chunks[idxs[i]]);
             mf_.chunks[idxs[i]] = refs[i];
             mark_stripe_dirty(idxs[i]);
+            pending_publish_.push_back(refs[i]);
         }
     }
     dirty_.clear();
@@ -370,6 +372,14 @@ St Object::commit(std::vector<uint8_t>* manifest_bytes) {
     if (s != St::ok) return s;
     mf_.logical_size = size_;
     if (manifest_bytes) mf_.encode(manifest_bytes);
+    // C4: publish the freshly committed refs so every GPU learns the new
+    // chunk->owner map (RCCL all-gather; no-op single-device)
+    for (size_t i = 0; i < pending_publish_.size(); i += 64) {
+        int nrec = (int)std::min<size_t>(64, pending_publish_.size() - i);
+        s = eng_->publish_manifest(pending_publish_.data() + i, nrec);
+        if (s != St::ok) return s;
+    }
+    pending_publish_.clear();
     return St::ok;
 }
 

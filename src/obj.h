@@ -58,6 +58,7 @@ class Object {
     Manifest mf_;                    // committed chunk refs (index = chunk idx)
     std::map<uint64_t, Dirty> dirty_;
     std::set<uint64_t> rs_dirty_;    // stripe indices with stale parity
+    std::vector<ChunkRef> pending_publish_;  // C4: refs committed since last publish
     uint64_t size_ = 0;              // logical size incl. dirty extension
 };
 
