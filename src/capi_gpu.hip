@@ -162,6 +162,33 @@ int eh_gpu_gf_matmul(const uint8_t* shards, int k, int m, const uint8_t* coef, u
     }
 }
 
+// MFMA bit-plane variant (RS 6+2 shapes; falls back to the table kernel
+// otherwise — tests compare both against the golden).
+int eh_gpu_gf_matmul_mfma(const uint8_t* shards, int k, int m, const uint8_t* coef,
+                          uint64_t len, uint8_t* out) {
+    try {
+        DevBuf d_in, d_out;
+        HIP_CHECK(d_in.alloc(len * k));
+        HIP_CHECK(d_out.alloc(len * m));
+        HIP_CHECK(hipMemcpy(d_in.p, shards, len * k, hipMemcpyHostToDevice));
+        RsDesc d{};
+        for (int j = 0; j < k; j++) d.in[j] = d_in.as<uint8_t>() + len * j;
+        for (int i = 0; i < m; i++) d.out[i] = d_out.as<uint8_t>() + len * i;
+        for (int i = 0; i < m; i++)
+            for (int j = 0; j < k; j++) d.coef[i][j] = coef[i * k + j];
+        d.k = k;
+        d.m = m;
+        d.len = len;
+        launch_gf_matmul_mfma(d, nullptr);
+        HIP_CHECK(hipDeviceSynchronize());
+        HIP_CHECK(hipMemcpy(out, d_out.p, len * m, hipMemcpyDeviceToHost));
+        return 0;
+    } catch (const std::exception& e) {
+        LOGE("eh_gpu_gf_matmul_mfma: %s", e.what());
+        return -1;
+    }
+}
+
 // GPU chunk crc (XOR of per-4KiB-page crc32c).
 int eh_gpu_chunk_crc(const uint8_t* data, uint32_t len, uint32_t* out) {
     try {

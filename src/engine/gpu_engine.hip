@@ -16,6 +16,7 @@
 //
 // Fails LOUDLY when no GPU is present — no silent CPU fallback.
 #include <algorithm>
+#include <condition_variable>
 #include <list>
 #include <mutex>
 #include <stdexcept>
@@ -65,10 +66,24 @@ struct ArenaLoc {
     std::list<Fp>::iterator lru;  // valid when !spilled
 };
 
-// Per-device context: arena + allocator + streams + staging.
+constexpr int kPipeSlots = 2;  // concurrent ingest pipelines per device
+
+// One ingest pipeline: stream + staging + scratch (double-buffered per device
+// so concurrent writers overlap H2D/kernels/D2D across slots — SURVEY §4.3).
+struct PipeSlot {
+    hipStream_t s{};
+    uint8_t *d_stage = nullptr, *d_comp = nullptr, *d_segscratch = nullptr,
+            *d_fps = nullptr;
+    DevChunk *d_chunks = nullptr, *d_copy = nullptr;
+    ulong2* d_segmeta = nullptr;
+    uint32_t *d_segsizes = nullptr, *d_sizes = nullptr, *d_crcs = nullptr;
+    uint8_t *pin_in = nullptr, *pin_meta = nullptr;
+};
+
+// Per-device context: arena + allocator + pipelines.
 struct DevCtx {
     int hip_dev = 0;
-    hipStream_t s_pipe{}, s_read{};
+    hipStream_t s_read{};
     uint8_t* arena = nullptr;
     uint64_t arena_sz = 0;
     std::mutex alloc_mu;
@@ -76,17 +91,35 @@ struct DevCtx {
     uint64_t bump = 0;
     std::atomic<uint64_t> used{0};
 
-    uint8_t *d_stage = nullptr, *d_comp = nullptr, *d_rdstage = nullptr,
-            *d_segscratch = nullptr, *d_fps = nullptr, *d_rs = nullptr;
+    PipeSlot slots[kPipeSlots];
+    std::mutex slot_mu;
+    std::condition_variable slot_cv;
+    uint32_t slot_free_mask = (1u << kPipeSlots) - 1;
+
+    uint8_t *d_rdstage = nullptr, *d_rs = nullptr;
     unsigned long long* d_index = nullptr;  // G7 fingerprint index (HBM)
     uint64_t index_mask = 0;
-    DevChunk *d_chunks = nullptr, *d_copy = nullptr;
-    ulong2* d_segmeta = nullptr;
-    uint32_t *d_segsizes = nullptr, *d_sizes = nullptr, *d_crcs = nullptr;
-    uint8_t *pin_in = nullptr, *pin_out = nullptr, *pin_meta = nullptr;
+    DevChunk* d_rchunks = nullptr;          // read-path descriptors
+    uint8_t *pin_out = nullptr, *pin_rmeta = nullptr;
+    uint32_t* d_rstatus = nullptr;
     uint64_t stage_sz = 0;
-    std::mutex pipe_mu;  // staging buffers + streams
+    std::mutex pipe_mu;  // read-path stream + staging
     std::mutex rs_mu;    // d_rs stripe buffer
+
+    PipeSlot* acquire_slot() {
+        std::unique_lock<std::mutex> lk(slot_mu);
+        slot_cv.wait(lk, [&] { return slot_free_mask != 0; });
+        int i = __builtin_ctz(slot_free_mask);
+        slot_free_mask &= ~(1u << i);
+        return &slots[i];
+    }
+    void release_slot(PipeSlot* s) {
+        {
+            std::lock_guard<std::mutex> lk(slot_mu);
+            slot_free_mask |= 1u << (uint32_t)(s - slots);
+        }
+        slot_cv.notify_one();
+    }
 
     bool arena_alloc(uint32_t bytes, uint64_t* off) {
         int cls = (int)((bytes + kGranule - 1) / kGranule) - 1;
@@ -143,23 +176,30 @@ class GpuEngine : public Engine {
                     LOGW("GpuEngine: peer access %d->%d: %s", D.hip_dev,
                          conf.gpu_first + o, hipGetErrorString(pe));
             }
-            HIP_CHECK(hipStreamCreateWithFlags(&D.s_pipe, hipStreamNonBlocking));
             HIP_CHECK(hipStreamCreateWithFlags(&D.s_read, hipStreamNonBlocking));
             D.arena_sz = conf.arena_bytes;
             HIP_CHECK(hipMalloc(&D.arena, D.arena_sz));
             D.stage_sz = (uint64_t)kMaxBatch * cs;
-            HIP_CHECK(hipMalloc(&D.d_stage, D.stage_sz));
-            HIP_CHECK(hipMalloc(&D.d_comp, D.stage_sz));
+            for (int sl = 0; sl < kPipeSlots; sl++) {
+                PipeSlot& S = D.slots[sl];
+                HIP_CHECK(hipStreamCreateWithFlags(&S.s, hipStreamNonBlocking));
+                HIP_CHECK(hipMalloc(&S.d_stage, D.stage_sz));
+                HIP_CHECK(hipMalloc(&S.d_comp, D.stage_sz));
+                HIP_CHECK(hipMalloc(&S.d_segscratch, (uint64_t)kMaxSegs * kSegStride));
+                HIP_CHECK(hipMalloc(&S.d_chunks, sizeof(DevChunk) * kMaxBatch * 2));
+                HIP_CHECK(hipMalloc(&S.d_segmeta, sizeof(ulong2) * kMaxSegs));
+                HIP_CHECK(hipMalloc(&S.d_segsizes, sizeof(uint32_t) * kMaxSegs));
+                HIP_CHECK(hipMalloc(&S.d_sizes, sizeof(uint32_t) * kMaxBatch));
+                HIP_CHECK(hipMalloc(&S.d_fps, 32 * kMaxBatch));
+                HIP_CHECK(hipMalloc(&S.d_crcs, sizeof(uint32_t) * kMaxBatch * 2));
+                HIP_CHECK(hipMalloc(&S.d_copy, sizeof(DevChunk) * kMaxBatch * 4));
+                HIP_CHECK(hipHostMalloc(&S.pin_in, D.stage_sz, hipHostMallocDefault));
+                HIP_CHECK(hipHostMalloc(&S.pin_meta, kPinMetaSz, hipHostMallocDefault));
+            }
             HIP_CHECK(hipMalloc(&D.d_rdstage, D.stage_sz));
             HIP_CHECK(hipMalloc(&D.d_rs, (uint64_t)(kRsMaxK + kRsMaxM) * cs));
-            HIP_CHECK(hipMalloc(&D.d_segscratch, (uint64_t)kMaxSegs * kSegStride));
-            HIP_CHECK(hipMalloc(&D.d_chunks, sizeof(DevChunk) * kMaxBatch * 2));
-            HIP_CHECK(hipMalloc(&D.d_segmeta, sizeof(ulong2) * kMaxSegs));
-            HIP_CHECK(hipMalloc(&D.d_segsizes, sizeof(uint32_t) * kMaxSegs));
-            HIP_CHECK(hipMalloc(&D.d_sizes, sizeof(uint32_t) * kMaxBatch));
-            HIP_CHECK(hipMalloc(&D.d_fps, 32 * kMaxBatch));
-            HIP_CHECK(hipMalloc(&D.d_crcs, sizeof(uint32_t) * kMaxBatch * 2));
-            HIP_CHECK(hipMalloc(&D.d_copy, sizeof(DevChunk) * kMaxBatch * 2));
+            HIP_CHECK(hipMalloc(&D.d_rchunks, sizeof(DevChunk) * kMaxBatch));
+            HIP_CHECK(hipMalloc(&D.d_rstatus, sizeof(uint32_t) * kMaxBatch));
             // G7: per-device fingerprint index shard in HBM (tags of chunks
             // this device owns); 1/1024 of the arena, power-of-two slots
             {
@@ -170,9 +210,8 @@ class GpuEngine : public Engine {
                 HIP_CHECK(hipMemset(D.d_index, 0, slots * 8));
                 D.index_mask = slots - 1;
             }
-            HIP_CHECK(hipHostMalloc(&D.pin_in, D.stage_sz, hipHostMallocDefault));
             HIP_CHECK(hipHostMalloc(&D.pin_out, D.stage_sz, hipHostMallocDefault));
-            HIP_CHECK(hipHostMalloc(&D.pin_meta, kPinMetaSz, hipHostMallocDefault));
+            HIP_CHECK(hipHostMalloc(&D.pin_rmeta, kPinMetaSz, hipHostMallocDefault));
         }
         if (ndev > 1) {
             if (dist_.init_local(ndev, conf.gpu_first) != St::ok)
@@ -187,16 +226,22 @@ class GpuEngine : public Engine {
         for (auto& D : devs_) {
             (void)hipSetDevice(D.hip_dev);
             (void)hipDeviceSynchronize();
-            for (void* p : {(void*)D.arena, (void*)D.d_stage, (void*)D.d_comp,
-                            (void*)D.d_rdstage, (void*)D.d_rs, (void*)D.d_segscratch,
-                            (void*)D.d_chunks, (void*)D.d_segmeta, (void*)D.d_segsizes,
-                            (void*)D.d_sizes, (void*)D.d_fps, (void*)D.d_crcs,
-                            (void*)D.d_copy, (void*)D.d_index})
+            for (int sl = 0; sl < kPipeSlots; sl++) {
+                PipeSlot& S = D.slots[sl];
+                for (void* p : {(void*)S.d_stage, (void*)S.d_comp, (void*)S.d_segscratch,
+                                (void*)S.d_chunks, (void*)S.d_segmeta, (void*)S.d_segsizes,
+                                (void*)S.d_sizes, (void*)S.d_fps, (void*)S.d_crcs,
+                                (void*)S.d_copy})
+                    (void)hipFree(p);
+                (void)hipHostFree(S.pin_in);
+                (void)hipHostFree(S.pin_meta);
+                (void)hipStreamDestroy(S.s);
+            }
+            for (void* p : {(void*)D.arena, (void*)D.d_rdstage, (void*)D.d_rs,
+                            (void*)D.d_rchunks, (void*)D.d_rstatus, (void*)D.d_index})
                 (void)hipFree(p);
-            (void)hipHostFree(D.pin_in);
             (void)hipHostFree(D.pin_out);
-            (void)hipHostFree(D.pin_meta);
-            (void)hipStreamDestroy(D.s_pipe);
+            (void)hipHostFree(D.pin_rmeta);
             (void)hipStreamDestroy(D.s_read);
         }
     }
@@ -281,13 +326,18 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
     const bool do_fp = conf_.fp == FpAlgo::blake3;
     const int ingest_dev = (int)(rr_.fetch_add(1) % devs_.size());
     DevCtx& D = devs_[ingest_dev];
-
-    std::lock_guard<std::mutex> pg(D.pipe_mu);
+    PipeSlot* SLP = D.acquire_slot();
+    PipeSlot& S = *SLP;
+    struct SlotGuard {
+        DevCtx& d;
+        PipeSlot* s;
+        ~SlotGuard() { d.release_slot(s); }
+    } slot_guard{D, SLP};
     HIP_CHECK(hipSetDevice(D.hip_dev));
 
-    auto* h_chunks = (DevChunk*)(D.pin_meta + kOffChunks);
-    auto* h_chunks2 = (DevChunk*)(D.pin_meta + kOffChunks2);
-    auto* h_segmeta = (ulong2*)(D.pin_meta + kOffSegMeta);
+    auto* h_chunks = (DevChunk*)(S.pin_meta + kOffChunks);
+    auto* h_chunks2 = (DevChunk*)(S.pin_meta + kOffChunks2);
+    auto* h_segmeta = (ulong2*)(S.pin_meta + kOffSegMeta);
     const bool pool_pinned = BufPool::inst().pinned();
     int nsegs = 0;
     uint64_t off = 0;
@@ -295,12 +345,12 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
     for (int i = 0; i < n; i++) {
         if (pool_pinned && put[i].donate && put[i].donate->p == put[i].data) {
             // dirty buffer is already pinned: DMA straight from it
-            HIP_CHECK(hipMemcpyAsync(D.d_stage + off, put[i].data, put[i].len,
-                                     hipMemcpyHostToDevice, D.s_pipe));
+            HIP_CHECK(hipMemcpyAsync(S.d_stage + off, put[i].data, put[i].len,
+                                     hipMemcpyHostToDevice, S.s));
         } else {
-            std::memcpy(D.pin_in + staged, put[i].data, put[i].len);
-            HIP_CHECK(hipMemcpyAsync(D.d_stage + off, D.pin_in + staged, put[i].len,
-                                     hipMemcpyHostToDevice, D.s_pipe));
+            std::memcpy(S.pin_in + staged, put[i].data, put[i].len);
+            HIP_CHECK(hipMemcpyAsync(S.d_stage + off, S.pin_in + staged, put[i].len,
+                                     hipMemcpyHostToDevice, S.s));
             staged += round_up(put[i].len, 64);
         }
         h_chunks[i] = DevChunk{off, off, put[i].len, (uint32_t)nsegs};
@@ -312,44 +362,44 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
         nsegs += ns;
         off += round_up(put[i].len, 64);
     }
-    HIP_CHECK(hipMemcpyAsync(D.d_chunks, h_chunks, sizeof(DevChunk) * n,
-                             hipMemcpyHostToDevice, D.s_pipe));
+    HIP_CHECK(hipMemcpyAsync(S.d_chunks, h_chunks, sizeof(DevChunk) * n,
+                             hipMemcpyHostToDevice, S.s));
     metrics().h2d_bytes += off;
 
-    if (do_fp) launch_blake3_chunks(D.d_stage, D.d_chunks, D.d_fps, n, D.s_pipe);
+    if (do_fp) launch_blake3_chunks(S.d_stage, S.d_chunks, S.d_fps, n, S.s);
     if (do_lz4) {
-        HIP_CHECK(hipMemcpyAsync(D.d_segmeta, h_segmeta, sizeof(ulong2) * nsegs,
-                                 hipMemcpyHostToDevice, D.s_pipe));
-        launch_lz4_compress_segs(D.d_stage, D.d_segmeta, D.d_segscratch, D.d_segsizes, nsegs,
-                                 D.s_pipe);
-        HIP_CHECK(hipMemcpyAsync(D.d_chunks + kMaxBatch, h_chunks2, sizeof(DevChunk) * n,
-                                 hipMemcpyHostToDevice, D.s_pipe));
-        launch_lz4_emit(D.d_stage, D.d_segscratch, D.d_segsizes, D.d_chunks + kMaxBatch,
-                        D.d_comp, D.d_sizes, n, D.s_pipe);
+        HIP_CHECK(hipMemcpyAsync(S.d_segmeta, h_segmeta, sizeof(ulong2) * nsegs,
+                                 hipMemcpyHostToDevice, S.s));
+        launch_lz4_compress_segs(S.d_stage, S.d_segmeta, S.d_segscratch, S.d_segsizes, nsegs,
+                                 S.s);
+        HIP_CHECK(hipMemcpyAsync(S.d_chunks + kMaxBatch, h_chunks2, sizeof(DevChunk) * n,
+                                 hipMemcpyHostToDevice, S.s));
+        launch_lz4_emit(S.d_stage, S.d_segscratch, S.d_segsizes, S.d_chunks + kMaxBatch,
+                        S.d_comp, S.d_sizes, n, S.s);
     }
 
-    auto* h_fps = (uint8_t*)(D.pin_meta + kOffFps);
-    auto* h_sizes = (uint32_t*)(D.pin_meta + kOffSizes);
-    auto* h_crcs = (uint32_t*)(D.pin_meta + kOffCrcs);
+    auto* h_fps = (uint8_t*)(S.pin_meta + kOffFps);
+    auto* h_sizes = (uint32_t*)(S.pin_meta + kOffSizes);
+    auto* h_crcs = (uint32_t*)(S.pin_meta + kOffCrcs);
     // G7: upsert the batch's fingerprints into THIS device's HBM index.
     // (Placement-owner sharding refines this: tags also get upserted into the
     // owner's shard below once owners are known; the ingest-dev upsert gives
     // the fast dup signal for the common single-device case.)
-    auto* h_dups = (uint32_t*)(D.pin_meta + kOffStatus);
+    auto* h_dups = (uint32_t*)(S.pin_meta + kOffStatus);
     if (do_fp && conf_.dedup) {
         // d_crcs doubles as the dup-out scratch here (n <= kMaxBatch)
-        launch_dedup_upsert(D.d_index, D.index_mask, D.d_fps, D.d_crcs, n, D.s_pipe);
-        HIP_CHECK(hipMemcpyAsync(h_dups, D.d_crcs, 4 * n, hipMemcpyDeviceToHost, D.s_pipe));
+        launch_dedup_upsert(D.d_index, D.index_mask, S.d_fps, S.d_crcs, n, S.s);
+        HIP_CHECK(hipMemcpyAsync(h_dups, S.d_crcs, 4 * n, hipMemcpyDeviceToHost, S.s));
     }
     if (do_fp)
-        HIP_CHECK(hipMemcpyAsync(h_fps, D.d_fps, 32 * n, hipMemcpyDeviceToHost, D.s_pipe));
+        HIP_CHECK(hipMemcpyAsync(h_fps, S.d_fps, 32 * n, hipMemcpyDeviceToHost, S.s));
     if (do_lz4)
-        HIP_CHECK(hipMemcpyAsync(h_sizes, D.d_sizes, 4 * n, hipMemcpyDeviceToHost, D.s_pipe));
-    HIP_CHECK(hipStreamSynchronize(D.s_pipe));
+        HIP_CHECK(hipMemcpyAsync(h_sizes, S.d_sizes, 4 * n, hipMemcpyDeviceToHost, S.s));
+    HIP_CHECK(hipStreamSynchronize(S.s));
 
     struct Pend {
         int i;
-        uint64_t src_off;   // in D.d_stage (raw) or D.d_comp (framed)
+        uint64_t src_off;   // in S.d_stage (raw) or S.d_comp (framed)
         bool from_comp;
         uint32_t stored;
         uint64_t arena_off;
@@ -425,9 +475,9 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
         std::vector<int> crc_maps[2];
         if (conf_.crc) {
             for (int pass = 0; pass < 2; pass++) {
-                auto* h_cd = (DevChunk*)(D.pin_meta + kOffCrcDesc) + pass * kMaxBatch;
-                DevChunk* d_cd = D.d_copy + pass * kMaxBatch;
-                uint32_t* d_out = D.d_crcs + 0;  // reused per pass via offsets below
+                auto* h_cd = (DevChunk*)(S.pin_meta + kOffCrcDesc) + pass * kMaxBatch;
+                DevChunk* d_cd = S.d_copy + pass * kMaxBatch;
+                uint32_t* d_out = S.d_crcs + 0;  // reused per pass via offsets below
                 (void)d_out;
                 int m = 0;
                 std::vector<int> idxmap;
@@ -439,20 +489,20 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
                 }
                 if (!m) continue;
                 HIP_CHECK(hipMemcpyAsync(d_cd, h_cd, sizeof(DevChunk) * m,
-                                         hipMemcpyHostToDevice, D.s_pipe));
-                launch_crc32c_chunks(pass ? D.d_comp : D.d_stage, d_cd,
-                                     D.d_crcs + pass * kMaxBatch, m, D.s_pipe);
+                                         hipMemcpyHostToDevice, S.s));
+                launch_crc32c_chunks(pass ? S.d_comp : S.d_stage, d_cd,
+                                     S.d_crcs + pass * kMaxBatch, m, S.s);
                 HIP_CHECK(hipMemcpyAsync(h_crcs + pass * kMaxBatch,
-                                         D.d_crcs + pass * kMaxBatch, 4 * m,
-                                         hipMemcpyDeviceToHost, D.s_pipe));
+                                         S.d_crcs + pass * kMaxBatch, 4 * m,
+                                         hipMemcpyDeviceToHost, S.s));
                 crc_maps[pass] = idxmap;
             }
         }
         // local-owner chunks: batch copy kernel; remote owners: xGMI peer copy
         // (independent of the crc kernels — one sync covers everything)
         for (int pass = 0; pass < 2; pass++) {
-            auto* h_copy = (DevChunk*)(D.pin_meta + kOffCopyDesc) + pass * kMaxBatch;
-            DevChunk* d_cp = D.d_copy + (2 + pass) * kMaxBatch;
+            auto* h_copy = (DevChunk*)(S.pin_meta + kOffCopyDesc) + pass * kMaxBatch;
+            DevChunk* d_cp = S.d_copy + (2 + pass) * kMaxBatch;
             int m = 0;
             for (auto& p : pend) {
                 if ((int)p.from_comp != pass || p.dev != ingest_dev) continue;
@@ -460,17 +510,17 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
             }
             if (!m) continue;
             HIP_CHECK(hipMemcpyAsync(d_cp, h_copy, sizeof(DevChunk) * m,
-                                     hipMemcpyHostToDevice, D.s_pipe));
-            launch_batch_copy(pass ? D.d_comp : D.d_stage, D.arena, d_cp, m, D.s_pipe);
+                                     hipMemcpyHostToDevice, S.s));
+            launch_batch_copy(pass ? S.d_comp : S.d_stage, D.arena, d_cp, m, S.s);
         }
         for (auto& p : pend) {
             if (p.dev == ingest_dev) continue;
             DevCtx& O = devs_[p.dev];
             HIP_CHECK(hipMemcpyPeerAsync(O.arena + p.arena_off, O.hip_dev,
-                                         (p.from_comp ? D.d_comp : D.d_stage) + p.src_off,
-                                         D.hip_dev, p.stored, D.s_pipe));
+                                         (p.from_comp ? S.d_comp : S.d_stage) + p.src_off,
+                                         D.hip_dev, p.stored, S.s));
         }
-        HIP_CHECK(hipStreamSynchronize(D.s_pipe));
+        HIP_CHECK(hipStreamSynchronize(S.s));
         if (conf_.crc) {
             std::lock_guard<std::mutex> g(map_mu_);
             for (int pass = 0; pass < 2; pass++) {
@@ -591,6 +641,19 @@ St GpuEngine::get_chunks(const GetChunk* get, int n) {
     }
     if (dev_reqs.empty()) return St::ok;
 
+    // staging blobs carried out of the per-device locks; user-buffer copies
+    // happen after the streams are released so concurrent readers overlap
+    struct OutCopy {
+        uint8_t* dst;
+        uint64_t src_off;
+        uint32_t len;
+    };
+    struct Staged {
+        Blob blob;
+        std::vector<OutCopy> copies;
+    };
+    std::vector<Staged> staged;
+
     for (size_t d = 0; d < devs_.size(); d++) {
         DevCtx& D = devs_[d];
         std::vector<DevReq*> mine;
@@ -599,8 +662,8 @@ St GpuEngine::get_chunks(const GetChunk* get, int n) {
         if (mine.empty()) continue;
         std::lock_guard<std::mutex> pg(D.pipe_mu);
         HIP_CHECK(hipSetDevice(D.hip_dev));
-        auto* h_dc = (DevChunk*)(D.pin_meta + kOffChunks);
-        auto* h_status = (uint32_t*)(D.pin_meta + kOffStatus);
+        auto* h_dc = (DevChunk*)(D.pin_rmeta + kOffChunks);
+        auto* h_status = (uint32_t*)(D.pin_rmeta + kOffStatus);
 
         size_t j0 = 0;
         while (j0 < mine.size()) {
@@ -624,41 +687,41 @@ St GpuEngine::get_chunks(const GetChunk* get, int n) {
                 mi++;
             }
             if (mi) {
-                HIP_CHECK(hipMemcpyAsync(D.d_chunks, h_dc, sizeof(DevChunk) * mi,
+                HIP_CHECK(hipMemcpyAsync(D.d_rchunks, h_dc, sizeof(DevChunk) * mi,
                                          hipMemcpyHostToDevice, D.s_read));
-                launch_lz4_decompress_chunks(D.arena, D.d_chunks, D.d_rdstage, D.d_sizes, mi,
-                                             D.s_read);
-                HIP_CHECK(hipMemcpyAsync(h_status, D.d_sizes, 4 * mi, hipMemcpyDeviceToHost,
-                                         D.s_read));
+                launch_lz4_decompress_chunks(D.arena, D.d_rchunks, D.d_rdstage, D.d_rstatus,
+                                             mi, D.s_read);
+                HIP_CHECK(hipMemcpyAsync(h_status, D.d_rstatus, 4 * mi,
+                                         hipMemcpyDeviceToHost, D.s_read));
             }
+            Staged st;
+            st.blob = BufPool::inst().get((uint32_t)std::max<uint64_t>(outb, 64));
             uint64_t pout = 0;
             mi = 0;
             for (size_t j = j0; j < j1; j++) {
                 const GetChunk& g = *mine[j]->g;
-                const uint8_t* src;
+                const uint8_t* srcp;
                 if (mine[j]->compressed) {
-                    src = D.d_rdstage + (uint64_t)mi * cs + g.off;
+                    srcp = D.d_rdstage + (uint64_t)mi * cs + g.off;
                     mi++;
                 } else {
-                    src = D.arena + mine[j]->arena_off + g.off;
+                    srcp = D.arena + mine[j]->arena_off + g.off;
                 }
-                HIP_CHECK(hipMemcpyAsync(D.pin_out + pout, src, g.len, hipMemcpyDeviceToHost,
-                                         D.s_read));
+                HIP_CHECK(hipMemcpyAsync(st.blob.p + pout, srcp, g.len,
+                                         hipMemcpyDeviceToHost, D.s_read));
+                st.copies.push_back({g.out, pout, g.len});
                 pout += round_up(g.len, 64);
             }
             HIP_CHECK(hipStreamSynchronize(D.s_read));
             metrics().d2h_bytes += pout;
             for (int x = 0; x < mi; x++)
                 if (h_status[x]) return St::bad_format;
-            pout = 0;
-            for (size_t j = j0; j < j1; j++) {
-                const GetChunk& g = *mine[j]->g;
-                std::memcpy(g.out, D.pin_out + pout, g.len);
-                pout += round_up(g.len, 64);
-            }
+            staged.push_back(std::move(st));
             j0 = j1;
         }
     }
+    for (auto& st : staged)
+        for (auto& c : st.copies) std::memcpy(c.dst, st.blob.p + c.src_off, c.len);
     return St::ok;
 }
 
@@ -670,7 +733,7 @@ St GpuEngine::gather_chunk_to_dev(const ChunkRef& ref, int ddev, uint8_t* dst) {
     DevCtx& T = devs_[ddev];
     HIP_CHECK(hipSetDevice(T.hip_dev));
     if (ref.fp.is_zero()) {  // hole = zero shard
-        HIP_CHECK(hipMemsetAsync(dst, 0, cs, T.s_pipe));
+        HIP_CHECK(hipMemsetAsync(dst, 0, cs, T.s_read));
         return St::ok;
     }
     uint64_t aoff = 0;
@@ -694,27 +757,27 @@ St GpuEngine::gather_chunk_to_dev(const ChunkRef& ref, int ddev, uint8_t* dst) {
         if (s != St::ok) return s;
         HIP_CHECK(hipSetDevice(T.hip_dev));
         HIP_CHECK(hipMemcpyAsync(dst, full.data(), ref.logical_len, hipMemcpyHostToDevice,
-                                 T.s_pipe));
+                                 T.s_read));
     } else if (!comp) {
         DevCtx& O = devs_[odev];
         if (odev == ddev) {
             HIP_CHECK(hipMemcpyAsync(dst, O.arena + aoff, stored, hipMemcpyDeviceToDevice,
-                                     T.s_pipe));
+                                     T.s_read));
         } else {
             HIP_CHECK(hipMemcpyPeerAsync(dst, T.hip_dev, O.arena + aoff, O.hip_dev, stored,
-                                         T.s_pipe));
+                                         T.s_read));
         }
     } else {
         DevCtx& O = devs_[odev];
         std::lock_guard<std::mutex> og(O.pipe_mu);
         HIP_CHECK(hipSetDevice(O.hip_dev));
-        auto* h_dc = (DevChunk*)(O.pin_meta + kOffChunks);
-        auto* h_status = (uint32_t*)(O.pin_meta + kOffStatus);
+        auto* h_dc = (DevChunk*)(O.pin_rmeta + kOffChunks);
+        auto* h_status = (uint32_t*)(O.pin_rmeta + kOffStatus);
         h_dc[0] = DevChunk{aoff, 0, ref.logical_len, stored};
-        HIP_CHECK(hipMemcpyAsync(O.d_chunks, h_dc, sizeof(DevChunk), hipMemcpyHostToDevice,
+        HIP_CHECK(hipMemcpyAsync(O.d_rchunks, h_dc, sizeof(DevChunk), hipMemcpyHostToDevice,
                                  O.s_read));
-        launch_lz4_decompress_chunks(O.arena, O.d_chunks, O.d_rdstage, O.d_sizes, 1, O.s_read);
-        HIP_CHECK(hipMemcpyAsync(h_status, O.d_sizes, 4, hipMemcpyDeviceToHost, O.s_read));
+        launch_lz4_decompress_chunks(O.arena, O.d_rchunks, O.d_rdstage, O.d_rstatus, 1, O.s_read);
+        HIP_CHECK(hipMemcpyAsync(h_status, O.d_rstatus, 4, hipMemcpyDeviceToHost, O.s_read));
         HIP_CHECK(hipStreamSynchronize(O.s_read));
         if (h_status[0]) return St::bad_format;
         if (odev == ddev) {
@@ -728,7 +791,7 @@ St GpuEngine::gather_chunk_to_dev(const ChunkRef& ref, int ddev, uint8_t* dst) {
     }
     if (ref.logical_len < cs) {
         HIP_CHECK(hipSetDevice(T.hip_dev));
-        HIP_CHECK(hipMemsetAsync(dst + ref.logical_len, 0, cs - ref.logical_len, T.s_pipe));
+        HIP_CHECK(hipMemsetAsync(dst + ref.logical_len, 0, cs - ref.logical_len, T.s_read));
     }
     return St::ok;
 }
@@ -748,7 +811,7 @@ St GpuEngine::rs_encode_stripe(const ChunkRef* data, int k, int m, uint32_t shar
         if (s != St::ok) return s;
     }
     HIP_CHECK(hipSetDevice(C.hip_dev));
-    HIP_CHECK(hipStreamSynchronize(C.s_pipe));
+    HIP_CHECK(hipStreamSynchronize(C.s_read));
     RsDesc d{};
     auto M = rs_parity_matrix(k, m);
     for (int j = 0; j < k; j++) d.in[j] = C.d_rs + (uint64_t)j * cs;
@@ -758,18 +821,15 @@ St GpuEngine::rs_encode_stripe(const ChunkRef* data, int k, int m, uint32_t shar
     d.k = k;
     d.m = m;
     d.len = cs;
-    launch_gf_matmul(d, C.s_pipe);
-    HIP_CHECK(hipStreamSynchronize(C.s_pipe));
+    launch_gf_matmul(d, C.s_read);
+    HIP_CHECK(hipStreamSynchronize(C.s_read));
     // D2H parity, then ingest through the normal pipeline (hash/compress/crc)
     std::vector<uint8_t> host((size_t)m * cs);
-    {
-        std::lock_guard<std::mutex> pg(C.pipe_mu);
-        for (int i = 0; i < m; i++)
-            HIP_CHECK(hipMemcpyAsync(host.data() + (size_t)i * cs,
-                                     C.d_rs + (uint64_t)(k + i) * cs, cs,
-                                     hipMemcpyDeviceToHost, C.s_read));
-        HIP_CHECK(hipStreamSynchronize(C.s_read));
-    }
+    for (int i = 0; i < m; i++)
+        HIP_CHECK(hipMemcpyAsync(host.data() + (size_t)i * cs,
+                                 C.d_rs + (uint64_t)(k + i) * cs, cs,
+                                 hipMemcpyDeviceToHost, C.s_read));
+    HIP_CHECK(hipStreamSynchronize(C.s_read));
     std::vector<PutChunk> puts((size_t)m);
     for (int i = 0; i < m; i++) puts[i] = {host.data() + (size_t)i * cs, cs, nullptr};
     St s = put_chunks(puts.data(), parity_out, m);

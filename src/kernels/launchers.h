@@ -40,5 +40,6 @@ struct RsDesc {
     uint64_t len;
 };
 void launch_gf_matmul(const RsDesc& d, hipStream_t stream);
+void launch_gf_matmul_mfma(const RsDesc& d, hipStream_t stream);  // bit-plane MFMA variant
 
 }  // namespace edgehip

@@ -164,4 +164,106 @@ void launch_gf_matmul(const RsDesc& d, hipStream_t stream) {
         hipLaunchKernelGGL(k_gf_matmul, dim3(blocks), dim3(256), 0, stream, d);
 }
 
+// ---------------------------------------------------------------------------
+// MFMA bit-plane variant (the north-star-mandated showcase): GF(256) is
+// linear over GF(2), so out_bits = BigM (8m x 8k bit matrix) x in_bits mod 2,
+// evaluated with v_mfma_i32_16x16x64_i8 on 0/1 operands — one MFMA computes
+// 16 output bit-rows x 16 byte positions over a 64-deep bit-plane K.
+// RS 6+2 maps perfectly: M = 8*2 = 16 rows, K = 8*6 = 48 (pad to 64).
+//
+// Per-wave step: lane l holds A row (l&15) k-chunk (l>>4)*16 (the constant
+// BigM fragment, built once), and B col n=(l&15): bit-expansion of bytes
+// shard[2g][n], shard[2g+1][n] with g=l>>4 (inverse-movmask multiply trick).
+// D lane l holds rows (l>>4)*4+q, col l&15 (q=0..3); parity bits collect via
+// __ballot(D&1): ballot q covers rows {q, 4+q, 8+q, 12+q} x cols 0..15.
+//
+// Honest expectation (SURVEY §8.4 hard part 2, Appendix A): RS 6+2 is only
+// ~2.33 GF-MACs/byte, so the bitslice/deslice VALU dominates and the MFMA
+// pipe is mostly idle; the v_perm table kernel above is the production
+// default, this variant exists for the measured comparison (profiles/).
+using i8x16 = __attribute__((ext_vector_type(16))) char;
+using i32x4 = __attribute__((ext_vector_type(4))) int;
+
+EH_DEV uint64_t bit_expand8(uint32_t x) {
+    // byte i of result = (x >> i) & 1   (little-endian inverse movmskb)
+    uint64_t y = ((uint64_t)x * 0x0101010101010101ull) & 0x8040201008040201ull;
+    uint64_t z = ((y + 0x7F7F7F7F7F7F7F7Full) | y) & 0x8080808080808080ull;
+    return z >> 7;
+}
+
+template <int K, int M>
+__global__ __launch_bounds__(256) void k_gf_matmul_mfma(RsDesc d) {
+#if defined(__gfx950__)
+    static_assert(8 * M <= 16 && 8 * K <= 64, "RS shape must fit one MFMA tile");
+    const int lane = threadIdx.x & 63;
+    const int row = lane & 15;        // A row / D col
+    const int g = lane >> 4;          // k-chunk (16 k per chunk)
+
+    // constant A fragment: A[row][k] for k in [16g, 16g+16)
+    i8x16 a_frag;
+#pragma unroll
+    for (int t = 0; t < 16; t++) {
+        int kk = g * 16 + t;
+        int j = kk >> 3, bi = kk & 7;
+        uint8_t v = 0;
+        if (row < 8 * M && j < K)
+            v = (gf_mul_slow(d.coef[row >> 3][j], (uint8_t)(1u << bi)) >> (row & 7)) & 1;
+        a_frag[t] = (char)v;
+    }
+
+    const uint64_t nsteps = d.len / 16;  // 16 byte positions per MFMA
+    const uint64_t stride = gridDim.x * (blockDim.x / 64);
+    const uint64_t wave = blockIdx.x * (blockDim.x / 64) + (threadIdx.x / 64);
+    for (uint64_t s = wave; s < nsteps; s += stride) {
+        const uint64_t base = s * 16;
+        // B fragment: col n=row, k in [16g,16g+16) = bits of shards 2g, 2g+1
+        uint32_t x0 = (2 * g + 0) < K ? d.in[2 * g + 0][base + row] : 0;
+        uint32_t x1 = (2 * g + 1) < K ? d.in[2 * g + 1][base + row] : 0;
+        uint64_t e0 = bit_expand8(x0);
+        uint64_t e1 = bit_expand8(x1);
+        i8x16 b_frag;
+        __builtin_memcpy(&b_frag, &e0, 8);
+        __builtin_memcpy((char*)&b_frag + 8, &e1, 8);
+
+        i32x4 acc = {0, 0, 0, 0};
+        acc = __builtin_amdgcn_mfma_i32_16x16x64_i8(
+            *(const __attribute__((ext_vector_type(4))) int*)&a_frag,
+            *(const __attribute__((ext_vector_type(4))) int*)&b_frag, acc, 0, 0, 0);
+
+        // collect parity bits: ballot q covers rows {q,4+q,8+q,12+q} cols 0-15
+        uint64_t bal[4];
+#pragma unroll
+        for (int q = 0; q < 4; q++) bal[q] = __ballot(acc[q] & 1);
+
+        // lanes 0..15 assemble parity bytes for their column n = lane
+        if (lane < 16 * M) {
+            int p = lane >> 4;  // parity shard (M<=1 handled by guard)
+            int n = lane & 15;
+            // parity byte bits b = row p*8+b; row r -> ballot q=r&3, group r>>2
+            uint8_t out = 0;
+#pragma unroll
+            for (int b = 0; b < 8; b++) {
+                int r = p * 8 + b;
+                out |= (uint8_t)(((bal[r & 3] >> (16 * (r >> 2) + n)) & 1) << b);
+            }
+            d.out[p][base + n] = out;
+        }
+        // M==2: lanes 16..31 cover parity 1 via the branch above (lane<32)
+    }
+#else
+    (void)d;
+#endif
+}
+
+void launch_gf_matmul_mfma(const RsDesc& d, hipStream_t stream) {
+    uint64_t waves = d.len / 16;
+    int blocks = (int)std::min<uint64_t>((waves + 3) / 4, 2048);
+    if (blocks < 1) blocks = 1;
+    if (d.k == 6 && d.m == 2 && d.len % 16 == 0) {
+        hipLaunchKernelGGL((k_gf_matmul_mfma<6, 2>), dim3(blocks), dim3(256), 0, stream, d);
+    } else {
+        launch_gf_matmul(d, stream);  // fallback: table kernel
+    }
+}
+
 }  // namespace edgehip
