@@ -16,7 +16,7 @@ HIPFLAGS ?= -O3 -std=c++17 --offload-arch=$(ARCH) -fPIC -MMD -MP
 LDFLAGS  ?= -L$(ROCM)/lib -lamdhip64 -lrccl -Wl,-rpath,$(ROCM)/lib -pthread
 
 HOST_SRCS := src/obs.cc src/bufpool.cc src/conf.cc src/manifest.cc src/blake3.cc src/lz4.cc \
-             src/gf256.cc src/journal.cc src/store.cc src/obj.cc src/vfs.cc src/fuse.cc \
+             src/cdc.cc src/gf256.cc src/journal.cc src/store.cc src/obj.cc src/vfs.cc src/fuse.cc \
              src/engine/engine.cc src/dist/dist.cc
 HIP_SRCS  := src/engine/gpu_engine.hip src/kernels/fingerprint.hip \
              src/kernels/lz4.hip src/kernels/rs_gf256.hip \

@@ -49,6 +49,7 @@ static bool parse_int(const std::string& v, int* out) {
 St Conf::apply_kv(const std::string& k, const std::string& v) {
     auto bad = [&]() { LOGE("conf: bad value for %s: '%s'", k.c_str(), v.c_str()); return St::inval; };
     if (k == "chunk_size") return parse_size(v, &chunk_size) && chunk_size >= 4 * KiB ? St::ok : bad();
+    if (k == "cdc") return parse_bool(v, &cdc) ? St::ok : bad();
     if (k == "fp") {
         if (v == "none") fp = FpAlgo::none;
         else if (v == "blake3") fp = FpAlgo::blake3;

@@ -19,6 +19,7 @@ enum class EngineKind : uint8_t { host = 0, gpu = 1 };
 struct Conf {
     // data layout
     uint64_t chunk_size = 1 * MiB;     // fixed 1 MiB default
+    bool cdc = false;                  // §9 Q2: content-defined chunking (gear)
     FpAlgo fp = FpAlgo::blake3;        // §9 Q3
     Compress compress = Compress::lz4;
     bool dedup = true;                 // fingerprint-index dedup on ingest

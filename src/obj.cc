@@ -1,5 +1,8 @@
 #include "obj.h"
 
+#include <algorithm>
+
+#include "cdc.h"
 #include "obs.h"
 
 namespace edgehip {
@@ -24,7 +27,8 @@ void Object::mark_stripe_dirty(uint64_t chunk_idx) {
 St Object::rs_update_locked() {
     const int k = conf_->rs_k, m = conf_->rs_m;
     if (k <= 0 || m <= 0) return St::ok;
-    const uint32_t cs = (uint32_t)conf_->chunk_size;
+    // CDC chunks may reach 4x the average size (cdc.cc max bound)
+    const uint32_t cs = (uint32_t)(conf_->cdc ? 4 * conf_->chunk_size : conf_->chunk_size);
     for (auto it = rs_dirty_.begin(); it != rs_dirty_.end();) {
         uint64_t s = *it;
         // any still-dirty data chunk in the stripe? leave for the next flush
@@ -69,7 +73,7 @@ St Object::rs_update_locked() {
 St Object::rs_reconstruct_chunk_locked(uint64_t idx, uint8_t* chunk_buf) {
     const int k = conf_->rs_k, m = conf_->rs_m;
     if (k <= 0 || m <= 0) return St::not_found;
-    const uint32_t cs = (uint32_t)conf_->chunk_size;
+    const uint32_t cs = (uint32_t)(conf_->cdc ? 4 * conf_->chunk_size : conf_->chunk_size);
     uint64_t s = idx / k;
     if (mf_.parity.size() < (s + 1) * (uint64_t)m) return St::not_found;  // no parity yet
     std::vector<ChunkRef> refs((size_t)(k + m));
@@ -115,6 +119,7 @@ St Object::load_chunk_for_rmw(uint64_t idx, Dirty* d) {
 
 St Object::write(uint64_t off, const uint8_t* data, uint64_t len) {
     std::lock_guard<std::mutex> g(mu_);
+    if (conf_->cdc) return cdc_write_locked(off, data, len);
     const uint64_t cs = conf_->chunk_size;
     uint64_t pos = off;
     const uint8_t* p = data;
@@ -211,6 +216,7 @@ St Object::read(uint64_t off, uint64_t len, uint8_t* out, uint64_t* got) {
     if (off >= size_) { *got = 0; return St::ok; }
     len = std::min(len, size_ - off);
     *got = len;
+    if (conf_->cdc) return cdc_read_locked(off, len, out);
     const uint64_t cs = conf_->chunk_size;
     uint64_t pos = off, end = off + len;
     uint8_t* o = out;
@@ -271,6 +277,7 @@ St Object::read(uint64_t off, uint64_t len, uint8_t* out, uint64_t* got) {
 }
 
 St Object::flush_locked() {
+    if (conf_->cdc) return cdc_cut_locked(true);
     if (dirty_.empty()) return St::ok;
     std::vector<uint64_t> idxs;
     std::vector<PutChunk> puts;
@@ -319,6 +326,20 @@ St Object::flush() {
 
 St Object::truncate(uint64_t size) {
     std::lock_guard<std::mutex> g(mu_);
+    if (conf_->cdc) {
+        // CDC objects: truncate(0) resets; other sizes unsupported (append
+        // streams — SURVEY §9 Q2 optional mode)
+        if (size != 0) return size == size_ ? St::ok : St::inval;
+        for (auto& c : mf_.chunks) drop_ref(c);
+        for (auto& c : mf_.parity) drop_ref(c);
+        mf_.chunks.clear();
+        mf_.parity.clear();
+        cum_.clear();
+        tail_.clear();
+        rs_dirty_.clear();
+        size_ = 0;
+        return St::ok;
+    }
     const uint64_t cs = conf_->chunk_size;
     if (size < size_) {
         uint64_t nchunks = div_up(size, cs);
@@ -385,9 +406,105 @@ St Object::commit(std::vector<uint8_t>* manifest_bytes) {
 
 St Object::set_manifest(const Manifest& m) {
     std::lock_guard<std::mutex> g(mu_);
-    if (!dirty_.empty()) return St::inval;
+    if (!dirty_.empty() || !tail_.empty()) return St::inval;
     mf_ = m;
     size_ = m.logical_size;
+    if (conf_->cdc) {
+        cum_.clear();
+        uint64_t acc = 0;
+        for (auto& c : mf_.chunks) {
+            acc += c.logical_len;
+            cum_.push_back(acc);
+        }
+    }
+    return St::ok;
+}
+
+// ---- CDC mode (G2): gear-boundary append stream ---------------------------
+St Object::cdc_write_locked(uint64_t off, const uint8_t* data, uint64_t len) {
+    if (off != size_) return St::inval;  // append-only (docs/options.md)
+    tail_.insert(tail_.end(), data, data + len);
+    size_ += len;
+    metrics().write_ops++;
+    metrics().write_bytes += len;
+    // cut complete chunks once enough stream is buffered (keep up to one
+    // max-chunk of tail so a boundary near the end can still move)
+    if (tail_.size() >= 8 * conf_->chunk_size) return cdc_cut_locked(false);
+    return St::ok;
+}
+
+St Object::cdc_cut_locked(bool final_tail) {
+    if (tail_.empty()) {
+        mf_.logical_size = size_;
+        return final_tail ? rs_update_locked() : St::ok;
+    }
+    std::vector<size_t> ends;
+    cdc_boundaries(tail_.data(), tail_.size(), conf_->chunk_size, final_tail, &ends);
+    // keep the last (possibly still-moving) piece buffered unless finalizing
+    size_t cut_upto = 0;
+    std::vector<PutChunk> puts;
+    for (size_t e : ends) {
+        puts.push_back({tail_.data() + cut_upto, (uint32_t)(e - cut_upto), nullptr});
+        cut_upto = e;
+    }
+    if (!puts.empty()) {
+        std::vector<ChunkRef> refs(puts.size());
+        St s = eng_->put_chunks(puts.data(), refs.data(), (int)puts.size());
+        if (s != St::ok) return s;
+        for (auto& r : refs) {
+            mf_.chunks.push_back(r);
+            cum_.push_back((cum_.empty() ? 0 : cum_.back()) + r.logical_len);
+            mark_stripe_dirty(mf_.chunks.size() - 1);
+            pending_publish_.push_back(r);
+        }
+        tail_.erase(tail_.begin(), tail_.begin() + cut_upto);
+    }
+    mf_.logical_size = size_;
+    if (final_tail) return rs_update_locked();
+    return St::ok;
+}
+
+St Object::cdc_read_locked(uint64_t off, uint64_t len, uint8_t* out) {
+    uint64_t committed = cum_.empty() ? 0 : cum_.back();
+    uint64_t pos = off;
+    uint8_t* o = out;
+    uint64_t end = off + len;
+    std::vector<GetChunk> gets;
+    while (pos < end && pos < committed) {
+        // binary search: first chunk whose cumulative end > pos
+        size_t lo = std::upper_bound(cum_.begin(), cum_.end(), pos) - cum_.begin();
+        uint64_t cstart = lo == 0 ? 0 : cum_[lo - 1];
+        const ChunkRef& r = mf_.chunks[lo];
+        uint32_t co = (uint32_t)(pos - cstart);
+        uint32_t n = (uint32_t)std::min<uint64_t>(r.logical_len - co, end - pos);
+        gets.push_back({&r, o, co, n});
+        pos += n;
+        o += n;
+    }
+    if (!gets.empty()) {
+        St s = eng_->get_chunks(gets.data(), (int)gets.size());
+        if (s == St::not_found && conf_->rs_k > 0) {
+            std::vector<uint8_t> cbuf(4 * conf_->chunk_size + 64);
+            for (auto& g2 : gets) {
+                St s2 = eng_->get_chunks(&g2, 1);
+                if (s2 == St::ok) continue;
+                if (s2 != St::not_found) return s2;
+                uint64_t idx = (uint64_t)(g2.ref - &mf_.chunks[0]);
+                s2 = rs_reconstruct_chunk_locked(idx, cbuf.data());
+                if (s2 != St::ok) return s2;
+                std::memcpy(g2.out, cbuf.data() + g2.off, g2.len);
+            }
+            s = St::ok;
+        }
+        if (s != St::ok) return s;
+    }
+    // tail bytes
+    if (pos < end) {
+        uint64_t toff = pos - committed;
+        std::memcpy(o, tail_.data() + toff, end - pos);
+    }
+    metrics().read_ops++;
+    metrics().read_bytes += len;
     return St::ok;
 }
 

@@ -59,6 +59,12 @@ class Object {
     std::map<uint64_t, Dirty> dirty_;
     std::set<uint64_t> rs_dirty_;    // stripe indices with stale parity
     std::vector<ChunkRef> pending_publish_;  // C4: refs committed since last publish
+    // CDC mode (conf.cdc): append-only stream, variable chunk boundaries
+    std::vector<uint64_t> cum_;   // cumulative END offset per committed chunk
+    std::vector<uint8_t> tail_;   // uncut stream tail
+    St cdc_write_locked(uint64_t off, const uint8_t* data, uint64_t len);
+    St cdc_cut_locked(bool final_tail);
+    St cdc_read_locked(uint64_t off, uint64_t len, uint8_t* out);
     uint64_t size_ = 0;              // logical size incl. dirty extension
 };
 

@@ -1,0 +1,94 @@
+"""Content-defined chunking (G2, cdc=on): gear boundaries, append streams,
+dedup across SHIFTED content (the property fixed chunking cannot give)."""
+
+import pytest
+
+from edgehip.lib import Engine
+from tests.conftest import rand_bytes
+
+CDC = "engine=host,fp=blake3,compress=none,dedup=1,cdc=1,chunk_size=256k"
+MB = 1 << 20
+
+
+def test_roundtrip_and_boundaries():
+    with Engine(CDC) as eng, eng.object() as o:
+        data = rand_bytes(8 * MB, seed=1)
+        o.write(0, data)
+        o.flush()
+        assert o.read(0, len(data)) == data
+        # chunk count should be near len/avg (256k) — between len/1M and len/64K
+        n = eng.stored_chunks
+        assert 8 <= n <= 128, n
+
+
+def test_unaligned_reads():
+    with Engine(CDC) as eng, eng.object() as o:
+        data = rand_bytes(4 * MB + 999, seed=2)
+        o.write(0, data)
+        o.flush()
+        import random
+        rng = random.Random(3)
+        for _ in range(50):
+            off = rng.randrange(0, len(data) - 1)
+            ln = rng.randrange(1, min(300000, len(data) - off))
+            assert o.read(off, ln) == data[off:off + ln]
+
+
+def test_tail_visible_before_flush():
+    with Engine(CDC) as eng, eng.object() as o:
+        o.write(0, b"hello ")
+        o.write(6, b"world")
+        assert o.read(0, 11) == b"hello world"
+        o.flush()
+        assert o.read(0, 11) == b"hello world"
+
+
+def test_append_only_enforced():
+    with Engine(CDC) as eng, eng.object() as o:
+        o.write(0, b"abc")
+        with pytest.raises(OSError):
+            o.write(100, b"gap")
+
+
+def test_dedup_across_shift():
+    """Insert bytes at the front; CDC re-syncs so most chunks still dedup.
+    Fixed chunking would dedup nothing."""
+    with Engine(CDC) as eng:
+        base = rand_bytes(16 * MB, seed=5)
+        with eng.object() as o1:
+            o1.write(0, base)
+            o1.flush()
+        chunks_before = eng.stored_chunks
+        bytes_before = eng.stored_bytes
+        with eng.object() as o2:
+            o2.write(0, b"X" * 777 + base)  # shift everything by 777 bytes
+            o2.flush()
+            assert o2.read(777, 1 << 20) == base[:1 << 20]
+        added = eng.stored_bytes - bytes_before
+        # only the first chunk(s) differ; the rest re-align and dedup
+        assert added < 4 * MB, f"CDC failed to re-sync: {added} new bytes"
+        assert eng.stored_chunks < chunks_before * 2
+
+
+def test_cdc_streaming_cut_equals_oneshot():
+    """Boundaries must not depend on write-call granularity."""
+    data = rand_bytes(12 * MB, seed=6)
+    with Engine(CDC) as e1, e1.object() as o1:
+        o1.write(0, data)
+        m1 = o1.commit()
+    with Engine(CDC) as e2, e2.object() as o2:
+        for off in range(0, len(data), 100000):
+            o2.write(off, data[off:off + 100000])
+        m2 = o2.commit()
+    # same content -> same chunk fingerprints (oid differs: zero here), so
+    # compare the ref arrays
+    assert m1[64:] == m2[64:]
+
+
+def test_cdc_with_rs():
+    with Engine("engine=host,fp=blake3,compress=none,dedup=1,cdc=1,"
+                "chunk_size=256k,rs=6+2") as eng, eng.object() as o:
+        data = rand_bytes(6 * MB, seed=7)
+        o.write(0, data)
+        o.commit()
+        assert o.read(0, len(data)) == data
