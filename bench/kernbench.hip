@@ -113,6 +113,9 @@ int main(int argc, char** argv) {
         r.m = 2;
         r.len = shard;
         bench("rs_6+2_decode", (double)shard * 8, [&] { launch_gf_matmul(r, nullptr); });
+        // the MFMA bit-plane variant (north-star mandated comparison)
+        bench("rs_6+2_encode_mfma", (double)shard * 8,
+              [&] { launch_gf_matmul_mfma(d, nullptr); });
     }
 
     // dedup index upsert

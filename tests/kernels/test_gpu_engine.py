@@ -95,3 +95,19 @@ def test_commit_manifest():
         m = o.commit()
         assert len(m) == 64 + 3 * 64 + 4
         assert o.read(0, len(data)) == data
+
+
+def test_gpu_cdc_mode():
+    from edgehip.lib import Engine
+    from tests.conftest import rand_bytes as rb
+    with Engine("engine=gpu,fp=blake3,compress=lz4,dedup=1,cdc=1,chunk_size=256k,"
+                "arena=2g,spill=0") as eng, eng.object() as o:
+        data = rb(16 << 20, seed=9)
+        o.write(0, data)
+        o.flush()
+        assert o.read(0, len(data)) == data
+        n1 = eng.stored_bytes
+        with eng.object() as o2:
+            o2.write(0, b"Y" * 123 + data)
+            o2.flush()
+        assert eng.stored_bytes - n1 < 5 << 20  # shifted content re-dedups

@@ -200,3 +200,21 @@ def test_dedup_index_hbm(lib):
     assert sum(out1) == n // 2
     # pass 2: everything is a dup
     assert sum(out2) == n
+
+
+def test_gf_matmul_mfma_vs_golden(lib):
+    """North-star-mandated MFMA bit-plane GF(256) variant: bit-exact vs the
+    golden AND vs the v_perm table kernel."""
+    so = lib._so
+    so.eh_gpu_gf_matmul_mfma.argtypes = [ctypes.c_void_p, ctypes.c_int, ctypes.c_int,
+                                         ctypes.c_void_p, ctypes.c_uint64, ctypes.c_void_p]
+    so.eh_gpu_gf_matmul_mfma.restype = ctypes.c_int
+    k, m, ln = 6, 2, 1 << 20
+    shards = [rand_bytes(ln, seed=200 + i) for i in range(k)]
+    M = ggf.parity_matrix(k, m)
+    coef = bytes(c for row in M for c in row)
+    out = ctypes.create_string_buffer(ln * m)
+    assert so.eh_gpu_gf_matmul_mfma(b"".join(shards), k, m, coef, ln, out) == 0
+    golden = ggf.encode(k, m, shards)
+    for i in range(m):
+        assert out.raw[i * ln:(i + 1) * ln] == golden[i], f"parity {i}"
