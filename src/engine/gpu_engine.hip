@@ -151,8 +151,12 @@ struct DevCtx {
 class GpuEngine : public Engine {
   public:
     explicit GpuEngine(const Conf& conf) : conf_(conf), spill_(conf.spill_bytes) {
-        if (conf.chunk_size > 1 * MiB)
-            throw std::runtime_error("GpuEngine: chunk_size > 1 MiB not supported");
+        // CDC chunks reach 4x the average (cdc.cc max bound); staging strides
+        // are sized for the largest chunk the object layer can produce.
+        max_chunk_ = conf.cdc ? 4 * conf.chunk_size : conf.chunk_size;
+        if (max_chunk_ > 1 * MiB)
+            throw std::runtime_error(
+                "GpuEngine: max chunk (chunk_size, or 4x for cdc) > 1 MiB unsupported");
         int nvis = 0;
         hipError_t e = hipGetDeviceCount(&nvis);
         if (e != hipSuccess || nvis <= conf.gpu_first)
@@ -163,7 +167,7 @@ class GpuEngine : public Engine {
             LOGW("GpuEngine: requested %d GPUs, only %d visible — using %d", conf.ngpus,
                  nvis - conf.gpu_first, ndev);
         devs_ = std::vector<DevCtx>(ndev);
-        const uint64_t cs = conf.chunk_size;
+        const uint64_t cs = max_chunk_;
         for (int d = 0; d < ndev; d++) {
             DevCtx& D = devs_[d];
             D.hip_dev = conf.gpu_first + d;
@@ -308,6 +312,7 @@ class GpuEngine : public Engine {
     St gather_chunk_to_dev(const ChunkRef& ref, int ddev, uint8_t* dst);
 
     Conf conf_;
+    uint64_t max_chunk_ = 0;
     std::vector<DevCtx> devs_;
     std::atomic<uint32_t> rr_{0};  // ingest/compute device rotation
     DistCtx dist_;
@@ -321,7 +326,7 @@ class GpuEngine : public Engine {
 };
 
 St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
-    const uint32_t cs = (uint32_t)conf_.chunk_size;
+    const uint32_t cs = (uint32_t)max_chunk_;
     const bool do_lz4 = conf_.compress == Compress::lz4;
     const bool do_fp = conf_.fp == FpAlgo::blake3;
     const int ingest_dev = (int)(rr_.fetch_add(1) % devs_.size());
@@ -569,7 +574,7 @@ St GpuEngine::evict_for(int dev, uint32_t need_bytes) {
 }
 
 St GpuEngine::get_chunks(const GetChunk* get, int n) {
-    const uint32_t cs = (uint32_t)conf_.chunk_size;
+    const uint32_t cs = (uint32_t)max_chunk_;
     struct DevReq {
         const GetChunk* g;
         uint64_t arena_off;
@@ -729,7 +734,7 @@ St GpuEngine::get_chunks(const GetChunk* get, int n) {
 // (zero-fill holes, on-owner decompress, xGMI peer copy). Part of the RS
 // stripe path (C2 analog).
 St GpuEngine::gather_chunk_to_dev(const ChunkRef& ref, int ddev, uint8_t* dst) {
-    const uint32_t cs = (uint32_t)conf_.chunk_size;
+    const uint32_t cs = (uint32_t)max_chunk_;
     DevCtx& T = devs_[ddev];
     HIP_CHECK(hipSetDevice(T.hip_dev));
     if (ref.fp.is_zero()) {  // hole = zero shard
@@ -800,12 +805,12 @@ St GpuEngine::gather_chunk_to_dev(const ChunkRef& ref, int ddev, uint8_t* dst) {
 // onto one device over xGMI, run the GF(256) kernel, ingest parity chunks.
 St GpuEngine::rs_encode_stripe(const ChunkRef* data, int k, int m, uint32_t shard_len,
                                ChunkRef* parity_out) {
-    if (shard_len != conf_.chunk_size || k > kRsMaxK || m > kRsMaxM)
+    if (shard_len != max_chunk_ || k > kRsMaxK || m > kRsMaxM)
         return Engine::rs_encode_stripe(data, k, m, shard_len, parity_out);
     const int cdev = (int)(rr_.fetch_add(1) % devs_.size());
     DevCtx& C = devs_[cdev];
     std::lock_guard<std::mutex> rg(C.rs_mu);
-    const uint32_t cs = (uint32_t)conf_.chunk_size;
+    const uint32_t cs = (uint32_t)max_chunk_;
     for (int j = 0; j < k; j++) {
         St s = gather_chunk_to_dev(data[j], cdev, C.d_rs + (uint64_t)j * cs);
         if (s != St::ok) return s;
