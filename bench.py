@@ -192,7 +192,7 @@ def main():
     opts = (f"engine=gpu,gpu_first={local_rank},fp=blake3,compress=lz4,dedup=1,"
             f"arena=32g,spill=0,threads=16,writeback=1"
             if use_gpu else
-            "engine=host,fp=blake3,compress=lz4,dedup=1,threads=16,writeback=1,prealloc=6g")
+            "engine=host,fp=blake3,compress=lz4,dedup=1,threads=16,writeback=1")
 
     from edgehip.mount import Mount
 

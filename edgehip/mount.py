@@ -32,7 +32,7 @@ class Mount:
             env.update(self.env)
         self.proc = subprocess.Popen(cmd, env=env)
         # wait for the mount to appear
-        deadline = time.time() + 15
+        deadline = time.time() + 120
         while time.time() < deadline:
             if self.proc.poll() is not None:
                 raise RuntimeError(f"edgehip mount exited rc={self.proc.returncode}")
@@ -40,7 +40,7 @@ class Mount:
                 return self
             time.sleep(0.02)
         self.kill()
-        raise RuntimeError("mount did not appear within 15s")
+        raise RuntimeError("mount did not appear within 120s")
 
     def kill(self):
         if self.proc and self.proc.poll() is None:

@@ -177,10 +177,11 @@ void launch_gf_matmul(const RsDesc& d, hipStream_t stream) {
 // D lane l holds rows (l>>4)*4+q, col l&15 (q=0..3); parity bits collect via
 // __ballot(D&1): ballot q covers rows {q, 4+q, 8+q, 12+q} x cols 0..15.
 //
-// Honest expectation (SURVEY §8.4 hard part 2, Appendix A): RS 6+2 is only
-// ~2.33 GF-MACs/byte, so the bitslice/deslice VALU dominates and the MFMA
-// pipe is mostly idle; the v_perm table kernel above is the production
-// default, this variant exists for the measured comparison (profiles/).
+// MEASURED on MI355X (profiles/kernbench_r01_rates.md): 494 GB/s vs the
+// v_perm table kernel's 3416 GB/s — RS 6+2 is only ~2.33 GF-MACs/byte, so
+// the bitslice/deslice VALU dominates and the MFMA pipe idles. The table
+// kernel is the production default; this variant is kept (golden-verified)
+// as the north-star-mandated comparison.
 using i8x16 = __attribute__((ext_vector_type(16))) char;
 using i32x4 = __attribute__((ext_vector_type(4))) int;
 
