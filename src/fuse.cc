@@ -25,7 +25,7 @@ constexpr const char* kStatsName = ".edgehip-stats";
 void fill_attr(const Inode* in, fuse_attr* a) {
     std::memset(a, 0, sizeof(*a));
     a->ino = in->ino;
-    a->size = in->size();
+    a->size = S_ISLNK(in->mode) ? in->symlink_target.size() : in->size();
     a->blocks = div_up(a->size, 512);
     a->atime = (uint64_t)(in->atime / 1000000000ll);
     a->mtime = (uint64_t)(in->mtime / 1000000000ll);
@@ -548,13 +548,39 @@ void FuseMount::handle(const uint8_t* req, size_t len, std::vector<uint8_t>* rep
             reply_err(0);
             stop_ = true;
             return;
+        case FUSE_SYMLINK: {
+            const char* name = (const char*)arg;
+            const char* target = name + strlen(name) + 1;
+            Inode* in;
+            St s = vfs_->symlink(h->nodeid, name, target, h->uid, h->gid, &in);
+            if (s != St::ok) { reply_st(s); return; }
+            fuse_entry_out e;
+            fill_entry(in, &e);
+            reply_body(&e, sizeof(e));
+            return;
+        }
+        case FUSE_READLINK: {
+            std::string t;
+            St s = vfs_->readlink(h->nodeid, &t);
+            if (s != St::ok) { reply_st(s); return; }
+            reply_body(t.data(), t.size());
+            return;
+        }
+        case FUSE_LINK: {
+            auto* l = (const fuse_link_in*)arg;
+            std::string name((const char*)(arg + sizeof(*l)));
+            Inode* in;
+            St s = vfs_->link(l->oldnodeid, h->nodeid, name, &in);
+            if (s != St::ok) { reply_st(s); return; }
+            fuse_entry_out e;
+            fill_entry(in, &e);
+            reply_body(&e, sizeof(e));
+            return;
+        }
         case FUSE_GETXATTR:
         case FUSE_SETXATTR:
         case FUSE_LISTXATTR:
         case FUSE_REMOVEXATTR:
-        case FUSE_READLINK:
-        case FUSE_SYMLINK:
-        case FUSE_LINK:
         case FUSE_GETLK:
         case FUSE_SETLK:
         case FUSE_SETLKW:

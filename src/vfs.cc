@@ -143,9 +143,43 @@ St Vfs::restore() {
                 p->nlink--;
                 break;
             }
-            case 4: {  // rename
+            case 4: {  // rename (also encodes symlink/link via \x01 prefix)
                 uint64_t np;
                 std::string nleaf;
+                if (r.path.rfind("\x01symlink:", 0) == 0) {
+                    std::string target = r.path.substr(10);
+                    if (resolve_parent_locked(r.path2, &parent, &leaf) != St::ok) break;
+                    Inode* p2 = get(parent);
+                    if (!p2 || p2->children.count(leaf)) break;
+                    auto in = std::make_unique<Inode>();
+                    in->ino = alloc_ino();
+                    in->mode = S_IFLNK | 0777;
+                    in->nlookup = 0;
+                    in->parent = parent;
+                    in->name = leaf;
+                    in->symlink_target = target;
+                    in->atime = in->mtime = in->ctime = (int64_t)wall_ns();
+                    p2->children[leaf] = in->ino;
+                    inodes_.emplace(in->ino, std::move(in));
+                    break;
+                }
+                if (r.path.rfind("\x01link:", 0) == 0) {
+                    std::string oldp = r.path.substr(6);
+                    uint64_t op2;
+                    std::string oleaf;
+                    if (resolve_parent_locked(oldp, &op2, &oleaf) != St::ok) break;
+                    Inode* opn = get(op2);
+                    if (!opn) break;
+                    auto oit = opn->children.find(oleaf);
+                    if (oit == opn->children.end()) break;
+                    if (resolve_parent_locked(r.path2, &parent, &leaf) != St::ok) break;
+                    Inode* p2 = get(parent);
+                    if (!p2 || p2->children.count(leaf)) break;
+                    p2->children[leaf] = oit->second;
+                    Inode* t = get(oit->second);
+                    if (t) t->nlink++;
+                    break;
+                }
                 if (resolve_parent_locked(r.path, &parent, &leaf) != St::ok) break;
                 if (resolve_parent_locked(r.path2, &np, &nleaf) != St::ok) break;
                 Inode* p = get(parent);
@@ -295,8 +329,8 @@ St Vfs::unlink(uint64_t parent, const std::string& name) {
     if (c->is_dir()) return St::is_dir;
     LOGD("vfs: unlink %s (ino %lu)", name.c_str(), (unsigned long)c->ino);
     if (journal_ && journal_->enabled()) journal_->unlink_rec(path_locked(c->ino));
-    c->unlinked = true;
     c->nlink--;
+    if (c->nlink == 0) c->unlinked = true;
     p->children.erase(it);
     p->mtime = p->ctime = (int64_t)wall_ns();
     maybe_free(c->ino);
@@ -377,7 +411,10 @@ St Vfs::readdir(uint64_t ino, std::vector<DirEntry>* out) {
     out->push_back({"..", in->parent, DT_DIR});
     for (auto& kv : in->children) {
         Inode* c = get(kv.second);
-        out->push_back({kv.first, kv.second, c && c->is_dir() ? (uint32_t)DT_DIR : (uint32_t)DT_REG});
+        uint32_t t = DT_REG;
+        if (c && c->is_dir()) t = DT_DIR;
+        else if (c && S_ISLNK(c->mode)) t = DT_LNK;
+        out->push_back({kv.first, kv.second, t});
     }
     return St::ok;
 }
@@ -462,6 +499,61 @@ St Vfs::fsync(uint64_t fh) {
     St s = flush(fh);
     if (s != St::ok) return s;
     if (journal_ && journal_->enabled()) return journal_->sync();
+    return St::ok;
+}
+
+St Vfs::symlink(uint64_t parent, const std::string& name, const std::string& target,
+                uint32_t uid, uint32_t gid, Inode** out) {
+    std::lock_guard<std::mutex> g(mu_);
+    Inode* p = get(parent);
+    if (!p) return St::not_found;
+    if (!p->is_dir()) return St::not_dir;
+    if (p->children.count(name)) return St::exists;
+    auto in = std::make_unique<Inode>();
+    in->ino = alloc_ino();
+    in->mode = S_IFLNK | 0777;
+    in->uid = uid;
+    in->gid = gid;
+    in->nlookup = 1;
+    in->parent = parent;
+    in->name = name;
+    in->symlink_target = target;
+    in->atime = in->mtime = in->ctime = (int64_t)wall_ns();
+    Inode* raw = in.get();
+    p->children[name] = in->ino;
+    p->mtime = p->ctime = in->ctime;
+    inodes_.emplace(raw->ino, std::move(in));
+    if (journal_ && journal_->enabled())
+        journal_->rename_rec("\x01symlink:" + target, path_locked(raw->ino));
+    *out = raw;
+    return St::ok;
+}
+
+St Vfs::readlink(uint64_t ino, std::string* target) {
+    std::lock_guard<std::mutex> g(mu_);
+    Inode* in = get(ino);
+    if (!in) return St::not_found;
+    if (!S_ISLNK(in->mode)) return St::inval;
+    *target = in->symlink_target;
+    return St::ok;
+}
+
+St Vfs::link(uint64_t ino, uint64_t newparent, const std::string& newname, Inode** out) {
+    std::lock_guard<std::mutex> g(mu_);
+    Inode* in = get(ino);
+    Inode* np = get(newparent);
+    if (!in || !np) return St::not_found;
+    if (in->is_dir()) return St::is_dir;
+    if (np->children.count(newname)) return St::exists;
+    np->children[newname] = ino;
+    in->nlink++;
+    in->nlookup++;
+    np->mtime = np->ctime = (int64_t)wall_ns();
+    if (journal_ && journal_->enabled())
+        journal_->rename_rec("\x01link:" + path_locked(ino), path_locked(newparent) == "/"
+                                                                 ? "/" + newname
+                                                                 : path_locked(newparent) + "/" + newname);
+    *out = in;
     return St::ok;
 }
 

@@ -29,6 +29,7 @@ struct Inode {
     int64_t atime = 0, mtime = 0, ctime = 0;  // ns since epoch
     std::map<std::string, uint64_t> children; // dirs only
     std::shared_ptr<Object> obj;              // files only
+    std::string symlink_target;               // symlinks only
     uint64_t parent = 0;
     std::string name;                         // dentry name (journal paths)
     bool unlinked = false;
@@ -63,6 +64,10 @@ class Vfs {
     St rmdir(uint64_t parent, const std::string& name);
     St rename(uint64_t parent, const std::string& name, uint64_t newparent,
               const std::string& newname, bool noreplace);
+    St symlink(uint64_t parent, const std::string& name, const std::string& target,
+               uint32_t uid, uint32_t gid, Inode** out);
+    St readlink(uint64_t ino, std::string* target);
+    St link(uint64_t ino, uint64_t newparent, const std::string& newname, Inode** out);
     St readdir(uint64_t ino, std::vector<DirEntry>* out);
 
     // handles

@@ -69,3 +69,16 @@ def test_remount_after_overwrite(tmp_mount, tmp_path):
         (tmp_mount / "f.bin").write_bytes(v2)
     with Mount(tmp_mount, opts(jdir)):
         assert (tmp_mount / "f.bin").read_bytes() == v2
+
+
+def test_symlink_hardlink_survive_remount(tmp_mount, tmp_path):
+    jdir = tmp_path / "jrnl"
+    with Mount(tmp_mount, opts(jdir)):
+        (tmp_mount / "f.txt").write_text("x\n")
+        (tmp_mount / "s").symlink_to("f.txt")
+        os.link(tmp_mount / "f.txt", tmp_mount / "h")
+    with Mount(tmp_mount, opts(jdir)):
+        assert os.readlink(tmp_mount / "s") == "f.txt"
+        assert (tmp_mount / "s").read_text() == "x\n"
+        assert (tmp_mount / "h").read_text() == "x\n"
+        assert (tmp_mount / "f.txt").stat().st_nlink == 2

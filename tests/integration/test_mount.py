@@ -113,3 +113,18 @@ def test_statfs(tmp_mount):
         st = os.statvfs(tmp_mount)
         assert st.f_bsize == 4096
         assert st.f_blocks > 0
+
+
+def test_symlink_and_hardlink(tmp_mount):
+    with Mount(tmp_mount, CONTENTOPTS):
+        p = tmp_mount / "orig.txt"
+        p.write_text("payload\n")
+        (tmp_mount / "sym").symlink_to("orig.txt")
+        assert (tmp_mount / "sym").read_text() == "payload\n"
+        assert os.readlink(tmp_mount / "sym") == "orig.txt"
+        os.link(p, tmp_mount / "hard")
+        assert (tmp_mount / "hard").read_text() == "payload\n"
+        assert p.stat().st_nlink == 2
+        p.unlink()
+        assert (tmp_mount / "hard").read_text() == "payload\n"  # survives
+        assert (tmp_mount / "hard").stat().st_nlink == 1
