@@ -147,7 +147,7 @@ St Vfs::restore() {
                 uint64_t np;
                 std::string nleaf;
                 if (r.path.rfind("\x01symlink:", 0) == 0) {
-                    std::string target = r.path.substr(10);
+                    std::string target = r.path.substr(9);
                     if (resolve_parent_locked(r.path2, &parent, &leaf) != St::ok) break;
                     Inode* p2 = get(parent);
                     if (!p2 || p2->children.count(leaf)) break;
