@@ -13,7 +13,7 @@ CXX      ?= g++
 ARCH     ?= gfx950
 CXXFLAGS ?= -O2 -g -std=c++17 -Wall -Wextra -fPIC -pthread -MMD -MP -I$(ROCM)/include -D__HIP_PLATFORM_AMD__
 HIPFLAGS ?= -O3 -std=c++17 --offload-arch=$(ARCH) -fPIC -MMD -MP
-LDFLAGS  ?= -L$(ROCM)/lib -lamdhip64 -lrccl -Wl,-rpath,$(ROCM)/lib -pthread
+LDFLAGS  ?= -L$(ROCM)/lib -lamdhip64 -lrccl -lroctx64 -Wl,-rpath,$(ROCM)/lib -pthread
 
 HOST_SRCS := src/obs.cc src/bufpool.cc src/conf.cc src/manifest.cc src/blake3.cc src/lz4.cc \
              src/cdc.cc src/gf256.cc src/journal.cc src/store.cc src/obj.cc src/vfs.cc src/fuse.cc \

@@ -326,6 +326,7 @@ class GpuEngine : public Engine {
 };
 
 St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
+    RoctxRange rr_ingest("edgehip::ingest_batch");
     const uint32_t cs = (uint32_t)max_chunk_;
     const bool do_lz4 = conf_.compress == Compress::lz4;
     const bool do_fp = conf_.fp == FpAlgo::blake3;
@@ -574,6 +575,7 @@ St GpuEngine::evict_for(int dev, uint32_t need_bytes) {
 }
 
 St GpuEngine::get_chunks(const GetChunk* get, int n) {
+    RoctxRange rr_get("edgehip::get_chunks");
     const uint32_t cs = (uint32_t)max_chunk_;
     struct DevReq {
         const GetChunk* g;
@@ -807,6 +809,7 @@ St GpuEngine::rs_encode_stripe(const ChunkRef* data, int k, int m, uint32_t shar
                                ChunkRef* parity_out) {
     if (shard_len != max_chunk_ || k > kRsMaxK || m > kRsMaxM)
         return Engine::rs_encode_stripe(data, k, m, shard_len, parity_out);
+    RoctxRange rr_rs("edgehip::rs_encode_stripe");
     const int cdev = (int)(rr_.fetch_add(1) % devs_.size());
     DevCtx& C = devs_[cdev];
     std::lock_guard<std::mutex> rg(C.rs_mu);

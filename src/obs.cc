@@ -50,6 +50,17 @@ Metrics& metrics() {
     return m;
 }
 
+#include <roctracer/roctx.h>
+bool roctx_on() {
+    static bool on = [] {
+        const char* e = getenv("EDGEHIP_ROCTX");
+        return e && *e && *e != '0';
+    }();
+    return on;
+}
+void roctx_push(const char* name) { roctxRangePushA(name); }
+void roctx_pop() { roctxRangePop(); }
+
 std::string Metrics::render() const {
     char buf[4096];
     snprintf(buf, sizeof(buf),

@@ -28,6 +28,17 @@ struct Metrics {
 
 Metrics& metrics();
 
+// roctx tracing: EDGEHIP_ROCTX=1 emits named ranges around
+// the pipeline stages so rocprofv3 --sys-trace attributes device time to
+// data-path phases. Zero overhead when off (one branch on a cached bool).
+bool roctx_on();
+void roctx_push(const char* name);
+void roctx_pop();
+struct RoctxRange {
+    explicit RoctxRange(const char* name) { if (roctx_on()) roctx_push(name); }
+    ~RoctxRange() { if (roctx_on()) roctx_pop(); }
+};
+
 // Fault injection: conf `fault=site:prob[,site:prob]`.
 // Sites wired into the data path: "get_drop" (chunk read returns not_found,
 // exercising RS reconstruct), "put_io" (ingest fails with EIO),
