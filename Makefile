@@ -34,7 +34,7 @@ bin/kernbench: bench/kernbench.o $(HIP_OBJS:src/engine/gpu_engine.o=) src/gf256.
 	@mkdir -p bin
 	$(HIPCC) -o $@ bench/kernbench.o src/kernels/fingerprint.o src/kernels/lz4.o \
 	    src/kernels/rs_gf256.o src/kernels/checksum.o src/kernels/memops.o \
-	    src/kernels/dedup_index.o src/gf256.o src/obs.o src/bufpool.o -lpthread
+	    src/kernels/dedup_index.o src/gf256.o src/obs.o src/bufpool.o -L$(ROCM)/lib -lroctx64 -lpthread
 
 bin/edgehip: $(OBJS) src/main.o
 	@mkdir -p bin
