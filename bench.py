@@ -53,6 +53,8 @@ class ObjectIo:
     def __init__(self, opts):
         from edgehip.lib import Engine
         self.eng = Engine(opts)
+        self.write_secs = 0.0
+        self.read_secs = 0.0
 
     def close(self):
         self.eng.close()
@@ -67,20 +69,27 @@ class ObjectIo:
         src = memoryview(rand_buf)
         errs = []
 
+        phase = [[0.0, 0.0] for _ in range(jobs)]
+
         def job(j):
             try:
                 o = self.eng.object()
                 try:
+                    t0 = time.perf_counter()
                     for i in range(per_job):
                         off = ((i * jobs + j) % 128) * unit
                         o.write(i * unit, src[off:off + unit])
                     o.flush()
+                    t1 = time.perf_counter()
                     total = 0
                     for i in range(per_job):
                         got = o.read(i * unit, unit)
                         total += len(got)
                     if total != per_job * unit:
                         raise RuntimeError("short read")
+                    t2 = time.perf_counter()
+                    phase[j][0] = t1 - t0
+                    phase[j][1] = t2 - t1
                 finally:
                     o.truncate(0)
                     o.close()
@@ -94,6 +103,8 @@ class ObjectIo:
             t.join()
         if errs:
             raise errs[0]
+        self.write_secs += max(p[0] for p in phase)
+        self.read_secs += max(p[1] for p in phase)
         return 2 * per_job * unit * jobs
 
 
@@ -248,6 +259,11 @@ def main():
 
     dt = t1 - t0
     local_gibps = moved / GIB / dt
+    phase_gibps = None
+    if not use_fuse and io.write_secs > 0:
+        half = moved / 2 / GIB
+        phase_gibps = {"write": round(half / io.write_secs, 2),
+                       "read": round(half / io.read_secs, 2)}
 
     # aggregate: value = sum over ranks; ms_per_step = MAX over ranks
     if dist:
@@ -289,6 +305,7 @@ def main():
                 "transport": "fuse-mount" if use_fuse else
                              "object-api (box denies /dev/fuse; same native "
                              "data path minus the kernel hop)",
+                "phase_gibps": phase_gibps,
             },
         }
         print(json.dumps(result))
