@@ -66,7 +66,7 @@ struct ArenaLoc {
     std::list<Fp>::iterator lru;  // valid when !spilled
 };
 
-constexpr int kPipeSlots = 2;  // concurrent ingest pipelines per device
+constexpr int kPipeSlots = 4;  // concurrent ingest pipelines per device
 
 // One ingest pipeline: stream + staging + scratch (double-buffered per device
 // so concurrent writers overlap H2D/kernels/D2D across slots — SURVEY §4.3).
