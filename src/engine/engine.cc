@@ -111,6 +111,16 @@ class HostEngine : public Engine {
                     std::memcpy(g.out, full.data() + g.off, g.len);
                 }
             }
+            if (Faults::inst().fire("corrupt")) g.out[g.len / 2] ^= 0x5A;
+            if (conf_.verify_read && g.off == 0 && g.len == r.logical_len &&
+                conf_.fp == FpAlgo::blake3) {
+                // full verification: re-hash and compare the content address
+                if (blake3_fp(g.out, g.len) != r.fp) {
+                    LOGW("verify_read: fingerprint mismatch (chunk %s)",
+                         r.fp.hex().c_str());
+                    return St::not_found;  // object layer reconstructs via RS
+                }
+            }
         }
         return St::ok;
     }

@@ -24,6 +24,7 @@
 #include <vector>
 
 #include "../gf256.h"
+#include "../blake3.h"
 #include "../journal.h"
 #include "../kernels/launchers.h"
 #include "../lz4.h"
@@ -729,6 +730,20 @@ St GpuEngine::get_chunks(const GetChunk* get, int n) {
     }
     for (auto& st : staged)
         for (auto& c : st.copies) std::memcpy(c.dst, st.blob.p + c.src_off, c.len);
+    if (Faults::inst().enabled() || conf_.verify_read) {
+        for (int i = 0; i < n; i++) {
+            const GetChunk& g = get[i];
+            if (g.len && Faults::inst().fire("corrupt")) g.out[g.len / 2] ^= 0x5A;
+            if (conf_.verify_read && g.off == 0 && g.len == g.ref->logical_len &&
+                conf_.fp == FpAlgo::blake3) {
+                if (blake3_fp(g.out, g.len) != g.ref->fp) {
+                    LOGW("verify_read: fingerprint mismatch (chunk %s)",
+                         g.ref->fp.hex().c_str());
+                    return St::not_found;
+                }
+            }
+        }
+    }
     return St::ok;
 }
 
