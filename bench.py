@@ -176,7 +176,7 @@ def main():
                     help="GiB written (and read back) per step per rank")
     ap.add_argument("--engine", default=None, choices=[None, "gpu", "host"],
                     help="override engine (default: gpu if a GPU is visible)")
-    ap.add_argument("--jobs", type=int, default=4, help="parallel I/O jobs (fio numjobs)")
+    ap.add_argument("--jobs", type=int, default=6, help="parallel I/O jobs (fio numjobs)")
     ap.add_argument("--mnt-base", default="/tmp/edgehip_bench")
     args = ap.parse_args()
 

@@ -97,14 +97,22 @@ struct DevCtx {
     std::condition_variable slot_cv;
     uint32_t slot_free_mask = (1u << kPipeSlots) - 1;
 
-    uint8_t *d_rdstage = nullptr, *d_rs = nullptr;
+    struct ReadSlot {
+        hipStream_t s{};
+        uint8_t* d_rdstage = nullptr;
+        DevChunk* d_rchunks = nullptr;
+        uint32_t* d_rstatus = nullptr;
+        uint8_t* pin_rmeta = nullptr;
+        std::mutex mu;
+    };
+    static constexpr int kReadSlots = 2;
+    ReadSlot rslots[kReadSlots];
+    std::atomic<uint32_t> rrr{0};
+
+    uint8_t* d_rs = nullptr;
     unsigned long long* d_index = nullptr;  // G7 fingerprint index (HBM)
     uint64_t index_mask = 0;
-    DevChunk* d_rchunks = nullptr;          // read-path descriptors
-    uint8_t *pin_out = nullptr, *pin_rmeta = nullptr;
-    uint32_t* d_rstatus = nullptr;
     uint64_t stage_sz = 0;
-    std::mutex pipe_mu;  // read-path stream + staging
     std::mutex rs_mu;    // d_rs stripe buffer
 
     PipeSlot* acquire_slot() {
@@ -201,10 +209,15 @@ class GpuEngine : public Engine {
                 HIP_CHECK(hipHostMalloc(&S.pin_in, D.stage_sz, hipHostMallocDefault));
                 HIP_CHECK(hipHostMalloc(&S.pin_meta, kPinMetaSz, hipHostMallocDefault));
             }
-            HIP_CHECK(hipMalloc(&D.d_rdstage, D.stage_sz));
+            for (int rs2 = 0; rs2 < DevCtx::kReadSlots; rs2++) {
+                auto& R = D.rslots[rs2];
+                HIP_CHECK(hipStreamCreateWithFlags(&R.s, hipStreamNonBlocking));
+                HIP_CHECK(hipMalloc(&R.d_rdstage, D.stage_sz));
+                HIP_CHECK(hipMalloc(&R.d_rchunks, sizeof(DevChunk) * kMaxBatch));
+                HIP_CHECK(hipMalloc(&R.d_rstatus, sizeof(uint32_t) * kMaxBatch));
+                HIP_CHECK(hipHostMalloc(&R.pin_rmeta, 64 * KiB, hipHostMallocDefault));
+            }
             HIP_CHECK(hipMalloc(&D.d_rs, (uint64_t)(kRsMaxK + kRsMaxM) * cs));
-            HIP_CHECK(hipMalloc(&D.d_rchunks, sizeof(DevChunk) * kMaxBatch));
-            HIP_CHECK(hipMalloc(&D.d_rstatus, sizeof(uint32_t) * kMaxBatch));
             // G7: per-device fingerprint index shard in HBM (tags of chunks
             // this device owns); 1/1024 of the arena, power-of-two slots
             {
@@ -215,8 +228,7 @@ class GpuEngine : public Engine {
                 HIP_CHECK(hipMemset(D.d_index, 0, slots * 8));
                 D.index_mask = slots - 1;
             }
-            HIP_CHECK(hipHostMalloc(&D.pin_out, D.stage_sz, hipHostMallocDefault));
-            HIP_CHECK(hipHostMalloc(&D.pin_rmeta, kPinMetaSz, hipHostMallocDefault));
+
         }
         if (ndev > 1) {
             if (dist_.init_local(ndev, conf.gpu_first) != St::ok)
@@ -242,11 +254,16 @@ class GpuEngine : public Engine {
                 (void)hipHostFree(S.pin_meta);
                 (void)hipStreamDestroy(S.s);
             }
-            for (void* p : {(void*)D.arena, (void*)D.d_rdstage, (void*)D.d_rs,
-                            (void*)D.d_rchunks, (void*)D.d_rstatus, (void*)D.d_index})
+            for (int rs2 = 0; rs2 < DevCtx::kReadSlots; rs2++) {
+                auto& R = D.rslots[rs2];
+                (void)hipFree(R.d_rdstage);
+                (void)hipFree(R.d_rchunks);
+                (void)hipFree(R.d_rstatus);
+                (void)hipHostFree(R.pin_rmeta);
+                (void)hipStreamDestroy(R.s);
+            }
+            for (void* p : {(void*)D.arena, (void*)D.d_rs, (void*)D.d_index})
                 (void)hipFree(p);
-            (void)hipHostFree(D.pin_out);
-            (void)hipHostFree(D.pin_rmeta);
             (void)hipStreamDestroy(D.s_read);
         }
     }
@@ -668,10 +685,11 @@ St GpuEngine::get_chunks(const GetChunk* get, int n) {
         for (auto& q : dev_reqs)
             if (q.dev == (int)d) mine.push_back(&q);
         if (mine.empty()) continue;
-        std::lock_guard<std::mutex> pg(D.pipe_mu);
+        auto& R = D.rslots[D.rrr.fetch_add(1) % DevCtx::kReadSlots];
+        std::lock_guard<std::mutex> pg(R.mu);
         HIP_CHECK(hipSetDevice(D.hip_dev));
-        auto* h_dc = (DevChunk*)(D.pin_rmeta + kOffChunks);
-        auto* h_status = (uint32_t*)(D.pin_rmeta + kOffStatus);
+        auto* h_dc = (DevChunk*)(R.pin_rmeta);
+        auto* h_status = (uint32_t*)(R.pin_rmeta + 32 * KiB);
 
         size_t j0 = 0;
         while (j0 < mine.size()) {
@@ -695,12 +713,12 @@ St GpuEngine::get_chunks(const GetChunk* get, int n) {
                 mi++;
             }
             if (mi) {
-                HIP_CHECK(hipMemcpyAsync(D.d_rchunks, h_dc, sizeof(DevChunk) * mi,
-                                         hipMemcpyHostToDevice, D.s_read));
-                launch_lz4_decompress_chunks(D.arena, D.d_rchunks, D.d_rdstage, D.d_rstatus,
-                                             mi, D.s_read);
-                HIP_CHECK(hipMemcpyAsync(h_status, D.d_rstatus, 4 * mi,
-                                         hipMemcpyDeviceToHost, D.s_read));
+                HIP_CHECK(hipMemcpyAsync(R.d_rchunks, h_dc, sizeof(DevChunk) * mi,
+                                         hipMemcpyHostToDevice, R.s));
+                launch_lz4_decompress_chunks(D.arena, R.d_rchunks, R.d_rdstage, R.d_rstatus,
+                                             mi, R.s);
+                HIP_CHECK(hipMemcpyAsync(h_status, R.d_rstatus, 4 * mi,
+                                         hipMemcpyDeviceToHost, R.s));
             }
             Staged st;
             st.blob = BufPool::inst().get((uint32_t)std::max<uint64_t>(outb, 64));
@@ -710,17 +728,17 @@ St GpuEngine::get_chunks(const GetChunk* get, int n) {
                 const GetChunk& g = *mine[j]->g;
                 const uint8_t* srcp;
                 if (mine[j]->compressed) {
-                    srcp = D.d_rdstage + (uint64_t)mi * cs + g.off;
+                    srcp = R.d_rdstage + (uint64_t)mi * cs + g.off;
                     mi++;
                 } else {
                     srcp = D.arena + mine[j]->arena_off + g.off;
                 }
                 HIP_CHECK(hipMemcpyAsync(st.blob.p + pout, srcp, g.len,
-                                         hipMemcpyDeviceToHost, D.s_read));
+                                         hipMemcpyDeviceToHost, R.s));
                 st.copies.push_back({g.out, pout, g.len});
                 pout += round_up(g.len, 64);
             }
-            HIP_CHECK(hipStreamSynchronize(D.s_read));
+            HIP_CHECK(hipStreamSynchronize(R.s));
             metrics().d2h_bytes += pout;
             for (int x = 0; x < mi; x++)
                 if (h_status[x]) return St::bad_format;
@@ -791,25 +809,26 @@ St GpuEngine::gather_chunk_to_dev(const ChunkRef& ref, int ddev, uint8_t* dst) {
         }
     } else {
         DevCtx& O = devs_[odev];
-        std::lock_guard<std::mutex> og(O.pipe_mu);
+        auto& R = O.rslots[0];
+        std::lock_guard<std::mutex> og(R.mu);
         HIP_CHECK(hipSetDevice(O.hip_dev));
-        auto* h_dc = (DevChunk*)(O.pin_rmeta + kOffChunks);
-        auto* h_status = (uint32_t*)(O.pin_rmeta + kOffStatus);
+        auto* h_dc = (DevChunk*)R.pin_rmeta;
+        auto* h_status = (uint32_t*)(R.pin_rmeta + 32 * KiB);
         h_dc[0] = DevChunk{aoff, 0, ref.logical_len, stored};
-        HIP_CHECK(hipMemcpyAsync(O.d_rchunks, h_dc, sizeof(DevChunk), hipMemcpyHostToDevice,
-                                 O.s_read));
-        launch_lz4_decompress_chunks(O.arena, O.d_rchunks, O.d_rdstage, O.d_rstatus, 1, O.s_read);
-        HIP_CHECK(hipMemcpyAsync(h_status, O.d_rstatus, 4, hipMemcpyDeviceToHost, O.s_read));
-        HIP_CHECK(hipStreamSynchronize(O.s_read));
+        HIP_CHECK(hipMemcpyAsync(R.d_rchunks, h_dc, sizeof(DevChunk), hipMemcpyHostToDevice,
+                                 R.s));
+        launch_lz4_decompress_chunks(O.arena, R.d_rchunks, R.d_rdstage, R.d_rstatus, 1, R.s);
+        HIP_CHECK(hipMemcpyAsync(h_status, R.d_rstatus, 4, hipMemcpyDeviceToHost, R.s));
+        HIP_CHECK(hipStreamSynchronize(R.s));
         if (h_status[0]) return St::bad_format;
         if (odev == ddev) {
-            HIP_CHECK(hipMemcpyAsync(dst, O.d_rdstage, ref.logical_len,
-                                     hipMemcpyDeviceToDevice, O.s_read));
+            HIP_CHECK(hipMemcpyAsync(dst, R.d_rdstage, ref.logical_len,
+                                     hipMemcpyDeviceToDevice, R.s));
         } else {
-            HIP_CHECK(hipMemcpyPeerAsync(dst, T.hip_dev, O.d_rdstage, O.hip_dev,
-                                         ref.logical_len, O.s_read));
+            HIP_CHECK(hipMemcpyPeerAsync(dst, T.hip_dev, R.d_rdstage, O.hip_dev,
+                                         ref.logical_len, R.s));
         }
-        HIP_CHECK(hipStreamSynchronize(O.s_read));
+        HIP_CHECK(hipStreamSynchronize(R.s));
     }
     if (ref.logical_len < cs) {
         HIP_CHECK(hipSetDevice(T.hip_dev));
