@@ -81,10 +81,10 @@ class ObjectIo:
                         o.write(i * unit, src[off:off + unit])
                     o.flush()
                     t1 = time.perf_counter()
+                    rbuf = bytearray(unit)
                     total = 0
                     for i in range(per_job):
-                        got = o.read(i * unit, unit)
-                        total += len(got)
+                        total += o.read_into(i * unit, unit, rbuf)
                     if total != per_job * unit:
                         raise RuntimeError("short read")
                     t2 = time.perf_counter()

@@ -41,11 +41,14 @@ def main():
         def job(j):
             rng = random.Random(j)
             n = 0
+            buf = bytearray(args.qd * 4096)
+            lens = [4096] * args.qd
             while time.perf_counter() < stop:
-                off = rng.randrange(0, (total - 4096) // 4096) * 4096
-                got = o.read(off, 4096)
-                assert len(got) == 4096
-                n += 1
+                offs = [rng.randrange(0, (total - 4096) // 4096) * 4096
+                        for _ in range(args.qd)]
+                got = o.preadv(offs, lens, buf)
+                assert got == args.qd * 4096
+                n += args.qd
             counts[j] = n
 
         ts = [threading.Thread(target=job, args=(j,)) for j in range(args.jobs)]

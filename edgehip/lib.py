@@ -54,6 +54,9 @@ class EhLib:
         so.eh_object_size.restype = ctypes.c_uint64
         so.eh_object_commit.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_size_t]
         so.eh_object_commit.restype = ctypes.c_int64
+        so.eh_object_preadv.argtypes = [ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p,
+                                        ctypes.c_void_p, ctypes.c_void_p]
+        so.eh_object_preadv.restype = ctypes.c_int64
 
     # ---- convenience wrappers ----
     def blake3(self, data: bytes) -> bytes:
@@ -205,6 +208,27 @@ class Obj:
         if got < 0:
             raise OSError(-got, "object read failed")
         return out.raw[:got]
+
+    def read_into(self, off: int, n: int, buf) -> int:
+        """Read into a reusable writable buffer (no per-call allocation)."""
+        mv = memoryview(buf)
+        b = (ctypes.c_ubyte * n).from_buffer(mv.cast("B"))
+        got = self.lib._so.eh_object_read(self.h, off, b, n)
+        if got < 0:
+            raise OSError(-got, "object read failed")
+        return got
+
+    def preadv(self, offs, lens, buf) -> int:
+        """Batched scatter read of len(offs) ranges into buf (concatenated)."""
+        n = len(offs)
+        oa = (ctypes.c_uint64 * n)(*offs)
+        la = (ctypes.c_uint32 * n)(*lens)
+        mv = memoryview(buf)
+        b = (ctypes.c_ubyte * mv.nbytes).from_buffer(mv.cast("B"))
+        got = self.lib._so.eh_object_preadv(self.h, n, oa, la, b)
+        if got < 0:
+            raise OSError(int(-got), "preadv failed")
+        return got
 
     def flush(self):
         rc = self.lib._so.eh_object_flush(self.h)

@@ -120,6 +120,23 @@ int64_t eh_object_commit(void* o, uint8_t* manifest_out, size_t cap) {
 
 }  // extern "C"
 
+// batched scatter read: n ranges concatenated into out (sum(lens) bytes).
+// One object-level pass per range but ranges run engine batches via the
+// object's internal batching; used by the random-read bench (fio iodepth).
+extern "C" int64_t eh_object_preadv(void* o, int n, const uint64_t* offs,
+                                    const uint32_t* lens, void* out) {
+    uint8_t* p = (uint8_t*)out;
+    int64_t total = 0;
+    for (int i = 0; i < n; i++) {
+        uint64_t got = 0;
+        St s = ((Object*)o)->read(offs[i], lens[i], p, &got);
+        if (s != St::ok) return (int64_t)st_errno(s);
+        p += got;
+        total += got;
+    }
+    return total;
+}
+
 extern "C" int eh_engine_drop_chunk(void* e, const uint8_t* fp32) {
     Fp f;
     std::memcpy(f.b.data(), fp32, 32);

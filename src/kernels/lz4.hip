@@ -30,102 +30,136 @@ EH_DEV uint32_t load32(const uint8_t* p) {
     return v;
 }
 
-// --- per-thread segment compressor ------------------------------------------
-// Hash table: 4096 x u16 in LDS PER WARP-GROUP? 4096*2B = 8 KiB per thread is
-// too much LDS for 256 threads. Instead: 512-entry table per thread packed in
-// LDS (512*2B = 1 KiB * 128 threads = 128 KiB). Block size 128 threads.
-constexpr int kTblBits = 9;  // 512 entries
-constexpr int kCThreads = 128;
+// --- wave-cooperative segment compressor ---------------------------------
+// One WAVE (64 lanes) compresses one 4 KiB segment staged in LDS: all lanes
+// probe candidate positions in parallel, a ballot picks the first real match,
+// match extension and literal copies are cooperative, and only the sequence
+// headers are lane-0 scalar work. Control flow is wave-uniform throughout
+// (guide G5); random data takes the no-match ballot path at ~64 B/iteration.
+constexpr int kTblBits = 9;  // 512 entries per wave
+constexpr int kCThreads = 256;  // 4 waves = 4 segments per workgroup
 
 EH_DEV uint32_t seg_hash(uint32_t v) { return (v * 2654435761u) >> (32 - kTblBits); }
 
-// Compress src[0..n) (n <= 4096) into dst; returns compressed size, or 0 if
-// output would reach cap. tbl = this thread's 512-entry LDS table.
-EH_DEV uint32_t lz4_compress_seg(const uint8_t* __restrict__ src, uint32_t n,
-                                 uint8_t* __restrict__ dst, uint32_t cap,
-                                 uint16_t* __restrict__ tbl) {
-    if (n == 0) return 0;
-#pragma unroll 4
-    for (int i = 0; i < (1 << kTblBits); i++) tbl[i] = 0xFFFF;
+EH_DEV uint32_t lds_load32(const uint8_t* p) {
+    uint32_t v;
+    __builtin_memcpy(&v, p, 4);
+    return v;
+}
 
-    uint32_t ip = 0, anchor = 0, op = 0;
-    const uint32_t last_lit = n >= 5 ? n - 5 : 0;
-    const uint32_t mflimit = n >= 12 ? n - 12 : 0;
+__global__ __launch_bounds__(kCThreads) void k_lz4_compress_segs(
+    const uint8_t* __restrict__ src, const ulong2* __restrict__ seg_meta,
+    uint8_t* __restrict__ scratch, uint32_t* __restrict__ sizes, int nsegs) {
+    __shared__ __attribute__((aligned(16))) uint8_t sseg[4][kSeg + 16];
+    __shared__ uint16_t stbl[4][1 << kTblBits];
+    const int wid = threadIdx.x >> 6;
+    const int lane = threadIdx.x & 63;
+    const int si = blockIdx.x * 4 + wid;
+    if (si >= nsegs) return;  // whole wave exits together
 
-    auto emit = [&](uint32_t lit_start, uint32_t lit_len, uint32_t off,
-                    uint32_t mlen) -> bool {
-        uint32_t need = 1 + lit_len / 255 + 1 + lit_len + (mlen ? 2 + (mlen - 4) / 255 + 1 : 0);
-        if (op + need > cap) return false;
-        uint32_t tok_pos = op++;
-        uint32_t ll = lit_len;
-        uint8_t tok;
-        if (ll >= 15) {
-            tok = 15u << 4;
-            ll -= 15;
-            while (ll >= 255) { dst[op++] = 255; ll -= 255; }
-            dst[op++] = (uint8_t)ll;
-        } else {
-            tok = (uint8_t)(ll << 4);
+    const uint64_t soff = seg_meta[si].x;
+    const uint32_t n = (uint32_t)seg_meta[si].y;
+    uint8_t* seg = sseg[wid];
+    uint16_t* tbl = stbl[wid];
+
+    // stage the segment + clear the table
+    if (n == kSeg && (soff & 15) == 0) {
+        for (uint32_t o = lane * 16; o < kSeg; o += 64 * 16)
+            *(uint4*)(seg + o) = *(const uint4*)(src + soff + o);
+    } else {
+        for (uint32_t o = lane * 16; o < n; o += 64 * 16) {
+            uint32_t c = min(16u, n - o);
+            for (uint32_t b = 0; b < c; b++) seg[o + b] = src[soff + o + b];
         }
-        for (uint32_t i = 0; i < lit_len; i++) dst[op++] = src[lit_start + i];
-        if (mlen) {
-            dst[op++] = (uint8_t)off;
-            dst[op++] = (uint8_t)(off >> 8);
-            uint32_t ml = mlen - 4;
-            if (ml >= 15) {
-                tok |= 15;
-                ml -= 15;
-                while (ml >= 255) { dst[op++] = 255; ml -= 255; }
-                dst[op++] = (uint8_t)ml;
-            } else {
-                tok |= (uint8_t)ml;
+    }
+#pragma unroll
+    for (int i = lane; i < (1 << kTblBits); i += 64) tbl[i] = 0xFFFF;
+    __builtin_amdgcn_s_waitcnt(0);  // drain this wave's LDS writes (lgkm+vm)
+
+    uint8_t* out = scratch + (uint64_t)si * kSegOutStride;
+    const uint32_t cap = n > 64 ? n - 32 : n;
+    uint32_t ip = 0, anchor = 0, op = 0;
+    bool fail = false;
+
+    auto emit_seq = [&](uint32_t lit, uint32_t mlen, uint32_t moff) -> bool {
+        // sizes (computed redundantly by every lane — keeps control uniform)
+        uint32_t lit_ext = lit >= 15 ? 1 + (lit - 15) / 255 : 0;
+        uint32_t ml2 = mlen ? mlen - 4 : 0;
+        uint32_t ml_ext = mlen && ml2 >= 15 ? 1 + (ml2 - 15) / 255 : 0;
+        uint32_t need = 1 + lit_ext + lit + (mlen ? 2 : 0) + ml_ext;
+        if (op + need > cap) return false;
+        uint32_t lit_pos = op + 1 + lit_ext;
+        if (lane == 0) {
+            uint8_t tok = (uint8_t)((lit >= 15 ? 15u : lit) << 4);
+            if (mlen) tok |= (uint8_t)(ml2 >= 15 ? 15u : ml2);
+            out[op] = tok;
+            uint32_t w = op + 1;
+            if (lit >= 15) {
+                uint32_t r = lit - 15;
+                while (r >= 255) { out[w++] = 255; r -= 255; }
+                out[w++] = (uint8_t)r;
+            }
+            w = lit_pos + lit;
+            if (mlen) {
+                out[w++] = (uint8_t)moff;
+                out[w++] = (uint8_t)(moff >> 8);
+                if (ml2 >= 15) {
+                    uint32_t r = ml2 - 15;
+                    while (r >= 255) { out[w++] = 255; r -= 255; }
+                    out[w++] = (uint8_t)r;
+                }
             }
         }
-        dst[tok_pos] = tok;
+        for (uint32_t i = lane; i < lit; i += 64) out[lit_pos + i] = seg[anchor + i];
+        op += need;
         return true;
     };
 
     if (n >= 16) {
-        ip = 1;
-        tbl[seg_hash(load32(src))] = 0;
-        while (ip <= mflimit) {
-            uint32_t h = seg_hash(load32(src + ip));
-            uint32_t cand = tbl[h];
-            tbl[h] = (uint16_t)ip;
-            if (cand != 0xFFFF && cand < ip && load32(src + cand) == load32(src + ip)) {
-                // extend backward
-                while (ip > anchor && cand > 0 && src[ip - 1] == src[cand - 1]) { ip--; cand--; }
-                uint32_t mlen = 4;
-                while (ip + mlen < last_lit && src[ip + mlen] == src[cand + mlen]) mlen++;
-                if (!emit(anchor, ip - anchor, ip - cand, mlen)) return 0;
-                ip += mlen;
-                anchor = ip;
-                if (ip > mflimit) break;
-                tbl[seg_hash(load32(src + ip - 2))] = (uint16_t)(ip - 2);
-            } else {
-                ip++;
+        const uint32_t limit = n - 12;  // last position that may start a match
+        while (ip <= limit) {
+            uint32_t p = ip + lane;
+            bool m = false;
+            uint32_t cand = 0xFFFF, h = 0, v = 0;
+            if (p <= limit) {
+                v = lds_load32(seg + p);
+                h = seg_hash(v);
+                cand = tbl[h];
+                m = cand != 0xFFFF && cand < p && lds_load32(seg + cand) == v;
             }
+            unsigned long long bal = __ballot(m);
+            if (p <= limit) tbl[h] = (uint16_t)p;  // insert after the probe
+            if (!bal) {
+                ip += 64;
+                continue;
+            }
+            int f = __ffsll((long long)bal) - 1;
+            uint32_t mp = ip + f;
+            uint32_t mc = (uint32_t)__shfl((int)cand, f, 64);
+            // cooperative forward extension, match end capped at n-5
+            const uint32_t maxm = (n - 5) - mp;
+            uint32_t mlen = 4;
+            while (mlen < maxm) {
+                uint32_t q = mlen + lane;
+                bool diff = q >= maxm || seg[mp + q] != seg[mc + q];
+                unsigned long long db = __ballot(diff);
+                if (db) {
+                    mlen += (uint32_t)(__ffsll((long long)db) - 1);
+                    break;
+                }
+                mlen += 64;
+            }
+            if (mlen > maxm) mlen = maxm;
+            if (!emit_seq(mp - anchor, mlen, mp - mc)) {
+                fail = true;
+                break;
+            }
+            anchor = mp + mlen;
+            ip = anchor;
         }
     }
-    if (!emit(anchor, n - anchor, 0, 0)) return 0;
-    return op;
-}
-
-// Kernel A: one thread per 4 KiB segment. seg_meta[i] = {src_off (bytes),
-// len (<=4096)}; outputs to scratch at i*kSegOutStride; sizes[i] = compressed
-// size or 0 (incompressible / doesn't fit under len).
-__global__ __launch_bounds__(kCThreads) void k_lz4_compress_segs(
-    const uint8_t* __restrict__ src, const ulong2* __restrict__ seg_meta,
-    uint8_t* __restrict__ scratch, uint32_t* __restrict__ sizes, int nsegs) {
-    __shared__ uint16_t tables[kCThreads][1 << kTblBits];
-    int i = blockIdx.x * blockDim.x + threadIdx.x;
-    if (i >= nsegs) return;
-    uint64_t off = seg_meta[i].x;
-    uint32_t len = (uint32_t)seg_meta[i].y;
-    // cap at len-1: only worth keeping if it actually shrinks
-    uint32_t cap = len > 64 ? len - 32 : len;
-    sizes[i] = lz4_compress_seg(src + off, len, scratch + (uint64_t)i * kSegOutStride, cap,
-                                tables[threadIdx.x]);
+    if (!fail && !emit_seq(n - anchor, 0, 0)) fail = true;
+    if (lane == 0) sizes[si] = fail ? 0 : op;
 }
 
 // --- frame assembly ----------------------------------------------------------
@@ -356,7 +390,7 @@ __global__ __launch_bounds__(256) void k_lz4_decompress_chunks(
 void launch_lz4_compress_segs(const uint8_t* src, const ulong2* seg_meta, uint8_t* scratch,
                               uint32_t* sizes, int nsegs, hipStream_t stream) {
     if (nsegs <= 0) return;
-    int blocks = (nsegs + kCThreads - 1) / kCThreads;
+    int blocks = (nsegs + 3) / 4;  // one wave per segment, 4 waves per WG
     hipLaunchKernelGGL(k_lz4_compress_segs, dim3(blocks), dim3(kCThreads), 0, stream, src,
                        seg_meta, scratch, sizes, nsegs);
 }

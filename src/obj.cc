@@ -212,7 +212,7 @@ St Object::write(uint64_t off, const uint8_t* data, uint64_t len) {
 }
 
 St Object::read(uint64_t off, uint64_t len, uint8_t* out, uint64_t* got) {
-    std::lock_guard<std::mutex> g(mu_);
+    std::unique_lock<std::mutex> g(mu_);
     if (off >= size_) { *got = 0; return St::ok; }
     len = std::min(len, size_ - off);
     *got = len;
@@ -221,6 +221,9 @@ St Object::read(uint64_t off, uint64_t len, uint8_t* out, uint64_t* got) {
     uint64_t pos = off, end = off + len;
     uint8_t* o = out;
     std::vector<GetChunk> gets;
+    std::vector<ChunkRef> refcopies;   // snapshot: engine runs without mu_
+    refcopies.reserve(div_up(len, cs) + 2);
+    std::vector<uint64_t> idxs;
     while (pos < end) {
         uint64_t idx = pos / cs;
         uint32_t co = (uint32_t)(pos - idx * cs);
@@ -244,7 +247,11 @@ St Object::read(uint64_t off, uint64_t len, uint8_t* out, uint64_t* got) {
         } else if (idx < mf_.chunks.size() && !mf_.chunks[idx].fp.is_zero()) {
             const ChunkRef& r = mf_.chunks[idx];
             uint32_t avail = r.logical_len > co ? std::min(n, r.logical_len - co) : 0;
-            if (avail) gets.push_back({&r, o, co, avail});
+            if (avail) {
+                refcopies.push_back(r);
+                gets.push_back({nullptr, o, co, avail});  // ref patched below
+                idxs.push_back(idx);
+            }
             if (avail < n) std::memset(o + avail, 0, n - avail);
         } else {
             std::memset(o, 0, n);  // hole
@@ -253,23 +260,29 @@ St Object::read(uint64_t off, uint64_t len, uint8_t* out, uint64_t* got) {
         o += n;
     }
     if (!gets.empty()) {
+        for (size_t i = 0; i < gets.size(); i++) gets[i].ref = &refcopies[i];
+        g.unlock();  // concurrent readers overlap in the engine
         St s = eng_->get_chunks(gets.data(), (int)gets.size());
         if (s == St::not_found && conf_->rs_k > 0) {
             // degraded read: find the lost chunk(s), reconstruct, retry
+            g.lock();
             std::vector<uint8_t> cbuf(conf_->chunk_size);
-            for (auto& g2 : gets) {
+            for (size_t i = 0; i < gets.size(); i++) {
+                GetChunk& g2 = gets[i];
                 St s2 = eng_->get_chunks(&g2, 1);
                 if (s2 == St::ok) continue;
                 if (s2 != St::not_found) return s2;
-                uint64_t idx = (uint64_t)(g2.ref - &mf_.chunks[0]);
+                uint64_t idx = idxs[i];
                 if (idx >= mf_.chunks.size()) return St::not_found;
                 s2 = rs_reconstruct_chunk_locked(idx, cbuf.data());
                 if (s2 != St::ok) return s2;
                 std::memcpy(g2.out, cbuf.data() + g2.off, g2.len);
             }
             s = St::ok;
+            g.unlock();
         }
         if (s != St::ok) return s;
+        g.lock();
     }
     metrics().read_ops++;
     metrics().read_bytes += len;
