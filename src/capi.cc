@@ -125,16 +125,10 @@ int64_t eh_object_commit(void* o, uint8_t* manifest_out, size_t cap) {
 // object's internal batching; used by the random-read bench (fio iodepth).
 extern "C" int64_t eh_object_preadv(void* o, int n, const uint64_t* offs,
                                     const uint32_t* lens, void* out) {
-    uint8_t* p = (uint8_t*)out;
-    int64_t total = 0;
-    for (int i = 0; i < n; i++) {
-        uint64_t got = 0;
-        St s = ((Object*)o)->read(offs[i], lens[i], p, &got);
-        if (s != St::ok) return (int64_t)st_errno(s);
-        p += got;
-        total += got;
-    }
-    return total;
+    uint64_t got = 0;
+    St s = ((Object*)o)->preadv(offs, lens, n, (uint8_t*)out, &got);
+    if (s != St::ok) return (int64_t)st_errno(s);
+    return (int64_t)got;
 }
 
 extern "C" int eh_engine_drop_chunk(void* e, const uint8_t* fp32) {

@@ -289,6 +289,90 @@ St Object::read(uint64_t off, uint64_t len, uint8_t* out, uint64_t* got) {
     return St::ok;
 }
 
+St Object::preadv(const uint64_t* offs, const uint32_t* lens, int n, uint8_t* out,
+                  uint64_t* got) {
+    std::unique_lock<std::mutex> g(mu_);
+    *got = 0;
+    if (conf_->cdc) {  // CDC: per-range path (still correct, less batched)
+        uint8_t* o = out;
+        for (int i = 0; i < n; i++) {
+            uint64_t off = offs[i];
+            if (off >= size_) continue;
+            uint64_t len = std::min<uint64_t>(lens[i], size_ - off);
+            St s = cdc_read_locked(off, len, o);
+            if (s != St::ok) return s;
+            o += len;
+            *got += len;
+        }
+        return St::ok;
+    }
+    const uint64_t cs = conf_->chunk_size;
+    std::vector<GetChunk> gets;
+    std::vector<ChunkRef> refcopies;
+    std::vector<uint64_t> idxs;
+    size_t maxchunks = 0;
+    for (int i = 0; i < n; i++) maxchunks += lens[i] / cs + 2;
+    refcopies.reserve(maxchunks);
+    uint8_t* o = out;
+    for (int i = 0; i < n; i++) {
+        uint64_t off = offs[i];
+        if (off >= size_) continue;
+        uint64_t len = std::min<uint64_t>(lens[i], size_ - off);
+        uint64_t pos = off, end = off + len;
+        while (pos < end) {
+            uint64_t idx = pos / cs;
+            uint32_t co = (uint32_t)(pos - idx * cs);
+            uint32_t nn = (uint32_t)std::min<uint64_t>(cs - co, end - pos);
+            auto it = dirty_.find(idx);
+            if (it != dirty_.end()) {
+                Dirty& d = it->second;
+                uint32_t avail = d.len > co ? std::min(nn, d.len - co) : 0;
+                if (avail) std::memcpy(o, d.data.p + co, avail);
+                if (avail < nn) std::memset(o + avail, 0, nn - avail);
+            } else if (idx < mf_.chunks.size() && !mf_.chunks[idx].fp.is_zero()) {
+                const ChunkRef& r = mf_.chunks[idx];
+                uint32_t avail = r.logical_len > co ? std::min(nn, r.logical_len - co) : 0;
+                if (avail) {
+                    refcopies.push_back(r);
+                    gets.push_back({nullptr, o, co, avail});
+                    idxs.push_back(idx);
+                }
+                if (avail < nn) std::memset(o + avail, 0, nn - avail);
+            } else {
+                std::memset(o, 0, nn);
+            }
+            pos += nn;
+            o += nn;
+        }
+        *got += len;
+    }
+    if (!gets.empty()) {
+        for (size_t i = 0; i < gets.size(); i++) gets[i].ref = &refcopies[i];
+        g.unlock();
+        St s = eng_->get_chunks(gets.data(), (int)gets.size());
+        if (s == St::not_found && conf_->rs_k > 0) {
+            g.lock();
+            std::vector<uint8_t> cbuf(conf_->chunk_size);
+            for (size_t i = 0; i < gets.size(); i++) {
+                St s2 = eng_->get_chunks(&gets[i], 1);
+                if (s2 == St::ok) continue;
+                if (s2 != St::not_found) return s2;
+                if (idxs[i] >= mf_.chunks.size()) return St::not_found;
+                s2 = rs_reconstruct_chunk_locked(idxs[i], cbuf.data());
+                if (s2 != St::ok) return s2;
+                std::memcpy(gets[i].out, cbuf.data() + gets[i].off, gets[i].len);
+            }
+            s = St::ok;
+            g.unlock();
+        }
+        if (s != St::ok) return s;
+        g.lock();
+    }
+    metrics().read_ops += n;
+    metrics().read_bytes += *got;
+    return St::ok;
+}
+
 St Object::flush_locked() {
     if (conf_->cdc) return cdc_cut_locked(true);
     if (dirty_.empty()) return St::ok;

@@ -25,6 +25,10 @@ class Object {
     ~Object();
 
     St read(uint64_t off, uint64_t len, uint8_t* out, uint64_t* got);
+    // Batched scatter read: n ranges concatenated into out — ONE engine batch
+    // (the 4 KiB random-read path: fio iodepth analog).
+    St preadv(const uint64_t* offs, const uint32_t* lens, int n, uint8_t* out,
+              uint64_t* got);
     St write(uint64_t off, const uint8_t* data, uint64_t len);
     St flush();                    // drain write buffers into the engine
     St truncate(uint64_t size);

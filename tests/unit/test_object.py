@@ -123,3 +123,18 @@ def test_compressible_data_shrinks():
         o.flush()
         assert eng.stored_bytes < (1 << 20) // 4
         assert o.read(0, 1 << 20) == data
+
+
+def test_preadv_batched():
+    from edgehip.lib import Engine as E
+    with E(HOST) as eng, eng.object() as o:
+        data = rand_bytes(8 << 20, seed=21)
+        o.write(0, data)
+        o.flush()
+        offs = [0, 4096, (3 << 20) + 8192, (7 << 20) + 12288, 5 << 20]
+        lens = [4096] * 5
+        buf = bytearray(5 * 4096)
+        got = o.preadv(offs, lens, buf)
+        assert got == 5 * 4096
+        for i, off in enumerate(offs):
+            assert bytes(buf[i * 4096:(i + 1) * 4096]) == data[off:off + 4096], i
