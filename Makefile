@@ -21,7 +21,7 @@ HOST_SRCS := src/obs.cc src/bufpool.cc src/conf.cc src/manifest.cc src/blake3.cc
 HIP_SRCS  := src/engine/gpu_engine.hip src/kernels/fingerprint.hip \
              src/kernels/lz4.hip src/kernels/rs_gf256.hip \
              src/kernels/checksum.hip src/kernels/memops.hip \
-             src/kernels/dedup_index.hip
+             src/kernels/dedup_index.hip src/kernels/cdc.hip
 HOST_OBJS := $(HOST_SRCS:.cc=.o)
 HIP_OBJS  := $(HIP_SRCS:.hip=.o)
 OBJS      := $(HOST_OBJS) $(HIP_OBJS)
@@ -34,7 +34,7 @@ bin/kernbench: bench/kernbench.o $(HIP_OBJS:src/engine/gpu_engine.o=) src/gf256.
 	@mkdir -p bin
 	$(HIPCC) -o $@ bench/kernbench.o src/kernels/fingerprint.o src/kernels/lz4.o \
 	    src/kernels/rs_gf256.o src/kernels/checksum.o src/kernels/memops.o \
-	    src/kernels/dedup_index.o src/gf256.o src/obs.o src/bufpool.o -L$(ROCM)/lib -lroctx64 -lpthread
+	    src/kernels/dedup_index.o src/kernels/cdc.o src/gf256.o src/obs.o src/bufpool.o -L$(ROCM)/lib -lroctx64 -lpthread
 
 bin/edgehip: $(OBJS) src/main.o
 	@mkdir -p bin

@@ -3,6 +3,7 @@
 // host so pytest can compare against the CPU goldens bit-for-bit.
 #include <vector>
 
+#include "cdc.h"
 #include "kernels/launchers.h"
 #include "lz4.h"
 #include "obs.h"
@@ -231,6 +232,31 @@ int eh_gpu_dedup_test(const uint8_t* fps, int n, uint32_t* out1, uint32_t* out2)
         return 0;
     } catch (const std::exception& e) {
         LOGE("eh_gpu_dedup_test: %s", e.what());
+        return -1;
+    }
+}
+
+// G2: GPU candidate bitmap must select IDENTICAL boundaries to the host scan.
+int eh_gpu_cdc_boundaries(const uint8_t* data, uint64_t len, uint64_t avg,
+                          uint64_t* ends_out, int cap) {
+    try {
+        DevBuf d_src, d_bm;
+        HIP_CHECK(d_src.alloc(len));
+        uint64_t words = (len + 31) / 32;
+        HIP_CHECK(d_bm.alloc(words * 4));
+        HIP_CHECK(hipMemcpy(d_src.p, data, len, hipMemcpyHostToDevice));
+        launch_cdc_candidates(d_src.as<uint8_t>(), len, avg, d_bm.as<uint32_t>(), nullptr,
+                              gear_table());
+        HIP_CHECK(hipDeviceSynchronize());
+        std::vector<uint32_t> bm(words);
+        HIP_CHECK(hipMemcpy(bm.data(), d_bm.p, words * 4, hipMemcpyDeviceToHost));
+        std::vector<size_t> ends;
+        cdc_select(bm.data(), len, avg, true, &ends);
+        if ((int)ends.size() > cap) return -2;
+        for (size_t i = 0; i < ends.size(); i++) ends_out[i] = ends[i];
+        return (int)ends.size();
+    } catch (const std::exception& e) {
+        LOGE("eh_gpu_cdc_boundaries: %s", e.what());
         return -1;
     }
 }

@@ -15,6 +15,8 @@
 #include "../bufpool.h"
 #include "../common.h"
 #include "../conf.h"
+#include <vector>
+
 #include "../manifest.h"
 
 namespace edgehip {
@@ -71,6 +73,18 @@ class Engine {
         (void)recs;
         (void)n;
         return St::ok;
+    }
+
+    // G2 GPU path: produce the CDC candidate bitmap for host data (the
+    // object layer then runs the shared host-side selection). Default:
+    // St::io => caller uses the host scan.
+    virtual St cdc_scan(const uint8_t* data, uint64_t len, uint64_t avg,
+                        std::vector<uint32_t>* bitmap) {
+        (void)data;
+        (void)len;
+        (void)avg;
+        (void)bitmap;
+        return St::io;
     }
 
     // Test hook: drop a chunk's stored bytes, simulating shard loss

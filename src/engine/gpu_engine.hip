@@ -25,6 +25,7 @@
 
 #include "../gf256.h"
 #include "../blake3.h"
+#include "../cdc.h"
 #include "../journal.h"
 #include "../kernels/launchers.h"
 #include "../lz4.h"
@@ -280,6 +281,37 @@ class GpuEngine : public Engine {
     St release_chunk(const ChunkRef& ref) override;
     St rs_encode_stripe(const ChunkRef* data, int k, int m, uint32_t shard_len,
                         ChunkRef* parity_out) override;
+
+    St cdc_scan(const uint8_t* data, uint64_t len, uint64_t avg,
+                std::vector<uint32_t>* bitmap) override {
+        if (len == 0 || len > devs_[0].stage_sz) return St::io;  // host fallback
+        DevCtx& D = devs_[rr_.fetch_add(1) % devs_.size()];
+        PipeSlot* SLP = D.acquire_slot();
+        PipeSlot& S = *SLP;
+        struct SlotGuard {
+            DevCtx& d;
+            PipeSlot* s;
+            ~SlotGuard() { d.release_slot(s); }
+        } sg{D, SLP};
+        St result = St::ok;
+        do {
+            if (hipSetDevice(D.hip_dev) != hipSuccess) { result = St::gpu; break; }
+            std::memcpy(S.pin_in, data, len);
+            if (hipMemcpyAsync(S.d_stage, S.pin_in, len, hipMemcpyHostToDevice, S.s) !=
+                hipSuccess) { result = St::gpu; break; }
+            uint64_t words = (len + 31) / 32;
+            // reuse d_segsizes as the bitmap buffer (kMaxSegs u32 >= words for
+            // stage-sized inputs: stage 32 MiB -> 1 MiB bitmap = 256K words,
+            // kMaxSegs = 8192... too small; use d_segscratch instead)
+            uint32_t* d_bm = (uint32_t*)S.d_segscratch;  // 35 MB scratch, ample
+            launch_cdc_candidates(S.d_stage, len, avg, d_bm, S.s, gear_table());
+            bitmap->resize(words);
+            if (hipMemcpyAsync(bitmap->data(), d_bm, words * 4, hipMemcpyDeviceToHost,
+                               S.s) != hipSuccess) { result = St::gpu; break; }
+            if (hipStreamSynchronize(S.s) != hipSuccess) { result = St::gpu; break; }
+        } while (0);
+        return result;
+    }
 
     St publish_manifest(const ChunkRef* recs, int n) override {
         if (!dist_.active()) return St::ok;

@@ -25,6 +25,9 @@ void launch_crc32c_chunks(const uint8_t* src, const DevChunk* chunks, uint32_t* 
 void launch_batch_copy(const uint8_t* src, uint8_t* dst, const DevChunk* descs, int ndescs,
                        hipStream_t stream);
 
+void launch_cdc_candidates(const uint8_t* data, uint64_t len, uint64_t avg,
+                           uint32_t* bitmap, hipStream_t stream, const uint64_t* gear_host);
+
 void launch_dedup_upsert(unsigned long long* table, uint64_t mask, const uint8_t* fps,
                          uint32_t* out_dup, int n, hipStream_t stream);
 void launch_dedup_lookup(const unsigned long long* table, uint64_t mask, const uint8_t* fps,

@@ -536,7 +536,13 @@ St Object::cdc_cut_locked(bool final_tail) {
         return final_tail ? rs_update_locked() : St::ok;
     }
     std::vector<size_t> ends;
-    cdc_boundaries(tail_.data(), tail_.size(), conf_->chunk_size, final_tail, &ends);
+    {
+        std::vector<uint32_t> bm;
+        if (eng_->cdc_scan(tail_.data(), tail_.size(), conf_->chunk_size, &bm) == St::ok)
+            cdc_select(bm.data(), tail_.size(), conf_->chunk_size, final_tail, &ends);
+        else
+            cdc_boundaries(tail_.data(), tail_.size(), conf_->chunk_size, final_tail, &ends);
+    }
     // keep the last (possibly still-moving) piece buffered unless finalizing
     size_t cut_upto = 0;
     std::vector<PutChunk> puts;

@@ -218,3 +218,36 @@ def test_gf_matmul_mfma_vs_golden(lib):
     golden = ggf.encode(k, m, shards)
     for i in range(m):
         assert out.raw[i * ln:(i + 1) * ln] == golden[i], f"parity {i}"
+
+
+# ---- CDC candidates (G2) ----------------------------------------------------
+def test_cdc_gpu_boundaries_match_host(lib):
+    from edgehip.golden import lz4 as _unused  # noqa: F401
+    import ctypes as ct
+    so = lib._so
+    so.eh_gpu_cdc_boundaries.argtypes = [ct.c_void_p, ct.c_uint64, ct.c_uint64,
+                                         ct.c_void_p, ct.c_int]
+    so.eh_gpu_cdc_boundaries.restype = ct.c_int
+    data = rand_bytes(8 << 20, seed=77)
+    avg = 256 * 1024
+    out = (ct.c_uint64 * 4096)()
+    n = so.eh_gpu_cdc_boundaries(data, len(data), avg, out, 4096)
+    assert n > 0
+    gpu_ends = [out[i] for i in range(n)]
+    # host reference through the same selection (capi has no host hook; use
+    # the object layer indirectly): recompute via python? compare count sanity
+    # + exact equality against host engine cut by writing through a host
+    # engine with identical avg:
+    from edgehip.lib import Engine
+    with Engine(f"engine=host,fp=blake3,compress=none,dedup=0,cdc=1,"
+                f"chunk_size={avg}") as eng, eng.object() as o:
+        o.write(0, data)
+        m = o.commit()
+    import struct
+    nchunks = struct.unpack_from("<Q", m, 40)[0]
+    host_ends, acc = [], 0
+    for i in range(nchunks):
+        ll = struct.unpack_from("<I", m, 64 + 64 * i + 32)[0]
+        acc += ll
+        host_ends.append(acc)
+    assert gpu_ends == host_ends
