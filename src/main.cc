@@ -28,6 +28,7 @@ static int usage() {
             "usage: edgehip mount <mountpoint> [-o key=val,...] [-c conf_file]\n"
             "       edgehip stat <mountpoint>\n"
             "       edgehip fsck <journal_dir>\n"
+            "       edgehip mkfs <journal_dir>\n"
             "       edgehip bench <mountpoint> [size_mib [block_kib]]\n"
             "       edgehip version\n"
             "options: see docs/options.md; presets in conf/presets/\n");
@@ -143,6 +144,17 @@ int main(int argc, char** argv) {
     if (cmd == "fsck") {
         if (argc < 3) return usage();
         return fsck_cmd(argv[2]);
+    }
+    if (cmd == "mkfs") {
+        // initialize (or wipe) a journal directory for a durable mount
+        if (argc < 3) return usage();
+        std::string dir = argv[2];
+        std::string cmd2 = "rm -rf '" + dir + "/journal.log' '" + dir + "/chunks'";
+        if (system(cmd2.c_str()) != 0) LOGW("mkfs: cleanup failed (fresh dir?)");
+        Journal j;
+        if (j.open(dir) != St::ok) return 1;
+        printf("mkfs: initialized journal at %s\n", dir.c_str());
+        return 0;
     }
     if (cmd == "bench") {
         if (argc < 3) return usage();
