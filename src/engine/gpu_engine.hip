@@ -597,31 +597,50 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
 
 St GpuEngine::evict_for(int dev, uint32_t need_bytes) {
     if (conf_.spill_bytes == 0) return St::no_space;
-    uint64_t freed = 0;
+    // Batch the eviction (config 4's 1 TiB object streams through spill):
+    // pick LRU victims up to ~256 MiB, launch ALL D2H copies async, one sync,
+    // then hand the blobs to the spill store and free the arena slots.
+    DevCtx& O = devs_[dev];
+    HIP_CHECK(hipSetDevice(O.hip_dev));
+    struct Victim {
+        Fp fp;
+        uint64_t off;
+        uint32_t len;
+        Blob blob;
+    };
+    std::vector<Victim> victims;
+    uint64_t planned = 0;
+    const uint64_t target =
+        std::max<uint64_t>((uint64_t)need_bytes * 4, 256 * MiB);
     auto it2 = lru_.end();
-    while (freed < (uint64_t)need_bytes * 4 && it2 != lru_.begin()) {
+    while (planned < target && it2 != lru_.begin()) {
         --it2;
-        Fp victim = *it2;
-        auto it = map_.find(victim);
+        auto it = map_.find(*it2);
         if (it == map_.end() || it->second.spilled || it->second.dev != dev) continue;
         ArenaLoc& loc = it->second;
-        DevCtx& O = devs_[dev];
-        Blob b = BufPool::inst().get(loc.len);
-        b.len = loc.len;
-        HIP_CHECK(hipSetDevice(O.hip_dev));
-        HIP_CHECK(hipMemcpyAsync(b.p, O.arena + loc.off, loc.len, hipMemcpyDeviceToHost,
-                                 O.s_read));
-        HIP_CHECK(hipStreamSynchronize(O.s_read));
-        St s = spill_.put(victim, std::move(b));
-        if (s != St::ok && s != St::exists) return s;
-        O.arena_free(loc.off, loc.len);
-        freed += round_up(loc.len, kGranule);
+        Victim v;
+        v.fp = *it2;
+        v.off = loc.off;
+        v.len = loc.len;
+        v.blob = BufPool::inst().get(loc.len);
+        v.blob.len = loc.len;
+        HIP_CHECK(hipMemcpyAsync(v.blob.p, O.arena + loc.off, loc.len,
+                                 hipMemcpyDeviceToHost, O.s_read));
+        planned += round_up(loc.len, kGranule);
         loc.spilled = true;
-        it2 = lru_.erase(loc.lru);  // returns next; loop steps back from it
-        metrics().spilled_chunks++;
-        metrics().spill_bytes += loc.len;
+        it2 = lru_.erase(loc.lru);
+        victims.push_back(std::move(v));
     }
-    return freed ? St::ok : St::no_space;
+    if (victims.empty()) return St::no_space;
+    HIP_CHECK(hipStreamSynchronize(O.s_read));
+    for (auto& v : victims) {
+        St s = spill_.put(v.fp, std::move(v.blob));
+        if (s != St::ok && s != St::exists) return s;
+        O.arena_free(v.off, v.len);
+        metrics().spilled_chunks++;
+        metrics().spill_bytes += v.len;
+    }
+    return St::ok;
 }
 
 St GpuEngine::get_chunks(const GetChunk* get, int n) {
