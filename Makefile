@@ -26,7 +26,23 @@ HOST_OBJS := $(HOST_SRCS:.cc=.o)
 HIP_OBJS  := $(HIP_SRCS:.hip=.o)
 OBJS      := $(HOST_OBJS) $(HIP_OBJS)
 
-.PHONY: all clean test-bins
+.PHONY: all clean test-bins tsan asan
+
+# Sanitizer flavors: host C++ only (kernels unchanged).
+#   make asan  -> bin/edgehip-asan      make tsan -> bin/edgehip-tsan
+tsan:
+	$(MAKE) clean >/dev/null
+	$(MAKE) CXXFLAGS="-O1 -g -std=c++17 -fPIC -pthread -MMD -MP -I$(ROCM)/include -D__HIP_PLATFORM_AMD__ -fsanitize=thread" \
+	        LDFLAGS="$(LDFLAGS) -fsanitize=thread" bin/edgehip
+	mv bin/edgehip bin/edgehip-tsan
+	$(MAKE) clean >/dev/null
+
+asan:
+	$(MAKE) clean >/dev/null
+	$(MAKE) CXXFLAGS="-O1 -g -std=c++17 -fPIC -pthread -MMD -MP -I$(ROCM)/include -D__HIP_PLATFORM_AMD__ -fsanitize=address" \
+	        LDFLAGS="$(LDFLAGS) -fsanitize=address" bin/edgehip
+	mv bin/edgehip bin/edgehip-asan
+	$(MAKE) clean >/dev/null
 
 all: bin/edgehip lib/libedgehip.so bin/kernbench
 

@@ -113,7 +113,10 @@ def fuse_usable():
     import stat
     try:
         if not os.path.exists("/dev/fuse"):
-            os.mknod("/dev/fuse", stat.S_IFCHR | 0o600, os.makedev(10, 229))
+            try:
+                os.mknod("/dev/fuse", stat.S_IFCHR | 0o600, os.makedev(10, 229))
+            except FileExistsError:
+                pass  # another rank won the race
         fd = os.open("/dev/fuse", os.O_RDWR)
         os.close(fd)
         return True
