@@ -26,6 +26,7 @@ HOST_OBJS := $(HOST_SRCS:.cc=.o)
 HIP_OBJS  := $(HIP_SRCS:.hip=.o)
 OBJS      := $(HOST_OBJS) $(HIP_OBJS)
 
+.DEFAULT_GOAL := all
 .PHONY: all clean test-bins tsan asan
 
 # Sanitizer flavors: host C++ only (kernels unchanged).

@@ -65,6 +65,7 @@ struct ArenaLoc {
     int16_t dev = 0;    // owning device slot
     bool spilled = false;
     bool compressed = false;
+    bool evicting = false;  // D2H staged; slot must not be freed mid-copy
     std::list<Fp>::iterator lru;  // valid when !spilled
 };
 
@@ -94,6 +95,7 @@ struct DevCtx {
     std::atomic<uint64_t> used{0};
 
     PipeSlot slots[kPipeSlots];
+    std::mutex evict_mu;
     std::mutex slot_mu;
     std::condition_variable slot_cv;
     uint32_t slot_free_mask = (1u << kPipeSlots) - 1;
@@ -486,7 +488,7 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
         bool index_hint_dup = do_fp && conf_.dedup && h_dups[i] != 0;
         (void)index_hint_dup;  // host map confirms (stale tags after deletes)
         {
-            std::lock_guard<std::mutex> g(map_mu_);
+            std::unique_lock<std::mutex> g(map_mu_);
             if (conf_.dedup && do_fp) {
                 auto it = map_.find(r.fp);
                 if (it != map_.end()) {
@@ -502,10 +504,12 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
             if (!dup) {
                 int odev = owner_of(r.fp);
                 uint64_t aoff;
-                if (!devs_[odev].arena_alloc(stored, &aoff)) {
+                int tries = 0;
+                while (!devs_[odev].arena_alloc(stored, &aoff)) {
+                    g.unlock();  // eviction D2H must not block other writers
                     St s = evict_for(odev, stored);
-                    if (s != St::ok) return s;
-                    if (!devs_[odev].arena_alloc(stored, &aoff)) return St::no_space;
+                    g.lock();
+                    if (s != St::ok || ++tries > 8) return s == St::ok ? St::no_space : s;
                 }
                 ArenaLoc loc;
                 loc.off = aoff;
@@ -596,11 +600,15 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
 }
 
 St GpuEngine::evict_for(int dev, uint32_t need_bytes) {
+    // Called WITHOUT map_mu_. Three phases so the eviction D2H never blocks
+    // concurrent dedup/alloc (config 4's 1 TiB object streams through spill):
+    //   1. under map_mu_: pick LRU victims (no state change yet)
+    //   2. unlocked: async D2H of all victims on this device, one sync
+    //   3. under map_mu_: victims still present+unspilled commit to spill;
+    //      ones released meanwhile just drop their staged blob.
     if (conf_.spill_bytes == 0) return St::no_space;
-    // Batch the eviction (config 4's 1 TiB object streams through spill):
-    // pick LRU victims up to ~256 MiB, launch ALL D2H copies async, one sync,
-    // then hand the blobs to the spill store and free the arena slots.
     DevCtx& O = devs_[dev];
+    std::lock_guard<std::mutex> eg(O.evict_mu);  // one eviction per device
     HIP_CHECK(hipSetDevice(O.hip_dev));
     struct Victim {
         Fp fp;
@@ -609,36 +617,45 @@ St GpuEngine::evict_for(int dev, uint32_t need_bytes) {
         Blob blob;
     };
     std::vector<Victim> victims;
-    uint64_t planned = 0;
-    const uint64_t target =
-        std::max<uint64_t>((uint64_t)need_bytes * 4, 256 * MiB);
-    auto it2 = lru_.end();
-    while (planned < target && it2 != lru_.begin()) {
-        --it2;
-        auto it = map_.find(*it2);
-        if (it == map_.end() || it->second.spilled || it->second.dev != dev) continue;
-        ArenaLoc& loc = it->second;
-        Victim v;
-        v.fp = *it2;
-        v.off = loc.off;
-        v.len = loc.len;
-        v.blob = BufPool::inst().get(loc.len);
-        v.blob.len = loc.len;
-        HIP_CHECK(hipMemcpyAsync(v.blob.p, O.arena + loc.off, loc.len,
-                                 hipMemcpyDeviceToHost, O.s_read));
-        planned += round_up(loc.len, kGranule);
-        loc.spilled = true;
-        it2 = lru_.erase(loc.lru);
-        victims.push_back(std::move(v));
+    {
+        std::lock_guard<std::mutex> g(map_mu_);
+        uint64_t planned = 0;
+        const uint64_t target = std::max<uint64_t>((uint64_t)need_bytes * 4, 256 * MiB);
+        auto it2 = lru_.end();
+        while (planned < target && it2 != lru_.begin()) {
+            --it2;
+            auto it = map_.find(*it2);
+            if (it == map_.end() || it->second.spilled || it->second.dev != dev ||
+                it->second.evicting)
+                continue;
+            it->second.evicting = true;
+            victims.push_back({*it2, it->second.off, it->second.len, Blob{}});
+            planned += round_up(it->second.len, kGranule);
+        }
     }
     if (victims.empty()) return St::no_space;
-    HIP_CHECK(hipStreamSynchronize(O.s_read));
     for (auto& v : victims) {
-        St s = spill_.put(v.fp, std::move(v.blob));
-        if (s != St::ok && s != St::exists) return s;
-        O.arena_free(v.off, v.len);
-        metrics().spilled_chunks++;
-        metrics().spill_bytes += v.len;
+        v.blob = BufPool::inst().get(v.len);
+        v.blob.len = v.len;
+        HIP_CHECK(hipMemcpyAsync(v.blob.p, O.arena + v.off, v.len, hipMemcpyDeviceToHost,
+                                 O.s_read));
+    }
+    HIP_CHECK(hipStreamSynchronize(O.s_read));
+    {
+        std::lock_guard<std::mutex> g(map_mu_);
+        for (auto& v : victims) {
+            auto it = map_.find(v.fp);
+            if (it == map_.end() || it->second.spilled) continue;  // released meanwhile
+            ArenaLoc& loc = it->second;
+            loc.evicting = false;
+            St s = spill_.put(v.fp, std::move(v.blob));
+            if (s != St::ok && s != St::exists) return s;
+            O.arena_free(loc.off, loc.len);
+            loc.spilled = true;
+            lru_.erase(loc.lru);
+            metrics().spilled_chunks++;
+            metrics().spill_bytes += loc.len;
+        }
     }
     return St::ok;
 }
