@@ -203,6 +203,8 @@ def main():
     use_gpu = args.engine == "gpu" or (args.engine is None and gpu_available())
     if args.engine == "gpu" and not gpu_available():
         raise SystemExit("engine=gpu requested but no GPU visible")
+    if not use_gpu and args.gib_per_step == 4 and "--gib-per-step" not in sys.argv:
+        args.gib_per_step = 1  # host engine is ~100x slower; keep defaults quick
     opts = (f"engine=gpu,gpu_first={local_rank},fp=blake3,compress=lz4,dedup=1,"
             f"arena=32g,spill=0,threads=16,writeback=1"
             if use_gpu else
