@@ -635,7 +635,9 @@ St GpuEngine::evict_for(int dev, uint32_t need_bytes) {
     }
     if (victims.empty()) return St::no_space;
     for (auto& v : victims) {
-        v.blob = BufPool::inst().get(v.len);
+        // pageable blobs: the spill tier must not consume pinned memory, and
+        // hipHostMalloc page-locking cost (~2 GB/s) was the eviction wall
+        v.blob = BufPool::make_exact(v.len);
         v.blob.len = v.len;
         HIP_CHECK(hipMemcpyAsync(v.blob.p, O.arena + v.off, v.len, hipMemcpyDeviceToHost,
                                  O.s_read));
