@@ -18,6 +18,7 @@
 #include <algorithm>
 #include <condition_variable>
 #include <list>
+#include <shared_mutex>
 #include <mutex>
 #include <stdexcept>
 #include <unordered_map>
@@ -96,6 +97,10 @@ struct DevCtx {
 
     PipeSlot slots[kPipeSlots];
     std::mutex evict_mu;
+    // readers of arena slots hold shared; the evictor takes exclusive before
+    // freeing slots, so an in-flight D2H can never read a recycled slot.
+    // Lock order: evict_rw BEFORE map_mu_ whenever both are held.
+    std::shared_mutex evict_rw;
     std::mutex slot_mu;
     std::condition_variable slot_cv;
     uint32_t slot_free_mask = (1u << kPipeSlots) - 1;
@@ -644,6 +649,7 @@ St GpuEngine::evict_for(int dev, uint32_t need_bytes) {
     }
     HIP_CHECK(hipStreamSynchronize(O.s_read));
     {
+        std::unique_lock<std::shared_mutex> rw(O.evict_rw);  // drain arena readers
         std::lock_guard<std::mutex> g(map_mu_);
         for (auto& v : victims) {
             auto it = map_.find(v.fp);
@@ -751,15 +757,54 @@ St GpuEngine::get_chunks(const GetChunk* get, int n) {
 
     for (size_t d = 0; d < devs_.size(); d++) {
         DevCtx& D = devs_[d];
-        std::vector<DevReq*> mine;
+        std::vector<DevReq*> mine0;
         for (auto& q : dev_reqs)
-            if (q.dev == (int)d) mine.push_back(&q);
-        if (mine.empty()) continue;
+            if (q.dev == (int)d) mine0.push_back(&q);
+        if (mine0.empty()) continue;
         auto& R = D.rslots[D.rrr.fetch_add(1) % DevCtx::kReadSlots];
         std::lock_guard<std::mutex> pg(R.mu);
+        std::shared_lock<std::shared_mutex> rw(D.evict_rw);  // pin arena slots
         HIP_CHECK(hipSetDevice(D.hip_dev));
+        // revalidate: a victim may have been spilled between classify and
+        // acquiring the shared lock — route those through the spill path
+        std::vector<DevReq*> respill;
+        {
+            std::lock_guard<std::mutex> lk(map_mu_);
+            for (auto& q : dev_reqs) {
+                if (q.dev != (int)d) continue;
+                auto it = map_.find(q.g->ref->fp);
+                if (it == map_.end()) return St::not_found;
+                if (it->second.spilled) {
+                    q.dev = -1;  // handled below from host spill
+                    respill.push_back(&q);
+                } else {
+                    q.arena_off = it->second.off;
+                    q.stored = it->second.len;
+                    q.compressed = it->second.compressed;
+                }
+            }
+        }
+        for (auto* qp : respill) {
+            const GetChunk& g = *qp->g;
+            if (!qp->compressed) {
+                St s = spill_.get(g.ref->fp, g.out, g.off, g.len);
+                if (s != St::ok) return s;
+            } else {
+                std::vector<uint8_t> tmp(qp->stored), full(g.ref->logical_len);
+                St s = spill_.get(g.ref->fp, tmp.data(), 0, qp->stored);
+                if (s != St::ok) return s;
+                s = lz4_chunk_decompress(tmp.data(), tmp.size(), full.data(),
+                                         g.ref->logical_len);
+                if (s != St::ok) return s;
+                std::memcpy(g.out, full.data() + g.off, g.len);
+            }
+        }
         auto* h_dc = (DevChunk*)(R.pin_rmeta);
         auto* h_status = (uint32_t*)(R.pin_rmeta + 32 * KiB);
+        std::vector<DevReq*> mine;
+        for (auto* qp : mine0)
+            if (qp->dev == (int)d) mine.push_back(qp);
+        if (mine.empty()) continue;
 
         size_t j0 = 0;
         while (j0 < mine.size()) {
@@ -870,6 +915,7 @@ St GpuEngine::gather_chunk_to_dev(const ChunkRef& ref, int ddev, uint8_t* dst) {
                                  T.s_read));
     } else if (!comp) {
         DevCtx& O = devs_[odev];
+        std::shared_lock<std::shared_mutex> rw(O.evict_rw);
         if (odev == ddev) {
             HIP_CHECK(hipMemcpyAsync(dst, O.arena + aoff, stored, hipMemcpyDeviceToDevice,
                                      T.s_read));
@@ -877,10 +923,13 @@ St GpuEngine::gather_chunk_to_dev(const ChunkRef& ref, int ddev, uint8_t* dst) {
             HIP_CHECK(hipMemcpyPeerAsync(dst, T.hip_dev, O.arena + aoff, O.hip_dev, stored,
                                          T.s_read));
         }
+        HIP_CHECK(hipSetDevice(T.hip_dev));
+        HIP_CHECK(hipStreamSynchronize(T.s_read));  // complete under the lock
     } else {
         DevCtx& O = devs_[odev];
         auto& R = O.rslots[0];
         std::lock_guard<std::mutex> og(R.mu);
+        std::shared_lock<std::shared_mutex> rw(O.evict_rw);
         HIP_CHECK(hipSetDevice(O.hip_dev));
         auto* h_dc = (DevChunk*)R.pin_rmeta;
         auto* h_status = (uint32_t*)(R.pin_rmeta + 32 * KiB);
