@@ -385,6 +385,20 @@ class GpuEngine : public Engine {
 St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
     RoctxRange rr_ingest("edgehip::ingest_batch");
     const uint32_t cs = (uint32_t)max_chunk_;
+    // proactive eviction: keep ~8% arena headroom so writers rarely hit a
+    // hard alloc failure (which stalls every job until a 256 MiB pass lands);
+    // evict_mu's try-lock makes exactly one job pay while the rest ingest
+    if (conf_.spill_bytes) {
+        for (auto& Dv : devs_) {
+            if (Dv.used.load() + (uint64_t)n * cs > Dv.arena_sz - Dv.arena_sz / 12) {
+                if (Dv.evict_mu.try_lock()) {
+                    Dv.evict_mu.unlock();
+                    (void)evict_for((int)(&Dv - devs_.data()), (uint32_t)(8 * cs));
+                }
+                break;
+            }
+        }
+    }
     const bool do_lz4 = conf_.compress == Compress::lz4;
     const bool do_fp = conf_.fp == FpAlgo::blake3;
     const int ingest_dev = (int)(rr_.fetch_add(1) % devs_.size());
