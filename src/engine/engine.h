@@ -26,6 +26,9 @@ struct PutChunk {
     uint32_t len;
     Blob* donate = nullptr;  // optional: pool buffer holding `data`; the engine
                              // ALWAYS consumes it (stores or recycles) when set
+    bool pinned_src = false;  // `data` is pinned (pool) memory the CALLER keeps
+                              // alive through the call: engine may DMA from it
+                              // directly without taking ownership
 };
 
 struct GetChunk {

@@ -420,7 +420,8 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
     uint64_t off = 0;
     uint64_t staged = 0;
     for (int i = 0; i < n; i++) {
-        if (pool_pinned && put[i].donate && put[i].donate->p == put[i].data) {
+        if (pool_pinned &&
+            ((put[i].donate && put[i].donate->p == put[i].data) || put[i].pinned_src)) {
             // dirty buffer is already pinned: DMA straight from it
             HIP_CHECK(hipMemcpyAsync(S.d_stage + off, put[i].data, put[i].len,
                                      hipMemcpyHostToDevice, S.s));

@@ -1,6 +1,9 @@
 #include "obj.h"
 
 #include <algorithm>
+#include <deque>
+#include <functional>
+#include <thread>
 
 #include "cdc.h"
 #include "obs.h"
@@ -11,7 +14,106 @@ namespace edgehip {
 // is what gives the GPU engine kernel-launch-amortizing work.
 static constexpr int kPutBatch = 32;
 
-Object::~Object() = default;
+// Background flushers (SURVEY §4.3 "overlap H2D(n+1) with hash/LZ4(n)"):
+// write() hands full-chunk batches here so the writer's memcpy stream
+// overlaps the engine pipeline; flush()/commit() drain.
+class FlushPool {
+  public:
+    static FlushPool& inst() {
+        static FlushPool* p = new FlushPool;  // leaked: workers outlive statics
+        return *p;
+    }
+    void submit(std::function<void()> fn) {
+        {
+            std::lock_guard<std::mutex> g(mu_);
+            if (!started_) {
+                started_ = true;
+                // process-lifetime workers: detached so static teardown of
+                // this singleton never destroys a joinable std::thread
+                for (int i = 0; i < 4; i++) std::thread([this] { loop(); }).detach();
+            }
+            q_.push_back(std::move(fn));
+        }
+        cv_.notify_one();
+    }
+
+  private:
+    void loop() {
+        for (;;) {
+            std::function<void()> fn;
+            {
+                std::unique_lock<std::mutex> g(mu_);
+                cv_.wait(g, [&] { return !q_.empty(); });
+                fn = std::move(q_.front());
+                q_.pop_front();
+            }
+            fn();
+        }
+    }
+    std::mutex mu_;
+    std::condition_variable cv_;
+    std::deque<std::function<void()>> q_;
+    bool started_ = false;
+};
+
+Object::~Object() {
+    std::unique_lock<std::mutex> g(mu_);
+    wait_inflight_locked(g);
+}
+
+void Object::wait_inflight_locked(std::unique_lock<std::mutex>& lk) {
+    flush_cv_.wait(lk, [&] { return inflight_ == 0; });
+}
+
+// Hand a set of FULL dirty chunks to the background flushers. The task OWNS
+// the blobs; readers see the bytes through inflight_view_ (non-owning, only
+// dereferenced under mu_ while the view exists — the task erases the view
+// under mu_ BEFORE its blobs die).
+void Object::submit_async_flush_locked(std::vector<uint64_t> idxs) {
+    struct Item {
+        uint64_t idx;
+        Blob data;
+        uint32_t len;
+    };
+    auto items = std::make_shared<std::vector<Item>>();
+    items->reserve(idxs.size());
+    for (uint64_t idx : idxs) {
+        auto it = dirty_.find(idx);
+        items->push_back({idx, std::move(it->second.data), it->second.len});
+        dirty_.erase(it);
+        inflight_view_[idx] = {items->back().data.p, items->back().len};
+    }
+    inflight_++;
+    FlushPool::inst().submit([this, items] {
+        std::vector<PutChunk> puts;
+        puts.reserve(items->size());
+        for (auto& it : *items)
+            puts.push_back({it.data.p, it.len, nullptr, /*pinned_src=*/true});
+        std::vector<ChunkRef> refs(puts.size());
+        St s = eng_->put_chunks(puts.data(), refs.data(), (int)puts.size());
+        {
+            std::lock_guard<std::mutex> g(mu_);
+            if (s != St::ok) {
+                if (async_err_ == St::ok) async_err_ = s;
+            } else {
+                uint64_t maxidx = 0;
+                for (auto& it : *items) maxidx = std::max(maxidx, it.idx);
+                if (mf_.chunks.size() <= maxidx) mf_.chunks.resize(maxidx + 1);
+                for (size_t i = 0; i < items->size(); i++) {
+                    uint64_t idx = (*items)[i].idx;
+                    drop_ref(mf_.chunks[idx]);
+                    mf_.chunks[idx] = refs[i];
+                    mark_stripe_dirty(idx);
+                    pending_publish_.push_back(refs[i]);
+                }
+            }
+            for (auto& it : *items) inflight_view_.erase(it.idx);
+            inflight_--;
+        }
+        flush_cv_.notify_all();
+        // `items` (and the blobs) die here, after the views are gone
+    });
+}
 
 void Object::drop_ref(const ChunkRef& r) {
     if (!r.fp.is_zero()) eng_->release_chunk(r);
@@ -118,9 +220,18 @@ St Object::load_chunk_for_rmw(uint64_t idx, Dirty* d) {
 }
 
 St Object::write(uint64_t off, const uint8_t* data, uint64_t len) {
-    std::lock_guard<std::mutex> g(mu_);
+    std::unique_lock<std::mutex> g(mu_);
     if (conf_->cdc) return cdc_write_locked(off, data, len);
     const uint64_t cs = conf_->chunk_size;
+    // rewriting a chunk that a background flush still holds: drain first
+    // (rare — only rewrites racing their own just-written data)
+    if (!inflight_view_.empty()) {
+        uint64_t first = off / cs, last = len ? (off + len - 1) / cs : first;
+        bool overlap = false;
+        for (uint64_t i = first; i <= last && !overlap; i++)
+            overlap = inflight_view_.count(i) != 0;
+        if (overlap) wait_inflight_locked(g);
+    }
     uint64_t pos = off;
     const uint8_t* p = data;
     uint64_t end = off + len;
@@ -182,31 +293,22 @@ St Object::write(uint64_t off, const uint8_t* data, uint64_t len) {
     size_ = std::max(size_, end);
     metrics().write_ops++;
     metrics().write_bytes += len;
+    if (async_err_ != St::ok) {
+        St e = async_err_;
+        async_err_ = St::ok;
+        return e;
+    }
 
-    // flush completed chunks (all but the one still growing at the tail)
+    // hand completed chunks (all but the still-growing tail) to the
+    // background flushers — the writer's memcpy stream overlaps the engine
     int full = 0;
     for (auto& kv : dirty_)
         if (kv.second.len == cs && kv.first != last_idx) full++;
-    if (full >= kPutBatch) {
+    if (full >= kPutBatch || (full > 0 && inflight_ == 0 && full >= kPutBatch / 4)) {
         std::vector<uint64_t> idxs;
-        std::vector<PutChunk> puts;
-        for (auto& kv : dirty_) {
-            if (kv.second.len == cs && kv.first != last_idx) {
-                idxs.push_back(kv.first);
-                puts.push_back({kv.second.data.p, kv.second.len, &kv.second.data});
-            }
-        }
-        std::vector<ChunkRef> refs(puts.size());
-        St s = eng_->put_chunks(puts.data(), refs.data(), (int)puts.size());
-        if (s != St::ok) return s;
-        if (mf_.chunks.size() <= idxs.back()) mf_.chunks.resize(idxs.back() + 1);
-        for (size_t i = 0; i < idxs.size(); i++) {
-            drop_ref(mf_.chunks[idxs[i]]);
-            mf_.chunks[idxs[i]] = refs[i];
-            dirty_.erase(idxs[i]);
-            mark_stripe_dirty(idxs[i]);
-            pending_publish_.push_back(refs[i]);
-        }
+        for (auto& kv : dirty_)
+            if (kv.second.len == cs && kv.first != last_idx) idxs.push_back(kv.first);
+        submit_async_flush_locked(std::move(idxs));
     }
     return St::ok;
 }
@@ -229,6 +331,7 @@ St Object::read(uint64_t off, uint64_t len, uint8_t* out, uint64_t* got) {
         uint32_t co = (uint32_t)(pos - idx * cs);
         uint32_t n = (uint32_t)std::min<uint64_t>(cs - co, end - pos);
         auto it = dirty_.find(idx);
+        auto iv = inflight_view_.find(idx);
         if (it != dirty_.end()) {
             Dirty& d = it->second;
             bool committed = idx < mf_.chunks.size() && !mf_.chunks[idx].fp.is_zero();
@@ -243,6 +346,11 @@ St Object::read(uint64_t off, uint64_t len, uint8_t* out, uint64_t* got) {
             }
             uint32_t avail = d.len > co ? std::min(n, d.len - co) : 0;
             if (avail) std::memcpy(o, d.data.p + co, avail);
+            if (avail < n) std::memset(o + avail, 0, n - avail);
+        } else if (iv != inflight_view_.end()) {
+            uint32_t vl = iv->second.second;
+            uint32_t avail = vl > co ? std::min(n, vl - co) : 0;
+            if (avail) std::memcpy(o, iv->second.first + co, avail);
             if (avail < n) std::memset(o + avail, 0, n - avail);
         } else if (idx < mf_.chunks.size() && !mf_.chunks[idx].fp.is_zero()) {
             const ChunkRef& r = mf_.chunks[idx];
@@ -324,10 +432,16 @@ St Object::preadv(const uint64_t* offs, const uint32_t* lens, int n, uint8_t* ou
             uint32_t co = (uint32_t)(pos - idx * cs);
             uint32_t nn = (uint32_t)std::min<uint64_t>(cs - co, end - pos);
             auto it = dirty_.find(idx);
+            auto iv = inflight_view_.find(idx);
             if (it != dirty_.end()) {
                 Dirty& d = it->second;
                 uint32_t avail = d.len > co ? std::min(nn, d.len - co) : 0;
                 if (avail) std::memcpy(o, d.data.p + co, avail);
+                if (avail < nn) std::memset(o + avail, 0, nn - avail);
+            } else if (iv != inflight_view_.end()) {
+                uint32_t vl = iv->second.second;
+                uint32_t avail = vl > co ? std::min(nn, vl - co) : 0;
+                if (avail) std::memcpy(o, iv->second.first + co, avail);
                 if (avail < nn) std::memset(o + avail, 0, nn - avail);
             } else if (idx < mf_.chunks.size() && !mf_.chunks[idx].fp.is_zero()) {
                 const ChunkRef& r = mf_.chunks[idx];
@@ -373,9 +487,15 @@ St Object::preadv(const uint64_t* offs, const uint32_t* lens, int n, uint8_t* ou
     return St::ok;
 }
 
-St Object::flush_locked() {
+St Object::flush_locked(std::unique_lock<std::mutex>& lk) {
     if (conf_->cdc) return cdc_cut_locked(true);
-    if (dirty_.empty()) return St::ok;
+    wait_inflight_locked(lk);
+    if (async_err_ != St::ok) {
+        St e = async_err_;
+        async_err_ = St::ok;
+        return e;
+    }
+    if (dirty_.empty()) return rs_update_locked();
     std::vector<uint64_t> idxs;
     std::vector<PutChunk> puts;
     const uint64_t cs = conf_->chunk_size;
@@ -417,12 +537,13 @@ St Object::flush_locked() {
 }
 
 St Object::flush() {
-    std::lock_guard<std::mutex> g(mu_);
-    return flush_locked();
+    std::unique_lock<std::mutex> g(mu_);
+    return flush_locked(g);
 }
 
 St Object::truncate(uint64_t size) {
-    std::lock_guard<std::mutex> g(mu_);
+    std::unique_lock<std::mutex> g(mu_);
+    wait_inflight_locked(g);
     if (conf_->cdc) {
         // CDC objects: truncate(0) resets; other sizes unsupported (append
         // streams — SURVEY §9 Q2 optional mode)
@@ -485,8 +606,8 @@ St Object::truncate(uint64_t size) {
 }
 
 St Object::commit(std::vector<uint8_t>* manifest_bytes) {
-    std::lock_guard<std::mutex> g(mu_);
-    St s = flush_locked();
+    std::unique_lock<std::mutex> g(mu_);
+    St s = flush_locked(g);
     if (s != St::ok) return s;
     mf_.logical_size = size_;
     if (manifest_bytes) mf_.encode(manifest_bytes);
@@ -502,7 +623,8 @@ St Object::commit(std::vector<uint8_t>* manifest_bytes) {
 }
 
 St Object::set_manifest(const Manifest& m) {
-    std::lock_guard<std::mutex> g(mu_);
+    std::unique_lock<std::mutex> g(mu_);
+    wait_inflight_locked(g);
     if (!dirty_.empty() || !tail_.empty()) return St::inval;
     mf_ = m;
     size_ = m.logical_size;

@@ -4,6 +4,7 @@
 // engine below parallelizes across objects and within batches).
 #pragma once
 
+#include <condition_variable>
 #include <map>
 #include <memory>
 #include <mutex>
@@ -47,7 +48,9 @@ class Object {
         bool loaded = false;  // true once committed content was merged
     };
 
-    St flush_locked();
+    St flush_locked(std::unique_lock<std::mutex>& lk);
+    void submit_async_flush_locked(std::vector<uint64_t> idxs);
+    void wait_inflight_locked(std::unique_lock<std::mutex>& lk);
     St load_chunk_for_rmw(uint64_t idx, Dirty* d);
     void drop_ref(const ChunkRef& r);
     // RS (judged config 4): refresh parity for stripes whose data changed;
@@ -62,6 +65,12 @@ class Object {
     Manifest mf_;                    // committed chunk refs (index = chunk idx)
     std::map<uint64_t, Dirty> dirty_;
     std::set<uint64_t> rs_dirty_;    // stripe indices with stale parity
+    int inflight_ = 0;               // background flush batches outstanding
+    std::condition_variable flush_cv_;
+    St async_err_ = St::ok;
+    // chunks handed to a background flush: non-owning views (task owns the
+    // blobs; valid while the entry exists — both maintained under mu_)
+    std::map<uint64_t, std::pair<const uint8_t*, uint32_t>> inflight_view_;
     std::vector<ChunkRef> pending_publish_;  // C4: refs committed since last publish
     // CDC mode (conf.cdc): append-only stream, variable chunk boundaries
     std::vector<uint64_t> cum_;   // cumulative END offset per committed chunk
