@@ -67,6 +67,8 @@ struct ArenaLoc {
     bool spilled = false;
     bool compressed = false;
     bool evicting = false;  // D2H staged; slot must not be freed mid-copy
+    bool pending = true;    // arena bytes not landed yet (batch copy deferred):
+                            // never evict, readers/dedup-hits wait on map_cv_
     std::list<Fp>::iterator lru;  // valid when !spilled
 };
 
@@ -376,6 +378,7 @@ class GpuEngine : public Engine {
     std::atomic<uint64_t> epoch_{0};
 
     mutable std::mutex map_mu_;
+    std::condition_variable map_cv_;
     std::unordered_map<Fp, ArenaLoc, FpHash> map_;
     std::list<Fp> lru_;  // front = hottest
 
@@ -511,6 +514,16 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
             std::unique_lock<std::mutex> g(map_mu_);
             if (conf_.dedup && do_fp) {
                 auto it = map_.find(r.fp);
+                if (it != map_.end() && it->second.pending) {
+                    // another putter published this fp but its copy hasn't
+                    // landed; wait so crc/stored_len below are final
+                    if (!map_cv_.wait_for(g, std::chrono::seconds(30), [&] {
+                            auto i2 = map_.find(r.fp);
+                            return i2 == map_.end() || !i2->second.pending;
+                        }))
+                        return St::io;
+                    it = map_.find(r.fp);
+                }
                 if (it != map_.end()) {
                     it->second.refs++;
                     r.stored_len = it->second.len;
@@ -529,7 +542,19 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
                     g.unlock();  // eviction D2H must not block other writers
                     St s = evict_for(odev, stored);
                     g.lock();
-                    if (s != St::ok || ++tries > 8) return s == St::ok ? St::no_space : s;
+                    if (s != St::ok || ++tries > 8) {
+                        // roll back this batch's published-but-unlanded chunks
+                        for (auto& p : pend) {
+                            auto it = map_.find(refs[p.i].fp);
+                            if (it == map_.end() || !it->second.pending) continue;
+                            devs_[it->second.dev].arena_free(it->second.off,
+                                                            it->second.len);
+                            lru_.erase(it->second.lru);
+                            map_.erase(it);
+                        }
+                        map_cv_.notify_all();
+                        return s == St::ok ? St::no_space : s;
+                    }
                 }
                 ArenaLoc loc;
                 loc.off = aoff;
@@ -601,6 +626,14 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
                                          D.hip_dev, p.stored, S.s));
         }
         HIP_CHECK(hipStreamSynchronize(S.s));
+        {   // bytes are in the owner arenas now: lift the publish barrier
+            std::lock_guard<std::mutex> g(map_mu_);
+            for (auto& p : pend) {
+                auto it = map_.find(refs[p.i].fp);
+                if (it != map_.end()) it->second.pending = false;
+            }
+        }
+        map_cv_.notify_all();
         if (conf_.crc) {
             std::lock_guard<std::mutex> g(map_mu_);
             for (int pass = 0; pass < 2; pass++) {
@@ -646,7 +679,7 @@ St GpuEngine::evict_for(int dev, uint32_t need_bytes) {
             --it2;
             auto it = map_.find(*it2);
             if (it == map_.end() || it->second.spilled || it->second.dev != dev ||
-                it->second.evicting)
+                it->second.evicting || it->second.pending)
                 continue;
             it->second.evicting = true;
             victims.push_back({*it2, it->second.off, it->second.len, Blob{}});
@@ -721,8 +754,17 @@ St GpuEngine::get_chunks(const GetChunk* get, int n) {
         bool comp = false;
         int odev = 0;
         {
-            std::lock_guard<std::mutex> lk(map_mu_);
+            std::unique_lock<std::mutex> lk(map_mu_);
             auto it = map_.find(r.fp);
+            if (it != map_.end() && it->second.pending) {
+                // publishing batch hasn't landed its arena copy yet
+                if (!map_cv_.wait_for(lk, std::chrono::seconds(30), [&] {
+                        auto i2 = map_.find(r.fp);
+                        return i2 == map_.end() || !i2->second.pending;
+                    }))
+                    return St::io;
+                it = map_.find(r.fp);
+            }
             if (it == map_.end()) return St::not_found;
             ArenaLoc& loc = it->second;
             spilled = loc.spilled;

@@ -305,10 +305,18 @@ St Object::write(uint64_t off, const uint8_t* data, uint64_t len) {
     for (auto& kv : dirty_)
         if (kv.second.len == cs && kv.first != last_idx) full++;
     if (full >= kPutBatch || (full > 0 && inflight_ == 0 && full >= kPutBatch / 4)) {
-        std::vector<uint64_t> idxs;
+        // backpressure: ≤2 batches in flight per object bounds pinned-pool
+        // growth (each in-flight chunk pins a pool blob until its put lands)
+        flush_cv_.wait(g, [&] { return inflight_ < 2; });
+        if (async_err_ != St::ok) {
+            St e = async_err_;
+            async_err_ = St::ok;
+            return e;
+        }
+        std::vector<uint64_t> idxs;  // recompute: dirty_ may change in wait
         for (auto& kv : dirty_)
             if (kv.second.len == cs && kv.first != last_idx) idxs.push_back(kv.first);
-        submit_async_flush_locked(std::move(idxs));
+        if (!idxs.empty()) submit_async_flush_locked(std::move(idxs));
     }
     return St::ok;
 }
