@@ -19,6 +19,7 @@
 #include <condition_variable>
 #include <list>
 #include <shared_mutex>
+#include <thread>
 #include <mutex>
 #include <stdexcept>
 #include <unordered_map>
@@ -487,6 +488,8 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
         int dev;            // owner device
     };
     std::vector<Pend> pend;
+    std::vector<int> crc_fixups;   // intra-batch dup hits on pending entries
+    std::vector<int> cross_fixups; // dup hits on another batch's pending entry
     for (int i = 0; i < n; i++) {
         ChunkRef& r = refs[i];
         r = ChunkRef{};
@@ -515,14 +518,15 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
             if (conf_.dedup && do_fp) {
                 auto it = map_.find(r.fp);
                 if (it != map_.end() && it->second.pending) {
-                    // another putter published this fp but its copy hasn't
-                    // landed; wait so crc/stored_len below are final
-                    if (!map_cv_.wait_for(g, std::chrono::seconds(30), [&] {
-                            auto i2 = map_.find(r.fp);
-                            return i2 == map_.end() || !i2->second.pending;
-                        }))
-                        return St::io;
-                    it = map_.find(r.fp);
+                    bool own = false;  // published by THIS batch? (own-stream
+                    for (auto& p : pend)  // wait would deadlock)
+                        if (refs[p.i].fp == r.fp) { own = true; break; }
+                    // don't block the batch: take the dedup hit now, finalize
+                    // the crc (and handle publisher rollback) after our sync
+                    if (own)
+                        crc_fixups.push_back(i);
+                    else
+                        cross_fixups.push_back(i);
                 }
                 if (it != map_.end()) {
                     it->second.refs++;
@@ -538,9 +542,15 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
                 int odev = owner_of(r.fp);
                 uint64_t aoff;
                 int tries = 0;
+                int waits = 0;
                 while (!devs_[odev].arena_alloc(stored, &aoff)) {
                     g.unlock();  // eviction D2H must not block other writers
                     St s = evict_for(odev, stored);
+                    if (s == St::exists && ++waits < 20000) {  // transient: another
+                        std::this_thread::sleep_for(std::chrono::microseconds(500));
+                        g.lock();                       // evictor / pending batch
+                        continue;
+                    }
                     g.lock();
                     if (s != St::ok || ++tries > 8) {
                         // roll back this batch's published-but-unlanded chunks
@@ -647,6 +657,33 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
             }
         }
     }
+    if (!crc_fixups.empty()) {
+        std::lock_guard<std::mutex> g(map_mu_);
+        for (int i : crc_fixups) {
+            auto it = map_.find(refs[i].fp);
+            if (it != map_.end()) refs[i].crc = it->second.crc;
+        }
+    }
+    for (int i : cross_fixups) {
+        // the publishing batch ran concurrently with ours; by now its sync
+        // has (almost always) landed — wait out the remainder, then take
+        // the final crc. If it rolled back (alloc failure), re-put ourselves.
+        std::unique_lock<std::mutex> g(map_mu_);
+        if (!map_cv_.wait_for(g, std::chrono::seconds(30), [&] {
+                auto i2 = map_.find(refs[i].fp);
+                return i2 == map_.end() || !i2->second.pending;
+            }))
+            return St::io;
+        auto it = map_.find(refs[i].fp);
+        if (it != map_.end()) {
+            refs[i].crc = it->second.crc;
+        } else {
+            // publisher rolled back (arena allocation failed) — the store is
+            // out of space; surface that instead of recursing into put (a
+            // recursive batch could self-deadlock on this PipeSlot)
+            return St::no_space;
+        }
+    }
     for (int i = 0; i < n; i++)
         if (put[i].donate) put[i].donate->release();
     return St::ok;
@@ -686,7 +723,17 @@ St GpuEngine::evict_for(int dev, uint32_t need_bytes) {
             planned += round_up(it->second.len, kGranule);
         }
     }
-    if (victims.empty()) return St::no_space;
+    if (victims.empty()) {
+        // Nothing selectable — but if another evictor has the LRU tail staged
+        // (evicting) or putter batches haven't landed (pending), space will
+        // appear: tell the caller to wait+retry instead of failing.
+        std::lock_guard<std::mutex> g(map_mu_);
+        for (auto& kv : map_)
+            if (kv.second.dev == dev && !kv.second.spilled &&
+                (kv.second.evicting || kv.second.pending))
+                return St::exists;  // sentinel: transient, retry
+        return St::no_space;
+    }
     for (auto& v : victims) {
         // pageable blobs: the spill tier must not consume pinned memory, and
         // hipHostMalloc page-locking cost (~2 GB/s) was the eviction wall
