@@ -735,10 +735,11 @@ St GpuEngine::evict_for(int dev, uint32_t need_bytes) {
         return St::no_space;
     }
     for (auto& v : victims) {
-        // pageable blobs: the spill tier must not consume pinned memory, and
-        // hipHostMalloc page-locking cost (~2 GB/s) was the eviction wall
-        v.blob = BufPool::make_exact(v.len);
-        v.blob.len = v.len;
+        // Pool (pinned) blobs, recycled: fresh-malloc spill paid first-touch
+        // page faults on every eviction (~3 GB/s wall). Pool blocks are
+        // faulted+locked once and reused forever — steady-state spill D2H
+        // runs at PCIe DMA rate. Locked-RAM use is bounded by the spill cap.
+        v.blob = BufPool::inst().get(v.len);
         HIP_CHECK(hipMemcpyAsync(v.blob.p, O.arena + v.off, v.len, hipMemcpyDeviceToHost,
                                  O.s_read));
     }
