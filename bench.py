@@ -180,6 +180,8 @@ def main():
     ap.add_argument("--engine", default=None, choices=[None, "gpu", "host"],
                     help="override engine (default: gpu if a GPU is visible)")
     ap.add_argument("--jobs", type=int, default=6, help="parallel I/O jobs (fio numjobs)")
+    ap.add_argument("--arena", default="32g", help="HBM arena size (GPU engine)")
+    ap.add_argument("--spill", default="0", help="host spill tier cap (0=off)")
     ap.add_argument("--mnt-base", default="/tmp/edgehip_bench")
     args = ap.parse_args()
 
@@ -206,7 +208,7 @@ def main():
     if not use_gpu and args.gib_per_step == 4 and "--gib-per-step" not in sys.argv:
         args.gib_per_step = 1  # host engine is ~100x slower; keep defaults quick
     opts = (f"engine=gpu,gpu_first={local_rank},fp=blake3,compress=lz4,dedup=1,"
-            f"arena=32g,spill=0,threads=16,writeback=1"
+            f"arena={args.arena},spill={args.spill},threads=16,writeback=1"
             if use_gpu else
             "engine=host,fp=blake3,compress=lz4,dedup=1,threads=16,writeback=1")
 
