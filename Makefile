@@ -45,7 +45,11 @@ asan:
 	mv bin/edgehip bin/edgehip-asan
 	$(MAKE) clean >/dev/null
 
-all: bin/edgehip lib/libedgehip.so bin/kernbench
+all: bin/edgehip lib/libedgehip.so bin/kernbench bin/engine_bench
+
+bin/engine_bench: bench/engine_bench.o lib/libedgehip.so
+	@mkdir -p bin
+	$(CXX) $(CXXFLAGS) -o $@ bench/engine_bench.o -Llib -ledgehip -Wl,-rpath,'$$ORIGIN/../lib' $(LDFLAGS)
 
 bin/kernbench: bench/kernbench.o $(HIP_OBJS:src/engine/gpu_engine.o=) src/gf256.o src/obs.o src/bufpool.o
 	@mkdir -p bin
@@ -68,7 +72,7 @@ lib/libedgehip.so: $(OBJS) src/capi.o src/capi_gpu.o
 	$(HIPCC) $(HIPFLAGS) -c -o $@ $<
 
 clean:
-	rm -f $(OBJS) src/main.o src/capi.o src/capi_gpu.o bench/kernbench.o bench/kernbench.d
+	rm -f $(OBJS) src/main.o src/capi.o src/capi_gpu.o bench/kernbench.o bench/kernbench.d bench/engine_bench.o bench/engine_bench.d
 	rm -f $(OBJS:.o=.d) src/main.d src/capi.d src/capi_gpu.d
 	rm -rf bin lib
 

@@ -1,0 +1,92 @@
+// Native ingest/read micro-bench: the bench.py object-api workload with zero
+// Python in the loop, for rocprofv3 attribution of the engine pipeline
+// (SURVEY.md §7). Usage: engine_bench [gib_per_job [jobs [opts]]]
+#include <chrono>
+#include <cstdint>
+#include <cstdio>
+#include <cstdlib>
+#include <cstring>
+#include <random>
+#include <thread>
+#include <vector>
+
+#include "../include/edgehip.h"
+
+static constexpr uint64_t MB = 1ull << 20;
+
+int main(int argc, char** argv) {
+    double gib = argc > 1 ? atof(argv[1]) : 2.0;
+    int jobs = argc > 2 ? atoi(argv[2]) : 6;
+    const char* opts = argc > 3 ? argv[3]
+                               : "engine=gpu,fp=blake3,compress=lz4,dedup=1,"
+                                 "arena=32g,spill=0";
+    void* eng = eh_engine_create(opts);
+    if (!eng) {
+        fprintf(stderr, "engine create failed (%s)\n", opts);
+        return 1;
+    }
+    // content ring shared across jobs, like bench.py: unique per MiB
+    uint64_t ring = 1024 * MB;
+    std::vector<uint8_t> buf(ring);
+    {
+        std::vector<std::thread> ts;
+        for (int t = 0; t < 8; t++)
+            ts.emplace_back([&, t] {
+                std::mt19937_64 g(t);
+                auto* p = (uint64_t*)(buf.data() + t * (ring / 8));
+                for (uint64_t i = 0; i < ring / 8 / 8; i++) p[i] = g();
+            });
+        for (auto& t : ts) t.join();
+    }
+
+    uint64_t per_job = (uint64_t)(gib * 1024.0) / jobs * MB;
+    const uint64_t unit = 8 * MB;
+    std::vector<double> wsec(jobs), rsec(jobs);
+    std::vector<int> errs(jobs, 0);
+
+    auto worker = [&](int j) {
+        void* o = eh_object_create(eng);
+        auto t0 = std::chrono::steady_clock::now();
+        for (uint64_t off = 0; off < per_job; off += unit) {
+            uint64_t src = ((off / unit) * jobs + j) % (ring / unit) * unit;
+            if (eh_object_write(o, off, buf.data() + src, unit) != 0) errs[j]++;
+        }
+        if (eh_object_flush(o) != 0) errs[j]++;
+        auto t1 = std::chrono::steady_clock::now();
+        std::vector<uint8_t> rbuf(unit);
+        for (uint64_t off = 0; off < per_job; off += unit)
+            if (eh_object_read(o, off, rbuf.data(), unit) != (int64_t)unit) errs[j]++;
+        auto t2 = std::chrono::steady_clock::now();
+        wsec[j] = std::chrono::duration<double>(t1 - t0).count();
+        rsec[j] = std::chrono::duration<double>(t2 - t1).count();
+        eh_object_truncate(o, 0);
+        eh_object_destroy(o);
+    };
+
+    // warmup: half the data once (grows the pinned pool to steady state)
+    {
+        std::vector<std::thread> ts;
+        for (int j = 0; j < jobs; j++) ts.emplace_back(worker, j);
+        for (auto& t : ts) t.join();
+    }
+    std::vector<std::thread> ts;
+    auto T0 = std::chrono::steady_clock::now();
+    for (int j = 0; j < jobs; j++) ts.emplace_back(worker, j);
+    for (auto& t : ts) t.join();
+    auto T1 = std::chrono::steady_clock::now();
+
+    double wall = std::chrono::duration<double>(T1 - T0).count();
+    double w = 0, r = 0;
+    int e = 0;
+    for (int j = 0; j < jobs; j++) {
+        w = std::max(w, wsec[j]);
+        r = std::max(r, rsec[j]);
+        e += errs[j];
+    }
+    double gibs = gib;
+    printf("engine_bench: write %.2f GiB/s  read %.2f GiB/s  combined %.2f GiB/s"
+           "  (%.2f GiB x2, %d jobs, errs %d)\n",
+           gibs / w, gibs / r, 2 * gibs / wall, gibs, jobs, e);
+    eh_engine_destroy(eng);
+    return e ? 1 : 0;
+}
