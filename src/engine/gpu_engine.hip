@@ -73,12 +73,14 @@ struct ArenaLoc {
     std::list<Fp>::iterator lru;  // valid when !spilled
 };
 
-constexpr int kPipeSlots = 4;  // concurrent ingest pipelines per device
+constexpr int kPipeSlots = 6;  // concurrent ingest pipelines per device
 
 // One ingest pipeline: stream + staging + scratch (double-buffered per device
 // so concurrent writers overlap H2D/kernels/D2D across slots — SURVEY §4.3).
 struct PipeSlot {
     hipStream_t s{};
+    hipStream_t s2{};    // second H2D lane: per-chunk DMA setup (~40 us/MiB
+    hipEvent_t ev2{};    // observed) halves when copies split across 2 SDMA queues
     uint8_t *d_stage = nullptr, *d_comp = nullptr, *d_segscratch = nullptr,
             *d_fps = nullptr;
     DevChunk *d_chunks = nullptr, *d_copy = nullptr;
@@ -116,7 +118,7 @@ struct DevCtx {
         uint8_t* pin_rmeta = nullptr;
         std::mutex mu;
     };
-    static constexpr int kReadSlots = 2;
+    static constexpr int kReadSlots = 4;
     ReadSlot rslots[kReadSlots];
     std::atomic<uint32_t> rrr{0};
 
@@ -207,6 +209,8 @@ class GpuEngine : public Engine {
             for (int sl = 0; sl < kPipeSlots; sl++) {
                 PipeSlot& S = D.slots[sl];
                 HIP_CHECK(hipStreamCreateWithFlags(&S.s, hipStreamNonBlocking));
+                HIP_CHECK(hipStreamCreateWithFlags(&S.s2, hipStreamNonBlocking));
+                HIP_CHECK(hipEventCreateWithFlags(&S.ev2, hipEventDisableTiming));
                 HIP_CHECK(hipMalloc(&S.d_stage, D.stage_sz));
                 HIP_CHECK(hipMalloc(&S.d_comp, D.stage_sz));
                 HIP_CHECK(hipMalloc(&S.d_segscratch, (uint64_t)kMaxSegs * kSegStride));
@@ -263,6 +267,8 @@ class GpuEngine : public Engine {
                     (void)hipFree(p);
                 (void)hipHostFree(S.pin_in);
                 (void)hipHostFree(S.pin_meta);
+                (void)hipEventDestroy(S.ev2);
+                (void)hipStreamDestroy(S.s2);
                 (void)hipStreamDestroy(S.s);
             }
             for (int rs2 = 0; rs2 < DevCtx::kReadSlots; rs2++) {
@@ -423,12 +429,16 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
     int nsegs = 0;
     uint64_t off = 0;
     uint64_t staged = 0;
+    bool used_s2 = false;
     for (int i = 0; i < n; i++) {
         if (pool_pinned &&
             ((put[i].donate && put[i].donate->p == put[i].data) || put[i].pinned_src)) {
-            // dirty buffer is already pinned: DMA straight from it
+            // dirty buffer is already pinned: DMA straight from it, copies
+            // alternating across both slot streams (2 SDMA queues)
+            hipStream_t hs = (i & 1) ? S.s2 : S.s;
+            used_s2 |= (i & 1) != 0;
             HIP_CHECK(hipMemcpyAsync(S.d_stage + off, put[i].data, put[i].len,
-                                     hipMemcpyHostToDevice, S.s));
+                                     hipMemcpyHostToDevice, hs));
         } else {
             std::memcpy(S.pin_in + staged, put[i].data, put[i].len);
             HIP_CHECK(hipMemcpyAsync(S.d_stage + off, S.pin_in + staged, put[i].len,
@@ -443,6 +453,10 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
                 ulong2{off + (uint64_t)s * kSegSz, std::min(kSegSz, put[i].len - s * kSegSz)};
         nsegs += ns;
         off += round_up(put[i].len, 64);
+    }
+    if (used_s2) {  // kernels on S.s must see the s2 copies
+        HIP_CHECK(hipEventRecord(S.ev2, S.s2));
+        HIP_CHECK(hipStreamWaitEvent(S.s, S.ev2, 0));
     }
     HIP_CHECK(hipMemcpyAsync(S.d_chunks, h_chunks, sizeof(DevChunk) * n,
                              hipMemcpyHostToDevice, S.s));
