@@ -11,7 +11,7 @@ __global__ __launch_bounds__(256) void k_crc32c_chunks(
     const uint8_t* __restrict__ src, const DevChunk* __restrict__ chunks,
     uint32_t* __restrict__ out /* one per chunk */, int nchunks) {
     (void)nchunks;
-    __shared__ uint32_t tab[4][256];
+    __shared__ uint32_t tab[8][256];  // slice-by-8: 8 B per iteration
     // build tables cooperatively
     for (int i = threadIdx.x; i < 256; i += blockDim.x) {
         uint32_t c = i;
@@ -19,7 +19,7 @@ __global__ __launch_bounds__(256) void k_crc32c_chunks(
         tab[0][i] = c;
     }
     __syncthreads();
-    for (int t = 1; t < 4; t++) {
+    for (int t = 1; t < 8; t++) {
         for (int i = threadIdx.x; i < 256; i += blockDim.x)
             tab[t][i] = (tab[t - 1][i] >> 8) ^ tab[0][tab[t - 1][i] & 0xff];
         __syncthreads();
@@ -35,13 +35,16 @@ __global__ __launch_bounds__(256) void k_crc32c_chunks(
         const uint8_t* q = base + off;
         uint32_t crc = 0xFFFFFFFFu;
         uint32_t i = 0;
-        if ((((uintptr_t)q) & 3) == 0) {
-            for (; i + 4 <= len; i += 4) {
-                uint32_t v;
-                __builtin_memcpy(&v, q + i, 4);
-                v ^= crc;
-                crc = tab[3][v & 0xff] ^ tab[2][(v >> 8) & 0xff] ^ tab[1][(v >> 16) & 0xff] ^
-                      tab[0][(v >> 24) & 0xff];
+        if ((((uintptr_t)q) & 7) == 0) {
+            for (; i + 8 <= len; i += 8) {
+                uint32_t w0, w1;
+                __builtin_memcpy(&w0, q + i, 4);
+                __builtin_memcpy(&w1, q + i + 4, 4);
+                uint32_t v = w0 ^ crc;
+                crc = tab[7][v & 0xff] ^ tab[6][(v >> 8) & 0xff] ^
+                      tab[5][(v >> 16) & 0xff] ^ tab[4][(v >> 24) & 0xff] ^
+                      tab[3][w1 & 0xff] ^ tab[2][(w1 >> 8) & 0xff] ^
+                      tab[1][(w1 >> 16) & 0xff] ^ tab[0][(w1 >> 24) & 0xff];
             }
         }
         for (; i < len; i++) crc = tab[0][(crc ^ q[i]) & 0xff] ^ (crc >> 8);
