@@ -418,7 +418,14 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
     struct SlotGuard {
         DevCtx& d;
         PipeSlot* s;
-        ~SlotGuard() { d.release_slot(s); }
+        ~SlotGuard() {
+            // error paths can return with async copies still in flight from
+            // caller-owned (about-to-die) blobs — drain before handing the
+            // slot (and the sources) back. No-op (~2 us) when already synced.
+            (void)hipStreamSynchronize(s->s);
+            (void)hipStreamSynchronize(s->s2);
+            d.release_slot(s);
+        }
     } slot_guard{D, SLP};
     HIP_CHECK(hipSetDevice(D.hip_dev));
 
