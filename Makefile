@@ -34,15 +34,18 @@ OBJS      := $(HOST_OBJS) $(HIP_OBJS)
 tsan:
 	$(MAKE) clean >/dev/null
 	$(MAKE) CXXFLAGS="-O1 -g -std=c++17 -fPIC -pthread -MMD -MP -I$(ROCM)/include -D__HIP_PLATFORM_AMD__ -fsanitize=thread" \
-	        LDFLAGS="$(LDFLAGS) -fsanitize=thread" bin/edgehip
-	mv bin/edgehip bin/edgehip-tsan
+	        LDFLAGS="$(LDFLAGS) -fsanitize=thread" bin/edgehip lib/libedgehip.so
+	@mkdir -p tsan
+	mv bin/edgehip tsan/edgehip-tsan
+	mv lib/libedgehip.so tsan/libedgehip-tsan.so
 	$(MAKE) clean >/dev/null
 
 asan:
 	$(MAKE) clean >/dev/null
 	$(MAKE) CXXFLAGS="-O1 -g -std=c++17 -fPIC -pthread -MMD -MP -I$(ROCM)/include -D__HIP_PLATFORM_AMD__ -fsanitize=address" \
 	        LDFLAGS="$(LDFLAGS) -fsanitize=address" bin/edgehip
-	mv bin/edgehip bin/edgehip-asan
+	@mkdir -p tsan
+	mv bin/edgehip tsan/edgehip-asan
 	$(MAKE) clean >/dev/null
 
 all: bin/edgehip lib/libedgehip.so bin/kernbench bin/engine_bench

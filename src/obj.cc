@@ -109,8 +109,10 @@ void Object::submit_async_flush_locked(std::vector<uint64_t> idxs) {
             }
             for (auto& it : *items) inflight_view_.erase(it.idx);
             inflight_--;
+            // notify under mu_: ~Object may destroy the cv the instant the
+            // last waiter sees inflight_ == 0 (TSan-caught teardown race)
+            flush_cv_.notify_all();
         }
-        flush_cv_.notify_all();
         // `items` (and the blobs) die here, after the views are gone
     });
 }
