@@ -536,7 +536,11 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
         (void)index_hint_dup;  // host map confirms (stale tags after deletes)
         {
             std::unique_lock<std::mutex> g(map_mu_);
-            if (conf_.dedup && do_fp) {
+            if (do_fp) {
+                // The store is content-addressed: same fp = same bytes, ONE
+                // map entry, even with dedup=0 (a second emplace would fail
+                // silently and corrupt refcounts). dedup=0 only skips the
+                // accounting, not the physical sharing.
                 auto it = map_.find(r.fp);
                 if (it != map_.end() && it->second.pending) {
                     bool own = false;  // published by THIS batch? (own-stream
@@ -554,8 +558,10 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
                     r.stored_len = it->second.len;
                     r.crc = it->second.crc;
                     r.flags = it->second.compressed ? ChunkRef::kFlagLz4 : 0;
-                    metrics().dedup_hits++;
-                    metrics().dedup_bytes_saved += put[i].len;
+                    if (conf_.dedup) {
+                        metrics().dedup_hits++;
+                        metrics().dedup_bytes_saved += put[i].len;
+                    }
                     dup = true;
                 }
             }
@@ -965,6 +971,17 @@ St GpuEngine::get_chunks(const GetChunk* get, int n) {
             st.blob = BufPool::inst().get((uint32_t)std::max<uint64_t>(outb, 64));
             uint64_t pout = 0;
             mi = 0;
+            // merge arena/blob-contiguous requests into one DMA: per-copy
+            // SDMA setup is ~40 us/MiB (profiles/engine_bench_r01) and the
+            // bump allocator makes sequentially-written chunks adjacent
+            const uint8_t* run_src = nullptr;
+            uint64_t run_out = 0, run_len = 0;
+            auto flush_run = [&]() {
+                if (!run_len) return;
+                HIP_CHECK(hipMemcpyAsync(st.blob.p + run_out, run_src, run_len,
+                                         hipMemcpyDeviceToHost, R.s));
+                run_len = 0;
+            };
             for (size_t j = j0; j < j1; j++) {
                 const GetChunk& g = *mine[j]->g;
                 const uint8_t* srcp;
@@ -974,11 +991,20 @@ St GpuEngine::get_chunks(const GetChunk* get, int n) {
                 } else {
                     srcp = D.arena + mine[j]->arena_off + g.off;
                 }
-                HIP_CHECK(hipMemcpyAsync(st.blob.p + pout, srcp, g.len,
-                                         hipMemcpyDeviceToHost, R.s));
+                if (run_len && srcp == run_src + run_len &&
+                    pout == run_out + run_len) {
+                    run_len += g.len;
+                } else {
+                    flush_run();
+                    run_src = srcp;
+                    run_out = pout;
+                    run_len = g.len;
+                }
                 st.copies.push_back({g.out, pout, g.len});
                 pout += round_up(g.len, 64);
+                if (pout != run_out + run_len) flush_run();  // padding gap
             }
+            flush_run();
             HIP_CHECK(hipStreamSynchronize(R.s));
             metrics().d2h_bytes += pout;
             for (int x = 0; x < mi; x++)

@@ -119,3 +119,27 @@ def test_readers_race_eviction():
             for t in cs:
                 t.join()
             assert not errs, errs[0]
+
+
+def test_dedup_off_duplicate_content_spill():
+    """dedup=0 with repeated content: same-fp chunks must share one map
+    entry (content-addressed) — a silent emplace failure here corrupted
+    refcounts and broke reads under spill (round-1 regression)."""
+    with Engine("engine=gpu,fp=blake3,compress=lz4,dedup=0,arena=128m,"
+                "spill=2g") as eng:
+        ring = rand_bytes(32 * MB, seed=77)
+        objs = [eng.object() for _ in range(3)]
+        try:
+            for j, o in enumerate(objs):
+                for off in range(0, 96 * MB, MB):
+                    src = (off + j * MB) % (32 * MB)
+                    o.write(off, ring[src:src + MB])
+                o.flush()
+            for j, o in enumerate(objs):
+                for off in range(0, 96 * MB, 7 * MB):
+                    src = (off + j * MB) % (32 * MB)
+                    assert o.read(off, MB) == ring[src:src + MB], (j, off)
+        finally:
+            for o in objs:
+                o.truncate(0)
+                o.close()
