@@ -43,9 +43,10 @@ tsan:
 asan:
 	$(MAKE) clean >/dev/null
 	$(MAKE) CXXFLAGS="-O1 -g -std=c++17 -fPIC -pthread -MMD -MP -I$(ROCM)/include -D__HIP_PLATFORM_AMD__ -fsanitize=address" \
-	        LDFLAGS="$(LDFLAGS) -fsanitize=address" bin/edgehip
+	        LDFLAGS="$(LDFLAGS) -fsanitize=address" bin/edgehip lib/libedgehip.so
 	@mkdir -p tsan
 	mv bin/edgehip tsan/edgehip-asan
+	mv lib/libedgehip.so tsan/libedgehip-asan.so
 	$(MAKE) clean >/dev/null
 
 all: bin/edgehip lib/libedgehip.so bin/kernbench bin/engine_bench
