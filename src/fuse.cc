@@ -103,6 +103,7 @@ St FuseMount::mount(const std::string& mountpoint) {
     rep.o.minor = FUSE_KERNEL_MINOR_VERSION;    // 34 (header's); kernel negotiates min
     rep.o.max_readahead = conf_->max_readahead;
     uint32_t want = FUSE_ASYNC_READ | FUSE_BIG_WRITES | FUSE_PARALLEL_DIROPS |
+                    FUSE_DO_READDIRPLUS | FUSE_READDIRPLUS_AUTO |
                     FUSE_MAX_PAGES | FUSE_AUTO_INVAL_DATA;
     if (conf_->writeback) want |= FUSE_WRITEBACK_CACHE;
     rep.o.flags = want & ii->flags;
@@ -487,6 +488,41 @@ void FuseMount::handle(const uint8_t* req, size_t len, std::vector<uint8_t>* rep
             reply_body(reply_buf->data(), reply_buf->size());
             return;
         }
+        case FUSE_READDIRPLUS: {
+            // READDIR + LOOKUP in one trip (FUSE_DO_READDIRPLUS): each entry
+            // carries a full fuse_entry_out, and the kernel takes an nlookup
+            // reference exactly as if it had sent a LOOKUP — except for
+            // "." / "..", which go out with nodeid 0 (no reference).
+            auto* r = (const fuse_read_in*)arg;
+            std::vector<DirEntry> ents;
+            St s = vfs_->readdir(h->nodeid, &ents);
+            if (s != St::ok) { reply_st(s); return; }
+            reply_buf->clear();
+            uint64_t idx = r->offset;
+            while (idx < ents.size()) {
+                const DirEntry& de = ents[idx];
+                size_t entlen = FUSE_NAME_OFFSET_DIRENTPLUS + de.name.size();
+                size_t entlen_aligned = FUSE_DIRENT_ALIGN(entlen);
+                if (reply_buf->size() + entlen_aligned > r->size) break;
+                size_t base = reply_buf->size();
+                reply_buf->resize(base + entlen_aligned, 0);
+                auto* dp = (fuse_direntplus*)(reply_buf->data() + base);
+                bool dotty = de.name == "." || de.name == "..";
+                Inode* in = nullptr;
+                if (!dotty && vfs_->getattr(de.ino, &in) == St::ok) {
+                    fill_entry(in, &dp->entry_out);
+                    vfs_->ref(in);  // kernel holds it until FORGET
+                }
+                dp->dirent.ino = de.ino;
+                dp->dirent.off = idx + 1;
+                dp->dirent.namelen = (uint32_t)de.name.size();
+                dp->dirent.type = de.type;
+                std::memcpy(dp->dirent.name, de.name.data(), de.name.size());
+                idx++;
+            }
+            reply_body(reply_buf->data(), reply_buf->size());
+            return;
+        }
         case FUSE_RELEASEDIR:
         case FUSE_FSYNCDIR:
             reply_err(0);
@@ -587,7 +623,6 @@ void FuseMount::handle(const uint8_t* req, size_t len, std::vector<uint8_t>* rep
         case FUSE_BMAP:
         case FUSE_IOCTL:
         case FUSE_POLL:
-        case FUSE_READDIRPLUS:
         default:
             reply_err(-ENOSYS);
             return;

@@ -128,3 +128,28 @@ def test_symlink_and_hardlink(tmp_mount):
         p.unlink()
         assert (tmp_mount / "hard").read_text() == "payload\n"  # survives
         assert (tmp_mount / "hard").stat().st_nlink == 1
+
+
+def test_readdirplus_listing_with_attrs(tmp_mount):
+    """`ls -l` pattern: READDIRPLUS (offered via FUSE_DO_READDIRPLUS) returns
+    entry+attr in one trip; every stat must be correct and later FORGETs must
+    not underflow (unmount at fixture teardown would hang/crash if so)."""
+    d = os.path.join(tmp_mount, "plusdir")
+    os.mkdir(d)
+    sizes = {}
+    for i in range(200):
+        p = os.path.join(d, f"f{i:03d}")
+        with open(p, "wb") as f:
+            f.write(b"x" * (i * 7))
+        sizes[f"f{i:03d}"] = i * 7
+    with os.scandir(d) as it:  # scandir consumes d_type + stat from the plus reply
+        seen = {}
+        for e in it:
+            st = e.stat()
+            assert e.is_file()
+            seen[e.name] = st.st_size
+    assert seen == sizes
+    # second pass straight after FORGET churn
+    assert sorted(os.listdir(d)) == sorted(sizes)
+    for name, sz in sizes.items():
+        assert os.stat(os.path.join(d, name)).st_size == sz
