@@ -613,10 +613,57 @@ void FuseMount::handle(const uint8_t* req, size_t len, std::vector<uint8_t>* rep
             reply_body(&e, sizeof(e));
             return;
         }
-        case FUSE_GETXATTR:
-        case FUSE_SETXATTR:
-        case FUSE_LISTXATTR:
-        case FUSE_REMOVEXATTR:
+        case FUSE_SETXATTR: {
+            // no FUSE_SETXATTR_EXT negotiated -> 8-byte compat header
+            auto* x = (const fuse_setxattr_in*)arg;
+            const char* name = (const char*)arg + FUSE_COMPAT_SETXATTR_IN_SIZE;
+            const char* val = name + strlen(name) + 1;
+            St s = vfs_->setxattr(h->nodeid, name, std::string(val, x->size),
+                                  x->flags);
+            if (s == St::not_found) { reply_err(-ENODATA); return; }
+            reply_st(s);
+            return;
+        }
+        case FUSE_GETXATTR: {
+            auto* x = (const fuse_getxattr_in*)arg;
+            const char* name = (const char*)arg + sizeof(fuse_getxattr_in);
+            std::string v;
+            St s = vfs_->getxattr(h->nodeid, name, &v);
+            if (s == St::not_found) { reply_err(-ENODATA); return; }
+            if (s != St::ok) { reply_st(s); return; }
+            if (x->size == 0) {  // size probe
+                fuse_getxattr_out go{};
+                go.size = (uint32_t)v.size();
+                reply_body(&go, sizeof(go));
+            } else if (v.size() > x->size) {
+                reply_err(-ERANGE);
+            } else {
+                reply_body(v.data(), v.size());
+            }
+            return;
+        }
+        case FUSE_LISTXATTR: {
+            auto* x = (const fuse_getxattr_in*)arg;
+            std::string names;
+            St s = vfs_->listxattr(h->nodeid, &names);
+            if (s != St::ok) { reply_st(s); return; }
+            if (x->size == 0) {
+                fuse_getxattr_out go{};
+                go.size = (uint32_t)names.size();
+                reply_body(&go, sizeof(go));
+            } else if (names.size() > x->size) {
+                reply_err(-ERANGE);
+            } else {
+                reply_body(names.data(), names.size());
+            }
+            return;
+        }
+        case FUSE_REMOVEXATTR: {
+            St s = vfs_->removexattr(h->nodeid, (const char*)arg);
+            if (s == St::not_found) { reply_err(-ENODATA); return; }
+            reply_st(s);
+            return;
+        }
         case FUSE_GETLK:
         case FUSE_SETLK:
         case FUSE_SETLKW:

@@ -180,6 +180,32 @@ St Vfs::restore() {
                     if (t) t->nlink++;
                     break;
                 }
+                if (r.path.rfind("\x01xattr:", 0) == 0 ||
+                    r.path.rfind("\x01xattr-:", 0) == 0) {
+                    bool rm = r.path[6] == '-';
+                    std::string body = r.path.substr(rm ? 8 : 7);
+                    Inode* in = nullptr;
+                    if (r.path2 == "/") {
+                        in = get(1);
+                    } else if (resolve_parent_locked(r.path2, &parent, &leaf) ==
+                               St::ok) {
+                        Inode* p2 = get(parent);
+                        if (p2) {
+                            auto cit = p2->children.find(leaf);
+                            if (cit != p2->children.end()) in = get(cit->second);
+                        }
+                    }
+                    if (in) {
+                        if (rm) {
+                            in->xattrs.erase(body);
+                        } else {
+                            size_t nul = body.find('\0');
+                            if (nul != std::string::npos)
+                                in->xattrs[body.substr(0, nul)] = body.substr(nul + 1);
+                        }
+                    }
+                    break;
+                }
                 if (resolve_parent_locked(r.path, &parent, &leaf) != St::ok) break;
                 if (resolve_parent_locked(r.path2, &np, &nleaf) != St::ok) break;
                 Inode* p = get(parent);
@@ -416,6 +442,54 @@ St Vfs::readdir(uint64_t ino, std::vector<DirEntry>* out) {
         else if (c && S_ISLNK(c->mode)) t = DT_LNK;
         out->push_back({kv.first, kv.second, t});
     }
+    return St::ok;
+}
+
+St Vfs::setxattr(uint64_t ino, const std::string& name, const std::string& value,
+                 uint32_t flags) {
+    std::lock_guard<std::mutex> g(mu_);
+    Inode* in = get(ino);
+    if (!in) return St::not_found;
+    bool has = in->xattrs.count(name) != 0;
+    if ((flags & 1 /*XATTR_CREATE*/) && has) return St::exists;
+    if ((flags & 2 /*XATTR_REPLACE*/) && !has) return St::not_found;
+    in->xattrs[name] = value;
+    in->ctime = (int64_t)wall_ns();
+    if (journal_)
+        journal_->rename_rec("\x01xattr:" + name + std::string(1, '\0') + value,
+                             path_locked(ino));
+    return St::ok;
+}
+
+St Vfs::getxattr(uint64_t ino, const std::string& name, std::string* out) {
+    std::lock_guard<std::mutex> g(mu_);
+    Inode* in = get(ino);
+    if (!in) return St::not_found;
+    auto it = in->xattrs.find(name);
+    if (it == in->xattrs.end()) return St::not_found;
+    *out = it->second;
+    return St::ok;
+}
+
+St Vfs::listxattr(uint64_t ino, std::string* out) {
+    std::lock_guard<std::mutex> g(mu_);
+    Inode* in = get(ino);
+    if (!in) return St::not_found;
+    out->clear();
+    for (auto& kv : in->xattrs) {
+        out->append(kv.first);
+        out->push_back('\0');
+    }
+    return St::ok;
+}
+
+St Vfs::removexattr(uint64_t ino, const std::string& name) {
+    std::lock_guard<std::mutex> g(mu_);
+    Inode* in = get(ino);
+    if (!in) return St::not_found;
+    if (!in->xattrs.erase(name)) return St::not_found;
+    in->ctime = (int64_t)wall_ns();
+    if (journal_) journal_->rename_rec("\x01xattr-:" + name, path_locked(ino));
     return St::ok;
 }
 

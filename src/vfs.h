@@ -20,6 +20,9 @@
 
 namespace edgehip {
 
+// extended attributes live on the inode (journal-persisted via pseudo-records)
+using XattrMap = std::map<std::string, std::string>;
+
 struct Inode {
     uint64_t ino = 0;
     uint32_t mode = 0;      // type + perms
@@ -36,6 +39,7 @@ struct Inode {
 
     bool is_dir() const { return S_ISDIR(mode); }
     uint64_t size() const { return obj ? obj->size() : 0; }
+    XattrMap xattrs;
 };
 
 struct DirEntry {
@@ -69,6 +73,13 @@ class Vfs {
     St readlink(uint64_t ino, std::string* target);
     St link(uint64_t ino, uint64_t newparent, const std::string& newname, Inode** out);
     St readdir(uint64_t ino, std::vector<DirEntry>* out);
+
+    // xattrs (flags: XATTR_CREATE=1 / XATTR_REPLACE=2 semantics)
+    St setxattr(uint64_t ino, const std::string& name, const std::string& value,
+                uint32_t flags);
+    St getxattr(uint64_t ino, const std::string& name, std::string* out);
+    St listxattr(uint64_t ino, std::string* out);  // NUL-joined names
+    St removexattr(uint64_t ino, const std::string& name);
 
     // handles
     St open(uint64_t ino, uint64_t* fh);

@@ -82,3 +82,18 @@ def test_symlink_hardlink_survive_remount(tmp_mount, tmp_path):
         assert (tmp_mount / "s").read_text() == "x\n"
         assert (tmp_mount / "h").read_text() == "x\n"
         assert (tmp_mount / "f.txt").stat().st_nlink == 2
+
+
+def test_xattrs_survive_remount(tmp_mount, tmp_path):
+    jdir = tmp_path / "jrnl"
+    with Mount(tmp_mount, opts(jdir)):
+        p = tmp_mount / "tagged.txt"
+        p.write_text("data")
+        os.setxattr(p, "user.keep", b"v1")
+        os.setxattr(p, "user.gone", b"tmp")
+        os.removexattr(p, "user.gone")
+        os.setxattr(p, "user.keep", b"v2")  # overwrite: last write wins
+    with Mount(tmp_mount, opts(jdir)):
+        p = tmp_mount / "tagged.txt"
+        assert os.listxattr(p) == ["user.keep"]
+        assert os.getxattr(p, "user.keep") == b"v2"

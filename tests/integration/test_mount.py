@@ -153,3 +153,22 @@ def test_readdirplus_listing_with_attrs(tmp_mount):
     assert sorted(os.listdir(d)) == sorted(sizes)
     for name, sz in sizes.items():
         assert os.stat(os.path.join(d, name)).st_size == sz
+
+
+def test_xattr_roundtrip(tmp_mount):
+    p = os.path.join(tmp_mount, "xf")
+    with open(p, "wb") as f:
+        f.write(b"hi")
+    os.setxattr(p, "user.color", b"blue")
+    os.setxattr(p, "user.blob", bytes(range(256)))
+    assert os.getxattr(p, "user.color") == b"blue"
+    assert os.getxattr(p, "user.blob") == bytes(range(256))
+    assert sorted(os.listxattr(p)) == ["user.blob", "user.color"]
+    os.setxattr(p, "user.color", b"red", os.XATTR_REPLACE)
+    assert os.getxattr(p, "user.color") == b"red"
+    with pytest.raises(OSError):
+        os.setxattr(p, "user.color", b"x", os.XATTR_CREATE)  # EEXIST
+    os.removexattr(p, "user.color")
+    with pytest.raises(OSError):
+        os.getxattr(p, "user.color")  # ENODATA
+    assert os.listxattr(p) == ["user.blob"]
