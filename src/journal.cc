@@ -13,7 +13,43 @@ namespace edgehip {
 constexpr uint32_t kJMagic = 0x314A4845;  // 'EHJ1'
 
 Journal::~Journal() {
+    {
+        std::unique_lock<std::mutex> g(jmu_);
+        jstop_ = true;
+        jcv_.notify_all();
+    }
+    if (jworker_.joinable()) jworker_.join();
     if (fd_ >= 0) ::close(fd_);
+}
+
+void Journal::mirror_loop() {
+    for (;;) {
+        uint64_t id;
+        Fp fp;
+        std::vector<uint8_t> data;
+        {
+            std::unique_lock<std::mutex> g(jmu_);
+            jcv_.wait(g, [&] { return jstop_ || !jq_.empty(); });
+            if (jq_.empty()) {
+                if (jstop_) return;
+                continue;
+            }
+            // copy, write, THEN pop: the entry stays visible to get_chunk
+            // and sync() drain until its bytes are actually on disk
+            id = jq_.front().id;
+            fp = jq_.front().fp;
+            data = jq_.front().data;
+        }
+        (void)write_chunk_file(fp, data.data(), (uint32_t)data.size());
+        {
+            std::unique_lock<std::mutex> g(jmu_);
+            if (!jq_.empty() && jq_.front().id == id) {  // del may have raced
+                jq_bytes_ -= jq_.front().data.size();
+                jq_.pop_front();
+            }
+            jcv_drain_.notify_all();
+        }
+    }
 }
 
 St Journal::open(const std::string& dir) {
@@ -74,13 +110,17 @@ St Journal::file_rec(const std::string& p, const std::vector<uint8_t>& manifest)
 }
 
 St Journal::sync() {
+    {   // drain the chunk-mirror queue first: fsync means "it is on disk"
+        std::unique_lock<std::mutex> g(jmu_);
+        jcv_drain_.wait(g, [&] { return jq_.empty(); });
+    }
     std::lock_guard<std::mutex> g(mu_);
     return ::fdatasync(fd_) == 0 ? St::ok : St::io;
 }
 
 std::string Journal::chunk_path(const Fp& fp) const { return dir_ + "/chunks/" + fp.hex(); }
 
-St Journal::put_chunk(const Fp& fp, const uint8_t* data, uint32_t len) {
+St Journal::write_chunk_file(const Fp& fp, const uint8_t* data, uint32_t len) {
     std::string tmp = chunk_path(fp) + ".tmp";
     int fd = ::open(tmp.c_str(), O_CREAT | O_WRONLY | O_TRUNC | O_CLOEXEC, 0644);
     if (fd < 0) return St::io;
@@ -94,7 +134,30 @@ St Journal::put_chunk(const Fp& fp, const uint8_t* data, uint32_t len) {
     return St::ok;
 }
 
+St Journal::put_chunk(const Fp& fp, const uint8_t* data, uint32_t len) {
+    std::unique_lock<std::mutex> g(jmu_);
+    if (!jstarted_) {
+        jstarted_ = true;
+        jworker_ = std::thread([this] { mirror_loop(); });
+    }
+    // bound the write-behind queue at 256 MiB
+    jcv_drain_.wait(g, [&] { return jq_bytes_ < (256ull << 20); });
+    jq_.push_back({jq_next_id_++, fp, std::vector<uint8_t>(data, data + len)});
+    jq_bytes_ += len;
+    jcv_.notify_one();
+    return St::ok;
+}
+
 St Journal::get_chunk(const Fp& fp, std::vector<uint8_t>* out) const {
+    {   // still queued? serve from memory
+        auto* self = const_cast<Journal*>(this);
+        std::lock_guard<std::mutex> g(self->jmu_);
+        for (auto& q : self->jq_)
+            if (q.fp == fp) {
+                *out = q.data;
+                return St::ok;
+            }
+    }
     int fd = ::open(chunk_path(fp).c_str(), O_RDONLY | O_CLOEXEC);
     if (fd < 0) return St::not_found;
     struct stat st;
@@ -109,6 +172,18 @@ St Journal::get_chunk(const Fp& fp, std::vector<uint8_t>* out) const {
 }
 
 St Journal::del_chunk(const Fp& fp) {
+    {   // cancel any queued write of this chunk
+        std::lock_guard<std::mutex> g(jmu_);
+        for (auto it = jq_.begin(); it != jq_.end();) {
+            if (it->fp == fp) {
+                jq_bytes_ -= it->data.size();
+                it = jq_.erase(it);
+            } else {
+                ++it;
+            }
+        }
+        jcv_drain_.notify_all();
+    }
     ::unlink(chunk_path(fp).c_str());
     return St::ok;
 }

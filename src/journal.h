@@ -11,7 +11,10 @@
 //        {path}  4 RENAME {u16 old_len, old, new}  5 RMDIR {path}
 #pragma once
 
+#include <condition_variable>
+#include <deque>
 #include <mutex>
+#include <thread>
 #include <string>
 #include <vector>
 
@@ -50,10 +53,29 @@ class Journal {
   private:
     St append(uint8_t type, const std::vector<uint8_t>& body);
     std::string chunk_path(const Fp& fp) const;
+    St write_chunk_file(const Fp& fp, const uint8_t* data, uint32_t len);
+    void mirror_loop();
 
     std::string dir_;
     int fd_ = -1;
     std::mutex mu_;
+
+    // chunk-mirror write-behind: the ingest path enqueues; one worker writes.
+    // sync() drains the queue, so fsync durability is unchanged — only the
+    // put_chunks hot path stops paying per-chunk file I/O inline.
+    std::mutex jmu_;
+    std::condition_variable jcv_, jcv_drain_;
+    struct JItem {
+        uint64_t id;
+        Fp fp;
+        std::vector<uint8_t> data;
+    };
+    std::deque<JItem> jq_;
+    uint64_t jq_bytes_ = 0;
+    uint64_t jq_next_id_ = 1;
+    bool jstop_ = false;
+    bool jstarted_ = false;
+    std::thread jworker_;
 };
 
 }  // namespace edgehip
