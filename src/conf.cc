@@ -96,6 +96,7 @@ St Conf::apply_kv(const std::string& k, const std::string& v) {
         return St::ok;
     }
     if (k == "writeback") return parse_bool(v, &writeback) ? St::ok : bad();
+    if (k == "read_direct") return parse_bool(v, &read_direct) ? St::ok : bad();
     if (k == "direct_io") return parse_bool(v, &direct_io) ? St::ok : bad();
     if (k == "kernel_cache") return parse_bool(v, &kernel_cache) ? St::ok : bad();
     if (k == "log") {

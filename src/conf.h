@@ -45,6 +45,8 @@ struct Conf {
     uint32_t max_write = 1 * MiB;
     uint32_t max_readahead = 16 * MiB;  // deep readahead pipelines seq reads
     bool writeback = true;             // FUSE writeback cache (ack from page cache)
+    bool read_direct = true;           // engine reads DMA straight into the caller
+                                       // buffer (skip pinned-blob staging + memcpy)
     bool direct_io = false;            // force FOPEN_DIRECT_IO on opens
     bool kernel_cache = true;          // keep_cache on open
 

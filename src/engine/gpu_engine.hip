@@ -968,18 +968,22 @@ St GpuEngine::get_chunks(const GetChunk* get, int n) {
                                          hipMemcpyDeviceToHost, R.s));
             }
             Staged st;
-            st.blob = BufPool::inst().get((uint32_t)std::max<uint64_t>(outb, 64));
+            const bool direct = conf_.read_direct;
+            if (!direct)
+                st.blob = BufPool::inst().get((uint32_t)std::max<uint64_t>(outb, 64));
             uint64_t pout = 0;
             mi = 0;
             // merge arena/blob-contiguous requests into one DMA: per-copy
             // SDMA setup is ~40 us/MiB (profiles/engine_bench_r01) and the
             // bump allocator makes sequentially-written chunks adjacent
             const uint8_t* run_src = nullptr;
+            uint8_t* run_dst = nullptr;  // direct mode: user buffer
             uint64_t run_out = 0, run_len = 0;
             auto flush_run = [&]() {
                 if (!run_len) return;
-                HIP_CHECK(hipMemcpyAsync(st.blob.p + run_out, run_src, run_len,
-                                         hipMemcpyDeviceToHost, R.s));
+                HIP_CHECK(hipMemcpyAsync(direct ? run_dst : st.blob.p + run_out,
+                                         run_src, run_len, hipMemcpyDeviceToHost,
+                                         R.s));
                 run_len = 0;
             };
             for (size_t j = j0; j < j1; j++) {
@@ -991,18 +995,21 @@ St GpuEngine::get_chunks(const GetChunk* get, int n) {
                 } else {
                     srcp = D.arena + mine[j]->arena_off + g.off;
                 }
-                if (run_len && srcp == run_src + run_len &&
-                    pout == run_out + run_len) {
+                bool contiguous = run_len && srcp == run_src + run_len &&
+                                  (direct ? g.out == run_dst + run_len
+                                          : pout == run_out + run_len);
+                if (contiguous) {
                     run_len += g.len;
                 } else {
                     flush_run();
                     run_src = srcp;
+                    run_dst = g.out;
                     run_out = pout;
                     run_len = g.len;
                 }
-                st.copies.push_back({g.out, pout, g.len});
+                if (!direct) st.copies.push_back({g.out, pout, g.len});
                 pout += round_up(g.len, 64);
-                if (pout != run_out + run_len) flush_run();  // padding gap
+                if (!direct && pout != run_out + run_len) flush_run();  // padding gap
             }
             flush_run();
             HIP_CHECK(hipStreamSynchronize(R.s));
