@@ -968,7 +968,11 @@ St GpuEngine::get_chunks(const GetChunk* get, int n) {
                                          hipMemcpyDeviceToHost, R.s));
             }
             Staged st;
-            const bool direct = conf_.read_direct;
+            // direct DMA wins for bulk reads (+3 GiB/s: no staging memcpy)
+            // but per-copy setup kills small-read IOPS (531k -> 231k measured)
+            // — stage small-request batches through one blob instead
+            const bool direct =
+                conf_.read_direct && outb >= (uint64_t)(j1 - j0) * (256 * KiB);
             if (!direct)
                 st.blob = BufPool::inst().get((uint32_t)std::max<uint64_t>(outb, 64));
             uint64_t pout = 0;
