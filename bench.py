@@ -275,18 +275,19 @@ def main():
         phase_gibps = {"write": round(half / io.write_secs, 2),
                        "read": round(half / io.read_secs, 2)}
 
-    # aggregate: value = sum over ranks; ms_per_step = MAX over ranks
+    # aggregate: whole-job value = total bytes / slowest-rank time (the
+    # barriers bracket the region, so max dt IS the job wall time)
     if dist:
         import torch
-        agg = torch.tensor([local_gibps, dt], dtype=torch.float64)
+        agg = torch.tensor([moved / GIB, dt], dtype=torch.float64)
         if torch.cuda.is_available():
             agg = agg.cuda()
         sum_v = agg.clone()
         dist.all_reduce(sum_v, op=dist.ReduceOp.SUM)
         max_v = agg.clone()
         dist.all_reduce(max_v, op=dist.ReduceOp.MAX)
-        total_gibps = float(sum_v[0].item())
         max_dt = float(max_v[1].item())
+        total_gibps = float(sum_v[0].item()) / max_dt
     else:
         total_gibps = local_gibps
         max_dt = dt
