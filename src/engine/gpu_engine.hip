@@ -41,6 +41,7 @@ namespace edgehip {
 namespace {
 
 constexpr int kMaxBatch = 32;            // chunks per pipeline pass
+constexpr int kMaxGather = 1024;         // small-read gather descriptors per pass
 constexpr uint64_t kGranule = 64 * KiB;  // arena allocation granule
 constexpr uint32_t kSegSz = 4096;        // LZ4 segment (lz4.hip kSeg)
 constexpr uint32_t kSegStride = 4352;
@@ -228,7 +229,8 @@ class GpuEngine : public Engine {
                 auto& R = D.rslots[rs2];
                 HIP_CHECK(hipStreamCreateWithFlags(&R.s, hipStreamNonBlocking));
                 HIP_CHECK(hipMalloc(&R.d_rdstage, D.stage_sz));
-                HIP_CHECK(hipMalloc(&R.d_rchunks, sizeof(DevChunk) * kMaxBatch));
+                HIP_CHECK(hipMalloc(&R.d_rchunks,
+                                    sizeof(DevChunk) * (kMaxBatch + kMaxGather)));
                 HIP_CHECK(hipMalloc(&R.d_rstatus, sizeof(uint32_t) * kMaxBatch));
                 HIP_CHECK(hipHostMalloc(&R.pin_rmeta, 64 * KiB, hipHostMallocDefault));
             }
@@ -947,7 +949,9 @@ St GpuEngine::get_chunks(const GetChunk* get, int n) {
                 const DevReq& q = *mine[j1];
                 int nm = m + (q.compressed ? 1 : 0);
                 uint64_t no = outb + round_up(q.g->len, 64);
-                if (nm > kMaxBatch || no > D.stage_sz) break;
+                if (nm > kMaxBatch || (int)(j1 - j0) >= kMaxGather ||
+                    no + (uint64_t)nm * cs > D.stage_sz)
+                    break;
                 m = nm;
                 outb = no;
                 j1++;
@@ -975,6 +979,30 @@ St GpuEngine::get_chunks(const GetChunk* get, int n) {
                 conf_.read_direct && outb >= (uint64_t)(j1 - j0) * (256 * KiB);
             if (!direct)
                 st.blob = BufPool::inst().get((uint32_t)std::max<uint64_t>(outb, 64));
+            // staged mode: G9 gather packs the scattered arena reads into
+            // d_rdstage (after the decompress outputs) so the host pulls ONE
+            // contiguous D2H instead of a tiny copy per request — the per-copy
+            // SDMA setup was the 4-KiB-random-read bound
+            const uint64_t gather_base = (uint64_t)m * cs;
+            if (!direct) {
+                auto* h_gd = (DevChunk*)(R.pin_rmeta + 16 * KiB);
+                int ng = 0;
+                uint64_t goff = gather_base;
+                for (size_t j = j0; j < j1; j++) {
+                    if (mine[j]->compressed) continue;
+                    const GetChunk& g = *mine[j]->g;
+                    h_gd[ng++] = DevChunk{mine[j]->arena_off + g.off, goff, g.len, 0};
+                    goff += round_up(g.len, 64);
+                }
+                if (ng) {
+                    uint32_t maxl = 0;
+                    for (int k = 0; k < ng; k++) maxl = std::max(maxl, h_gd[k].len);
+                    DevChunk* d_gd = R.d_rchunks + kMaxBatch;
+                    HIP_CHECK(hipMemcpyAsync(d_gd, h_gd, sizeof(DevChunk) * ng,
+                                             hipMemcpyHostToDevice, R.s));
+                    launch_batch_copy(D.arena, R.d_rdstage, d_gd, ng, R.s, maxl);
+                }
+            }
             uint64_t pout = 0;
             mi = 0;
             // merge arena/blob-contiguous requests into one DMA: per-copy
@@ -990,12 +1018,16 @@ St GpuEngine::get_chunks(const GetChunk* get, int n) {
                                          R.s));
                 run_len = 0;
             };
+            uint64_t gpack = gather_base;
             for (size_t j = j0; j < j1; j++) {
                 const GetChunk& g = *mine[j]->g;
                 const uint8_t* srcp;
                 if (mine[j]->compressed) {
                     srcp = R.d_rdstage + (uint64_t)mi * cs + g.off;
                     mi++;
+                } else if (!direct) {
+                    srcp = R.d_rdstage + gpack;  // packed by the gather kernel
+                    gpack += round_up(g.len, 64);
                 } else {
                     srcp = D.arena + mine[j]->arena_off + g.off;
                 }

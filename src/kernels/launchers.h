@@ -23,7 +23,7 @@ void launch_crc32c_chunks(const uint8_t* src, const DevChunk* chunks, uint32_t* 
                           int nchunks, hipStream_t stream);
 
 void launch_batch_copy(const uint8_t* src, uint8_t* dst, const DevChunk* descs, int ndescs,
-                       hipStream_t stream);
+                       hipStream_t stream, uint32_t max_len = 1 << 20);
 
 void launch_cdc_candidates(const uint8_t* data, uint64_t len, uint64_t avg,
                            uint32_t* bitmap, hipStream_t stream, const uint64_t* gear_host);

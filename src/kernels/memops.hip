@@ -28,10 +28,14 @@ __global__ __launch_bounds__(256) void k_batch_copy(
 }
 
 void launch_batch_copy(const uint8_t* src, uint8_t* dst, const DevChunk* descs, int ndescs,
-                       hipStream_t stream) {
+                       hipStream_t stream, uint32_t max_len) {
     if (ndescs <= 0) return;
-    // x-dim sized for ~1 MiB descriptors; y-dim = descriptor index
-    hipLaunchKernelGGL(k_batch_copy, dim3(32, ndescs), dim3(256), 0, stream, src, dst,
+    // x-dim sized for the largest descriptor (4096 B per WG per pass);
+    // y-dim = descriptor index
+    uint32_t bx = (max_len + 256 * 16 - 1) / (256 * 16);
+    if (bx < 1) bx = 1;
+    if (bx > 32) bx = 32;
+    hipLaunchKernelGGL(k_batch_copy, dim3(bx, ndescs), dim3(256), 0, stream, src, dst,
                        descs, ndescs);
 }
 
