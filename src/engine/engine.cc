@@ -2,6 +2,7 @@
 #include "engine.h"
 
 #include <atomic>
+#include <thread>
 
 #include "../blake3.h"
 #include "../gf256.h"
@@ -22,13 +23,47 @@ class HostEngine : public Engine {
 
     St put_chunks(const PutChunk* put, ChunkRef* refs, int n) override {
         if (Faults::inst().fire("put_io")) return St::io;
+        // CPU-heavy per-chunk work (blake3 + lz4) fans out across threads;
+        // the store/dedup bookkeeping below stays sequential (it is cheap
+        // and batch-order dedup accounting is easier to reason about).
+        const bool do_fp = conf_.fp == FpAlgo::blake3;
+        const bool do_lz4 = conf_.compress == Compress::lz4;
+        struct Pre {
+            Fp fp;
+            std::vector<uint8_t> comp;
+            bool compressed = false;
+        };
+        std::vector<Pre> pre((size_t)n);
+        auto prep = [&](int i) {
+            const PutChunk& p = put[i];
+            if (do_fp) pre[i].fp = blake3_fp(p.data, p.len);
+            // probe-only dedup peek: on a hit the compression is wasted work,
+            // so skip it (racy reads are fine — worst case we compress)
+            bool likely_dup =
+                do_fp && conf_.dedup && store_.has(pre[i].fp);
+            pre[i].compressed = !likely_dup && do_lz4 && p.len >= 4096 &&
+                                lz4_chunk_compress(p.data, p.len, &pre[i].comp);
+        };
+        if (n > 1) {
+            int nt = std::min(8, n);
+            std::vector<std::thread> ts;
+            std::atomic<int> next{0};
+            for (int t = 0; t < nt; t++)
+                ts.emplace_back([&] {
+                    for (int i = next.fetch_add(1); i < n; i = next.fetch_add(1))
+                        prep(i);
+                });
+            for (auto& t : ts) t.join();
+        } else if (n == 1) {
+            prep(0);
+        }
         for (int i = 0; i < n; i++) {
             const PutChunk& p = put[i];
             ChunkRef& r = refs[i];
             r = ChunkRef{};
             r.logical_len = p.len;
-            if (conf_.fp == FpAlgo::blake3) {
-                r.fp = blake3_fp(p.data, p.len);
+            if (do_fp) {
+                r.fp = pre[i].fp;
             } else {
                 // fp=none: synthesized unique id (no content addressing, no dedup)
                 static std::atomic<uint64_t> seq{1};
@@ -47,9 +82,13 @@ class HostEngine : public Engine {
                     continue;
                 }
             }
-            std::vector<uint8_t> comp;
-            bool compressed = conf_.compress == Compress::lz4 && p.len >= 4096 &&
-                              lz4_chunk_compress(p.data, p.len, &comp);
+            std::vector<uint8_t> comp = std::move(pre[i].comp);
+            bool compressed = pre[i].compressed;
+            if (!compressed && comp.empty() && conf_.dedup && do_fp && do_lz4 &&
+                p.len >= 4096) {
+                // the dedup peek guessed "dup" but it is not stored: compress now
+                compressed = lz4_chunk_compress(p.data, p.len, &comp);
+            }
             metrics().lz4_in_bytes += p.len;
             St s;
             if (compressed) {
