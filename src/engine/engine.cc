@@ -199,6 +199,16 @@ class HostEngine : public Engine {
 
 // ---- Engine base: RS via the generic chunk APIs (host GF(256) math) --------
 // Works for any engine; GpuEngine overrides with the on-device kernel path.
+St Engine::rs_encode_stripes(const ChunkRef* data, int k, int m, uint32_t shard_len,
+                             int nstripes, ChunkRef* parity_out) {
+    for (int s = 0; s < nstripes; s++) {
+        St st = rs_encode_stripe(data + (size_t)s * k, k, m, shard_len,
+                                 parity_out + (size_t)s * m);
+        if (st != St::ok) return st;
+    }
+    return St::ok;
+}
+
 St Engine::rs_encode_stripe(const ChunkRef* data, int k, int m, uint32_t shard_len,
                             ChunkRef* parity_out) {
     std::vector<std::vector<uint8_t>> shards((size_t)k);

@@ -60,6 +60,12 @@ class Engine {
     virtual St rs_encode_stripe(const ChunkRef* data, int k, int m, uint32_t shard_len,
                                 ChunkRef* parity_out);
 
+    // Batched encode: nstripes stripes (data = nstripes*k refs, parity_out =
+    // nstripes*m). Engines override to pipeline; the base loops.
+    virtual St rs_encode_stripes(const ChunkRef* data, int k, int m,
+                                 uint32_t shard_len, int nstripes,
+                                 ChunkRef* parity_out);
+
     // Reconstruct stripe member `target` (0..k+m) from the present members.
     // refs[k+m] = k data + m parity; present[i] false = lost. Writes the
     // target's UNCOMPRESSED bytes (logical_len of the target, zero-padded

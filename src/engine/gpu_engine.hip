@@ -299,6 +299,8 @@ class GpuEngine : public Engine {
     St release_chunk(const ChunkRef& ref) override;
     St rs_encode_stripe(const ChunkRef* data, int k, int m, uint32_t shard_len,
                         ChunkRef* parity_out) override;
+    St rs_encode_stripes(const ChunkRef* data, int k, int m, uint32_t shard_len,
+                         int nstripes, ChunkRef* parity_out) override;
 
     St cdc_scan(const uint8_t* data, uint64_t len, uint64_t avg,
                 std::vector<uint32_t>* bitmap) override {
@@ -1153,6 +1155,67 @@ St GpuEngine::gather_chunk_to_dev(const ChunkRef& ref, int ddev, uint8_t* dst) {
 
 // On-device RS encode (C1/C2 + G6): gather the stripe's uncompressed shards
 // onto one device over xGMI, run the GF(256) kernel, ingest parity chunks.
+St GpuEngine::rs_encode_stripes(const ChunkRef* data, int k, int m,
+                                uint32_t shard_len, int nstripes,
+                                ChunkRef* parity_out) {
+    if (shard_len != max_chunk_ || k > kRsMaxK || m > kRsMaxM)
+        return Engine::rs_encode_stripes(data, k, m, shard_len, nstripes,
+                                         parity_out);
+    RoctxRange rr_rs("edgehip::rs_encode_stripes");
+    const uint32_t cs = (uint32_t)max_chunk_;
+    const int cdev = (int)(rr_.fetch_add(1) % devs_.size());
+    DevCtx& C = devs_[cdev];
+    auto M = rs_parity_matrix(k, m);
+    // pipeline in slabs: encode stripe i, queue its parity D2H on the same
+    // stream (ordering makes the d_rs reuse safe), sync ONCE per slab, then
+    // ingest all parity chunks in one pinned zero-copy batch — the
+    // per-stripe sync + 2-chunk put was an ~0.8 ms/stripe wall.
+    int slab = kMaxBatch / m;  // parity chunks per put batch
+    // pool blobs cap at 16 MiB — bound the slab's pinned staging to that
+    while (slab > 1 && (uint64_t)slab * m * cs > 16 * MiB) slab--;
+    for (int s0 = 0; s0 < nstripes; s0 += slab) {
+        int ns = std::min(slab, nstripes - s0);
+        Blob host = BufPool::inst().get((uint32_t)((uint64_t)ns * m * cs));
+        {
+            std::lock_guard<std::mutex> rg(C.rs_mu);
+            HIP_CHECK(hipSetDevice(C.hip_dev));
+            for (int s = 0; s < ns; s++) {
+                const ChunkRef* d0 = data + (size_t)(s0 + s) * k;
+                for (int j = 0; j < k; j++) {
+                    St st = gather_chunk_to_dev(d0[j], cdev, C.d_rs + (uint64_t)j * cs);
+                    if (st != St::ok) return st;
+                }
+                HIP_CHECK(hipStreamSynchronize(C.s_read));  // gathers may cross streams
+                RsDesc d{};
+                for (int j = 0; j < k; j++) d.in[j] = C.d_rs + (uint64_t)j * cs;
+                for (int i = 0; i < m; i++) d.out[i] = C.d_rs + (uint64_t)(k + i) * cs;
+                for (int i = 0; i < m; i++)
+                    for (int j = 0; j < k; j++) d.coef[i][j] = M[(size_t)i * k + j];
+                d.k = k;
+                d.m = m;
+                d.len = cs;
+                launch_gf_matmul(d, C.s_read);
+                for (int i = 0; i < m; i++)
+                    HIP_CHECK(hipMemcpyAsync(
+                        host.p + ((size_t)s * m + i) * cs,
+                        C.d_rs + (uint64_t)(k + i) * cs, cs,
+                        hipMemcpyDeviceToHost, C.s_read));
+            }
+            HIP_CHECK(hipStreamSynchronize(C.s_read));
+        }
+        std::vector<PutChunk> puts((size_t)ns * m);
+        for (int i = 0; i < ns * m; i++)
+            puts[i] = {host.p + (size_t)i * cs, cs, nullptr, /*pinned_src=*/true};
+        St st = put_chunks(puts.data(), parity_out + (size_t)s0 * m, ns * m);
+        if (st != St::ok) return st;
+        for (int i = 0; i < ns * m; i++)
+            parity_out[(size_t)s0 * m + i].flags |=
+                ChunkRef::kFlagParity | ChunkRef::kFlagRs;
+        metrics().rs_encoded_stripes += ns;
+    }
+    return St::ok;
+}
+
 St GpuEngine::rs_encode_stripe(const ChunkRef* data, int k, int m, uint32_t shard_len,
                                ChunkRef* parity_out) {
     if (shard_len != max_chunk_ || k > kRsMaxK || m > kRsMaxM)
