@@ -124,6 +124,7 @@ struct DevCtx {
     std::atomic<uint32_t> rrr{0};
 
     uint8_t* d_rs = nullptr;
+    uint8_t* d_rspar = nullptr;  // batched-parity staging (rs_encode_stripes)
     unsigned long long* d_index = nullptr;  // G7 fingerprint index (HBM)
     uint64_t index_mask = 0;
     uint64_t stage_sz = 0;
@@ -235,6 +236,7 @@ class GpuEngine : public Engine {
                 HIP_CHECK(hipHostMalloc(&R.pin_rmeta, 64 * KiB, hipHostMallocDefault));
             }
             HIP_CHECK(hipMalloc(&D.d_rs, (uint64_t)(kRsMaxK + kRsMaxM) * cs));
+            HIP_CHECK(hipMalloc(&D.d_rspar, (uint64_t)kMaxBatch * cs));
             // G7: per-device fingerprint index shard in HBM (tags of chunks
             // this device owns); 1/1024 of the arena, power-of-two slots
             {
@@ -281,7 +283,8 @@ class GpuEngine : public Engine {
                 (void)hipHostFree(R.pin_rmeta);
                 (void)hipStreamDestroy(R.s);
             }
-            for (void* p : {(void*)D.arena, (void*)D.d_rs, (void*)D.d_index})
+            for (void* p : {(void*)D.arena, (void*)D.d_rs, (void*)D.d_rspar,
+                            (void*)D.d_index})
                 (void)hipFree(p);
             (void)hipStreamDestroy(D.s_read);
         }
@@ -378,7 +381,8 @@ class GpuEngine : public Engine {
     int owner_of(const Fp& fp) const { return fp.b[4] % (int)devs_.size(); }
 
     St evict_for(int dev, uint32_t need_bytes);  // map_mu_ held
-    St flush_batch(const PutChunk* put, ChunkRef* refs, int n);
+    St flush_batch(const PutChunk* put, ChunkRef* refs, int n,
+                   bool device_src = false, int force_dev = -1);
     St gather_chunk_to_dev(const ChunkRef& ref, int ddev, uint8_t* dst);
 
     Conf conf_;
@@ -396,7 +400,8 @@ class GpuEngine : public Engine {
     HostStore spill_;
 };
 
-St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
+St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n,
+                          bool device_src, int force_dev) {
     RoctxRange rr_ingest("edgehip::ingest_batch");
     const uint32_t cs = (uint32_t)max_chunk_;
     // proactive eviction: keep ~8% arena headroom so writers rarely hit a
@@ -415,7 +420,8 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
     }
     const bool do_lz4 = conf_.compress == Compress::lz4;
     const bool do_fp = conf_.fp == FpAlgo::blake3;
-    const int ingest_dev = (int)(rr_.fetch_add(1) % devs_.size());
+    const int ingest_dev =
+        force_dev >= 0 ? force_dev : (int)(rr_.fetch_add(1) % devs_.size());
     DevCtx& D = devs_[ingest_dev];
     PipeSlot* SLP = D.acquire_slot();
     PipeSlot& S = *SLP;
@@ -442,7 +448,12 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
     uint64_t staged = 0;
     bool used_s2 = false;
     for (int i = 0; i < n; i++) {
-        if (pool_pinned &&
+        if (device_src) {
+            // source already in this device's memory (e.g. freshly encoded
+            // RS parity): D2D into staging, no PCIe roundtrip
+            HIP_CHECK(hipMemcpyAsync(S.d_stage + off, put[i].data, put[i].len,
+                                     hipMemcpyDeviceToDevice, S.s));
+        } else if (pool_pinned &&
             ((put[i].donate && put[i].donate->p == put[i].data) || put[i].pinned_src)) {
             // dirty buffer is already pinned: DMA straight from it, copies
             // alternating across both slot streams (2 SDMA queues)
@@ -610,7 +621,7 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n) {
                 metrics().chunks_put++;
             }
         }
-        if (!dup && conf_.journal_ptr)
+        if (!dup && conf_.journal_ptr && !device_src)
             conf_.journal_ptr->put_chunk(r.fp, put[i].data, put[i].len);
         // donated (possibly pinned) buffers are released AFTER the final
         // stream sync at the end of this batch — see below
@@ -1170,43 +1181,36 @@ St GpuEngine::rs_encode_stripes(const ChunkRef* data, int k, int m,
     // stream (ordering makes the d_rs reuse safe), sync ONCE per slab, then
     // ingest all parity chunks in one pinned zero-copy batch — the
     // per-stripe sync + 2-chunk put was an ~0.8 ms/stripe wall.
-    int slab = kMaxBatch / m;  // parity chunks per put batch
-    // pool blobs cap at 16 MiB — bound the slab's pinned staging to that
-    while (slab > 1 && (uint64_t)slab * m * cs > 16 * MiB) slab--;
+    const int slab = kMaxBatch / m;  // parity chunks per put batch
     for (int s0 = 0; s0 < nstripes; s0 += slab) {
         int ns = std::min(slab, nstripes - s0);
-        Blob host = BufPool::inst().get((uint32_t)((uint64_t)ns * m * cs));
-        {
-            std::lock_guard<std::mutex> rg(C.rs_mu);
-            HIP_CHECK(hipSetDevice(C.hip_dev));
-            for (int s = 0; s < ns; s++) {
-                const ChunkRef* d0 = data + (size_t)(s0 + s) * k;
-                for (int j = 0; j < k; j++) {
-                    St st = gather_chunk_to_dev(d0[j], cdev, C.d_rs + (uint64_t)j * cs);
-                    if (st != St::ok) return st;
-                }
-                HIP_CHECK(hipStreamSynchronize(C.s_read));  // gathers may cross streams
-                RsDesc d{};
-                for (int j = 0; j < k; j++) d.in[j] = C.d_rs + (uint64_t)j * cs;
-                for (int i = 0; i < m; i++) d.out[i] = C.d_rs + (uint64_t)(k + i) * cs;
-                for (int i = 0; i < m; i++)
-                    for (int j = 0; j < k; j++) d.coef[i][j] = M[(size_t)i * k + j];
-                d.k = k;
-                d.m = m;
-                d.len = cs;
-                launch_gf_matmul(d, C.s_read);
-                for (int i = 0; i < m; i++)
-                    HIP_CHECK(hipMemcpyAsync(
-                        host.p + ((size_t)s * m + i) * cs,
-                        C.d_rs + (uint64_t)(k + i) * cs, cs,
-                        hipMemcpyDeviceToHost, C.s_read));
+        std::lock_guard<std::mutex> rg(C.rs_mu);  // held across the put: it
+        HIP_CHECK(hipSetDevice(C.hip_dev));       // reads d_rspar via D2D
+        for (int s = 0; s < ns; s++) {
+            const ChunkRef* d0 = data + (size_t)(s0 + s) * k;
+            for (int j = 0; j < k; j++) {
+                St st = gather_chunk_to_dev(d0[j], cdev, C.d_rs + (uint64_t)j * cs);
+                if (st != St::ok) return st;
             }
-            HIP_CHECK(hipStreamSynchronize(C.s_read));
+            HIP_CHECK(hipStreamSynchronize(C.s_read));  // gathers may cross streams
+            RsDesc d{};
+            for (int j = 0; j < k; j++) d.in[j] = C.d_rs + (uint64_t)j * cs;
+            for (int i = 0; i < m; i++)
+                d.out[i] = C.d_rspar + ((uint64_t)s * m + i) * cs;
+            for (int i = 0; i < m; i++)
+                for (int j = 0; j < k; j++) d.coef[i][j] = M[(size_t)i * k + j];
+            d.k = k;
+            d.m = m;
+            d.len = cs;
+            launch_gf_matmul(d, C.s_read);
         }
+        HIP_CHECK(hipStreamSynchronize(C.s_read));
+        // parity never leaves the GPU: device-source put (D2D staging)
         std::vector<PutChunk> puts((size_t)ns * m);
         for (int i = 0; i < ns * m; i++)
-            puts[i] = {host.p + (size_t)i * cs, cs, nullptr, /*pinned_src=*/true};
-        St st = put_chunks(puts.data(), parity_out + (size_t)s0 * m, ns * m);
+            puts[i] = {C.d_rspar + (uint64_t)i * cs, cs, nullptr};
+        St st = flush_batch(puts.data(), parity_out + (size_t)s0 * m, ns * m,
+                            /*device_src=*/true, /*force_dev=*/cdev);
         if (st != St::ok) return st;
         for (int i = 0; i < ns * m; i++)
             parity_out[(size_t)s0 * m + i].flags |=
