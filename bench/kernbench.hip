@@ -75,8 +75,10 @@ int main(int argc, char** argv) {
         return best;
     };
 
+    uint32_t* d_fpg = nullptr;
+    (void)hipMalloc(&d_fpg, 32ull * 4 * nchunks);
     bench("blake3_fingerprint", (double)total,
-          [&] { launch_blake3_chunks(d_src, d_chunks, d_fps, nchunks, nullptr); });
+          [&] { launch_blake3_chunks(d_src, d_chunks, d_fps, nchunks, nullptr, d_fpg); });
 
     bench("crc32c_pages", (double)total,
           [&] { launch_crc32c_chunks(d_src, d_chunks, d_sizes, nchunks, nullptr); });

@@ -35,6 +35,8 @@ int eh_gpu_blake3(const uint8_t* data, uint64_t chunk_len, int nchunks, uint8_t*
         HIP_CHECK(d_src.alloc(total ? total : 1));
         HIP_CHECK(d_chunks.alloc(sizeof(DevChunk) * nchunks));
         HIP_CHECK(d_out.alloc(32 * nchunks));
+        DevBuf d_scr;
+        HIP_CHECK(d_scr.alloc(32ull * 4 * nchunks));
         HIP_CHECK(hipMemcpy(d_src.p, data, total, hipMemcpyHostToDevice));
         std::vector<DevChunk> descs(nchunks);
         for (int i = 0; i < nchunks; i++)
@@ -42,7 +44,7 @@ int eh_gpu_blake3(const uint8_t* data, uint64_t chunk_len, int nchunks, uint8_t*
         HIP_CHECK(hipMemcpy(d_chunks.p, descs.data(), sizeof(DevChunk) * nchunks,
                             hipMemcpyHostToDevice));
         launch_blake3_chunks(d_src.as<uint8_t>(), d_chunks.as<DevChunk>(), d_out.as<uint8_t>(),
-                             nchunks, nullptr);
+                             nchunks, nullptr, d_scr.as<uint32_t>());
         HIP_CHECK(hipDeviceSynchronize());
         HIP_CHECK(hipMemcpy(out_fps, d_out.p, 32 * nchunks, hipMemcpyDeviceToHost));
         return 0;
@@ -62,8 +64,10 @@ int eh_gpu_blake3_one(const uint8_t* data, uint64_t len, uint8_t* out_fp) {
         if (len) HIP_CHECK(hipMemcpy(d_src.p, data, len, hipMemcpyHostToDevice));
         DevChunk desc{0, 0, (uint32_t)len, 0};
         HIP_CHECK(hipMemcpy(d_chunks.p, &desc, sizeof(desc), hipMemcpyHostToDevice));
+        DevBuf d_scr;
+        HIP_CHECK(d_scr.alloc(32 * 4));
         launch_blake3_chunks(d_src.as<uint8_t>(), d_chunks.as<DevChunk>(), d_out.as<uint8_t>(),
-                             1, nullptr);
+                             1, nullptr, d_scr.as<uint32_t>());
         HIP_CHECK(hipDeviceSynchronize());
         HIP_CHECK(hipMemcpy(out_fp, d_out.p, 32, hipMemcpyDeviceToHost));
         return 0;

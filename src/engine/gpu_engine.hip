@@ -84,6 +84,7 @@ struct PipeSlot {
     hipEvent_t ev2{};    // observed) halves when copies split across 2 SDMA queues
     uint8_t *d_stage = nullptr, *d_comp = nullptr, *d_segscratch = nullptr,
             *d_fps = nullptr;
+    uint32_t* d_fpg = nullptr;  // blake3 subtree CVs (4 per chunk)
     DevChunk *d_chunks = nullptr, *d_copy = nullptr;
     ulong2* d_segmeta = nullptr;
     uint32_t *d_segsizes = nullptr, *d_sizes = nullptr, *d_crcs = nullptr;
@@ -221,6 +222,7 @@ class GpuEngine : public Engine {
                 HIP_CHECK(hipMalloc(&S.d_segsizes, sizeof(uint32_t) * kMaxSegs));
                 HIP_CHECK(hipMalloc(&S.d_sizes, sizeof(uint32_t) * kMaxBatch));
                 HIP_CHECK(hipMalloc(&S.d_fps, 32 * kMaxBatch));
+                HIP_CHECK(hipMalloc(&S.d_fpg, 32 * kMaxBatch * 4));
                 HIP_CHECK(hipMalloc(&S.d_crcs, sizeof(uint32_t) * kMaxBatch * 2));
                 HIP_CHECK(hipMalloc(&S.d_copy, sizeof(DevChunk) * kMaxBatch * 4));
                 HIP_CHECK(hipHostMalloc(&S.pin_in, D.stage_sz, hipHostMallocDefault));
@@ -266,8 +268,8 @@ class GpuEngine : public Engine {
                 PipeSlot& S = D.slots[sl];
                 for (void* p : {(void*)S.d_stage, (void*)S.d_comp, (void*)S.d_segscratch,
                                 (void*)S.d_chunks, (void*)S.d_segmeta, (void*)S.d_segsizes,
-                                (void*)S.d_sizes, (void*)S.d_fps, (void*)S.d_crcs,
-                                (void*)S.d_copy})
+                                (void*)S.d_sizes, (void*)S.d_fps, (void*)S.d_fpg,
+                                (void*)S.d_crcs, (void*)S.d_copy})
                     (void)hipFree(p);
                 (void)hipHostFree(S.pin_in);
                 (void)hipHostFree(S.pin_meta);
@@ -486,7 +488,7 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n,
                              hipMemcpyHostToDevice, S.s));
     metrics().h2d_bytes += off;
 
-    if (do_fp) launch_blake3_chunks(S.d_stage, S.d_chunks, S.d_fps, n, S.s);
+    if (do_fp) launch_blake3_chunks(S.d_stage, S.d_chunks, S.d_fps, n, S.s, S.d_fpg);
     if (do_lz4) {
         HIP_CHECK(hipMemcpyAsync(S.d_segmeta, h_segmeta, sizeof(ulong2) * nsegs,
                                  hipMemcpyHostToDevice, S.s));

@@ -1,52 +1,72 @@
 // G3: batched BLAKE3-256 chunk fingerprints on gfx950.
-// Grid = one 256-thread workgroup per chunk; each lane hashes leaves
-// (1 KiB) with stride 256, CVs tree-reduce through double-buffered LDS.
-// Max chunk 1 MiB (1024 leaves; the engine enforces this). Integer-ALU
-// bound: ~12 ops/byte puts the ceiling near the HBM read rate.
+// Grid = (groups_per_chunk, nchunks): each 256-thread workgroup hashes a
+// 256-leaf (256 KiB) aligned subtree — 4 WGs per 1 MiB chunk, so a 32-chunk
+// pipeline batch launches 128 WGs instead of 32 (the one-WG-per-chunk shape
+// filled 1/8 of the 256 CUs and ran at ~254 GB/s in-pipeline vs ~2.5 TB/s
+// at full occupancy — profiles/engine_bench_r01_pipeline.md). A trailing
+// finisher kernel combines the ≤4 subtree CVs (parent chain, ROOT on top).
+// Subtree split correctness: BLAKE3's tree splits at the largest power of
+// two ≤ n leaves, so 256-aligned groups are exact subtrees and the group
+// CVs combine with the same pairwise-with-promotion rule (verified against
+// the official vectors in tests/kernels/test_kernels.py).
+// Max chunk 1 MiB (1024 leaves; the engine enforces this).
 #include "blake3_dev.h"
 
 namespace edgehip {
 
-constexpr int kMaxLeavesLds = 1024;
+constexpr int kLeavesPerGroup = 256;
+constexpr int kGroupsPerChunk = 4;
 
 __global__ __launch_bounds__(256) void k_blake3_chunks(
     const uint8_t* __restrict__ src, const DevChunk* __restrict__ chunks,
-    uint8_t* __restrict__ out_fps /* 32 B per chunk */) {
-    __shared__ uint32_t cvA[kMaxLeavesLds][8];
-    __shared__ uint32_t cvB[kMaxLeavesLds / 2][8];
+    uint8_t* __restrict__ out_fps /* 32 B per chunk */,
+    uint32_t* __restrict__ group_cvs /* 8 u32 per (chunk, group) */) {
+    __shared__ uint32_t cvA[kLeavesPerGroup][8];
+    __shared__ uint32_t cvB[kLeavesPerGroup / 2][8];
 
-    const DevChunk c = chunks[blockIdx.x];
+    const int chunk = blockIdx.y;
+    const int group = blockIdx.x;
+    const DevChunk c = chunks[chunk];
     const uint8_t* base = src + c.src_off;
     const uint32_t len = c.len;
     const int tid = threadIdx.x;
 
     uint32_t nleaves = len == 0 ? 1 : (len + 1023) / 1024;
-    if (nleaves > kMaxLeavesLds) {  // engine caps chunk_size at 1 MiB; guard
-        if (tid == 0)
-            for (int i = 0; i < 32; i++) out_fps[blockIdx.x * 32 + i] = 0xEE;
+    uint32_t ngroups = (nleaves + kLeavesPerGroup - 1) / kLeavesPerGroup;
+    if (ngroups > kGroupsPerChunk) {  // engine caps chunk_size at 1 MiB; guard
+        if (group == 0 && tid == 0)
+            for (int i = 0; i < 32; i++) out_fps[chunk * 32 + i] = 0xEE;
         return;
     }
+    if ((uint32_t)group >= ngroups) return;
+
+    uint32_t lo = (uint32_t)group * kLeavesPerGroup;
+    uint32_t hi = min(nleaves, lo + kLeavesPerGroup);
+    uint32_t gl = hi - lo;  // leaves in this group
 
     const bool single = nleaves == 1;
-    for (uint32_t leaf = tid; leaf < nleaves; leaf += blockDim.x) {
+    for (uint32_t leaf = lo + tid; leaf < hi; leaf += blockDim.x) {
         uint32_t cv[8];
         uint32_t loff = leaf * 1024;
         uint32_t llen = len == 0 ? 0 : min(1024u, len - loff);
         b3_leaf_cv(base + loff, llen, leaf, single, cv);
 #pragma unroll
-        for (int i = 0; i < 8; i++) cvA[leaf][i] = cv[i];
+        for (int i = 0; i < 8; i++) cvA[leaf - lo][i] = cv[i];
     }
 
-    // Pairwise reduce with promotion of a trailing odd node; the final parent
-    // carries ROOT. (Equal to BLAKE3's split-at-largest-power-of-two tree.)
+    // Pairwise reduce with promotion of a trailing odd node. When this group
+    // is the WHOLE chunk (ngroups == 1) the top parent carries ROOT and the
+    // result goes straight to out_fps; otherwise the group CV is an internal
+    // subtree CV (never ROOT) for the finisher kernel.
+    const bool whole = ngroups == 1;
     uint32_t (*from)[8] = cvA;
     uint32_t (*to)[8] = cvB;
-    uint32_t n = nleaves;
+    uint32_t n = gl;
     while (n > 1) {
         __syncthreads();
         uint32_t pairs = n / 2;
         bool promote = n & 1;
-        bool root_level = (pairs == 1) && !promote;
+        bool root_level = whole && (pairs == 1) && !promote;
         for (uint32_t i = tid; i < pairs; i += blockDim.x)
             b3_parent_cv(from[2 * i], from[2 * i + 1], root_level, to[i]);
         if (promote && tid == 0) {
@@ -60,21 +80,66 @@ __global__ __launch_bounds__(256) void k_blake3_chunks(
     }
     __syncthreads();
 
-    if (tid < 8) {
-        uint32_t w = from[0][tid];
-        uint8_t* o = out_fps + blockIdx.x * 32 + tid * 4;
-        o[0] = w & 0xff;
-        o[1] = (w >> 8) & 0xff;
-        o[2] = (w >> 16) & 0xff;
-        o[3] = (w >> 24) & 0xff;
+    if (whole) {
+        if (tid < 8) {
+            uint32_t w = from[0][tid];
+            uint8_t* o = out_fps + chunk * 32 + tid * 4;
+            o[0] = w & 0xff;
+            o[1] = (w >> 8) & 0xff;
+            o[2] = (w >> 16) & 0xff;
+            o[3] = (w >> 24) & 0xff;
+        }
+    } else if (tid < 8) {
+        group_cvs[(chunk * kGroupsPerChunk + group) * 8 + tid] = from[0][tid];
+    }
+}
+
+// Combine ≤4 subtree CVs per chunk: pairwise with promotion, ROOT on top.
+__global__ __launch_bounds__(64) void k_blake3_finish(
+    const DevChunk* __restrict__ chunks, const uint32_t* __restrict__ group_cvs,
+    uint8_t* __restrict__ out_fps, int nchunks) {
+    int chunk = blockIdx.x * blockDim.x + threadIdx.x;
+    if (chunk >= nchunks) return;
+    const DevChunk c = chunks[chunk];
+    uint32_t nleaves = c.len == 0 ? 1 : (c.len + 1023) / 1024;
+    uint32_t ngroups = (nleaves + kLeavesPerGroup - 1) / kLeavesPerGroup;
+    if (ngroups <= 1 || ngroups > kGroupsPerChunk) return;  // done in pass 1
+
+    uint32_t cv[kGroupsPerChunk][8];
+    for (uint32_t g = 0; g < ngroups; g++)
+        for (int i = 0; i < 8; i++)
+            cv[g][i] = group_cvs[(chunk * kGroupsPerChunk + g) * 8 + i];
+    uint32_t n = ngroups;
+    while (n > 1) {
+        uint32_t pairs = n / 2;
+        bool promote = n & 1;
+        bool root_level = (pairs == 1) && !promote;
+        for (uint32_t i = 0; i < pairs; i++) {
+            uint32_t out[8];
+            b3_parent_cv(cv[2 * i], cv[2 * i + 1], root_level, out);
+            for (int k = 0; k < 8; k++) cv[i][k] = out[k];
+        }
+        if (promote)
+            for (int k = 0; k < 8; k++) cv[pairs][k] = cv[n - 1][k];
+        n = pairs + (promote ? 1 : 0);
+    }
+    uint8_t* o = out_fps + chunk * 32;
+    for (int i = 0; i < 8; i++) {
+        uint32_t w = cv[0][i];
+        o[4 * i] = w & 0xff;
+        o[4 * i + 1] = (w >> 8) & 0xff;
+        o[4 * i + 2] = (w >> 16) & 0xff;
+        o[4 * i + 3] = (w >> 24) & 0xff;
     }
 }
 
 void launch_blake3_chunks(const uint8_t* src, const DevChunk* chunks, uint8_t* out_fps,
-                          int nchunks, hipStream_t stream) {
+                          int nchunks, hipStream_t stream, uint32_t* group_scratch) {
     if (nchunks <= 0) return;
-    hipLaunchKernelGGL(k_blake3_chunks, dim3(nchunks), dim3(256), 0, stream, src, chunks,
-                       out_fps);
+    hipLaunchKernelGGL(k_blake3_chunks, dim3(kGroupsPerChunk, nchunks), dim3(256), 0,
+                       stream, src, chunks, out_fps, group_scratch);
+    hipLaunchKernelGGL(k_blake3_finish, dim3((nchunks + 63) / 64), dim3(64), 0, stream,
+                       chunks, group_scratch, out_fps, nchunks);
 }
 
 }  // namespace edgehip

@@ -8,7 +8,7 @@
 namespace edgehip {
 
 void launch_blake3_chunks(const uint8_t* src, const DevChunk* chunks, uint8_t* out_fps,
-                          int nchunks, hipStream_t stream);
+                          int nchunks, hipStream_t stream, uint32_t* group_scratch);
 
 void launch_lz4_compress_segs(const uint8_t* src, const ulong2* seg_meta, uint8_t* scratch,
                               uint32_t* sizes, int nsegs, hipStream_t stream);
