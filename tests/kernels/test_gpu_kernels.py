@@ -47,6 +47,8 @@ def mk_compressible(n, seed=1):
 
 # ---- BLAKE3 -----------------------------------------------------------------
 @pytest.mark.parametrize("n", [0, 1, 63, 64, 65, 1023, 1024, 1025, 4096, 65536,
+                               (1 << 18) - 1, 1 << 18, (1 << 18) + 1,  # 1/2-group edge
+                               (1 << 19) + 12345, 700000, (3 << 18) + 1,  # partial groups
                                (1 << 20) - 1, 1 << 20])
 def test_blake3_one_vs_host(lib, n):
     data = rand_bytes(n, seed=n)
