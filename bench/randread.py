@@ -20,7 +20,7 @@ def main():
     ap.add_argument("--size-gib", type=int, default=1)
     ap.add_argument("--seconds", type=float, default=10.0)
     ap.add_argument("--jobs", type=int, default=8)
-    ap.add_argument("--qd", type=int, default=16, help="reads per engine batch per job")
+    ap.add_argument("--qd", type=int, default=32, help="reads per engine batch per job")
     args = ap.parse_args()
 
     import threading
