@@ -184,6 +184,8 @@ class HostEngine : public Engine {
         return s;
     }
 
+    St addref_chunk(const ChunkRef& ref) override { return store_.addref(ref.fp); }
+
     St debug_drop_chunk(const Fp& fp) override { return store_.remove(fp); }
 
     uint64_t stored_bytes() const override { return store_.bytes(); }

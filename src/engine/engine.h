@@ -51,6 +51,8 @@ class Engine {
 
     // Drop one reference (unlink / overwrite).
     virtual St release_chunk(const ChunkRef& ref) = 0;
+    // Take an extra reference on a stored chunk (server-side copy / reflink).
+    virtual St addref_chunk(const ChunkRef& ref) = 0;
 
     // ---- Reed-Solomon stripes (judged config 4) ----
     // Encode m parity chunks over k stored data chunks (RS runs over the

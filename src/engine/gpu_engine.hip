@@ -302,6 +302,7 @@ class GpuEngine : public Engine {
 
     St get_chunks(const GetChunk* get, int n) override;
     St release_chunk(const ChunkRef& ref) override;
+    St addref_chunk(const ChunkRef& ref) override;
     St rs_encode_stripe(const ChunkRef* data, int k, int m, uint32_t shard_len,
                         ChunkRef* parity_out) override;
     St rs_encode_stripes(const ChunkRef* data, int k, int m, uint32_t shard_len,
@@ -1264,6 +1265,14 @@ St GpuEngine::rs_encode_stripe(const ChunkRef* data, int k, int m, uint32_t shar
     for (int i = 0; i < m; i++)
         parity_out[i].flags |= ChunkRef::kFlagParity | ChunkRef::kFlagRs;
     metrics().rs_encoded_stripes++;
+    return St::ok;
+}
+
+St GpuEngine::addref_chunk(const ChunkRef& ref) {
+    std::unique_lock<std::mutex> g(map_mu_);
+    auto it = map_.find(ref.fp);
+    if (it == map_.end()) return St::not_found;
+    it->second.refs++;
     return St::ok;
 }
 

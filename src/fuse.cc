@@ -664,6 +664,20 @@ void FuseMount::handle(const uint8_t* req, size_t len, std::vector<uint8_t>* rep
             reply_st(s);
             return;
         }
+        case FUSE_COPY_FILE_RANGE: {
+            // server-side copy: content-addressed chunks make aligned copies
+            // instant (the object layer shares chunk refs — no data moves)
+            auto* x = (const fuse_copy_file_range_in*)arg;
+            if (x->flags) { reply_err(-EINVAL); return; }
+            uint64_t copied = 0;
+            St s = vfs_->copy_range(x->fh_in, x->off_in, x->fh_out, x->off_out,
+                                    x->len, &copied);
+            if (s != St::ok) { reply_st(s); return; }
+            fuse_write_out wo{};
+            wo.size = (uint32_t)std::min<uint64_t>(copied, UINT32_MAX);
+            reply_body(&wo, sizeof(wo));
+            return;
+        }
         case FUSE_GETLK:
         case FUSE_SETLK:
         case FUSE_SETLKW:

@@ -128,6 +128,76 @@ void Object::mark_stripe_dirty(uint64_t chunk_idx) {
 // Refresh parity for dirty stripes whose data chunks are all committed.
 // Stripe s covers chunks [s*k, s*k+k); chunks beyond the object are holes
 // (zero shards). Parity refs live at mf_.parity[s*m + i].
+St Object::copy_range(Object* src, uint64_t soff, uint64_t doff, uint64_t len,
+                      uint64_t* copied) {
+    *copied = 0;
+    const uint64_t cs = conf_->chunk_size;
+    const bool can_share = !conf_->cdc && !src->conf_->cdc && eng_ == src->eng_;
+    // clamp at source EOF (copy_file_range semantics)
+    {
+        std::unique_lock<std::mutex> sg(src->mu_);
+        uint64_t ssz = src->size_;
+        if (soff >= ssz) return St::ok;
+        len = std::min(len, ssz - soff);
+    }
+    while (*copied < len) {
+        uint64_t s = soff + *copied, d = doff + *copied;
+        uint64_t rem = len - *copied;
+        bool shared = false;
+        if (can_share && s % cs == 0 && d % cs == 0 && rem >= cs) {
+            // lock both objects in address order (self-copy: one lock)
+            std::unique_lock<std::mutex> a, b;
+            if (src == this) {
+                a = std::unique_lock<std::mutex>(mu_);
+                wait_inflight_locked(a);
+            } else if (src < this) {
+                a = std::unique_lock<std::mutex>(src->mu_);
+                b = std::unique_lock<std::mutex>(mu_);
+                src->wait_inflight_locked(a);
+                wait_inflight_locked(b);
+            } else {
+                b = std::unique_lock<std::mutex>(mu_);
+                a = std::unique_lock<std::mutex>(src->mu_);
+                wait_inflight_locked(b);
+                src->wait_inflight_locked(a);
+            }
+            uint64_t sidx = s / cs, didx = d / cs;
+            bool src_clean = !src->dirty_.count(sidx) && !src->tail_.size() &&
+                             sidx < src->mf_.chunks.size() &&
+                             !src->mf_.chunks[sidx].fp.is_zero() &&
+                             src->mf_.chunks[sidx].logical_len == cs;
+            if (src_clean && !dirty_.count(didx) && sidx != (src == this ? didx : ~0ull)) {
+                ChunkRef r = src->mf_.chunks[sidx];
+                if (eng_->addref_chunk(r) == St::ok) {
+                    if (mf_.chunks.size() <= didx) mf_.chunks.resize(didx + 1);
+                    drop_ref(mf_.chunks[didx]);
+                    r.flags &= ~ChunkRef::kFlagRs;  // dst stripe re-encodes
+                    mf_.chunks[didx] = r;
+                    mark_stripe_dirty(didx);
+                    pending_publish_.push_back(r);
+                    size_ = std::max(size_, d + cs);
+                    shared = true;
+                    *copied += cs;
+                }
+            }
+        }
+        if (!shared) {
+            // byte fallback up to the next chunk boundary on either side
+            uint64_t upto = std::min({rem, cs - (d % cs), cs - (s % cs)});
+            std::vector<uint8_t> buf(upto);
+            uint64_t got = 0;
+            St st = src->read(s, upto, buf.data(), &got);
+            if (st != St::ok) return st;
+            if (got == 0) break;
+            st = write(d, buf.data(), got);
+            if (st != St::ok) return st;
+            *copied += got;
+            if (got < upto) break;
+        }
+    }
+    return St::ok;
+}
+
 St Object::rs_update_locked() {
     const int k = conf_->rs_k, m = conf_->rs_m;
     if (k <= 0 || m <= 0) return St::ok;

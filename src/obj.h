@@ -28,6 +28,12 @@ class Object {
     St read(uint64_t off, uint64_t len, uint8_t* out, uint64_t* got);
     // Batched scatter read: n ranges concatenated into out — ONE engine batch
     // (the 4 KiB random-read path: fio iodepth analog).
+    // Server-side copy (FUSE_COPY_FILE_RANGE): chunk-aligned spans share the
+    // source's chunk refs (content-addressed — no data moves); edges fall
+    // back to read+write. Clamps at the source EOF; returns bytes copied.
+    St copy_range(Object* src, uint64_t soff, uint64_t doff, uint64_t len,
+                  uint64_t* copied);
+
     St preadv(const uint64_t* offs, const uint32_t* lens, int n, uint8_t* out,
               uint64_t* got);
     St write(uint64_t off, const uint8_t* data, uint64_t len);

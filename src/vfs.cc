@@ -539,6 +539,25 @@ St Vfs::read(uint64_t fh, uint64_t off, uint64_t len, uint8_t* out, uint64_t* go
     return obj->read(off, len, out, got);
 }
 
+St Vfs::copy_range(uint64_t fh_in, uint64_t off_in, uint64_t fh_out,
+                   uint64_t off_out, uint64_t len, uint64_t* copied) {
+    std::shared_ptr<Object> src, dst;
+    Inode* din;
+    {
+        std::lock_guard<std::mutex> g(mu_);
+        auto i1 = handles_.find(fh_in);
+        auto i2 = handles_.find(fh_out);
+        if (i1 == handles_.end() || i2 == handles_.end()) return St::inval;
+        Inode* sin = get(i1->second);
+        din = get(i2->second);
+        if (!sin || !sin->obj || !din || !din->obj) return St::io;
+        src = sin->obj;
+        dst = din->obj;
+        din->mtime = din->ctime = (int64_t)wall_ns();
+    }
+    return dst->copy_range(src.get(), off_in, off_out, len, copied);
+}
+
 St Vfs::write(uint64_t fh, uint64_t off, const uint8_t* data, uint64_t len) {
     std::shared_ptr<Object> obj;
     Inode* in;

@@ -87,6 +87,8 @@ class Vfs {
     Inode* handle_inode(uint64_t fh);
 
     St read(uint64_t fh, uint64_t off, uint64_t len, uint8_t* out, uint64_t* got);
+    St copy_range(uint64_t fh_in, uint64_t off_in, uint64_t fh_out, uint64_t off_out,
+                  uint64_t len, uint64_t* copied);
     St write(uint64_t fh, uint64_t off, const uint8_t* data, uint64_t len);
     St flush(uint64_t fh);
     St fsync(uint64_t fh);  // flush + journal fdatasync

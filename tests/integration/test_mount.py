@@ -134,41 +134,83 @@ def test_readdirplus_listing_with_attrs(tmp_mount):
     """`ls -l` pattern: READDIRPLUS (offered via FUSE_DO_READDIRPLUS) returns
     entry+attr in one trip; every stat must be correct and later FORGETs must
     not underflow (unmount at fixture teardown would hang/crash if so)."""
-    d = os.path.join(tmp_mount, "plusdir")
-    os.mkdir(d)
-    sizes = {}
-    for i in range(200):
-        p = os.path.join(d, f"f{i:03d}")
-        with open(p, "wb") as f:
-            f.write(b"x" * (i * 7))
-        sizes[f"f{i:03d}"] = i * 7
-    with os.scandir(d) as it:  # scandir consumes d_type + stat from the plus reply
-        seen = {}
-        for e in it:
-            st = e.stat()
-            assert e.is_file()
-            seen[e.name] = st.st_size
-    assert seen == sizes
-    # second pass straight after FORGET churn
-    assert sorted(os.listdir(d)) == sorted(sizes)
-    for name, sz in sizes.items():
-        assert os.stat(os.path.join(d, name)).st_size == sz
+    with Mount(tmp_mount, CONTENTOPTS):
+        d = os.path.join(tmp_mount, "plusdir")
+        os.mkdir(d)
+        sizes = {}
+        for i in range(200):
+            p = os.path.join(d, f"f{i:03d}")
+            with open(p, "wb") as f:
+                f.write(b"x" * (i * 7))
+            sizes[f"f{i:03d}"] = i * 7
+        with os.scandir(d) as it:  # scandir consumes d_type + stat per entry
+            seen = {}
+            for e in it:
+                st = e.stat()
+                assert e.is_file()
+                seen[e.name] = st.st_size
+        assert seen == sizes
+        # second pass straight after FORGET churn
+        assert sorted(os.listdir(d)) == sorted(sizes)
+        for name, sz in sizes.items():
+            assert os.stat(os.path.join(d, name)).st_size == sz
 
 
 def test_xattr_roundtrip(tmp_mount):
-    p = os.path.join(tmp_mount, "xf")
-    with open(p, "wb") as f:
-        f.write(b"hi")
-    os.setxattr(p, "user.color", b"blue")
-    os.setxattr(p, "user.blob", bytes(range(256)))
-    assert os.getxattr(p, "user.color") == b"blue"
-    assert os.getxattr(p, "user.blob") == bytes(range(256))
-    assert sorted(os.listxattr(p)) == ["user.blob", "user.color"]
-    os.setxattr(p, "user.color", b"red", os.XATTR_REPLACE)
-    assert os.getxattr(p, "user.color") == b"red"
-    with pytest.raises(OSError):
-        os.setxattr(p, "user.color", b"x", os.XATTR_CREATE)  # EEXIST
-    os.removexattr(p, "user.color")
-    with pytest.raises(OSError):
-        os.getxattr(p, "user.color")  # ENODATA
-    assert os.listxattr(p) == ["user.blob"]
+    with Mount(tmp_mount, CONTENTOPTS):
+        p = os.path.join(tmp_mount, "xf")
+        with open(p, "wb") as f:
+            f.write(b"hi")
+        os.setxattr(p, "user.color", b"blue")
+        os.setxattr(p, "user.blob", bytes(range(256)))
+        assert os.getxattr(p, "user.color") == b"blue"
+        assert os.getxattr(p, "user.blob") == bytes(range(256))
+        assert sorted(os.listxattr(p)) == ["user.blob", "user.color"]
+        os.setxattr(p, "user.color", b"red", os.XATTR_REPLACE)
+        assert os.getxattr(p, "user.color") == b"red"
+        with pytest.raises(OSError):
+            os.setxattr(p, "user.color", b"x", os.XATTR_CREATE)  # EEXIST
+        os.removexattr(p, "user.color")
+        with pytest.raises(OSError):
+            os.getxattr(p, "user.color")  # ENODATA
+        assert os.listxattr(p) == ["user.blob"]
+
+
+def test_copy_file_range_instant_dedup(tmp_mount):
+    """FUSE_COPY_FILE_RANGE: content-addressed server-side copy — aligned
+    ranges share chunk refs (no data movement), edges fall back to RMW."""
+    with Mount(tmp_mount, CONTENTOPTS):
+        a = os.path.join(tmp_mount, "cfr_a")
+        b = os.path.join(tmp_mount, "cfr_b")
+        data = os.urandom(5 * (1 << 20) + 12345)
+        with open(a, "wb") as f:
+            f.write(data)
+        os.sync()
+        fin = os.open(a, os.O_RDONLY)
+        fout = os.open(b, os.O_WRONLY | os.O_CREAT)
+        try:
+            total, off = 0, 0
+            while total < len(data):
+                n = os.copy_file_range(fin, fout, len(data) - total, off, off)
+                assert n > 0
+                total += n
+                off += n
+        finally:
+            os.close(fin)
+            os.close(fout)
+        with open(b, "rb") as f:
+            assert f.read() == data
+        # unaligned cross-copy into the middle of b
+        fin = os.open(a, os.O_RDONLY)
+        fout = os.open(b, os.O_WRONLY)
+        try:
+            n = os.copy_file_range(fin, fout, 100000, 4096 + 7, (1 << 20) + 13)
+            assert n == 100000
+        finally:
+            os.close(fin)
+            os.close(fout)
+        expect = bytearray(data)
+        expect[(1 << 20) + 13:(1 << 20) + 13 + 100000] = \
+            data[4096 + 7:4096 + 7 + 100000]
+        with open(b, "rb") as f:
+            assert f.read() == bytes(expect)
