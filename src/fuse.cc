@@ -546,15 +546,24 @@ void FuseMount::handle(const uint8_t* req, size_t len, std::vector<uint8_t>* rep
             return;
         case FUSE_FALLOCATE: {
             auto* f = (const fuse_fallocate_in*)arg;
-            if (f->mode != 0) { reply_err(-EOPNOTSUPP); return; }
             Inode* in = vfs_->handle_inode(f->fh);
             if (!in) { reply_err(-EBADF); return; }
-            uint64_t end = f->offset + f->length;
-            if (end > in->size()) {
-                St s = vfs_->setattr_size(in->ino, end);
-                if (s != St::ok) { reply_st(s); return; }
+            if (f->mode == 0) {  // reserve: extend the logical size
+                uint64_t end = f->offset + f->length;
+                if (end > in->size()) {
+                    St s = vfs_->setattr_size(in->ino, end);
+                    if (s != St::ok) { reply_st(s); return; }
+                }
+                reply_err(0);
+                return;
             }
-            reply_err(0);
+            // FALLOC_FL_PUNCH_HOLE (1) requires KEEP_SIZE (2)
+            if (f->mode == 0x3) {
+                if (!in->obj) { reply_err(-EBADF); return; }
+                reply_st(in->obj->punch_hole(f->offset, f->length));
+                return;
+            }
+            reply_err(-EOPNOTSUPP);
             return;
         }
         case FUSE_LSEEK: {

@@ -128,6 +128,50 @@ void Object::mark_stripe_dirty(uint64_t chunk_idx) {
 // Refresh parity for dirty stripes whose data chunks are all committed.
 // Stripe s covers chunks [s*k, s*k+k); chunks beyond the object are holes
 // (zero shards). Parity refs live at mf_.parity[s*m + i].
+St Object::punch_hole(uint64_t off, uint64_t len) {
+    const uint64_t cs = conf_->chunk_size;
+    if (conf_->cdc) return St::inval;  // append-only CDC streams: unsupported
+    uint64_t zero_head = 0, zero_head_len = 0, zero_tail = 0, zero_tail_len = 0;
+    {
+        std::unique_lock<std::mutex> g(mu_);
+        wait_inflight_locked(g);
+        if (off >= size_ || len == 0) return St::ok;
+        len = std::min(len, size_ - off);
+        uint64_t first_full = div_up(off, cs);
+        uint64_t end_full = (off + len) / cs;  // exclusive
+        for (uint64_t idx = first_full; idx < end_full; idx++) {
+            dirty_.erase(idx);
+            if (idx < mf_.chunks.size() && !mf_.chunks[idx].fp.is_zero()) {
+                drop_ref(mf_.chunks[idx]);
+                mf_.chunks[idx] = ChunkRef{};  // hole: reads produce zeros
+                mark_stripe_dirty(idx);
+            }
+        }
+        if (off % cs && (first_full > end_full || off / cs < first_full)) {
+            zero_head = off;
+            zero_head_len = std::min(len, cs - off % cs);
+        }
+        uint64_t tail_start = end_full * cs;
+        if (off + len > tail_start && tail_start >= off) {
+            zero_tail = std::max(off, tail_start);
+            zero_tail_len = off + len - zero_tail;
+        }
+    }
+    // edge zeroing reuses the normal RMW write path (no locks held here)
+    std::vector<uint8_t> z;
+    if (zero_head_len) {
+        z.assign(zero_head_len, 0);
+        St s = write(zero_head, z.data(), zero_head_len);
+        if (s != St::ok) return s;
+    }
+    if (zero_tail_len && zero_tail != zero_head) {
+        z.assign(zero_tail_len, 0);
+        St s = write(zero_tail, z.data(), zero_tail_len);
+        if (s != St::ok) return s;
+    }
+    return St::ok;
+}
+
 St Object::copy_range(Object* src, uint64_t soff, uint64_t doff, uint64_t len,
                       uint64_t* copied) {
     *copied = 0;

@@ -28,6 +28,10 @@ class Object {
     St read(uint64_t off, uint64_t len, uint8_t* out, uint64_t* got);
     // Batched scatter read: n ranges concatenated into out — ONE engine batch
     // (the 4 KiB random-read path: fio iodepth analog).
+    // FALLOC_FL_PUNCH_HOLE: whole chunks in range drop their refs (space
+    // reclaimed, reads return zeros); partial edge chunks are zeroed in place.
+    St punch_hole(uint64_t off, uint64_t len);
+
     // Server-side copy (FUSE_COPY_FILE_RANGE): chunk-aligned spans share the
     // source's chunk refs (content-addressed — no data moves); edges fall
     // back to read+write. Clamps at the source EOF; returns bytes copied.

@@ -214,3 +214,29 @@ def test_copy_file_range_instant_dedup(tmp_mount):
             data[4096 + 7:4096 + 7 + 100000]
         with open(b, "rb") as f:
             assert f.read() == bytes(expect)
+
+
+def test_fallocate_punch_hole(tmp_mount):
+    import ctypes
+    libc = ctypes.CDLL("libc.so.6", use_errno=True)
+    FALLOC_FL_KEEP_SIZE, FALLOC_FL_PUNCH_HOLE = 0x1, 0x2
+    with Mount(tmp_mount, CONTENTOPTS):
+        p = os.path.join(tmp_mount, "holey")
+        data = bytearray(rand_bytes(4 * (1 << 20) + 999, seed=31))
+        with open(p, "wb") as f:
+            f.write(data)
+        os.sync()
+        fd = os.open(p, os.O_RDWR)
+        try:
+            # punch an unaligned 2.5 MiB hole
+            off, ln = 123456, (5 << 19) + 777
+            rc = libc.fallocate(fd, FALLOC_FL_PUNCH_HOLE | FALLOC_FL_KEEP_SIZE,
+                                ctypes.c_long(off), ctypes.c_long(ln))
+            assert rc == 0, ctypes.get_errno()
+        finally:
+            os.close(fd)
+        data[off:off + ln] = b"\x00" * ln
+        with open(p, "rb") as f:
+            got = f.read()
+        assert len(got) == len(data)
+        assert got == bytes(data)
