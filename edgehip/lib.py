@@ -57,6 +57,10 @@ class EhLib:
         so.eh_object_preadv.argtypes = [ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p,
                                         ctypes.c_void_p, ctypes.c_void_p]
         so.eh_object_preadv.restype = ctypes.c_int64
+        so.eh_object_copy_range.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                            ctypes.c_uint64, ctypes.c_uint64,
+                                            ctypes.c_uint64]
+        so.eh_object_copy_range.restype = ctypes.c_int64
 
     # ---- convenience wrappers ----
     def blake3(self, data: bytes) -> bytes:
@@ -216,6 +220,14 @@ class Obj:
         got = self.lib._so.eh_object_read(self.h, off, b, n)
         if got < 0:
             raise OSError(-got, "object read failed")
+        return got
+
+    def copy_range(self, src, soff: int, doff: int, n: int) -> int:
+        """Server-side copy from `src` object: aligned spans share chunk refs
+        (no data movement)."""
+        got = self.lib._so.eh_object_copy_range(self.h, src.h, soff, doff, n)
+        if got < 0:
+            raise OSError(int(-got), "copy_range failed")
         return got
 
     def preadv(self, offs, lens, buf) -> int:

@@ -131,6 +131,14 @@ extern "C" int64_t eh_object_preadv(void* o, int n, const uint64_t* offs,
     return (int64_t)got;
 }
 
+extern "C" int64_t eh_object_copy_range(void* dst, void* src, uint64_t soff,
+                                        uint64_t doff, uint64_t len) {
+    uint64_t copied = 0;
+    St s = ((Object*)dst)->copy_range((Object*)src, soff, doff, len, &copied);
+    if (s != St::ok) return (int64_t)st_errno(s);
+    return (int64_t)copied;
+}
+
 extern "C" int eh_engine_drop_chunk(void* e, const uint8_t* fp32) {
     Fp f;
     std::memcpy(f.b.data(), fp32, 32);

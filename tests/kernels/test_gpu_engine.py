@@ -111,3 +111,21 @@ def test_gpu_cdc_mode():
             o2.write(0, b"Y" * 123 + data)
             o2.flush()
         assert eng.stored_bytes - n1 < 5 << 20  # shifted content re-dedups
+
+
+def test_copy_range_gpu_addref():
+    """GPU engine addref path: server-side copy shares HBM-resident chunks."""
+    from edgehip.lib import Engine
+    with Engine("engine=gpu,fp=blake3,compress=lz4,dedup=1,arena=2g,"
+                "spill=0") as eng:
+        with eng.object() as a, eng.object() as b:
+            data = rand_bytes(8 << 20, seed=44)
+            a.write(0, data)
+            a.flush()
+            before = eng.stored_chunks
+            assert b.copy_range(a, 0, 0, len(data)) == len(data)
+            assert eng.stored_chunks == before
+            assert b.read(0, len(data)) == data
+            a.truncate(0)  # b's refs keep the chunks alive
+            assert b.read(0, len(data)) == data
+            b.truncate(0)
