@@ -570,16 +570,19 @@ void FuseMount::handle(const uint8_t* req, size_t len, std::vector<uint8_t>* rep
             auto* l = (const fuse_lseek_in*)arg;
             Inode* in = vfs_->handle_inode(l->fh);
             if (!in) { reply_err(-EBADF); return; }
-            // SEEK_DATA=3, SEEK_HOLE=4; no holes tracked at byte level: DATA
-            // everywhere before EOF.
+            // SEEK_DATA=3, SEEK_HOLE=4 — chunk-granular: punched/never-
+            // written chunks are holes, everything else is data.
             uint64_t sz = in->size();
             fuse_lseek_out lo{};
             if (l->whence == 3) {
                 if (l->offset >= sz) { reply_err(-ENXIO); return; }
-                lo.offset = l->offset;
+                lo.offset = in->obj ? in->obj->next_boundary(l->offset, true)
+                                    : l->offset;
+                if (lo.offset >= sz) { reply_err(-ENXIO); return; }
             } else if (l->whence == 4) {
                 if (l->offset >= sz) { reply_err(-ENXIO); return; }
-                lo.offset = sz;
+                lo.offset = in->obj ? in->obj->next_boundary(l->offset, false)
+                                    : sz;
             } else {
                 reply_err(-EINVAL);
                 return;

@@ -128,6 +128,20 @@ void Object::mark_stripe_dirty(uint64_t chunk_idx) {
 // Refresh parity for dirty stripes whose data chunks are all committed.
 // Stripe s covers chunks [s*k, s*k+k); chunks beyond the object are holes
 // (zero shards). Parity refs live at mf_.parity[s*m + i].
+uint64_t Object::next_boundary(uint64_t off, bool want_data) {
+    std::unique_lock<std::mutex> g(mu_);
+    const uint64_t cs = conf_->chunk_size;
+    if (conf_->cdc) return want_data ? off : size_;  // CDC streams: no holes
+    if (off >= size_) return size_;
+    uint64_t idx = off / cs;
+    for (;; idx++, off = idx * cs) {
+        if (off >= size_) return size_;
+        bool data = dirty_.count(idx) != 0 || inflight_view_.count(idx) != 0 ||
+                    (idx < mf_.chunks.size() && !mf_.chunks[idx].fp.is_zero());
+        if (data == want_data) return std::max(off, idx * cs);
+    }
+}
+
 St Object::punch_hole(uint64_t off, uint64_t len) {
     const uint64_t cs = conf_->chunk_size;
     if (conf_->cdc) return St::inval;  // append-only CDC streams: unsupported

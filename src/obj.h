@@ -28,6 +28,10 @@ class Object {
     St read(uint64_t off, uint64_t len, uint8_t* out, uint64_t* got);
     // Batched scatter read: n ranges concatenated into out — ONE engine batch
     // (the 4 KiB random-read path: fio iodepth analog).
+    // SEEK_DATA/SEEK_HOLE support: next offset >= off that is (or is not)
+    // backed by stored/dirty bytes. Returns size_ when no further data.
+    uint64_t next_boundary(uint64_t off, bool want_data);
+
     // FALLOC_FL_PUNCH_HOLE: whole chunks in range drop their refs (space
     // reclaimed, reads return zeros); partial edge chunks are zeroed in place.
     St punch_hole(uint64_t off, uint64_t len);

@@ -240,3 +240,23 @@ def test_fallocate_punch_hole(tmp_mount):
             got = f.read()
         assert len(got) == len(data)
         assert got == bytes(data)
+
+
+def test_lseek_data_hole(tmp_mount):
+    with Mount(tmp_mount, CONTENTOPTS):
+        p = os.path.join(tmp_mount, "sparse")
+        with open(p, "wb") as f:
+            f.write(b"A" * (1 << 20))       # chunk 0: data
+            f.seek(3 << 20)
+            f.write(b"B" * (1 << 20))       # chunk 3: data; 1-2 never written
+        os.sync()
+        fd = os.open(p, os.O_RDONLY)
+        try:
+            assert os.lseek(fd, 0, os.SEEK_DATA) == 0
+            hole = os.lseek(fd, 0, os.SEEK_HOLE)
+            assert hole == (1 << 20)
+            data = os.lseek(fd, hole, os.SEEK_DATA)
+            assert data == (3 << 20)
+            assert os.lseek(fd, data, os.SEEK_HOLE) == (4 << 20)  # EOF
+        finally:
+            os.close(fd)
