@@ -61,6 +61,9 @@ class EhLib:
                                             ctypes.c_uint64, ctypes.c_uint64,
                                             ctypes.c_uint64]
         so.eh_object_copy_range.restype = ctypes.c_int64
+        so.eh_object_punch_hole.argtypes = [ctypes.c_void_p, ctypes.c_uint64,
+                                            ctypes.c_uint64]
+        so.eh_object_punch_hole.restype = ctypes.c_int
 
     # ---- convenience wrappers ----
     def blake3(self, data: bytes) -> bytes:
@@ -221,6 +224,11 @@ class Obj:
         if got < 0:
             raise OSError(-got, "object read failed")
         return got
+
+    def punch_hole(self, off: int, n: int) -> None:
+        rc = self.lib._so.eh_object_punch_hole(self.h, off, n)
+        if rc != 0:
+            raise OSError(-rc, "punch_hole failed")
 
     def copy_range(self, src, soff: int, doff: int, n: int) -> int:
         """Server-side copy from `src` object: aligned spans share chunk refs

@@ -47,6 +47,7 @@ int eh_object_flush(void* obj);
 int eh_object_truncate(void* obj, uint64_t size);
 uint64_t eh_object_size(void* obj);
 int64_t eh_object_commit(void* obj, uint8_t* manifest_out, size_t cap);
+int eh_object_punch_hole(void* obj, uint64_t off, uint64_t len);
 /* server-side copy: returns bytes copied (aligned spans share chunk refs) */
 int64_t eh_object_copy_range(void* dst_obj, void* src_obj, uint64_t src_off,
                              uint64_t dst_off, uint64_t len);

@@ -131,6 +131,10 @@ extern "C" int64_t eh_object_preadv(void* o, int n, const uint64_t* offs,
     return (int64_t)got;
 }
 
+extern "C" int eh_object_punch_hole(void* o, uint64_t off, uint64_t len) {
+    return (int)st_errno(((Object*)o)->punch_hole(off, len));
+}
+
 extern "C" int64_t eh_object_copy_range(void* dst, void* src, uint64_t soff,
                                         uint64_t doff, uint64_t len) {
     uint64_t copied = 0;

@@ -22,6 +22,8 @@ ops = st.lists(
         st.tuples(st.just("read"), st.integers(0, SPAN + MB), st.integers(1, MB)),
         st.tuples(st.just("truncate"), st.integers(0, SPAN), st.just(0)),
         st.tuples(st.just("flush"), st.just(0), st.just(0)),
+        st.tuples(st.just("punch"), st.integers(0, SPAN), st.integers(1, 2 * MB)),
+        st.tuples(st.just("selfcopy"), st.integers(0, SPAN), st.integers(1, 2 * MB)),
     ),
     min_size=1,
     max_size=12,
@@ -57,5 +59,19 @@ def test_object_matches_model(ops, seed):
                     model.extend(b"\x00" * (a - len(model)))
             elif op == "flush":
                 o.flush()
+            elif op == "punch":
+                o.punch_hole(a, b)
+                end = min(a + b, len(model))
+                if a < len(model):
+                    model[a:end] = b"\x00" * (end - a)
+            elif op == "selfcopy":
+                dst = (a + 2 * MB) % (SPAN + MB)
+                got = o.copy_range(o, a, dst, b)
+                want = min(b, max(0, len(model) - a)) if a < len(model) else 0
+                assert got == want, f"selfcopy({a},{dst},{b}) -> {got} != {want}"
+                if got:
+                    if len(model) < dst + got:
+                        model.extend(b"\x00" * (dst + got - len(model)))
+                    model[dst:dst + got] = model[a:a + got]
         assert o.size == len(model)
         assert o.read(0, len(model) + 10) == bytes(model)
