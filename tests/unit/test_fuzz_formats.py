@@ -56,6 +56,10 @@ def test_lz4_chunk_decode_fuzz():
 
 
 @pytest.mark.fuse
+@pytest.mark.skipif("asan" in os.environ.get("EDGEHIP_LIB", "")
+                    or "tsan" in os.environ.get("EDGEHIP_LIB", ""),
+                    reason="spawns bin/edgehip; LD_PRELOADed sanitizer "
+                           "runtime breaks the un-instrumented child")
 def test_journal_replay_truncation_fuzz(tmp_path):
     """Replay of a journal truncated at any byte must not crash fsck and must
     yield a consistent (prefix) namespace."""
