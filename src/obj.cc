@@ -224,7 +224,8 @@ St Object::copy_range(Object* src, uint64_t soff, uint64_t doff, uint64_t len,
                              sidx < src->mf_.chunks.size() &&
                              !src->mf_.chunks[sidx].fp.is_zero() &&
                              src->mf_.chunks[sidx].logical_len == cs;
-            if (src_clean && !dirty_.count(didx) && sidx != (src == this ? didx : ~0ull)) {
+            bool self_same = src == this && sidx == didx;  // no-op span
+            if (src_clean && !dirty_.count(didx) && !self_same) {
                 ChunkRef r = src->mf_.chunks[sidx];
                 if (eng_->addref_chunk(r) == St::ok) {
                     if (mf_.chunks.size() <= didx) mf_.chunks.resize(didx + 1);
