@@ -67,7 +67,6 @@ St DistCtx::allgather_manifests(int dev_slot, const ChunkRef* recs, int n, int s
     std::memcpy(h.data(), recs, n * sizeof(ChunkRef));
     if (hipSetDevice(first_dev_ + dev_slot) != hipSuccess) return St::gpu;
     uint8_t* d_in = (uint8_t*)d_scratch_[dev_slot];
-    uint8_t* d_out = d_in + slots * sizeof(ChunkRef);
     if (hipMemcpyAsync(d_in, h.data(), h.size(), hipMemcpyHostToDevice,
                        (hipStream_t)streams_[dev_slot]) != hipSuccess)
         return St::gpu;
