@@ -360,7 +360,12 @@ void FuseMount::handle(const uint8_t* req, size_t len, std::vector<uint8_t>* rep
             auto* r = (const fuse_rename2_in*)arg;
             const char* oldname = (const char*)(arg + sizeof(*r));
             const char* newname = oldname + strlen(oldname) + 1;
-            bool noreplace = r->flags & (1u << 0);  // RENAME_NOREPLACE
+            if (r->flags & ~3u) { reply_err(-EINVAL); return; }  // no WHITEOUT
+            if (r->flags & 2u) {  // RENAME_EXCHANGE
+                reply_st(vfs_->exchange(h->nodeid, oldname, r->newdir, newname));
+                return;
+            }
+            bool noreplace = r->flags & 1u;  // RENAME_NOREPLACE
             reply_st(vfs_->rename(h->nodeid, oldname, r->newdir, newname, noreplace));
             return;
         }

@@ -595,6 +595,30 @@ St Vfs::fsync(uint64_t fh) {
     return St::ok;
 }
 
+St Vfs::exchange(uint64_t parent, const std::string& name, uint64_t newparent,
+                 const std::string& newname) {
+    std::lock_guard<std::mutex> g(mu_);
+    Inode* p = get(parent);
+    Inode* q = get(newparent);
+    if (!p || !q || !p->is_dir() || !q->is_dir()) return St::not_dir;
+    auto it1 = p->children.find(name);
+    auto it2 = q->children.find(newname);
+    if (it1 == p->children.end() || it2 == q->children.end()) return St::not_found;
+    std::swap(it1->second, it2->second);
+    Inode* a = get(it1->second);
+    Inode* b = get(it2->second);
+    if (a) { a->parent = parent; a->name = name; a->ctime = (int64_t)wall_ns(); }
+    if (b) { b->parent = newparent; b->name = newname; b->ctime = (int64_t)wall_ns(); }
+    if (journal_) {  // two cross renames reproduce the swap on replay via a temp name
+        std::string pa = path_locked(parent), pb = path_locked(newparent);
+        auto j = [&](const std::string& x) { return x == "/" ? std::string() : x; };
+        journal_->rename_rec(j(pa) + "/" + name, j(pb) + "/.eh_xchg_tmp");
+        journal_->rename_rec(j(pb) + "/" + newname, j(pa) + "/" + name);
+        journal_->rename_rec(j(pb) + "/.eh_xchg_tmp", j(pb) + "/" + newname);
+    }
+    return St::ok;
+}
+
 St Vfs::symlink(uint64_t parent, const std::string& name, const std::string& target,
                 uint32_t uid, uint32_t gid, Inode** out) {
     std::lock_guard<std::mutex> g(mu_);

@@ -68,6 +68,9 @@ class Vfs {
     St rmdir(uint64_t parent, const std::string& name);
     St rename(uint64_t parent, const std::string& name, uint64_t newparent,
               const std::string& newname, bool noreplace);
+    // RENAME_EXCHANGE: atomically swap two existing entries.
+    St exchange(uint64_t parent, const std::string& name, uint64_t newparent,
+                const std::string& newname);
     St symlink(uint64_t parent, const std::string& name, const std::string& target,
                uint32_t uid, uint32_t gid, Inode** out);
     St readlink(uint64_t ino, std::string* target);

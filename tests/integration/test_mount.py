@@ -260,3 +260,28 @@ def test_lseek_data_hole(tmp_mount):
             assert os.lseek(fd, data, os.SEEK_HOLE) == (4 << 20)  # EOF
         finally:
             os.close(fd)
+
+
+def test_rename_exchange(tmp_mount):
+    import ctypes
+    libc = ctypes.CDLL("libc.so.6", use_errno=True)
+    RENAME_EXCHANGE, AT_FDCWD = 2, -100
+    with Mount(tmp_mount, CONTENTOPTS):
+        a = os.path.join(tmp_mount, "xa")
+        b = os.path.join(tmp_mount, "sub")
+        os.mkdir(b)
+        bf = os.path.join(b, "xb")
+        with open(a, "w") as f:
+            f.write("AAA")
+        with open(bf, "w") as f:
+            f.write("BBBB")
+        rc = libc.renameat2(AT_FDCWD, a.encode(), AT_FDCWD, bf.encode(),
+                            RENAME_EXCHANGE)
+        assert rc == 0, ctypes.get_errno()
+        assert open(a).read() == "BBBB"
+        assert open(bf).read() == "AAA"
+        # exchanging with a missing target must fail (ENOENT)
+        rc = libc.renameat2(AT_FDCWD, a.encode(), AT_FDCWD,
+                            os.path.join(tmp_mount, "nope").encode(),
+                            RENAME_EXCHANGE)
+        assert rc != 0
