@@ -5,6 +5,9 @@
 #include <malloc.h>
 #include <unistd.h>
 #include <vector>
+#include <dirent.h>
+#include <map>
+#include <set>
 
 #include "manifest.h" 
 
@@ -27,7 +30,7 @@ static int usage() {
     fprintf(stderr,
             "usage: edgehip mount <mountpoint> [-o key=val,...] [-c conf_file]\n"
             "       edgehip stat <mountpoint>\n"
-            "       edgehip fsck <journal_dir>\n"
+            "       edgehip fsck <journal_dir> [--gc]\n"
             "       edgehip mkfs <journal_dir>\n"
             "       edgehip bench <mountpoint> [size_mib [block_kib]]\n"
             "       edgehip version\n"
@@ -81,7 +84,7 @@ static int bench_cmd(const char* mnt, long size_mib, long block_kib) {
 
 // Verify a journal: every record decodes, every manifest's chunks exist in
 // the chunk mirror (or are holes).
-static int fsck_cmd(const char* dir) {
+static int fsck_cmd(const char* dir, bool gc) {
     Journal j;
     if (j.open(dir) != St::ok) return 1;
     std::vector<Journal::Replay> recs;
@@ -89,33 +92,88 @@ static int fsck_cmd(const char* dir) {
         fprintf(stderr, "fsck: journal replay failed (corrupt log)\n");
         return 1;
     }
-    size_t files = 0, chunks = 0, missing = 0, bad_manifest = 0;
+    // last-wins live view: overwritten/unlinked versions legitimately freed
+    // their chunks, so only the surviving manifest per path is checked
+    std::map<std::string, std::vector<uint8_t>> live;
     for (auto& r : recs) {
-        if (r.type != 2) continue;
+        switch (r.type) {
+            case 2:
+                live[r.path] = r.manifest;
+                break;
+            case 3:
+                live.erase(r.path);
+                break;
+            case 4: {
+                if (!r.path.empty() && r.path[0] == '\x01') break;  // pseudo
+                auto it = live.find(r.path);
+                if (it != live.end()) {
+                    live[r.path2] = std::move(it->second);
+                    live.erase(it);
+                }
+                // directory rename: re-prefix every file underneath
+                std::string pre = r.path + "/";
+                std::vector<std::pair<std::string, std::vector<uint8_t>>> moved;
+                for (auto i2 = live.begin(); i2 != live.end();) {
+                    if (i2->first.rfind(pre, 0) == 0) {
+                        moved.emplace_back(r.path2 + "/" + i2->first.substr(pre.size()),
+                                           std::move(i2->second));
+                        i2 = live.erase(i2);
+                    } else {
+                        ++i2;
+                    }
+                }
+                for (auto& mv : moved) live[mv.first] = std::move(mv.second);
+                break;
+            }
+            default:
+                break;
+        }
+    }
+    size_t files = 0, chunks = 0, missing = 0, bad_manifest = 0;
+    std::set<std::string> referenced;
+    for (auto& kv : live) {
         files++;
         Manifest m;
-        if (Manifest::decode(r.manifest.data(), r.manifest.size(), &m) != St::ok) {
-            fprintf(stderr, "fsck: bad manifest for %s\n", r.path.c_str());
+        if (Manifest::decode(kv.second.data(), kv.second.size(), &m) != St::ok) {
+            fprintf(stderr, "fsck: bad manifest for %s\n", kv.first.c_str());
             bad_manifest++;
             continue;
         }
         for (auto& c : m.chunks) {
             if (c.fp.is_zero()) continue;
             chunks++;
+            referenced.insert(c.fp.hex());
             std::vector<uint8_t> bytes;
             if (j.get_chunk(c.fp, &bytes) != St::ok) {
-                fprintf(stderr, "fsck: %s: missing chunk %s\n", r.path.c_str(),
+                fprintf(stderr, "fsck: %s: missing chunk %s\n", kv.first.c_str(),
                         c.fp.hex().c_str());
                 missing++;
             } else if (bytes.size() != c.logical_len) {
-                fprintf(stderr, "fsck: %s: chunk %s size %zu != %u\n", r.path.c_str(),
-                        c.fp.hex().c_str(), bytes.size(), c.logical_len);
+                fprintf(stderr, "fsck: %s: chunk %s size %zu != %u\n",
+                        kv.first.c_str(), c.fp.hex().c_str(), bytes.size(),
+                        c.logical_len);
                 missing++;
             }
         }
     }
-    printf("fsck: %zu records, %zu file versions, %zu chunks, %zu missing, %zu bad\n",
-           recs.size(), files, chunks, missing, bad_manifest);
+    // GC: chunk-mirror files not referenced by any live manifest (e.g. from a
+    // kill -9 between chunk write and manifest commit) are safe to remove
+    size_t orphans = 0;
+    std::string cdir = std::string(dir) + "/chunks";
+    if (DIR* d = opendir(cdir.c_str())) {
+        while (struct dirent* e = readdir(d)) {
+            if (e->d_name[0] == '.') continue;
+            std::string name(e->d_name);
+            if (referenced.count(name)) continue;
+            orphans++;
+            if (gc) ::unlink((cdir + "/" + name).c_str());
+        }
+        closedir(d);
+    }
+    printf("fsck: %zu records, %zu live files, %zu chunks, %zu missing, %zu bad, "
+           "%zu orphan chunk files%s\n",
+           recs.size(), files, chunks, missing, bad_manifest, orphans,
+           gc && orphans ? " (removed)" : "");
     return (missing || bad_manifest) ? 1 : 0;
 }
 
@@ -143,7 +201,8 @@ int main(int argc, char** argv) {
     }
     if (cmd == "fsck") {
         if (argc < 3) return usage();
-        return fsck_cmd(argv[2]);
+        bool gc = argc > 3 && std::string(argv[3]) == "--gc";
+        return fsck_cmd(argv[2], gc);
     }
     if (cmd == "mkfs") {
         // initialize (or wipe) a journal directory for a durable mount

@@ -97,3 +97,28 @@ def test_xattrs_survive_remount(tmp_mount, tmp_path):
         p = tmp_mount / "tagged.txt"
         assert os.listxattr(p) == ["user.keep"]
         assert os.getxattr(p, "user.keep") == b"v2"
+
+
+def test_fsck_gc_orphans(tmp_mount, tmp_path):
+    """fsck --gc removes chunk-mirror files no live manifest references
+    (overwritten data and planted orphans), and live data survives."""
+    import subprocess
+    jdir = tmp_path / "jrnl"
+    data = rand_bytes(3 << 20, seed=60)
+    with Mount(tmp_mount, opts(jdir)):
+        (tmp_mount / "keep.bin").write_bytes(data)
+        (tmp_mount / "gone.bin").write_bytes(rand_bytes(2 << 20, seed=61))
+        (tmp_mount / "gone.bin").unlink()
+    # plant a fake orphan chunk file
+    orphan = jdir / "chunks" / ("ff" * 32)
+    orphan.write_bytes(b"junk")
+    r = subprocess.run(["./bin/edgehip", "fsck", str(jdir)], capture_output=True,
+                       text=True, timeout=60)
+    assert r.returncode == 0, r.stderr
+    assert "orphan" in r.stdout
+    r = subprocess.run(["./bin/edgehip", "fsck", str(jdir), "--gc"],
+                       capture_output=True, text=True, timeout=60)
+    assert r.returncode == 0, r.stderr
+    assert not orphan.exists()
+    with Mount(tmp_mount, opts(jdir)):
+        assert (tmp_mount / "keep.bin").read_bytes() == data
