@@ -28,7 +28,7 @@ static void on_signal(int) {
 
 static int usage() {
     fprintf(stderr,
-            "usage: edgehip mount <mountpoint> [-o key=val,...] [-c conf_file]\n"
+            "usage: edgehip mount <mountpoint> [-o key=val,...] [-c conf_file] [-d]\n"
             "       edgehip stat <mountpoint>\n"
             "       edgehip fsck <journal_dir> [--gc]\n"
             "       edgehip mkfs <journal_dir>\n"
@@ -225,14 +225,33 @@ int main(int argc, char** argv) {
         if (argc < 3) return usage();
         std::string mnt = argv[2];
         Conf conf;
+        bool daemonize = false;
         if (conf.apply_env() != St::ok) return 2;
         for (int i = 3; i < argc; i++) {
             if (!strcmp(argv[i], "-o") && i + 1 < argc) {
                 if (conf.apply_opts(argv[++i]) != St::ok) return 2;
             } else if (!strcmp(argv[i], "-c") && i + 1 < argc) {
                 if (conf.load_file(argv[++i]) != St::ok) return 2;
+            } else if (!strcmp(argv[i], "-d") || !strcmp(argv[i], "--daemon")) {
+                daemonize = true;
             } else {
                 return usage();
+            }
+        }
+        if (daemonize) {
+            // classic double-fork AFTER arg parsing, BEFORE engine init (HIP
+            // contexts do not survive fork) — parent exits once the child is off
+            pid_t pid = fork();
+            if (pid < 0) { perror("fork"); return 1; }
+            if (pid > 0) return 0;
+            setsid();
+            pid = fork();
+            if (pid < 0) _exit(1);
+            if (pid > 0) _exit(0);
+            int devnull = ::open("/dev/null", O_RDWR);
+            if (devnull >= 0) {
+                dup2(devnull, 0);
+                if (conf.log_level < 3) { dup2(devnull, 1); dup2(devnull, 2); }
             }
         }
         g_log_level = (LogLevel)conf.log_level;
