@@ -122,3 +122,21 @@ def test_fsck_gc_orphans(tmp_mount, tmp_path):
     assert not orphan.exists()
     with Mount(tmp_mount, opts(jdir)):
         assert (tmp_mount / "keep.bin").read_bytes() == data
+
+
+def test_rename_exchange_survives_remount(tmp_mount, tmp_path):
+    import ctypes
+    libc = ctypes.CDLL("libc.so.6", use_errno=True)
+    RENAME_EXCHANGE, AT_FDCWD = 2, -100
+    jdir = tmp_path / "jrnl"
+    with Mount(tmp_mount, opts(jdir)):
+        (tmp_mount / "p").write_text("PPP")
+        (tmp_mount / "q").write_text("QQ")
+        rc = libc.renameat2(AT_FDCWD, str(tmp_mount / "p").encode(),
+                            AT_FDCWD, str(tmp_mount / "q").encode(),
+                            RENAME_EXCHANGE)
+        assert rc == 0, ctypes.get_errno()
+        assert (tmp_mount / "p").read_text() == "QQ"
+    with Mount(tmp_mount, opts(jdir)):
+        assert (tmp_mount / "p").read_text() == "QQ"
+        assert (tmp_mount / "q").read_text() == "PPP"
