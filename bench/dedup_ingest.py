@@ -25,20 +25,36 @@ def main():
     per = args.gib_per_file << 30
     base = rand_bytes(256 * mb, seed=7)  # shared content ring
 
+    import threading
+
     with Engine(f"engine=gpu,ngpus={args.ngpus},fp=blake3,compress=lz4,dedup=1,"
                 "arena=16g,spill=0") as eng:
         t0 = time.perf_counter()
-        objs = []
-        for f in range(args.files):
-            o = eng.object()
-            # distinct 1 MiB head, identical body => ~100% dedup after file 0
-            o.write(0, rand_bytes(mb, seed=1000 + f))
-            for off in range(mb, per, 64 * mb):
-                n = min(64 * mb, per - off)
-                src_off = (off - mb) % (192 * mb)
-                o.write(off, base[src_off:src_off + n])
-            o.flush()
-            objs.append(o)
+        objs = [eng.object() for _ in range(args.files)]
+        errs = []
+
+        def ingest(f):
+            # fio-style: all files ingest concurrently (ctypes drops the GIL,
+            # so writer memcpys and engine batches overlap across files)
+            try:
+                o = objs[f]
+                # distinct 1 MiB head, identical body => ~100% dedup vs file 0
+                o.write(0, rand_bytes(mb, seed=1000 + f))
+                for off in range(mb, per, 64 * mb):
+                    n = min(64 * mb, per - off)
+                    src_off = (off - mb) % (192 * mb)
+                    o.write(off, base[src_off:src_off + n])
+                o.flush()
+            except Exception as e:  # pragma: no cover
+                errs.append(e)
+
+        ts = [threading.Thread(target=ingest, args=(f,)) for f in range(args.files)]
+        for t in ts:
+            t.start()
+        for t in ts:
+            t.join()
+        if errs:
+            raise errs[0]
         dt = time.perf_counter() - t0
         logical = args.files * per
         stored = eng.stored_bytes
