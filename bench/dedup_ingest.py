@@ -23,7 +23,9 @@ def main():
 
     mb = 1 << 20
     per = args.gib_per_file << 30
-    base = rand_bytes(256 * mb, seed=7)  # shared content ring
+    base = memoryview(bytearray(rand_bytes(256 * mb, seed=7)))  # shared ring
+    # (memoryview slices pass to ctypes zero-copy; bytes slicing would copy
+    # under the GIL and serialize the concurrent writers)
 
     import threading
 
