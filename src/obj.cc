@@ -30,7 +30,7 @@ class FlushPool {
                 started_ = true;
                 // process-lifetime workers: detached so static teardown of
                 // this singleton never destroys a joinable std::thread
-                for (int i = 0; i < 4; i++) std::thread([this] { loop(); }).detach();
+                for (int i = 0; i < 8; i++) std::thread([this] { loop(); }).detach();
             }
             q_.push_back(std::move(fn));
         }
