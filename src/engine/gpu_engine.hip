@@ -490,16 +490,6 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n,
     metrics().h2d_bytes += off;
 
     if (do_fp) launch_blake3_chunks(S.d_stage, S.d_chunks, S.d_fps, n, S.s, S.d_fpg);
-    if (do_lz4) {
-        HIP_CHECK(hipMemcpyAsync(S.d_segmeta, h_segmeta, sizeof(ulong2) * nsegs,
-                                 hipMemcpyHostToDevice, S.s));
-        launch_lz4_compress_segs(S.d_stage, S.d_segmeta, S.d_segscratch, S.d_segsizes, nsegs,
-                                 S.s);
-        HIP_CHECK(hipMemcpyAsync(S.d_chunks + kMaxBatch, h_chunks2, sizeof(DevChunk) * n,
-                                 hipMemcpyHostToDevice, S.s));
-        launch_lz4_emit(S.d_stage, S.d_segscratch, S.d_segsizes, S.d_chunks + kMaxBatch,
-                        S.d_comp, S.d_sizes, n, S.s);
-    }
 
     auto* h_fps = (uint8_t*)(S.pin_meta + kOffFps);
     auto* h_sizes = (uint32_t*)(S.pin_meta + kOffSizes);
@@ -516,9 +506,61 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n,
     }
     if (do_fp)
         HIP_CHECK(hipMemcpyAsync(h_fps, S.d_fps, 32 * n, hipMemcpyDeviceToHost, S.s));
-    if (do_lz4)
-        HIP_CHECK(hipMemcpyAsync(h_sizes, S.d_sizes, 4 * n, hipMemcpyDeviceToHost, S.s));
     HIP_CHECK(hipStreamSynchronize(S.s));
+
+    // LZ4 runs only for chunks the host map does NOT already store: on
+    // dedup-heavy ingest (config 5) compressing duplicate bytes was ~30% of
+    // batch kernel time for nothing. Cost: one extra sync on mixed batches.
+    bool lz4_ran[kMaxBatch] = {};
+    if (do_lz4) {
+        int nlive = 0, lsegs = 0;
+        auto* h_segmeta2 = (ulong2*)(S.pin_meta + kOffSegMeta);  // rebuilt in place
+        auto* h_chunks2b = (DevChunk*)(S.pin_meta + kOffChunks2);
+        for (int i = 0; i < n; i++) {
+            bool maybe_dup = false;
+            if (do_fp && conf_.dedup) {
+                Fp f;
+                std::memcpy(f.b.data(), h_fps + 32 * i, 32);
+                std::lock_guard<std::mutex> g(map_mu_);
+                maybe_dup = map_.count(f) != 0;
+            }
+            if (maybe_dup) continue;  // stored (or pending) copy exists
+            lz4_ran[i] = true;
+            uint32_t ns2 = (put[i].len + kSegSz - 1) / kSegSz;
+            uint64_t coff = h_chunks[i].src_off;
+            h_chunks2b[nlive] =
+                DevChunk{coff, (uint64_t)i * cs, put[i].len, (uint32_t)lsegs};
+            for (uint32_t s2 = 0; s2 < ns2; s2++)
+                h_segmeta2[lsegs + s2] =
+                    ulong2{coff + (uint64_t)s2 * kSegSz,
+                           std::min(kSegSz, put[i].len - s2 * kSegSz)};
+            lsegs += ns2;
+            nlive++;
+        }
+        if (nlive) {
+            HIP_CHECK(hipMemcpyAsync(S.d_segmeta, h_segmeta2, sizeof(ulong2) * lsegs,
+                                     hipMemcpyHostToDevice, S.s));
+            launch_lz4_compress_segs(S.d_stage, S.d_segmeta, S.d_segscratch,
+                                     S.d_segsizes, lsegs, S.s);
+            HIP_CHECK(hipMemcpyAsync(S.d_chunks + kMaxBatch, h_chunks2b,
+                                     sizeof(DevChunk) * nlive, hipMemcpyHostToDevice,
+                                     S.s));
+            launch_lz4_emit(S.d_stage, S.d_segscratch, S.d_segsizes,
+                            S.d_chunks + kMaxBatch, S.d_comp, S.d_sizes, nlive, S.s);
+            HIP_CHECK(hipMemcpyAsync(h_sizes, S.d_sizes, 4 * nlive,
+                                     hipMemcpyDeviceToHost, S.s));
+            HIP_CHECK(hipStreamSynchronize(S.s));
+            // scatter sizes back to per-chunk order
+            uint32_t tmp[kMaxBatch];
+            int k2 = 0;
+            for (int i = 0; i < n; i++) tmp[i] = 0;
+            for (int i = 0; i < n; i++)
+                if (lz4_ran[i]) tmp[i] = h_sizes[k2++];
+            std::memcpy(h_sizes, tmp, sizeof(uint32_t) * n);
+        } else {
+            std::memset(h_sizes, 0, sizeof(uint32_t) * n);
+        }
+    }
 
     struct Pend {
         int i;
@@ -543,7 +585,7 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n,
             std::memcpy(r.fp.b.data(), &s, 8);
             r.fp.b[31] = 0xFE;
         }
-        uint32_t framed = do_lz4 ? h_sizes[i] : 0;
+        uint32_t framed = (do_lz4 && lz4_ran[i]) ? h_sizes[i] : 0;
         bool compressed = framed != 0 && framed < put[i].len;
         uint32_t stored = compressed ? framed : put[i].len;
         if (compressed) r.flags |= ChunkRef::kFlagLz4;
