@@ -112,6 +112,28 @@ struct DevCtx {
     std::condition_variable slot_cv;
     uint32_t slot_free_mask = (1u << kPipeSlots) - 1;
 
+    // Admission control: bound the arena bytes held by not-yet-landed
+    // (pending) batches. Without it, concurrent batches can each hold a
+    // partial allocation and starve each other on a small arena (classic
+    // partial-hold deadlock: everything pending, nothing evictable).
+    std::mutex adm_mu;
+    std::condition_variable adm_cv;
+    uint64_t adm_used = 0;
+    uint64_t adm_cap = 0;  // set at init: max(arena/2, 2 batches)
+
+    void admit(uint64_t bytes) {
+        std::unique_lock<std::mutex> g(adm_mu);
+        adm_cv.wait(g, [&] { return adm_used + bytes <= adm_cap || adm_used == 0; });
+        adm_used += bytes;
+    }
+    void unadmit(uint64_t bytes) {
+        {
+            std::lock_guard<std::mutex> g(adm_mu);
+            adm_used -= bytes;
+        }
+        adm_cv.notify_all();
+    }
+
     struct ReadSlot {
         hipStream_t s{};
         uint8_t* d_rdstage = nullptr;
@@ -209,6 +231,8 @@ class GpuEngine : public Engine {
             D.arena_sz = conf.arena_bytes;
             HIP_CHECK(hipMalloc(&D.arena, D.arena_sz));
             D.stage_sz = (uint64_t)kMaxBatch * cs;
+            D.adm_cap = std::max<uint64_t>(D.arena_sz / 2,
+                                           2 * (uint64_t)kMaxBatch * cs);
             for (int sl = 0; sl < kPipeSlots; sl++) {
                 PipeSlot& S = D.slots[sl];
                 HIP_CHECK(hipStreamCreateWithFlags(&S.s, hipStreamNonBlocking));
@@ -428,6 +452,19 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n,
     const int ingest_dev =
         force_dev >= 0 ? force_dev : (int)(rr_.fetch_add(1) % devs_.size());
     DevCtx& D = devs_[ingest_dev];
+    // reserve worst-case pending bytes on every device this batch can place
+    // chunks on (placement is by fingerprint; reserving the full batch on
+    // each is conservative but deadlock-free — released once bytes land)
+    const uint64_t adm_bytes = (uint64_t)n * cs;
+    for (auto& Dv : devs_) Dv.admit(adm_bytes);
+    struct AdmGuard {
+        std::vector<DevCtx>& devs;
+        uint64_t bytes;
+        ~AdmGuard() {
+            for (auto& Dv : devs) Dv.unadmit(bytes);
+        }
+    } adm_guard{devs_, adm_bytes};
+
     PipeSlot* SLP = D.acquire_slot();
     PipeSlot& S = *SLP;
     struct SlotGuard {
