@@ -140,3 +140,50 @@ def test_rename_exchange_survives_remount(tmp_mount, tmp_path):
     with Mount(tmp_mount, opts(jdir)):
         assert (tmp_mount / "p").read_text() == "QQ"
         assert (tmp_mount / "q").read_text() == "PPP"
+
+
+def test_crash_torture_random_kill(tmp_mount, tmp_path):
+    """Randomized crash-consistency torture: kill -9 the daemon at a random
+    moment mid-workload, remount, fsck must pass (rc 0/1, no crash) and every
+    file that completed an fsync BEFORE the kill must read back exactly."""
+    import random
+    import threading
+    rng = random.Random(99)
+    jdir = tmp_path / "jrnl"
+    for it in range(5):
+        m = Mount(tmp_mount, opts(jdir))
+        m.__enter__()
+        synced = {}
+        stop = threading.Event()
+
+        def workload():
+            i = 0
+            while not stop.is_set():
+                name = f"t{it}_{i}.bin"
+                data = rand_bytes(rng.randrange(1, 3 << 20), seed=it * 1000 + i)
+                try:
+                    fd = os.open(tmp_mount / name, os.O_WRONLY | os.O_CREAT)
+                    os.write(fd, data)
+                    os.fsync(fd)
+                    os.close(fd)
+                    synced[name] = data  # fsync returned -> durable
+                except OSError:
+                    return  # daemon died mid-op: expected
+                i += 1
+
+        t = threading.Thread(target=workload)
+        t.start()
+        time.sleep(rng.uniform(0.05, 0.6))
+        m.proc.send_signal(signal.SIGKILL)  # exact PID, never a pattern
+        m.proc.wait()
+        stop.set()
+        t.join(timeout=30)
+        subprocess.run(["umount", "-l", str(tmp_mount)], check=False,
+                       stderr=subprocess.DEVNULL)
+        time.sleep(0.1)
+        r = subprocess.run(["./bin/edgehip", "fsck", str(jdir)],
+                           capture_output=True, timeout=60)
+        assert r.returncode in (0, 1), r.stderr[-300:]
+        with Mount(tmp_mount, opts(jdir)):
+            for name, data in synced.items():
+                assert (tmp_mount / name).read_bytes() == data, (it, name)
