@@ -129,3 +129,19 @@ def test_copy_range_gpu_addref():
             a.truncate(0)  # b's refs keep the chunks alive
             assert b.read(0, len(data)) == data
             b.truncate(0)
+
+
+def test_punch_hole_gpu_reclaims():
+    from edgehip.lib import Engine
+    with Engine("engine=gpu,fp=blake3,compress=none,dedup=0,arena=1g,"
+                "spill=0") as eng, eng.object() as o:
+        data = rand_bytes(8 << 20, seed=45)
+        o.write(0, data)
+        o.flush()
+        before = eng.stored_chunks
+        o.punch_hole(1 << 20, 3 << 20)  # chunks 1..3 dropped
+        assert eng.stored_chunks == before - 3
+        expect = bytearray(data)
+        expect[1 << 20:4 << 20] = b"\x00" * (3 << 20)
+        assert o.read(0, len(data)) == bytes(expect)
+        o.truncate(0)
