@@ -32,7 +32,7 @@ static int usage() {
             "       edgehip stat <mountpoint>\n"
             "       edgehip fsck <journal_dir> [--gc]\n"
             "       edgehip mkfs <journal_dir>\n"
-            "       edgehip bench <mountpoint> [size_mib [block_kib]]\n"
+            "       edgehip bench <mountpoint> [size_mib [block_kib [jobs]]]\n"
             "       edgehip version\n"
             "options: see docs/options.md; presets in conf/presets/\n");
     return 2;
@@ -40,50 +40,69 @@ static int usage() {
 
 // fio-style sequential write+read through an existing mount (native driver,
 // no Python overhead): reports GiB/s per phase.
-static int bench_cmd(const char* mnt, long size_mib, long block_kib) {
-    std::string path = std::string(mnt) + "/.edgehip-bench.tmp";
+static int bench_cmd(const char* mnt, long size_mib, long block_kib, int jobs) {
     size_t bs = (size_t)block_kib * 1024;
-    size_t total = (size_t)size_mib * 1024 * 1024;
-    std::vector<uint8_t> buf(bs);
-    for (size_t i = 0; i < bs; i += 8) {
-        uint64_t v = 0x9E3779B97F4A7C15ull * (i + 1);
-        memcpy(&buf[i], &v, std::min((size_t)8, bs - i));
+    size_t per_job = (size_t)size_mib * 1024 * 1024 / jobs;
+    std::vector<double> wsec(jobs), rsec(jobs);
+    std::vector<int> errs(jobs, 0);
+    auto worker = [&](int j) {
+        std::string path =
+            std::string(mnt) + "/.edgehip-bench." + std::to_string(j) + ".tmp";
+        std::vector<uint8_t> buf(bs);
+        for (size_t i = 0; i < bs; i += 8) {
+            uint64_t v = 0x9E3779B97F4A7C15ull * (i + 1 + j * 77);
+            memcpy(&buf[i], &v, std::min((size_t)8, bs - i));
+        }
+        int fd = ::open(path.c_str(), O_WRONLY | O_CREAT | O_TRUNC, 0644);
+        if (fd < 0) { errs[j]++; return; }
+        uint64_t t0 = now_ns();
+        for (size_t off = 0; off < per_job; off += bs) {
+            uint64_t tag = (off + j) * 0x2545F4914F6CDD1Dull;
+            memcpy(buf.data(), &tag, 8);  // vary content: dedup finds nothing
+            if (::write(fd, buf.data(), bs) != (ssize_t)bs) { errs[j]++; return; }
+        }
+        ::fsync(fd);
+        ::close(fd);
+        uint64_t t1 = now_ns();
+        wsec[j] = (t1 - t0) * 1e-9;
+        fd = ::open(path.c_str(), O_RDONLY | O_DIRECT);
+        if (fd < 0) fd = ::open(path.c_str(), O_RDONLY);
+        void* rbuf = nullptr;
+        if (posix_memalign(&rbuf, 4096, bs)) { errs[j]++; return; }
+        t0 = now_ns();
+        size_t got = 0;
+        while (got < per_job) {
+            ssize_t r = ::read(fd, rbuf, bs);
+            if (r <= 0) { errs[j]++; break; }
+            got += (size_t)r;
+        }
+        t1 = now_ns();
+        ::close(fd);
+        free(rbuf);
+        rsec[j] = (t1 - t0) * 1e-9;
+        ::unlink(path.c_str());
+    };
+    std::vector<std::thread> ts;
+    for (int j = 0; j < jobs; j++) ts.emplace_back(worker, j);
+    for (auto& t : ts) t.join();
+    double w = 0, r = 0;
+    int e = 0;
+    for (int j = 0; j < jobs; j++) {
+        w = std::max(w, wsec[j]);
+        r = std::max(r, rsec[j]);
+        e += errs[j];
     }
-    int fd = ::open(path.c_str(), O_WRONLY | O_CREAT | O_TRUNC, 0644);
-    if (fd < 0) { perror("open"); return 1; }
-    uint64_t t0 = now_ns();
-    for (size_t off = 0; off < total; off += bs) {
-        // vary content so dedup finds nothing
-        uint64_t tag = off * 0x2545F4914F6CDD1Dull;
-        memcpy(buf.data(), &tag, 8);
-        if (::write(fd, buf.data(), bs) != (ssize_t)bs) { perror("write"); return 1; }
+    if (e || w == 0 || r == 0) {
+        fprintf(stderr, "bench: %d job errors\n", e);
+        return 1;
     }
-    ::fsync(fd);
-    ::close(fd);
-    uint64_t t1 = now_ns();
-    printf("seq write: %.2f GiB/s (%ld MiB, %ld KiB blocks)\n",
-           total / ((t1 - t0) * 1e-9) / (1ull << 30), size_mib, block_kib);
-    fd = ::open(path.c_str(), O_RDONLY | O_DIRECT);
-    if (fd < 0) fd = ::open(path.c_str(), O_RDONLY);
-    void* rbuf = nullptr;
-    if (posix_memalign(&rbuf, 4096, bs)) return 1;
-    t0 = now_ns();
-    size_t got = 0;
-    while (got < total) {
-        ssize_t n = ::read(fd, rbuf, bs);
-        if (n <= 0) { perror("read"); return 1; }
-        got += (size_t)n;
-    }
-    t1 = now_ns();
-    ::close(fd);
-    free(rbuf);
-    printf("seq read:  %.2f GiB/s\n", total / ((t1 - t0) * 1e-9) / (1ull << 30));
-    ::unlink(path.c_str());
+    double total = (double)per_job * jobs;
+    printf("seq write: %.2f GiB/s (%ld MiB, %ld KiB blocks, %d jobs)\n",
+           total / w / (1ull << 30), size_mib, block_kib, jobs);
+    printf("seq read:  %.2f GiB/s\n", total / r / (1ull << 30));
     return 0;
 }
 
-// Verify a journal: every record decodes, every manifest's chunks exist in
-// the chunk mirror (or are holes).
 static int fsck_cmd(const char* dir, bool gc) {
     Journal j;
     if (j.open(dir) != St::ok) return 1;
@@ -219,7 +238,8 @@ int main(int argc, char** argv) {
         if (argc < 3) return usage();
         long mib = argc > 3 ? atol(argv[3]) : 2048;
         long bk = argc > 4 ? atol(argv[4]) : 1024;
-        return bench_cmd(argv[2], mib, bk);
+        int jobs = argc > 5 ? atoi(argv[5]) : 4;
+        return bench_cmd(argv[2], mib, bk, jobs > 0 ? jobs : 1);
     }
     if (cmd == "mount") {
         if (argc < 3) return usage();
