@@ -8,6 +8,7 @@
 #include <dirent.h>
 #include <map>
 #include <set>
+#include <thread>
 
 #include "manifest.h" 
 
