@@ -104,7 +104,9 @@ St FuseMount::mount(const std::string& mountpoint) {
     rep.o.max_readahead = conf_->max_readahead;
     uint32_t want = FUSE_ASYNC_READ | FUSE_BIG_WRITES | FUSE_PARALLEL_DIROPS |
                     FUSE_DO_READDIRPLUS | FUSE_READDIRPLUS_AUTO |
-                    FUSE_MAX_PAGES | FUSE_AUTO_INVAL_DATA;
+                    FUSE_MAX_PAGES | FUSE_AUTO_INVAL_DATA |
+                    FUSE_CACHE_SYMLINKS |       // kernel caches readlink results
+                    FUSE_NO_OPENDIR_SUPPORT;    // our OPENDIR is stateless
     if (conf_->writeback) want |= FUSE_WRITEBACK_CACHE;
     rep.o.flags = want & ii->flags;
     rep.o.max_background = 64;
