@@ -52,11 +52,16 @@ class Mount:
                 self.proc.wait()
 
     def __exit__(self, *a):
-        subprocess.run(["umount", self.mountpoint], check=False,
-                       stderr=subprocess.DEVNULL)
+        # EBUSY is possible while kernel writeback drains — retry briefly
+        for _ in range(20):
+            r = subprocess.run(["umount", self.mountpoint], check=False,
+                               stderr=subprocess.DEVNULL)
+            if r.returncode == 0 or not os.path.ismount(self.mountpoint):
+                break
+            time.sleep(0.1)
         if self.proc:
             try:
-                self.proc.wait(timeout=10)
+                self.proc.wait(timeout=30)
             except subprocess.TimeoutExpired:
                 self.kill()
         return False
