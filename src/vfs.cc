@@ -239,12 +239,15 @@ St Vfs::commit_and_journal(uint64_t ino, const std::shared_ptr<Object>& obj) {
     St s = obj->commit(journal_ && journal_->enabled() ? &bytes : nullptr);
     if (s != St::ok) return s;
     if (journal_ && journal_->enabled()) {
-        std::string p;
-        {
-            std::lock_guard<std::mutex> g(mu_);
-            p = path_locked(ino);
-        }
-        journal_->file_rec(p, bytes);
+        // Path resolution AND the journal append happen under mu_, in program
+        // order with rename/unlink records. FUSE RELEASE is asynchronous: a
+        // commit can run AFTER the file was unlinked or renamed, and a
+        // file_rec appended out of order would resurrect the old path on
+        // replay (observed as a rare remount flake).
+        std::lock_guard<std::mutex> g(mu_);
+        Inode* in = get(ino);
+        if (!in || (in->unlinked && in->nlink == 0)) return St::ok;
+        journal_->file_rec(path_locked(ino), bytes);
     }
     return St::ok;
 }
