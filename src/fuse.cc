@@ -425,7 +425,9 @@ void FuseMount::handle(const uint8_t* req, size_t len, std::vector<uint8_t>* rep
             }
             reply_buf->resize(r->size);
             uint64_t got = 0;
+            uint64_t t0 = now_ns();
             St s = vfs_->read(r->fh, r->offset, r->size, reply_buf->data(), &got);
+            metrics().read_lat.record(now_ns() - t0);
             if (s != St::ok) { reply_st(s); return; }
             reply_body(reply_buf->data(), got);
             return;
@@ -434,7 +436,9 @@ void FuseMount::handle(const uint8_t* req, size_t len, std::vector<uint8_t>* rep
             auto* w = (const fuse_write_in*)arg;
             const uint8_t* data = arg + sizeof(*w);
             if (sizeof(fuse_in_header) + sizeof(*w) + w->size > len) { reply_err(-EINVAL); return; }
+            uint64_t t0 = now_ns();
             St s = vfs_->write(w->fh, w->offset, data, w->size);
+            metrics().write_lat.record(now_ns() - t0);
             if (s != St::ok) { reply_st(s); return; }
             fuse_write_out wo{};
             wo.size = w->size;

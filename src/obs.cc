@@ -61,6 +61,30 @@ bool roctx_on() {
 void roctx_push(const char* name) { roctxRangePushA(name); }
 void roctx_pop() { roctxRangePop(); }
 
+std::string LatHist::render(const char* name) const {
+    // cumulative percentile summary from power-of-two buckets
+    uint64_t total = 0;
+    for (int i = 0; i < kBuckets; i++) total += b[i].load();
+    char buf[256];
+    if (!total) {
+        snprintf(buf, sizeof(buf), "%s p50=- p90=- p99=- n=0\n", name);
+        return buf;
+    }
+    auto pct = [&](double q) {
+        uint64_t need = (uint64_t)(q * total);
+        if (need == 0) need = 1;
+        uint64_t acc = 0;
+        for (int i = 0; i < kBuckets; i++) {
+            acc += b[i].load();
+            if (acc >= need) return (double)(1ull << (i + 1)) / 1e3;  // -> us
+        }
+        return (double)(1ull << kBuckets) / 1e3;
+    };
+    snprintf(buf, sizeof(buf), "%s p50<=%.0fus p90<=%.0fus p99<=%.0fus n=%llu\n",
+             name, pct(0.50), pct(0.90), pct(0.99), (unsigned long long)total);
+    return buf;
+}
+
 std::string Metrics::render() const {
     char buf[4096];
     snprintf(buf, sizeof(buf),
@@ -84,7 +108,10 @@ std::string Metrics::render() const {
              (unsigned long long)cache_hits.load(), (unsigned long long)cache_misses.load(),
              (unsigned long long)h2d_bytes.load(), (unsigned long long)d2h_bytes.load(),
              (unsigned long long)spilled_chunks.load(), (unsigned long long)spill_bytes.load());
-    return std::string(buf);
+    std::string out(buf);
+    out += read_lat.render("read_latency");
+    out += write_lat.render("write_latency");
+    return out;
 }
 
 Faults& Faults::inst() {

@@ -11,6 +11,19 @@
 
 namespace edgehip {
 
+// Power-of-two latency histogram (ns): bucket b counts samples in
+// [2^b, 2^(b+1)). Lock-free; rendered as cumulative micro/millisecond lines.
+struct LatHist {
+    static constexpr int kBuckets = 36;  // up to ~64 s
+    std::atomic<uint64_t> b[kBuckets]{};
+    void record(uint64_t ns) {
+        int i = 63 - __builtin_clzll(ns | 1);
+        if (i >= kBuckets) i = kBuckets - 1;
+        b[i].fetch_add(1, std::memory_order_relaxed);
+    }
+    std::string render(const char* name) const;
+};
+
 struct Metrics {
     std::atomic<uint64_t> fuse_reqs{0};
     std::atomic<uint64_t> read_ops{0}, read_bytes{0};
@@ -22,6 +35,7 @@ struct Metrics {
     std::atomic<uint64_t> cache_hits{0}, cache_misses{0};
     std::atomic<uint64_t> h2d_bytes{0}, d2h_bytes{0};
     std::atomic<uint64_t> spilled_chunks{0}, spill_bytes{0};
+    LatHist read_lat, write_lat;  // per-FUSE-request service time
 
     std::string render() const;  // text for the in-mount .edgehip/stats file
 };

@@ -285,3 +285,15 @@ def test_rename_exchange(tmp_mount):
                             os.path.join(tmp_mount, "nope").encode(),
                             RENAME_EXCHANGE)
         assert rc != 0
+
+
+def test_stats_latency_histograms(tmp_mount):
+    with Mount(tmp_mount, CONTENTOPTS):
+        p = tmp_mount / "lat.bin"
+        p.write_bytes(rand_bytes(2 << 20, seed=70))
+        with open(p, "rb") as f:
+            f.read()
+        stats = (tmp_mount / ".edgehip-stats").read_text()
+        assert "read_latency p50<=" in stats
+        assert "write_latency p50<=" in stats
+        assert "n=0" not in stats.split("write_latency")[1]
