@@ -291,8 +291,14 @@ def test_stats_latency_histograms(tmp_mount):
     with Mount(tmp_mount, CONTENTOPTS):
         p = tmp_mount / "lat.bin"
         p.write_bytes(rand_bytes(2 << 20, seed=70))
-        with open(p, "rb") as f:
-            f.read()
+        # O_DIRECT bypasses the page cache so the read reaches the daemon
+        import mmap as _mmap
+        fd = os.open(p, os.O_RDONLY | os.O_DIRECT)
+        buf = _mmap.mmap(-1, 1 << 20)
+        try:
+            os.readv(fd, [buf])
+        finally:
+            os.close(fd)
         stats = (tmp_mount / ".edgehip-stats").read_text()
         assert "read_latency p50<=" in stats
         assert "write_latency p50<=" in stats
