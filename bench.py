@@ -183,6 +183,7 @@ def main():
     ap.add_argument("--arena", default="32g", help="HBM arena size (GPU engine)")
     ap.add_argument("--spill", default="0", help="host spill tier cap (0=off)")
     ap.add_argument("--rs", default="0", help="erasure coding, e.g. 6+2 (config 4)")
+    ap.add_argument("--journal", default="", help="journal dir (durable mode)")
     ap.add_argument("--dedup", type=int, default=1, choices=[0, 1],
                     help="dedup off gives honest spill numbers (the 1 GiB "
                          "content ring otherwise dedups into HBM)")
@@ -213,6 +214,7 @@ def main():
         args.gib_per_step = 1  # host engine is ~100x slower; keep defaults quick
     opts = (f"engine=gpu,gpu_first={local_rank},fp=blake3,compress=lz4,dedup={args.dedup},"
             f"arena={args.arena},spill={args.spill},rs={args.rs},threads=16,writeback=1"
+            + (f",journal={args.journal}_r{local_rank}" if args.journal else "")
             if use_gpu else
             "engine=host,fp=blake3,compress=lz4,dedup=1,threads=16,writeback=1")
 
