@@ -795,11 +795,35 @@ St Object::set_manifest(const Manifest& m) {
 
 // ---- CDC mode (G2): gear-boundary append stream ---------------------------
 St Object::cdc_write_locked(uint64_t off, const uint8_t* data, uint64_t len) {
-    if (off != size_) return St::inval;  // append-only (docs/options.md)
+    // Append-only stream, but FUSE writeback may deliver requests out of
+    // order across reader threads: stash ahead-of-stream pieces (bounded)
+    // and drain once contiguous. Rewrites below the stream head stay EINVAL
+    // (docs/options.md).
+    if (off > size_) {
+        constexpr uint64_t kStashCap = 512ull << 20;
+        if (cdc_stash_bytes_ + len > kStashCap) return St::inval;
+        auto& v = cdc_stash_[off];
+        if (!v.empty()) return St::inval;  // duplicate delivery
+        v.assign(data, data + len);
+        cdc_stash_bytes_ += len;
+        metrics().write_ops++;
+        metrics().write_bytes += len;
+        return St::ok;
+    }
+    if (off != size_) return St::inval;
     tail_.insert(tail_.end(), data, data + len);
     size_ += len;
     metrics().write_ops++;
     metrics().write_bytes += len;
+    // drain any stashed pieces the new bytes made contiguous
+    for (auto it = cdc_stash_.begin();
+         it != cdc_stash_.end() && it->first == size_;) {
+        tail_.insert(tail_.end(), it->second.begin(), it->second.end());
+        size_ += it->second.size();
+        cdc_stash_bytes_ -= it->second.size();
+        it = cdc_stash_.erase(it);
+        it = cdc_stash_.begin();  // head may have advanced past several
+    }
     // cut complete chunks once enough stream is buffered (keep up to one
     // max-chunk of tail so a boundary near the end can still move)
     if (tail_.size() >= 8 * conf_->chunk_size) return cdc_cut_locked(false);
@@ -807,6 +831,8 @@ St Object::cdc_write_locked(uint64_t off, const uint8_t* data, uint64_t len) {
 }
 
 St Object::cdc_cut_locked(bool final_tail) {
+    if (final_tail && !cdc_stash_.empty())
+        return St::inval;  // stream has an unfilled gap (sparse write)
     if (tail_.empty()) {
         mf_.logical_size = size_;
         return final_tail ? rs_update_locked() : St::ok;

@@ -88,6 +88,11 @@ class Object {
     std::vector<ChunkRef> pending_publish_;  // C4: refs committed since last publish
     // CDC mode (conf.cdc): append-only stream, variable chunk boundaries
     std::vector<uint64_t> cum_;   // cumulative END offset per committed chunk
+    // CDC reorder window: FUSE writeback can deliver a sequentially-written
+    // file's 1 MiB requests out of order across reader threads; stash
+    // ahead-of-stream writes until the gap fills (bounded).
+    std::map<uint64_t, std::vector<uint8_t>> cdc_stash_;
+    uint64_t cdc_stash_bytes_ = 0;
     std::vector<uint8_t> tail_;   // uncut stream tail
     St cdc_write_locked(uint64_t off, const uint8_t* data, uint64_t len);
     St cdc_cut_locked(bool final_tail);

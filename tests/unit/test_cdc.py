@@ -46,8 +46,18 @@ def test_tail_visible_before_flush():
 def test_append_only_enforced():
     with Engine(CDC) as eng, eng.object() as o:
         o.write(0, b"abc")
+        # rewriting below the stream head is rejected outright
         with pytest.raises(OSError):
-            o.write(100, b"gap")
+            o.write(1, b"x")
+        # a forward gap is stashed (FUSE writeback reorders requests) but an
+        # unfilled gap must fail at flush
+        o.write(100, b"gap")
+        with pytest.raises(OSError):
+            o.flush()
+        # filling the gap makes the stream whole again
+        o.write(3, b"y" * 97)
+        o.flush()
+        assert o.size == 103
 
 
 def test_dedup_across_shift():
