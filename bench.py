@@ -184,6 +184,8 @@ def main():
     ap.add_argument("--spill", default="0", help="host spill tier cap (0=off)")
     ap.add_argument("--rs", default="0", help="erasure coding, e.g. 6+2 (config 4)")
     ap.add_argument("--journal", default="", help="journal dir (durable mode)")
+    ap.add_argument("--cdc", type=int, default=0, choices=[0, 1],
+                    help="content-defined chunking mode (append-only streams)")
     ap.add_argument("--dedup", type=int, default=1, choices=[0, 1],
                     help="dedup off gives honest spill numbers (the 1 GiB "
                          "content ring otherwise dedups into HBM)")
@@ -215,6 +217,7 @@ def main():
     opts = (f"engine=gpu,gpu_first={local_rank},fp=blake3,compress=lz4,dedup={args.dedup},"
             f"arena={args.arena},spill={args.spill},rs={args.rs},threads=16,writeback=1"
             + (f",journal={args.journal}_r{local_rank}" if args.journal else "")
+            + (",cdc=1,chunk_size=256k" if args.cdc else "")
             if use_gpu else
             "engine=host,fp=blake3,compress=lz4,dedup=1,threads=16,writeback=1")
 
