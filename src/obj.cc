@@ -816,13 +816,12 @@ St Object::cdc_write_locked(uint64_t off, const uint8_t* data, uint64_t len) {
     metrics().write_ops++;
     metrics().write_bytes += len;
     // drain any stashed pieces the new bytes made contiguous
-    for (auto it = cdc_stash_.begin();
-         it != cdc_stash_.end() && it->first == size_;) {
+    while (!cdc_stash_.empty() && cdc_stash_.begin()->first == size_) {
+        auto it = cdc_stash_.begin();
         tail_.insert(tail_.end(), it->second.begin(), it->second.end());
         size_ += it->second.size();
         cdc_stash_bytes_ -= it->second.size();
-        it = cdc_stash_.erase(it);
-        it = cdc_stash_.begin();  // head may have advanced past several
+        cdc_stash_.erase(it);
     }
     // cut complete chunks once enough stream is buffered (keep up to one
     // max-chunk of tail so a boundary near the end can still move)
