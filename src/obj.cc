@@ -178,7 +178,7 @@ St Object::punch_hole(uint64_t off, uint64_t len) {
         St s = write(zero_head, z.data(), zero_head_len);
         if (s != St::ok) return s;
     }
-    if (zero_tail_len && zero_tail != zero_head) {
+    if (zero_tail_len && (zero_head_len == 0 || zero_tail != zero_head)) {
         z.assign(zero_tail_len, 0);
         St s = write(zero_tail, z.data(), zero_tail_len);
         if (s != St::ok) return s;

@@ -176,3 +176,16 @@ def test_copy_range_then_source_delete():
         assert b.read(0, len(data)) == data  # copy keeps the chunks alive
         b.truncate(0)
         b.close()
+
+
+def test_punch_hole_subchunk_at_zero():
+    # regression (hypothesis-found): aligned sub-chunk punch at offset 0 was
+    # skipped by a bogus head/tail dedup guard
+    with Engine(HOST) as eng, eng.object() as o:
+        o.write(0, b"A")
+        o.punch_hole(0, 1)
+        assert o.read(0, 1) == b"\x00"
+        data = rand_bytes(1 << 20, seed=77)
+        o.write(0, data)
+        o.punch_hole(0, 1000)
+        assert o.read(0, 2000) == b"\x00" * 1000 + data[1000:2000]
