@@ -2,6 +2,7 @@
 sequences must match a plain bytearray model byte-for-byte (SURVEY §5 unit
 tier, hypothesis)."""
 
+import os
 import pytest
 from hypothesis import HealthCheck, given, settings
 from hypothesis import strategies as st
@@ -75,3 +76,22 @@ def test_object_matches_model(ops, seed):
                     model[dst:dst + got] = model[a:a + got]
         assert o.size == len(model)
         assert o.read(0, len(model) + 10) == bytes(model)
+
+
+@pytest.mark.skipif(not os.environ.get("EDGEHIP_DEEP"),
+                    reason="deep fuzz: set EDGEHIP_DEEP=1 (minutes, not CI-default)")
+@settings(max_examples=400, deadline=None,
+          suppress_health_check=[HealthCheck.too_slow])
+@given(ops=st.lists(
+    st.one_of(
+        st.tuples(st.just("write"), st.integers(0, SPAN), st.integers(1, MB)),
+        st.tuples(st.just("read"), st.integers(0, SPAN + MB), st.integers(1, MB)),
+        st.tuples(st.just("truncate"), st.integers(0, SPAN), st.just(0)),
+        st.tuples(st.just("flush"), st.just(0), st.just(0)),
+        st.tuples(st.just("punch"), st.integers(0, SPAN), st.integers(1, 2 * MB)),
+        st.tuples(st.just("selfcopy"), st.integers(0, SPAN), st.integers(1, 2 * MB)),
+    ), min_size=4, max_size=30), seed=st.integers(0, 2**31))
+def test_object_matches_model_deep(ops, seed):
+    """Long-sequence variant (found the punch-at-0 bug that 25 short examples
+    missed). Run with EDGEHIP_DEEP=1."""
+    test_object_matches_model.hypothesis.inner_test(ops, seed)
