@@ -143,3 +143,77 @@ def test_dedup_off_duplicate_content_spill():
             for o in objs:
                 o.truncate(0)
                 o.close()
+
+
+def test_threaded_random_ops_stress_gpu():
+    """Randomized threaded op mix against the GPU engine with a small arena
+    and spill: write/read/punch/truncate/copy_range per-object models under
+    constant eviction. Seeds in failure messages."""
+    import random
+
+    with Engine("engine=gpu,fp=blake3,compress=lz4,dedup=1,arena=256m,"
+                "spill=2g") as eng:
+        errs = []
+
+        def job(seed):
+            rng = random.Random(seed)
+            try:
+                with eng.object() as o:
+                    model = bytearray()
+                    for _ in range(60):
+                        op = rng.choice(["write", "read", "punch", "trunc",
+                                         "flush", "copy"])
+                        if op == "write":
+                            off = rng.randrange(0, 24 * MB)
+                            n = rng.randrange(1, 2 * MB)
+                            pat = rand_bytes(min(n, 8192), seed=seed ^ n)
+                            pat = (pat * (n // len(pat) + 1))[:n]
+                            o.write(off, pat)
+                            if len(model) < off + n:
+                                model.extend(b"\x00" * (off + n - len(model)))
+                            model[off:off + n] = pat
+                        elif op == "read":
+                            off = rng.randrange(0, 26 * MB)
+                            n = rng.randrange(1, 2 * MB)
+                            assert o.read(off, n) == bytes(model[off:off + n]), \
+                                f"seed={seed}"
+                        elif op == "punch":
+                            off = rng.randrange(0, 24 * MB)
+                            n = rng.randrange(1, 2 * MB)
+                            o.punch_hole(off, n)
+                            end = min(off + n, len(model))
+                            if off < len(model):
+                                model[off:end] = b"\x00" * (end - off)
+                        elif op == "trunc":
+                            sz = rng.randrange(0, 24 * MB)
+                            o.truncate(sz)
+                            if sz <= len(model):
+                                del model[sz:]
+                            else:
+                                model.extend(b"\x00" * (sz - len(model)))
+                        elif op == "flush":
+                            o.flush()
+                        elif op == "copy":
+                            src = rng.randrange(0, 12 * MB)
+                            dst = rng.randrange(0, 24 * MB)
+                            n = rng.randrange(1, 2 * MB)
+                            if abs(dst - src) >= n:
+                                got = o.copy_range(o, src, dst, n)
+                                want = min(n, max(0, len(model) - src)) \
+                                    if src < len(model) else 0
+                                assert got == want, f"seed={seed}"
+                                if got:
+                                    if len(model) < dst + got:
+                                        model.extend(b"\x00" * (dst + got - len(model)))
+                                    model[dst:dst + got] = model[src:src + got]
+                    assert o.read(0, len(model) + 1) == bytes(model), f"seed={seed}"
+                    o.truncate(0)
+            except Exception as e:  # pragma: no cover
+                errs.append(e)
+
+        ts = [threading.Thread(target=job, args=(2000 + j,)) for j in range(6)]
+        for t in ts:
+            t.start()
+        for t in ts:
+            t.join()
+        assert not errs, errs[0]
