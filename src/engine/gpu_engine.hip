@@ -1306,45 +1306,14 @@ St GpuEngine::rs_encode_stripes(const ChunkRef* data, int k, int m,
 
 St GpuEngine::rs_encode_stripe(const ChunkRef* data, int k, int m, uint32_t shard_len,
                                ChunkRef* parity_out) {
+    // Non-standard shapes go straight to the host implementation here —
+    // via the batched entry they would bounce back through this virtual
+    // (base rs_encode_stripes loops rs_encode_stripe) and recurse forever.
     if (shard_len != max_chunk_ || k > kRsMaxK || m > kRsMaxM)
         return Engine::rs_encode_stripe(data, k, m, shard_len, parity_out);
-    RoctxRange rr_rs("edgehip::rs_encode_stripe");
-    const int cdev = (int)(rr_.fetch_add(1) % devs_.size());
-    DevCtx& C = devs_[cdev];
-    std::lock_guard<std::mutex> rg(C.rs_mu);
-    const uint32_t cs = (uint32_t)max_chunk_;
-    for (int j = 0; j < k; j++) {
-        St s = gather_chunk_to_dev(data[j], cdev, C.d_rs + (uint64_t)j * cs);
-        if (s != St::ok) return s;
-    }
-    HIP_CHECK(hipSetDevice(C.hip_dev));
-    HIP_CHECK(hipStreamSynchronize(C.s_read));
-    RsDesc d{};
-    auto M = rs_parity_matrix(k, m);
-    for (int j = 0; j < k; j++) d.in[j] = C.d_rs + (uint64_t)j * cs;
-    for (int i = 0; i < m; i++) d.out[i] = C.d_rs + (uint64_t)(k + i) * cs;
-    for (int i = 0; i < m; i++)
-        for (int j = 0; j < k; j++) d.coef[i][j] = M[(size_t)i * k + j];
-    d.k = k;
-    d.m = m;
-    d.len = cs;
-    launch_gf_matmul(d, C.s_read);
-    HIP_CHECK(hipStreamSynchronize(C.s_read));
-    // D2H parity, then ingest through the normal pipeline (hash/compress/crc)
-    std::vector<uint8_t> host((size_t)m * cs);
-    for (int i = 0; i < m; i++)
-        HIP_CHECK(hipMemcpyAsync(host.data() + (size_t)i * cs,
-                                 C.d_rs + (uint64_t)(k + i) * cs, cs,
-                                 hipMemcpyDeviceToHost, C.s_read));
-    HIP_CHECK(hipStreamSynchronize(C.s_read));
-    std::vector<PutChunk> puts((size_t)m);
-    for (int i = 0; i < m; i++) puts[i] = {host.data() + (size_t)i * cs, cs, nullptr};
-    St s = put_chunks(puts.data(), parity_out, m);
-    if (s != St::ok) return s;
-    for (int i = 0; i < m; i++)
-        parity_out[i].flags |= ChunkRef::kFlagParity | ChunkRef::kFlagRs;
-    metrics().rs_encoded_stripes++;
-    return St::ok;
+    // single-stripe callers (reconstruct-repair, tests) share the batched
+    // path: encode on-device, parity ingested via D2D — no PCIe roundtrip
+    return rs_encode_stripes(data, k, m, shard_len, 1, parity_out);
 }
 
 St GpuEngine::addref_chunk(const ChunkRef& ref) {
