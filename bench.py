@@ -30,7 +30,12 @@ MIB = 1 << 20
 
 
 def build_native():
-    if not (REPO / "bin" / "edgehip").exists() or not (REPO / "lib" / "libedgehip.so").exists():
+    # always run make: incremental no-op when up to date, and a source edit
+    # can never ship behind a stale prebuilt binary.
+    # flock serializes concurrent ranks (torchrun launches N processes).
+    import fcntl
+    with open("/tmp/edgehip_build.lock", "w") as lk:
+        fcntl.flock(lk, fcntl.LOCK_EX)
         subprocess.run(["make", "-j8"], cwd=REPO, check=True, stdout=subprocess.DEVNULL)
 
 

@@ -20,6 +20,7 @@ Journal::~Journal() {
     }
     if (jworker_.joinable()) jworker_.join();
     if (fd_ >= 0) ::close(fd_);
+    if (chunks_dirfd_ >= 0) ::close(chunks_dirfd_);
 }
 
 void Journal::mirror_loop() {
@@ -40,7 +41,15 @@ void Journal::mirror_loop() {
             fp = jq_.front().fp;
             data = jq_.front().data;
         }
-        (void)write_chunk_file(fp, data.data(), (uint32_t)data.size());
+        if (write_chunk_file(fp, data.data(), (uint32_t)data.size()) != St::ok) {
+            // ENOSPC/EIO: the chunk is NOT durable. Mark the journal failed so
+            // sync()/fsync report St::io instead of acknowledging lost data
+            // (ADVICE r1 high #1); the entry is still popped (retrying forever
+            // on a full disk would wedge the bounded queue and every writer).
+            werr_.store(true);
+            LOGE("journal: chunk mirror write failed for %s — journal marked "
+                 "failed, fsync will return EIO", fp.hex().c_str());
+        }
         {
             std::unique_lock<std::mutex> g(jmu_);
             if (!jq_.empty() && jq_.front().id == id) {  // del may have raced
@@ -61,6 +70,13 @@ St Journal::open(const std::string& dir) {
         LOGE("journal: open %s/journal.log failed: %s", dir.c_str(), strerror(errno));
         return St::io;
     }
+    chunks_dirfd_ = ::open((dir + "/chunks").c_str(), O_RDONLY | O_DIRECTORY | O_CLOEXEC);
+    if (chunks_dirfd_ < 0) {
+        LOGE("journal: open %s/chunks dir failed: %s", dir.c_str(), strerror(errno));
+        ::close(fd_);
+        fd_ = -1;
+        return St::io;
+    }
     return St::ok;
 }
 
@@ -79,8 +95,27 @@ St Journal::append(uint8_t type, const std::vector<uint8_t>& body) {
     rec.insert(rec.end(), body.begin(), body.end());
     put32(crc32c(rec.data(), rec.size()));
     std::lock_guard<std::mutex> g(mu_);
-    ssize_t w = ::write(fd_, rec.data(), rec.size());
-    return w == (ssize_t)rec.size() ? St::ok : St::io;
+    // Loop on EINTR/short writes; on failure ftruncate back to the record
+    // start so no torn bytes stay mid-log ahead of later records (a torn
+    // record would make replay silently discard everything after it —
+    // ADVICE r1 medium #2).
+    struct stat st;
+    if (fstat(fd_, &st)) return St::io;
+    const off_t start = st.st_size;  // O_APPEND: writes land at EOF
+    size_t done = 0;
+    while (done < rec.size()) {
+        ssize_t w = ::write(fd_, rec.data() + done, rec.size() - done);
+        if (w < 0) {
+            if (errno == EINTR) continue;
+            LOGE("journal: log write failed at %zu/%zu: %s", done, rec.size(),
+                 strerror(errno));
+            if (::ftruncate(fd_, start))
+                werr_.store(true);  // can't even roll back: journal is toast
+            return St::io;
+        }
+        done += (size_t)w;
+    }
+    return St::ok;
 }
 
 static std::vector<uint8_t> path_body(const std::string& p) {
@@ -109,11 +144,61 @@ St Journal::file_rec(const std::string& p, const std::vector<uint8_t>& manifest)
     return append(2, b);
 }
 
+// two length-prefixed strings -> body {u16 a_len, a, b}
+static std::vector<uint8_t> pair_body(const std::string& a, const std::string& b) {
+    std::vector<uint8_t> v;
+    v.push_back((uint8_t)(a.size() & 0xff));
+    v.push_back((uint8_t)(a.size() >> 8));
+    v.insert(v.end(), a.begin(), a.end());
+    v.insert(v.end(), b.begin(), b.end());
+    return v;
+}
+
+St Journal::symlink_rec(const std::string& target, const std::string& p) {
+    return append(6, pair_body(target, p));
+}
+St Journal::link_rec(const std::string& oldp, const std::string& newp) {
+    return append(7, pair_body(oldp, newp));
+}
+St Journal::xattr_set_rec(const std::string& p, const std::string& name,
+                          const std::string& value) {
+    // {u16 path_len, path, u16 name_len, name, value}
+    std::vector<uint8_t> b = pair_body(p, "");
+    b.push_back((uint8_t)(name.size() & 0xff));
+    b.push_back((uint8_t)(name.size() >> 8));
+    b.insert(b.end(), name.begin(), name.end());
+    b.insert(b.end(), value.begin(), value.end());
+    return append(8, b);
+}
+St Journal::xattr_del_rec(const std::string& p, const std::string& name) {
+    return append(9, pair_body(p, name));
+}
+St Journal::exchange_rec(const std::string& a, const std::string& b) {
+    return append(10, pair_body(a, b));
+}
+
 St Journal::sync() {
     {   // drain the chunk-mirror queue first: fsync means "it is on disk"
         std::unique_lock<std::mutex> g(jmu_);
         jcv_drain_.wait(g, [&] { return jq_.empty(); });
     }
+    if (werr_.load()) return St::io;  // a mirror/log write was lost — say so
+    // fsync chunk files written since the last sync (deferred from the
+    // mirror loop), then the chunks/ dir once for the renames
+    std::vector<std::string> files;
+    {
+        std::lock_guard<std::mutex> g(jmu_);
+        files.swap(pending_fsync_);
+    }
+    for (auto& p : files) {
+        int fd = ::open(p.c_str(), O_RDONLY | O_CLOEXEC);
+        if (fd < 0) continue;  // deleted meanwhile: nothing to make durable
+        int rc = ::fsync(fd);
+        ::close(fd);
+        if (rc) return St::io;
+    }
+    if (!files.empty() && chunks_dirfd_ >= 0 && ::fsync(chunks_dirfd_))
+        return St::io;
     std::lock_guard<std::mutex> g(mu_);
     return ::fdatasync(fd_) == 0 ? St::ok : St::io;
 }
@@ -124,13 +209,31 @@ St Journal::write_chunk_file(const Fp& fp, const uint8_t* data, uint32_t len) {
     std::string tmp = chunk_path(fp) + ".tmp";
     int fd = ::open(tmp.c_str(), O_CREAT | O_WRONLY | O_TRUNC | O_CLOEXEC, 0644);
     if (fd < 0) return St::io;
-    ssize_t w = ::write(fd, data, len);
-    ::close(fd);
-    if (w != (ssize_t)len) {
-        ::unlink(tmp.c_str());
-        return St::io;
+    size_t done = 0;
+    while (done < len) {
+        ssize_t w = ::write(fd, data + done, len - done);
+        if (w < 0) {
+            if (errno == EINTR) continue;
+            ::close(fd);
+            ::unlink(tmp.c_str());
+            return St::io;
+        }
+        done += (size_t)w;
     }
+    ::close(fd);
     if (::rename(tmp.c_str(), chunk_path(fp).c_str())) return St::io;
+    // Durability is deferred to sync(): the file lands unfsynced here (the
+    // mirror loop is the ingest write-behind — an fsync per chunk would
+    // serialize it to disk latency), and sync() fsyncs every file written
+    // since the last sync plus the chunks/ dir before acknowledging
+    // (fsync must not return ok with chunk bytes only in
+    // page cache). A crash between rename and sync can leave a torn chunk
+    // file; restore detects that (size + content-address mismatch) and
+    // treats the chunk as absent — correct, since sync never covered it.
+    {
+        std::lock_guard<std::mutex> g(jmu_);
+        pending_fsync_.push_back(chunk_path(fp));
+    }
     return St::ok;
 }
 
@@ -199,13 +302,28 @@ St Journal::replay(std::vector<Replay>* out) const {
         return (uint32_t)data[at] | (uint32_t)data[at + 1] << 8 |
                (uint32_t)data[at + 2] << 16 | (uint32_t)data[at + 3] << 24;
     };
+    size_t resyncs = 0;
     while (pos + 16 <= data.size()) {
-        if (get32(pos) != kJMagic) break;
+        if (get32(pos) != kJMagic ||
+            12 + (uint64_t)get32(pos + 8) + 4 > data.size() - pos ||
+            crc32c(data.data() + pos, 12 + get32(pos + 8)) !=
+                get32(pos + 12 + get32(pos + 8))) {
+            // Bad magic / length / crc. A torn TAIL (crash mid-append) is
+            // normal and ends replay; mid-log damage must not silently drop
+            // every later fsync-acknowledged record:
+            // scan forward for the next magic and resync.
+            size_t next = pos + 1;
+            while (next + 16 <= data.size() && get32(next) != kJMagic) next++;
+            if (next + 16 > data.size()) break;  // tail — clean stop
+            LOGE("journal: corrupt record at offset %zu, resynced at %zu "
+                 "(records between are lost)", pos, next);
+            resyncs++;
+            pos = next;
+            continue;
+        }
         uint8_t type = data[pos + 4];
         uint32_t blen = get32(pos + 8);
         size_t total = 12 + blen + 4;
-        if (pos + total > data.size()) break;  // torn tail
-        if (crc32c(data.data() + pos, 12 + blen) != get32(pos + 12 + blen)) break;
         const uint8_t* body = data.data() + pos + 12;
         Replay r;
         r.type = type;
@@ -215,18 +333,32 @@ St Journal::replay(std::vector<Replay>* out) const {
             case 5:
                 r.path.assign((const char*)body, blen);
                 break;
-            case 4: {
+            case 4:    // rename   {u16 old_len, old, new}
+            case 6:    // symlink  {u16 target_len, target, path}
+            case 7:    // link     {u16 old_len, old, new}
+            case 9:    // xattr-del{u16 path_len, path, name}
+            case 10: { // exchange {u16 a_len, a, b}
                 uint16_t ol = body[0] | body[1] << 8;
                 if (2u + ol > blen) return St::bad_format;
                 r.path.assign((const char*)body + 2, ol);
                 r.path2.assign((const char*)body + 2 + ol, blen - 2 - ol);
                 break;
             }
-            case 2: {
+            case 2: {  // file {u16 path_len, path, manifest}
                 uint16_t pl = body[0] | body[1] << 8;
                 if (2u + pl > blen) return St::bad_format;
                 r.path.assign((const char*)body + 2, pl);
                 r.manifest.assign(body + 2 + pl, body + blen);
+                break;
+            }
+            case 8: {  // xattr-set {u16 path_len, path, u16 name_len, name, value}
+                uint16_t pl = body[0] | body[1] << 8;
+                if (2u + pl + 2u > blen) return St::bad_format;
+                r.path.assign((const char*)body + 2, pl);
+                uint16_t nl = body[2 + pl] | body[3 + pl] << 8;
+                if (4u + pl + nl > blen) return St::bad_format;
+                r.path2.assign((const char*)body + 4 + pl, nl);
+                r.manifest.assign(body + 4 + pl + nl, body + blen);
                 break;
             }
             default:
@@ -235,6 +367,7 @@ St Journal::replay(std::vector<Replay>* out) const {
         out->push_back(std::move(r));
         pos += total;
     }
+    if (resyncs) metrics().journal_resyncs += resyncs;
     return St::ok;
 }
 

@@ -709,6 +709,8 @@ St Object::truncate(uint64_t size) {
         cum_.clear();
         tail_.clear();
         rs_dirty_.clear();
+        cdc_stash_.clear();   // a gapped stream must not wedge the object
+        cdc_stash_bytes_ = 0; // past its reset
         size_ = 0;
         return St::ok;
     }
@@ -762,7 +764,18 @@ St Object::truncate(uint64_t size) {
 St Object::commit(std::vector<uint8_t>* manifest_bytes) {
     std::unique_lock<std::mutex> g(mu_);
     St s = flush_locked(g);
-    if (s != St::ok) return s;
+    if (s != St::ok) {
+        if (conf_->cdc && !cdc_stash_.empty()) {
+            // unfilled gap in the append stream: surface the error ONCE,
+            // then drop the stranded pieces so the object recovers instead
+            // of failing every later flush forever
+            LOGW("obj: cdc commit dropping %llu stashed bytes past a gap",
+                 (unsigned long long)cdc_stash_bytes_);
+            cdc_stash_.clear();
+            cdc_stash_bytes_ = 0;
+        }
+        return s;
+    }
     mf_.logical_size = size_;
     if (manifest_bytes) mf_.encode(manifest_bytes);
     // C4: publish the freshly committed refs so every GPU learns the new
@@ -815,12 +828,21 @@ St Object::cdc_write_locked(uint64_t off, const uint8_t* data, uint64_t len) {
     size_ += len;
     metrics().write_ops++;
     metrics().write_bytes += len;
-    // drain any stashed pieces the new bytes made contiguous
-    while (!cdc_stash_.empty() && cdc_stash_.begin()->first == size_) {
+    // drain any stashed pieces the new bytes made contiguous; a piece whose
+    // offset the stream has already passed (an in-order write overlapped it)
+    // contributes only its non-overlapped suffix — an exact-match-only drain
+    // left such entries permanently stuck and wedged every later flush
+    // 
+    while (!cdc_stash_.empty() && cdc_stash_.begin()->first <= size_) {
         auto it = cdc_stash_.begin();
-        tail_.insert(tail_.end(), it->second.begin(), it->second.end());
-        size_ += it->second.size();
-        cdc_stash_bytes_ -= it->second.size();
+        uint64_t soff = it->first;
+        uint64_t slen = it->second.size();
+        if (soff + slen > size_) {
+            uint64_t skip = size_ - soff;  // 0 when exactly contiguous
+            tail_.insert(tail_.end(), it->second.begin() + skip, it->second.end());
+            size_ += slen - skip;
+        }  // else: entirely covered by in-order bytes — drop it
+        cdc_stash_bytes_ -= slen;
         cdc_stash_.erase(it);
     }
     // cut complete chunks once enough stream is buffered (keep up to one

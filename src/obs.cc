@@ -97,7 +97,8 @@ std::string Metrics::render() const {
              "rs_encoded_stripes %llu\nrs_reconstructs %llu\n"
              "cache_hits %llu\ncache_misses %llu\n"
              "h2d_bytes %llu\nd2h_bytes %llu\n"
-             "spilled_chunks %llu\nspill_bytes %llu\n",
+             "spilled_chunks %llu\nspill_bytes %llu\n"
+             "journal_resyncs %llu\n",
              (unsigned long long)fuse_reqs.load(),
              (unsigned long long)read_ops.load(), (unsigned long long)read_bytes.load(),
              (unsigned long long)write_ops.load(), (unsigned long long)write_bytes.load(),
@@ -107,7 +108,8 @@ std::string Metrics::render() const {
              (unsigned long long)rs_encoded_stripes.load(), (unsigned long long)rs_reconstructs.load(),
              (unsigned long long)cache_hits.load(), (unsigned long long)cache_misses.load(),
              (unsigned long long)h2d_bytes.load(), (unsigned long long)d2h_bytes.load(),
-             (unsigned long long)spilled_chunks.load(), (unsigned long long)spill_bytes.load());
+             (unsigned long long)spilled_chunks.load(), (unsigned long long)spill_bytes.load(),
+             (unsigned long long)journal_resyncs.load());
     std::string out(buf);
     out += read_lat.render("read_latency");
     out += write_lat.render("write_latency");

@@ -35,6 +35,7 @@ struct Metrics {
     std::atomic<uint64_t> cache_hits{0}, cache_misses{0};
     std::atomic<uint64_t> h2d_bytes{0}, d2h_bytes{0};
     std::atomic<uint64_t> spilled_chunks{0}, spill_bytes{0};
+    std::atomic<uint64_t> journal_resyncs{0};  // mid-log corruption skips on replay
     LatHist read_lat, write_lat;  // per-FUSE-request service time
 
     std::string render() const;  // text for the in-mount .edgehip/stats file

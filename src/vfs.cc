@@ -143,69 +143,81 @@ St Vfs::restore() {
                 p->nlink--;
                 break;
             }
-            case 4: {  // rename (also encodes symlink/link via \x01 prefix)
+            case 6: {  // symlink {target, path}
+                const std::string& target = r.path;
+                if (resolve_parent_locked(r.path2, &parent, &leaf) != St::ok) break;
+                Inode* p2 = get(parent);
+                if (!p2 || p2->children.count(leaf)) break;
+                auto in = std::make_unique<Inode>();
+                in->ino = alloc_ino();
+                in->mode = S_IFLNK | 0777;
+                in->nlookup = 0;
+                in->parent = parent;
+                in->name = leaf;
+                in->symlink_target = target;
+                in->atime = in->mtime = in->ctime = (int64_t)wall_ns();
+                p2->children[leaf] = in->ino;
+                inodes_.emplace(in->ino, std::move(in));
+                break;
+            }
+            case 7: {  // link {oldpath, newpath}
+                uint64_t op2;
+                std::string oleaf;
+                if (resolve_parent_locked(r.path, &op2, &oleaf) != St::ok) break;
+                Inode* opn = get(op2);
+                if (!opn) break;
+                auto oit = opn->children.find(oleaf);
+                if (oit == opn->children.end()) break;
+                if (resolve_parent_locked(r.path2, &parent, &leaf) != St::ok) break;
+                Inode* p2 = get(parent);
+                if (!p2 || p2->children.count(leaf)) break;
+                p2->children[leaf] = oit->second;
+                Inode* t = get(oit->second);
+                if (t) t->nlink++;
+                break;
+            }
+            case 8:    // xattr-set {path, name, value(manifest field)}
+            case 9: {  // xattr-del {path, name}
+                Inode* in = nullptr;
+                if (r.path == "/") {
+                    in = get(1);
+                } else if (resolve_parent_locked(r.path, &parent, &leaf) == St::ok) {
+                    Inode* p2 = get(parent);
+                    if (p2) {
+                        auto cit = p2->children.find(leaf);
+                        if (cit != p2->children.end()) in = get(cit->second);
+                    }
+                }
+                if (in) {
+                    if (r.type == 9)
+                        in->xattrs.erase(r.path2);
+                    else
+                        in->xattrs[r.path2] =
+                            std::string(r.manifest.begin(), r.manifest.end());
+                }
+                break;
+            }
+            case 10: {  // exchange {path_a, path_b}: swap the two entries
+                uint64_t pb;
+                std::string bleaf;
+                if (resolve_parent_locked(r.path, &parent, &leaf) != St::ok) break;
+                if (resolve_parent_locked(r.path2, &pb, &bleaf) != St::ok) break;
+                Inode* pa = get(parent);
+                Inode* pbn = get(pb);
+                if (!pa || !pbn) break;
+                auto ia = pa->children.find(leaf);
+                auto ib = pbn->children.find(bleaf);
+                if (ia == pa->children.end() || ib == pbn->children.end()) break;
+                std::swap(ia->second, ib->second);
+                Inode* a = get(ia->second);
+                Inode* b = get(ib->second);
+                if (a) { a->parent = parent; a->name = leaf; }
+                if (b) { b->parent = pb; b->name = bleaf; }
+                break;
+            }
+            case 4: {  // rename {old, new}
                 uint64_t np;
                 std::string nleaf;
-                if (r.path.rfind("\x01symlink:", 0) == 0) {
-                    std::string target = r.path.substr(9);
-                    if (resolve_parent_locked(r.path2, &parent, &leaf) != St::ok) break;
-                    Inode* p2 = get(parent);
-                    if (!p2 || p2->children.count(leaf)) break;
-                    auto in = std::make_unique<Inode>();
-                    in->ino = alloc_ino();
-                    in->mode = S_IFLNK | 0777;
-                    in->nlookup = 0;
-                    in->parent = parent;
-                    in->name = leaf;
-                    in->symlink_target = target;
-                    in->atime = in->mtime = in->ctime = (int64_t)wall_ns();
-                    p2->children[leaf] = in->ino;
-                    inodes_.emplace(in->ino, std::move(in));
-                    break;
-                }
-                if (r.path.rfind("\x01link:", 0) == 0) {
-                    std::string oldp = r.path.substr(6);
-                    uint64_t op2;
-                    std::string oleaf;
-                    if (resolve_parent_locked(oldp, &op2, &oleaf) != St::ok) break;
-                    Inode* opn = get(op2);
-                    if (!opn) break;
-                    auto oit = opn->children.find(oleaf);
-                    if (oit == opn->children.end()) break;
-                    if (resolve_parent_locked(r.path2, &parent, &leaf) != St::ok) break;
-                    Inode* p2 = get(parent);
-                    if (!p2 || p2->children.count(leaf)) break;
-                    p2->children[leaf] = oit->second;
-                    Inode* t = get(oit->second);
-                    if (t) t->nlink++;
-                    break;
-                }
-                if (r.path.rfind("\x01xattr:", 0) == 0 ||
-                    r.path.rfind("\x01xattr-:", 0) == 0) {
-                    bool rm = r.path[6] == '-';
-                    std::string body = r.path.substr(rm ? 8 : 7);
-                    Inode* in = nullptr;
-                    if (r.path2 == "/") {
-                        in = get(1);
-                    } else if (resolve_parent_locked(r.path2, &parent, &leaf) ==
-                               St::ok) {
-                        Inode* p2 = get(parent);
-                        if (p2) {
-                            auto cit = p2->children.find(leaf);
-                            if (cit != p2->children.end()) in = get(cit->second);
-                        }
-                    }
-                    if (in) {
-                        if (rm) {
-                            in->xattrs.erase(body);
-                        } else {
-                            size_t nul = body.find('\0');
-                            if (nul != std::string::npos)
-                                in->xattrs[body.substr(0, nul)] = body.substr(nul + 1);
-                        }
-                    }
-                    break;
-                }
                 if (resolve_parent_locked(r.path, &parent, &leaf) != St::ok) break;
                 if (resolve_parent_locked(r.path2, &np, &nleaf) != St::ok) break;
                 Inode* p = get(parent);
@@ -458,9 +470,7 @@ St Vfs::setxattr(uint64_t ino, const std::string& name, const std::string& value
     if ((flags & 2 /*XATTR_REPLACE*/) && !has) return St::not_found;
     in->xattrs[name] = value;
     in->ctime = (int64_t)wall_ns();
-    if (journal_)
-        journal_->rename_rec("\x01xattr:" + name + std::string(1, '\0') + value,
-                             path_locked(ino));
+    if (journal_) journal_->xattr_set_rec(path_locked(ino), name, value);
     return St::ok;
 }
 
@@ -492,7 +502,7 @@ St Vfs::removexattr(uint64_t ino, const std::string& name) {
     if (!in) return St::not_found;
     if (!in->xattrs.erase(name)) return St::not_found;
     in->ctime = (int64_t)wall_ns();
-    if (journal_) journal_->rename_rec("\x01xattr-:" + name, path_locked(ino));
+    if (journal_) journal_->xattr_del_rec(path_locked(ino), name);
     return St::ok;
 }
 
@@ -612,12 +622,10 @@ St Vfs::exchange(uint64_t parent, const std::string& name, uint64_t newparent,
     Inode* b = get(it2->second);
     if (a) { a->parent = parent; a->name = name; a->ctime = (int64_t)wall_ns(); }
     if (b) { b->parent = newparent; b->name = newname; b->ctime = (int64_t)wall_ns(); }
-    if (journal_) {  // two cross renames reproduce the swap on replay via a temp name
+    if (journal_) {  // dedicated EXCHANGE record (no reserved temp names)
         std::string pa = path_locked(parent), pb = path_locked(newparent);
         auto j = [&](const std::string& x) { return x == "/" ? std::string() : x; };
-        journal_->rename_rec(j(pa) + "/" + name, j(pb) + "/.eh_xchg_tmp");
-        journal_->rename_rec(j(pb) + "/" + newname, j(pa) + "/" + name);
-        journal_->rename_rec(j(pb) + "/.eh_xchg_tmp", j(pb) + "/" + newname);
+        journal_->exchange_rec(j(pa) + "/" + name, j(pb) + "/" + newname);
     }
     return St::ok;
 }
@@ -644,7 +652,7 @@ St Vfs::symlink(uint64_t parent, const std::string& name, const std::string& tar
     p->mtime = p->ctime = in->ctime;
     inodes_.emplace(raw->ino, std::move(in));
     if (journal_ && journal_->enabled())
-        journal_->rename_rec("\x01symlink:" + target, path_locked(raw->ino));
+        journal_->symlink_rec(target, path_locked(raw->ino));
     *out = raw;
     return St::ok;
 }
@@ -670,9 +678,9 @@ St Vfs::link(uint64_t ino, uint64_t newparent, const std::string& newname, Inode
     in->nlookup++;
     np->mtime = np->ctime = (int64_t)wall_ns();
     if (journal_ && journal_->enabled())
-        journal_->rename_rec("\x01link:" + path_locked(ino), path_locked(newparent) == "/"
-                                                                 ? "/" + newname
-                                                                 : path_locked(newparent) + "/" + newname);
+        journal_->link_rec(path_locked(ino), path_locked(newparent) == "/"
+                                                 ? "/" + newname
+                                                 : path_locked(newparent) + "/" + newname);
     *out = in;
     return St::ok;
 }
