@@ -64,6 +64,14 @@ class EhLib:
         so.eh_object_punch_hole.argtypes = [ctypes.c_void_p, ctypes.c_uint64,
                                             ctypes.c_uint64]
         so.eh_object_punch_hole.restype = ctypes.c_int
+        so.eh_engine_ndev.argtypes = [ctypes.c_void_p]
+        so.eh_engine_ndev.restype = ctypes.c_int
+        so.eh_engine_chunk_device.argtypes = [ctypes.c_void_p, ctypes.c_void_p]
+        so.eh_engine_chunk_device.restype = ctypes.c_int
+        so.eh_engine_mark_device_failed.argtypes = [ctypes.c_void_p, ctypes.c_int]
+        so.eh_engine_mark_device_failed.restype = ctypes.c_int
+        so.eh_metric.argtypes = [ctypes.c_char_p]
+        so.eh_metric.restype = ctypes.c_uint64
 
     # ---- convenience wrappers ----
     def blake3(self, data: bytes) -> bytes:
@@ -146,6 +154,12 @@ def get_lib() -> EhLib:
     return _lib
 
 
+def metric(name: str) -> int:
+    """Global metrics counter by name (e.g. dedup_hits); -1 if unknown."""
+    v = get_lib()._so.eh_metric(name.encode())
+    return -1 if v == (1 << 64) - 1 else v
+
+
 class Engine:
     """Engine+Object harness over ctypes (tests only)."""
 
@@ -177,6 +191,19 @@ class Engine:
     @property
     def is_gpu(self):
         return bool(self.lib._so.eh_engine_is_gpu(self.h))
+
+    @property
+    def ndev(self):
+        return self.lib._so.eh_engine_ndev(self.h)
+
+    def chunk_device(self, fp: bytes) -> int:
+        """Device slot storing this chunk (-2 host/spill tier, -1 absent)."""
+        return self.lib._so.eh_engine_chunk_device(self.h, fp)
+
+    def mark_device_failed(self, dev: int) -> None:
+        rc = self.lib._so.eh_engine_mark_device_failed(self.h, dev)
+        if rc != 0:
+            raise OSError(rc, f"mark_device_failed({dev}) rc={rc}")
 
     def object(self):
         return Obj(self)

@@ -55,6 +55,12 @@ int64_t eh_object_copy_range(void* dst_obj, void* src_obj, uint64_t src_off,
 /* ---- GPU kernel test hooks (tests/kernels) ---- */
 int eh_gpu_available(void);
 
+/* multi-device topology (striped store) */
+int eh_engine_ndev(void* engine);
+int eh_engine_chunk_device(void* engine, const uint8_t fp32[32]); /* -2 host, -1 absent */
+int eh_engine_mark_device_failed(void* engine, int dev);
+uint64_t eh_metric(const char* name); /* global counter, (uint64_t)-1 unknown */
+
 #ifdef __cplusplus
 }
 #endif

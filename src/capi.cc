@@ -148,3 +148,37 @@ extern "C" int eh_engine_drop_chunk(void* e, const uint8_t* fp32) {
     std::memcpy(f.b.data(), fp32, 32);
     return (int)((EhEngine*)e)->eng->debug_drop_chunk(f);
 }
+
+// ---- multi-device topology (tests/multigpu + bench striped mode) ----
+extern "C" int eh_engine_ndev(void* e) { return ((EhEngine*)e)->eng->ndev(); }
+// >=0 device slot; -2 host/spill tier; -1 absent
+extern "C" int eh_engine_chunk_device(void* e, const uint8_t* fp32) {
+    Fp f;
+    std::memcpy(f.b.data(), fp32, 32);
+    return ((EhEngine*)e)->eng->chunk_device(f);
+}
+extern "C" int eh_engine_mark_device_failed(void* e, int dev) {
+    return (int)((EhEngine*)e)->eng->mark_device_failed(dev);
+}
+
+// global metrics counter by name (bench honesty: exports dedup_hits etc.)
+extern "C" uint64_t eh_metric(const char* name) {
+    Metrics& m = metrics();
+    std::string n(name ? name : "");
+    if (n == "dedup_hits") return m.dedup_hits.load();
+    if (n == "dedup_bytes_saved") return m.dedup_bytes_saved.load();
+    if (n == "chunks_put") return m.chunks_put.load();
+    if (n == "chunks_get") return m.chunks_get.load();
+    if (n == "cache_hits") return m.cache_hits.load();
+    if (n == "cache_misses") return m.cache_misses.load();
+    if (n == "spilled_chunks") return m.spilled_chunks.load();
+    if (n == "spill_bytes") return m.spill_bytes.load();
+    if (n == "rs_encoded_stripes") return m.rs_encoded_stripes.load();
+    if (n == "rs_reconstructs") return m.rs_reconstructs.load();
+    if (n == "h2d_bytes") return m.h2d_bytes.load();
+    if (n == "d2h_bytes") return m.d2h_bytes.load();
+    if (n == "lz4_in_bytes") return m.lz4_in_bytes.load();
+    if (n == "lz4_out_bytes") return m.lz4_out_bytes.load();
+    if (n == "journal_resyncs") return m.journal_resyncs.load();
+    return (uint64_t)-1;
+}

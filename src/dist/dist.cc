@@ -30,7 +30,7 @@ DistCtx::~DistCtx() {
     }
 }
 
-St DistCtx::init_local(int ngpus, int first_dev) {
+St DistCtx::init_local(int ngpus, int first_dev, uint64_t table_slots) {
     ndev_ = ngpus;
     first_dev_ = first_dev;
     if (ngpus <= 1) return St::ok;  // single device: no collectives needed
@@ -38,8 +38,8 @@ St DistCtx::init_local(int ngpus, int first_dev) {
     streams_.assign(ngpus, nullptr);
     d_tables_.assign(ngpus, nullptr);
     d_scratch_.assign(ngpus, nullptr);
-    counts_.assign(ngpus, 0);
-    table_cap_ = kTableSlots;
+    totals_.assign(ngpus, 0);
+    table_cap_ = table_slots ? table_slots : kTableSlots;
     std::vector<int> devs(ngpus);
     for (int i = 0; i < ngpus; i++) devs[i] = first_dev + i;
     // one communicator spanning the node's GPUs (RCCL over xGMI)
@@ -49,7 +49,7 @@ St DistCtx::init_local(int ngpus, int first_dev) {
         hipStream_t s;
         if (hipStreamCreateWithFlags(&s, hipStreamNonBlocking) != hipSuccess) return St::gpu;
         streams_[i] = s;
-        if (hipMalloc(&d_tables_[i], kTableSlots * sizeof(ChunkRef)) != hipSuccess)
+        if (hipMalloc(&d_tables_[i], table_cap_ * sizeof(ChunkRef)) != hipSuccess)
             return St::gpu;
         if (hipMalloc(&d_scratch_[i], 64 * sizeof(ChunkRef) * (ngpus + 1)) != hipSuccess)
             return St::gpu;
@@ -81,17 +81,32 @@ St DistCtx::allgather_manifests(int dev_slot, const ChunkRef* recs, int n, int s
                                  (ncclComm_t)comms_[i], (hipStream_t)streams_[i]));
     }
     NCCL_CHECK(ncclGroupEnd());
-    // append the gathered records into each device's manifest table
+    // Append the gathered records into each device's manifest RING. The
+    // write head wraps at capacity (oldest records overwritten) — totals_
+    // stays monotonic so manifest_count() saturates instead of zeroing.
     for (int i = 0; i < ndev_; i++) {
         if (hipSetDevice(first_dev_ + i) != hipSuccess) return St::gpu;
-        uint64_t take = std::min<uint64_t>((uint64_t)ndev_ * slots,
-                                           table_cap_ - counts_[i]);
-        uint8_t* out = (uint8_t*)d_scratch_[i] + slots * sizeof(ChunkRef);
-        if (hipMemcpyAsync((uint8_t*)d_tables_[i] + counts_[i] * sizeof(ChunkRef), out,
-                           take * sizeof(ChunkRef), hipMemcpyDeviceToDevice,
-                           (hipStream_t)streams_[i]) != hipSuccess)
-            return St::gpu;
-        counts_[i] = (counts_[i] + take) % table_cap_;
+        uint64_t nrec = (uint64_t)ndev_ * slots;
+        const uint8_t* src = (uint8_t*)d_scratch_[i] + slots * sizeof(ChunkRef);
+        uint64_t head = totals_[i] % table_cap_;
+        if (!warned_wrap_ && totals_[i] + nrec > table_cap_) {
+            warned_wrap_ = true;
+            LOGW("dist: manifest table wrapped at %llu records — oldest "
+                 "entries are recycled (ring semantics)",
+                 (unsigned long long)table_cap_);
+        }
+        while (nrec) {
+            uint64_t take = std::min(nrec, table_cap_ - head);
+            if (hipMemcpyAsync((uint8_t*)d_tables_[i] + head * sizeof(ChunkRef),
+                               src, take * sizeof(ChunkRef),
+                               hipMemcpyDeviceToDevice,
+                               (hipStream_t)streams_[i]) != hipSuccess)
+                return St::gpu;
+            src += take * sizeof(ChunkRef);
+            totals_[i] += take;
+            head = (head + take) % table_cap_;
+            nrec -= take;
+        }
     }
     for (int i = 0; i < ndev_; i++) {
         if (hipSetDevice(first_dev_ + i) != hipSuccess) return St::gpu;
@@ -153,7 +168,11 @@ void* DistCtx::manifest_table(int dev_slot) const {
     return dev_slot < (int)d_tables_.size() ? d_tables_[dev_slot] : nullptr;
 }
 uint64_t DistCtx::manifest_count(int dev_slot) const {
-    return dev_slot < (int)counts_.size() ? counts_[dev_slot] : 0;
+    if (dev_slot >= (int)totals_.size()) return 0;
+    return std::min(totals_[dev_slot], table_cap_);
+}
+uint64_t DistCtx::manifest_total(int dev_slot) const {
+    return dev_slot < (int)totals_.size() ? totals_[dev_slot] : 0;
 }
 
 }  // namespace edgehip

@@ -191,6 +191,9 @@ class HostEngine : public Engine {
     uint64_t stored_bytes() const override { return store_.bytes(); }
     uint64_t stored_chunks() const override { return store_.count(); }
     bool is_gpu() const override { return false; }
+    int chunk_device(const Fp& fp) const override {
+        return store_.has(fp) ? kDevHost : kDevAbsent;
+    }
 
   private:
     Conf conf_;
@@ -202,17 +205,20 @@ class HostEngine : public Engine {
 // ---- Engine base: RS via the generic chunk APIs (host GF(256) math) --------
 // Works for any engine; GpuEngine overrides with the on-device kernel path.
 St Engine::rs_encode_stripes(const ChunkRef* data, int k, int m, uint32_t shard_len,
-                             int nstripes, ChunkRef* parity_out) {
+                             int nstripes, ChunkRef* parity_out,
+                             const int16_t* parity_place) {
     for (int s = 0; s < nstripes; s++) {
         St st = rs_encode_stripe(data + (size_t)s * k, k, m, shard_len,
-                                 parity_out + (size_t)s * m);
+                                 parity_out + (size_t)s * m,
+                                 parity_place ? parity_place + (size_t)s * m
+                                              : nullptr);
         if (st != St::ok) return st;
     }
     return St::ok;
 }
 
 St Engine::rs_encode_stripe(const ChunkRef* data, int k, int m, uint32_t shard_len,
-                            ChunkRef* parity_out) {
+                            ChunkRef* parity_out, const int16_t* parity_place) {
     std::vector<std::vector<uint8_t>> shards((size_t)k);
     std::vector<GetChunk> gets;
     for (int j = 0; j < k; j++) {
@@ -231,7 +237,10 @@ St Engine::rs_encode_stripe(const ChunkRef* data, int k, int m, uint32_t shard_l
     for (int i = 0; i < m; i++) out[i] = par[i].data();
     rs_encode(k, m, in.data(), out.data(), shard_len);
     std::vector<PutChunk> puts((size_t)m);
-    for (int i = 0; i < m; i++) puts[i] = {par[i].data(), shard_len, nullptr};
+    for (int i = 0; i < m; i++) {
+        puts[i] = {par[i].data(), shard_len, nullptr};
+        if (parity_place) puts[i].place_dev = parity_place[i];
+    }
     St s = put_chunks(puts.data(), parity_out, m);
     if (s != St::ok) return s;
     for (int i = 0; i < m; i++)

@@ -328,9 +328,40 @@ class GpuEngine : public Engine {
     St release_chunk(const ChunkRef& ref) override;
     St addref_chunk(const ChunkRef& ref) override;
     St rs_encode_stripe(const ChunkRef* data, int k, int m, uint32_t shard_len,
-                        ChunkRef* parity_out) override;
+                        ChunkRef* parity_out,
+                        const int16_t* parity_place = nullptr) override;
     St rs_encode_stripes(const ChunkRef* data, int k, int m, uint32_t shard_len,
-                         int nstripes, ChunkRef* parity_out) override;
+                         int nstripes, ChunkRef* parity_out,
+                         const int16_t* parity_place = nullptr) override;
+
+    int ndev() const override { return (int)devs_.size(); }
+
+    int chunk_device(const Fp& fp) const override {
+        std::lock_guard<std::mutex> g(map_mu_);
+        auto it = map_.find(fp);
+        if (it == map_.end()) return kDevAbsent;
+        return it->second.spilled ? kDevHost : (int)it->second.dev;
+    }
+
+    // SURVEY §6.3 device-level failure: mark the slot; its chunks read as
+    // not_found (the object layer reconstructs via surviving RS shards) and
+    // new placements skip it. The hip context may be dead — nothing here
+    // touches the device.
+    St mark_device_failed(int dev_slot) override {
+        if (dev_slot < 0 || dev_slot >= (int)devs_.size()) return St::inval;
+        uint32_t m = failed_mask_.fetch_or(1u << dev_slot);
+        if (!(m & (1u << dev_slot)))
+            LOGE("engine: device slot %d (hip %d) marked FAILED — reads of its "
+                 "chunks degrade to RS reconstruct, writes re-stripe over %d "
+                 "surviving devices",
+                 dev_slot, devs_[dev_slot].hip_dev,
+                 (int)devs_.size() - __builtin_popcount(m | (1u << dev_slot)));
+        if (failed_mask_.load() == (1u << devs_.size()) - 1) return St::gpu;
+        return St::ok;
+    }
+    bool device_failed(int dev_slot) const override {
+        return dev_slot >= 0 && (failed_mask_.load() >> dev_slot) & 1;
+    }
 
     St cdc_scan(const uint8_t* data, uint64_t len, uint64_t avg,
                 std::vector<uint32_t>* bitmap) override {
@@ -407,6 +438,26 @@ class GpuEngine : public Engine {
     // (EP-style sharding; dedup-consistent because fp is content-derived)
     int owner_of(const Fp& fp) const { return fp.b[4] % (int)devs_.size(); }
 
+    // Remap a desired owner slot away from failed devices: walk forward to
+    // the next surviving slot (deterministic, so concurrent writers agree).
+    int alive_slot(int want) const {
+        uint32_t failed = failed_mask_.load(std::memory_order_relaxed);
+        if (!failed) return want;
+        int n = (int)devs_.size();
+        for (int i = 0; i < n; i++) {
+            int d = (want + i) % n;
+            if (!((failed >> d) & 1)) return d;
+        }
+        return want;  // everything failed: caller will hit the hip error
+    }
+
+    // Effective owner for a new chunk: the stripe-placement hint when given,
+    // fingerprint hash otherwise, always remapped off failed devices.
+    int place_chunk(const Fp& fp, int16_t hint) const {
+        int want = (hint >= 0 && hint < (int)devs_.size()) ? hint : owner_of(fp);
+        return alive_slot(want);
+    }
+
     St evict_for(int dev, uint32_t need_bytes);  // map_mu_ held
     St flush_batch(const PutChunk* put, ChunkRef* refs, int n,
                    bool device_src = false, int force_dev = -1);
@@ -416,6 +467,7 @@ class GpuEngine : public Engine {
     uint64_t max_chunk_ = 0;
     std::vector<DevCtx> devs_;
     std::atomic<uint32_t> rr_{0};  // ingest/compute device rotation
+    std::atomic<uint32_t> failed_mask_{0};  // bit d: device slot d failed
     DistCtx dist_;
     std::atomic<uint64_t> epoch_{0};
 
@@ -450,7 +502,8 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n,
     const bool do_lz4 = conf_.compress == Compress::lz4 && !device_src;
     const bool do_fp = conf_.fp == FpAlgo::blake3;
     const int ingest_dev =
-        force_dev >= 0 ? force_dev : (int)(rr_.fetch_add(1) % devs_.size());
+        force_dev >= 0 ? force_dev
+                       : alive_slot((int)(rr_.fetch_add(1) % devs_.size()));
     DevCtx& D = devs_[ingest_dev];
     // reserve worst-case pending bytes on every device this batch can place
     // chunks on (placement is by fingerprint; reserving the full batch on
@@ -665,7 +718,7 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n,
                 }
             }
             if (!dup) {
-                int odev = owner_of(r.fp);
+                int odev = place_chunk(r.fp, put[i].place_dev);
                 uint64_t aoff;
                 int tries = 0;
                 int waits = 0;
@@ -946,6 +999,9 @@ St GpuEngine::get_chunks(const GetChunk* get, int n) {
             stored = loc.len;
             comp = loc.compressed;
             odev = loc.dev;
+            // a failed device's arena is gone: report the chunk lost so the
+            // object layer reconstructs it from surviving RS shards
+            if (!spilled && device_failed(odev)) return St::not_found;
             if (!spilled) {
                 metrics().cache_hits++;
                 lru_.erase(loc.lru);
@@ -1196,6 +1252,7 @@ St GpuEngine::gather_chunk_to_dev(const ChunkRef& ref, int ddev, uint8_t* dst) {
         stored = it->second.len;
         comp = it->second.compressed;
         odev = it->second.dev;
+        if (!spilled && device_failed(odev)) return St::not_found;
     }
     if (spilled) {
         std::vector<uint8_t> full(ref.logical_len);
@@ -1252,13 +1309,14 @@ St GpuEngine::gather_chunk_to_dev(const ChunkRef& ref, int ddev, uint8_t* dst) {
 // onto one device over xGMI, run the GF(256) kernel, ingest parity chunks.
 St GpuEngine::rs_encode_stripes(const ChunkRef* data, int k, int m,
                                 uint32_t shard_len, int nstripes,
-                                ChunkRef* parity_out) {
+                                ChunkRef* parity_out,
+                                const int16_t* parity_place) {
     if (shard_len != max_chunk_ || k > kRsMaxK || m > kRsMaxM)
         return Engine::rs_encode_stripes(data, k, m, shard_len, nstripes,
-                                         parity_out);
+                                         parity_out, parity_place);
     RoctxRange rr_rs("edgehip::rs_encode_stripes");
     const uint32_t cs = (uint32_t)max_chunk_;
-    const int cdev = (int)(rr_.fetch_add(1) % devs_.size());
+    const int cdev = alive_slot((int)(rr_.fetch_add(1) % devs_.size()));
     DevCtx& C = devs_[cdev];
     auto M = rs_parity_matrix(k, m);
     // pipeline in slabs: encode stripe i, queue its parity D2H on the same
@@ -1289,10 +1347,13 @@ St GpuEngine::rs_encode_stripes(const ChunkRef* data, int k, int m,
             launch_gf_matmul(d, C.s_read);
         }
         HIP_CHECK(hipStreamSynchronize(C.s_read));
-        // parity never leaves the GPU: device-source put (D2D staging)
+        // parity never leaves the GPU: device-source put (D2D staging);
+        // stripe-aware hints place each parity shard on its own device
         std::vector<PutChunk> puts((size_t)ns * m);
-        for (int i = 0; i < ns * m; i++)
+        for (int i = 0; i < ns * m; i++) {
             puts[i] = {C.d_rspar + (uint64_t)i * cs, cs, nullptr};
+            if (parity_place) puts[i].place_dev = parity_place[(size_t)s0 * m + i];
+        }
         St st = flush_batch(puts.data(), parity_out + (size_t)s0 * m, ns * m,
                             /*device_src=*/true, /*force_dev=*/cdev);
         if (st != St::ok) return st;
@@ -1305,15 +1366,16 @@ St GpuEngine::rs_encode_stripes(const ChunkRef* data, int k, int m,
 }
 
 St GpuEngine::rs_encode_stripe(const ChunkRef* data, int k, int m, uint32_t shard_len,
-                               ChunkRef* parity_out) {
+                               ChunkRef* parity_out, const int16_t* parity_place) {
     // Non-standard shapes go straight to the host implementation here —
     // via the batched entry they would bounce back through this virtual
     // (base rs_encode_stripes loops rs_encode_stripe) and recurse forever.
     if (shard_len != max_chunk_ || k > kRsMaxK || m > kRsMaxM)
-        return Engine::rs_encode_stripe(data, k, m, shard_len, parity_out);
+        return Engine::rs_encode_stripe(data, k, m, shard_len, parity_out,
+                                        parity_place);
     // single-stripe callers (reconstruct-repair, tests) share the batched
     // path: encode on-device, parity ingested via D2D — no PCIe roundtrip
-    return rs_encode_stripes(data, k, m, shard_len, 1, parity_out);
+    return rs_encode_stripes(data, k, m, shard_len, 1, parity_out, parity_place);
 }
 
 St GpuEngine::addref_chunk(const ChunkRef& ref) {

@@ -19,7 +19,11 @@
 //     u32 stored_len   stored (possibly compressed) length
 //     u32 flags        bit0 = lz4-compressed; bit1 = rs-striped
 //     u32 crc32c       of the stored bytes (verify-lite on read)
-//     u64 shard_map    RS: 8 × u8 device index per shard (k data + m parity)
+//     u64 shard_map    RS: 8 × u8 ACTUAL device slot per shard (k data then
+//                      m parity; byte j = shard j's device; 0xFE = host/spill
+//                      tier, 0xFF = hole/unknown). Recorded at parity-encode
+//                      time from the store's placement (obj.cc
+//                      rs_update_locked), identical across a stripe's refs.
 //     u64 reserved     = 0
 //   trailer (4 B):
 //     u32 crc32c over header + refs

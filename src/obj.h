@@ -72,6 +72,10 @@ class Object {
     void mark_stripe_dirty(uint64_t chunk_idx);
     St rs_update_locked();
     St rs_reconstruct_chunk_locked(uint64_t idx, uint8_t* chunk_buf);
+    // stripe-aware placement: device slot for shard `shard` of `stripe`
+    // (data shards 0..k-1, parity k..k+m-1); -1 = no hint (fp placement)
+    int16_t shard_place_(uint64_t stripe, int shard) const;
+    int16_t data_place_(uint64_t chunk_idx) const;
 
     Engine* eng_;
     const Conf* conf_;
