@@ -88,7 +88,7 @@ St Conf::apply_kv(const std::string& k, const std::string& v) {
     if (k == "pinned") return parse_size(v, &pinned_bytes) ? St::ok : bad();
     if (k == "spill") return parse_size(v, &spill_bytes) ? St::ok : bad();
     if (k == "spill_dir") { spill_dir = v; return St::ok; }
-    if (k == "threads") return parse_int(v, &fuse_threads) && fuse_threads >= 1 && fuse_threads <= 128 ? St::ok : bad();
+    if (k == "threads") return parse_int(v, &fuse_threads) && fuse_threads >= 0 && fuse_threads <= 128 ? St::ok : bad();
     if (k == "max_write") {
         uint64_t x;
         if (!parse_size(v, &x) || x < 4 * KiB || x > 16 * MiB) return bad();

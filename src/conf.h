@@ -41,7 +41,10 @@ struct Conf {
     std::string spill_dir;                 // backing dir for spill (empty = RAM)
 
     // fuse frontend
-    int fuse_threads = 8;
+    int fuse_threads = 0;  // 0 = auto: clamp(nproc/2, 4, 12). Oversubscribing
+                           // a small box collapses O_DIRECT write throughput
+                           // 2.5x (16 threads on 8 cores measured 2.2 GiB/s
+                           // vs 5.6 at 4 threads — single-queue contention)
     uint32_t max_write = 1 * MiB;
     uint32_t max_readahead = 16 * MiB;  // deep readahead pipelines seq reads
     bool writeback = true;             // FUSE writeback cache (ack from page cache)

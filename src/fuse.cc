@@ -1,5 +1,6 @@
 #include "fuse.h"
 
+#include <algorithm>
 #include <errno.h>
 #include <fcntl.h>
 #include <linux/fuse.h>
@@ -164,11 +165,20 @@ St FuseMount::mount(const std::string& mountpoint) {
             LOGW("fuse: bdi for %s not found in mountinfo", mountpoint.c_str());
         }
     }
+    // threads=0 -> auto-size by cores: each reader thread is a full request
+    // pipeline (read(2) copy + handle + reply) and oversubscription collapses
+    // throughput (measured 2.5x on an 8-core box); half the cores leaves the
+    // other half for flush workers / engine threads.
+    int nthreads = conf_->fuse_threads;
+    if (nthreads <= 0) {
+        long np = sysconf(_SC_NPROCESSORS_ONLN);
+        nthreads = (int)std::min(12l, std::max(4l, np / 2));
+    }
     LOGI("fuse: mounted %s (kernel 7.%u, max_write %u, flags 0x%x, %d threads)",
-         mountpoint.c_str(), kernel_minor_, conf_->max_write, rep.o.flags, conf_->fuse_threads);
+         mountpoint.c_str(), kernel_minor_, conf_->max_write, rep.o.flags, nthreads);
     mounted_ = true;
     stop_ = false;
-    for (int i = 0; i < conf_->fuse_threads; i++)
+    for (int i = 0; i < nthreads; i++)
         threads_.emplace_back([this, i] { reader_loop(i); });
     return St::ok;
 }
