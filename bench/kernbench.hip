@@ -76,7 +76,7 @@ int main(int argc, char** argv) {
     };
 
     uint32_t* d_fpg = nullptr;
-    (void)hipMalloc(&d_fpg, 32ull * 4 * nchunks);
+    (void)hipMalloc(&d_fpg, 32ull * kB3Groups * nchunks);
     bench("blake3_fingerprint", (double)total,
           [&] { launch_blake3_chunks(d_src, d_chunks, d_fps, nchunks, nullptr, d_fpg); });
 

@@ -6,8 +6,17 @@
 
 namespace edgehip {
 
-// slice-by-4 tables built in LDS once per block (4 KiB)
-__global__ __launch_bounds__(256) void k_crc32c_chunks(
+// Grid = (kCrcSlices, nchunks), block = one wave (64): the page-XOR
+// structure makes the whole checksum order-free, so each chunk splits
+// across kCrcSlices workgroups that atomicXor their partial into
+// out[chunk] (zeroed by the launcher). A 1 MiB chunk has 256 pages =
+// 4 slices x 64 threads, exactly one page per lane. r1's one-256-thread-
+// WG-per-chunk shape put 32 WGs on the 256-CU chip and ran at 137 GB/s
+// in-pipeline; 4 x 64 chunks = 256 single-wave WGs.
+constexpr int kCrcSlices = 4;
+constexpr int kCrcBlock = 64;
+
+__global__ __launch_bounds__(kCrcBlock) void k_crc32c_chunks(
     const uint8_t* __restrict__ src, const DevChunk* __restrict__ chunks,
     uint32_t* __restrict__ out /* one per chunk */, int nchunks) {
     (void)nchunks;
@@ -25,11 +34,12 @@ __global__ __launch_bounds__(256) void k_crc32c_chunks(
         __syncthreads();
     }
 
-    const DevChunk c = chunks[blockIdx.x];
+    const DevChunk c = chunks[blockIdx.y];
     const uint8_t* base = src + c.src_off;
     uint32_t npages = (c.len + 4095) / 4096;
     uint32_t acc = 0;
-    for (uint32_t p = threadIdx.x; p < npages; p += blockDim.x) {
+    for (uint32_t p = blockIdx.x * blockDim.x + threadIdx.x; p < npages;
+         p += gridDim.x * blockDim.x) {
         uint32_t off = p * 4096;
         uint32_t len = min(4096u, c.len - off);
         const uint8_t* q = base + off;
@@ -50,24 +60,23 @@ __global__ __launch_bounds__(256) void k_crc32c_chunks(
         for (; i < len; i++) crc = tab[0][(crc ^ q[i]) & 0xff] ^ (crc >> 8);
         acc ^= ~crc;
     }
-    // wave + block XOR-reduce, then one atomic per block
+    // single-wave block: the wave XOR-reduce IS the block reduce
 #pragma unroll
     for (int s = 32; s >= 1; s >>= 1) acc ^= __shfl_xor(acc, s, 64);
-    __shared__ uint32_t wsum[4];
-    int wave = threadIdx.x / 64;
-    if ((threadIdx.x & 63) == 0) wsum[wave] = acc;
-    __syncthreads();
     if (threadIdx.x == 0) {
-        uint32_t r = wsum[0] ^ wsum[1] ^ wsum[2] ^ wsum[3];
-        out[blockIdx.x] = r;
+        if (gridDim.x == 1)
+            out[blockIdx.y] = acc;
+        else if (acc)  // out is pre-zeroed; XOR-combine slice partials
+            atomicXor(out + blockIdx.y, acc);
     }
 }
 
 void launch_crc32c_chunks(const uint8_t* src, const DevChunk* chunks, uint32_t* out,
                           int nchunks, hipStream_t stream) {
     if (nchunks <= 0) return;
-    hipLaunchKernelGGL(k_crc32c_chunks, dim3(nchunks), dim3(256), 0, stream, src, chunks,
-                       out, nchunks);
+    (void)hipMemsetAsync(out, 0, sizeof(uint32_t) * nchunks, stream);
+    hipLaunchKernelGGL(k_crc32c_chunks, dim3(kCrcSlices, nchunks), dim3(kCrcBlock), 0,
+                       stream, src, chunks, out, nchunks);
 }
 
 }  // namespace edgehip

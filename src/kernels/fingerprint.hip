@@ -1,23 +1,25 @@
 // G3: batched BLAKE3-256 chunk fingerprints on gfx950.
-// Grid = (groups_per_chunk, nchunks): each 256-thread workgroup hashes a
-// 256-leaf (256 KiB) aligned subtree — 4 WGs per 1 MiB chunk, so a 32-chunk
-// pipeline batch launches 128 WGs instead of 32 (the one-WG-per-chunk shape
-// filled 1/8 of the 256 CUs and ran at ~254 GB/s in-pipeline vs ~2.5 TB/s
-// at full occupancy — profiles/engine_bench_r01_pipeline.md). A trailing
-// finisher kernel combines the ≤4 subtree CVs (parent chain, ROOT on top).
-// Subtree split correctness: BLAKE3's tree splits at the largest power of
-// two ≤ n leaves, so 256-aligned groups are exact subtrees and the group
-// CVs combine with the same pairwise-with-promotion rule (verified against
-// the official vectors in tests/kernels/test_kernels.py).
+// Grid = (groups_per_chunk, nchunks): each 128-thread workgroup hashes a
+// 128-leaf (128 KiB) aligned subtree — 8 WGs per 1 MiB chunk, so a 64-chunk
+// pipeline batch launches 512 WGs (r1's 4-WG split at 32-chunk batches put
+// 128 WGs on the 256-CU chip and ran at ~254 GB/s in-pipeline vs ~2.2 TB/s
+// standalone — profiles/engine_bench_r01_pipeline.md).
+// A trailing finisher kernel combines the ≤8 subtree CVs (parent chain,
+// ROOT on top). Subtree split correctness: BLAKE3's tree splits at the
+// largest power of two ≤ n leaves, so 128-aligned groups are exact subtrees
+// and the group CVs combine with the same pairwise-with-promotion rule
+// (bit-exact against the official vectors in tests/kernels).
 // Max chunk 1 MiB (1024 leaves; the engine enforces this).
 #include "blake3_dev.h"
 
+#include "launchers.h"
+
 namespace edgehip {
 
-constexpr int kLeavesPerGroup = 256;
-constexpr int kGroupsPerChunk = 4;
+constexpr int kLeavesPerGroup = 128;
+constexpr int kGroupsPerChunk = kB3Groups;  // 8
 
-__global__ __launch_bounds__(256) void k_blake3_chunks(
+__global__ __launch_bounds__(128) void k_blake3_chunks(
     const uint8_t* __restrict__ src, const DevChunk* __restrict__ chunks,
     uint8_t* __restrict__ out_fps /* 32 B per chunk */,
     uint32_t* __restrict__ group_cvs /* 8 u32 per (chunk, group) */) {
@@ -136,8 +138,9 @@ __global__ __launch_bounds__(64) void k_blake3_finish(
 void launch_blake3_chunks(const uint8_t* src, const DevChunk* chunks, uint8_t* out_fps,
                           int nchunks, hipStream_t stream, uint32_t* group_scratch) {
     if (nchunks <= 0) return;
-    hipLaunchKernelGGL(k_blake3_chunks, dim3(kGroupsPerChunk, nchunks), dim3(256), 0,
-                       stream, src, chunks, out_fps, group_scratch);
+    hipLaunchKernelGGL(k_blake3_chunks, dim3(kGroupsPerChunk, nchunks),
+                       dim3(kLeavesPerGroup), 0, stream, src, chunks, out_fps,
+                       group_scratch);
     hipLaunchKernelGGL(k_blake3_finish, dim3((nchunks + 63) / 64), dim3(64), 0, stream,
                        chunks, group_scratch, out_fps, nchunks);
 }

@@ -7,6 +7,8 @@
 
 namespace edgehip {
 
+// group_scratch: 32 B x kB3Groups per chunk (subtree chaining values)
+constexpr int kB3Groups = 8;
 void launch_blake3_chunks(const uint8_t* src, const DevChunk* chunks, uint8_t* out_fps,
                           int nchunks, hipStream_t stream, uint32_t* group_scratch);
 

@@ -12,7 +12,7 @@ namespace edgehip {
 
 // Full chunks accumulated per object before one batched engine call; batching
 // is what gives the GPU engine kernel-launch-amortizing work.
-static constexpr int kPutBatch = 32;
+static constexpr int kPutBatch = 64;  // matches the engine's kMaxBatch
 
 // Background flushers (SURVEY §4.3 "overlap H2D(n+1) with hash/LZ4(n)"):
 // write() hands full-chunk batches here so the writer's memcpy stream
