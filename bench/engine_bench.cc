@@ -1,6 +1,7 @@
 // Native ingest/read micro-bench: the bench.py object-api workload with zero
 // Python in the loop, for rocprofv3 attribution of the engine pipeline
 // (SURVEY.md §7). Usage: engine_bench [gib_per_job [jobs [opts]]]
+#include <atomic>
 #include <chrono>
 #include <cstdint>
 #include <cstdio>
@@ -44,11 +45,21 @@ int main(int argc, char** argv) {
     std::vector<double> wsec(jobs), rsec(jobs);
     std::vector<int> errs(jobs, 0);
 
+    // globally-unique content stamping (bench.py stamp_blocks analog): each
+    // job owns a disjoint ring region and stamps every 1 MiB block with a
+    // fresh counter before writing — zero dedup hits by construction
+    std::atomic<uint64_t> stamp{1};
+    const uint64_t region_units = ring / unit / jobs;
     auto worker = [&](int j) {
         void* o = eh_object_create(eng);
         auto t0 = std::chrono::steady_clock::now();
         for (uint64_t off = 0; off < per_job; off += unit) {
-            uint64_t src = ((off / unit) * jobs + j) % (ring / unit) * unit;
+            uint64_t src = ((uint64_t)j * region_units +
+                            (off / unit) % region_units) * unit;
+            for (uint64_t b = 0; b < unit; b += MB) {
+                uint64_t s = stamp.fetch_add(1);
+                std::memcpy(buf.data() + src + b, &s, 8);
+            }
             if (eh_object_write(o, off, buf.data() + src, unit) != 0) errs[j]++;
         }
         if (eh_object_flush(o) != 0) errs[j]++;

@@ -62,6 +62,7 @@ constexpr uint64_t kOffCopy = 276 * KiB;
 constexpr uint64_t kOffStatus = 280 * KiB;
 constexpr uint64_t kOffCrcDesc = 284 * KiB;   // DevChunk[2*kMaxBatch] = 4 KiB
 constexpr uint64_t kOffCopyDesc = 292 * KiB;  // DevChunk[2*kMaxBatch] = 4 KiB
+constexpr uint64_t kOffPtrDesc = 300 * KiB;   // PtrChunk[kMaxBatch] = 1.5 KiB
 constexpr uint64_t kPinMetaSz = 320 * KiB;
 static_assert(kOffSegMeta + sizeof(ulong2) * kMaxSegs <= kOffFps, "segmeta overflow");
 static_assert(kOffCopyDesc + sizeof(DevChunk) * 2 * kMaxBatch <= kPinMetaSz,
@@ -93,6 +94,7 @@ struct PipeSlot {
             *d_fps = nullptr;
     uint32_t* d_fpg = nullptr;  // blake3 subtree CVs (kB3Groups per chunk)
     DevChunk *d_chunks = nullptr, *d_copy = nullptr;
+    PtrChunk* d_ptrdesc = nullptr;  // k_gather_host descriptors
     ulong2* d_segmeta = nullptr;
     uint32_t *d_segsizes = nullptr, *d_sizes = nullptr, *d_crcs = nullptr;
     uint8_t *pin_in = nullptr, *pin_meta = nullptr;
@@ -256,6 +258,7 @@ class GpuEngine : public Engine {
                 HIP_CHECK(hipMalloc(&S.d_fpg, 32 * kMaxBatch * kB3Groups));
                 HIP_CHECK(hipMalloc(&S.d_crcs, sizeof(uint32_t) * kMaxBatch * 2));
                 HIP_CHECK(hipMalloc(&S.d_copy, sizeof(DevChunk) * kMaxBatch * 4));
+                HIP_CHECK(hipMalloc(&S.d_ptrdesc, sizeof(PtrChunk) * kMaxBatch));
                 HIP_CHECK(hipHostMalloc(&S.pin_in, D.stage_sz, hipHostMallocDefault));
                 HIP_CHECK(hipHostMalloc(&S.pin_meta, kPinMetaSz, hipHostMallocDefault));
             }
@@ -286,6 +289,23 @@ class GpuEngine : public Engine {
             if (dist_.init_local(ndev, conf.gpu_first) != St::ok)
                 throw std::runtime_error("GpuEngine: RCCL communicator init failed");
         }
+        // Probe zero-copy: can kernels read hipHostMalloc'd memory at the
+        // host address? (True on ROCm large-BAR systems — SVA.) When yes,
+        // the ingest pulls pinned dirty buffers with ONE k_gather_host
+        // launch instead of a per-chunk hipMemcpyAsync (whose ~40 us SDMA
+        // setup per 1 MiB was the r1 write wall).
+        {
+            uint8_t* probe = nullptr;
+            void* dp = nullptr;
+            if (hipHostMalloc(&probe, 64, hipHostMallocDefault) == hipSuccess) {
+                host_zero_copy_ =
+                    hipHostGetDevicePointer(&dp, probe, 0) == hipSuccess &&
+                    dp == probe;
+                (void)hipHostFree(probe);
+            }
+            LOGI("GpuEngine: pinned zero-copy gather %s",
+                 host_zero_copy_ ? "enabled" : "unavailable (per-chunk DMA)");
+        }
         LOGI("GpuEngine: %d dev (first %d), arena %.1f GiB/GPU, spill cap %.1f GiB",
              ndev, conf.gpu_first, conf.arena_bytes / (double)GiB,
              conf.spill_bytes / (double)GiB);
@@ -300,7 +320,7 @@ class GpuEngine : public Engine {
                 for (void* p : {(void*)S.d_stage, (void*)S.d_comp, (void*)S.d_segscratch,
                                 (void*)S.d_chunks, (void*)S.d_segmeta, (void*)S.d_segsizes,
                                 (void*)S.d_sizes, (void*)S.d_fps, (void*)S.d_fpg,
-                                (void*)S.d_crcs, (void*)S.d_copy})
+                                (void*)S.d_crcs, (void*)S.d_copy, (void*)S.d_ptrdesc})
                     (void)hipFree(p);
                 (void)hipHostFree(S.pin_in);
                 (void)hipHostFree(S.pin_meta);
@@ -475,6 +495,7 @@ class GpuEngine : public Engine {
     std::vector<DevCtx> devs_;
     std::atomic<uint32_t> rr_{0};  // ingest/compute device rotation
     std::atomic<uint32_t> failed_mask_{0};  // bit d: device slot d failed
+    bool host_zero_copy_ = false;  // kernels can read pinned host memory
     DistCtx dist_;
     std::atomic<uint64_t> epoch_{0};
 
@@ -544,20 +565,32 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n,
     auto* h_chunks = (DevChunk*)(S.pin_meta + kOffChunks);
     auto* h_chunks2 = (DevChunk*)(S.pin_meta + kOffChunks2);
     auto* h_segmeta = (ulong2*)(S.pin_meta + kOffSegMeta);
+    auto* h_ptrdesc = (PtrChunk*)(S.pin_meta + kOffPtrDesc);
     const bool pool_pinned = BufPool::inst().pinned();
     int nsegs = 0;
+    int nptr = 0;
+    uint32_t ptr_maxlen = 0;
     uint64_t off = 0;
     uint64_t staged = 0;
     bool used_s2 = false;
     for (int i = 0; i < n; i++) {
+        const bool pinned_src =
+            pool_pinned &&
+            ((put[i].donate && put[i].donate->p == put[i].data) || put[i].pinned_src);
         if (device_src) {
             // source already in this device's memory (e.g. freshly encoded
             // RS parity): D2D into staging, no PCIe roundtrip
             HIP_CHECK(hipMemcpyAsync(S.d_stage + off, put[i].data, put[i].len,
                                      hipMemcpyDeviceToDevice, S.s));
-        } else if (pool_pinned &&
-            ((put[i].donate && put[i].donate->p == put[i].data) || put[i].pinned_src)) {
-            // dirty buffer is already pinned: DMA straight from it, copies
+        } else if (pinned_src && host_zero_copy_) {
+            // pinned dirty buffer, device-readable at the host address: queue
+            // for the ONE-launch k_gather_host pull (per-chunk hipMemcpyAsync
+            // SDMA setup was the r1 write wall — ~40 us per 1 MiB)
+            h_ptrdesc[nptr] = PtrChunk{put[i].data, off, put[i].len, 0};
+            ptr_maxlen = std::max(ptr_maxlen, put[i].len);
+            nptr++;
+        } else if (pinned_src) {
+            // pinned but not device-mapped: DMA straight from it, copies
             // alternating across both slot streams (2 SDMA queues)
             hipStream_t hs = (i & 1) ? S.s2 : S.s;
             used_s2 |= (i & 1) != 0;
@@ -577,6 +610,11 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n,
                 ulong2{off + (uint64_t)s * kSegSz, std::min(kSegSz, put[i].len - s * kSegSz)};
         nsegs += ns;
         off += round_up(put[i].len, 64);
+    }
+    if (nptr) {
+        HIP_CHECK(hipMemcpyAsync(S.d_ptrdesc, h_ptrdesc, sizeof(PtrChunk) * nptr,
+                                 hipMemcpyHostToDevice, S.s));
+        launch_gather_host(S.d_ptrdesc, S.d_stage, nptr, ptr_maxlen, S.s);
     }
     if (used_s2) {  // kernels on S.s must see the s2 copies
         HIP_CHECK(hipEventRecord(S.ev2, S.s2));

@@ -20,6 +20,14 @@ struct DevChunk {
     uint32_t aux;       // kernel-specific
 };
 
+// pointer-source copy descriptor (k_gather_host: pinned host -> device)
+struct PtrChunk {
+    const void* src;    // device-accessible (pinned) host pointer
+    uint64_t dst_off;   // byte offset into the destination base
+    uint32_t len;
+    uint32_t pad;
+};
+
 #define HIP_CHECK(expr)                                                              \
     do {                                                                             \
         hipError_t _e = (expr);                                                      \

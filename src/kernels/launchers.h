@@ -26,6 +26,10 @@ void launch_crc32c_chunks(const uint8_t* src, const DevChunk* chunks, uint32_t* 
 
 void launch_batch_copy(const uint8_t* src, uint8_t* dst, const DevChunk* descs, int ndescs,
                        hipStream_t stream, uint32_t max_len = 1 << 20);
+// one-launch pull of scattered pinned host buffers into device staging
+// (replaces per-chunk hipMemcpyAsync SDMA setup on the ingest path)
+void launch_gather_host(const PtrChunk* descs, uint8_t* dst, int ndescs,
+                        uint32_t max_len, hipStream_t stream);
 
 void launch_cdc_candidates(const uint8_t* data, uint64_t len, uint64_t avg,
                            uint32_t* bitmap, hipStream_t stream, const uint64_t* gear_host);
