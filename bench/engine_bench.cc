@@ -65,8 +65,12 @@ int main(int argc, char** argv) {
         if (eh_object_flush(o) != 0) errs[j]++;
         auto t1 = std::chrono::steady_clock::now();
         std::vector<uint8_t> rbuf(unit);
+        // pinned destination: direct-DMA reads land at link rate (D2H into
+        // pageable memory degrades to runtime-staged copies)
+        bool pinned = eh_mem_pin(rbuf.data(), unit) == 0;
         for (uint64_t off = 0; off < per_job; off += unit)
             if (eh_object_read(o, off, rbuf.data(), unit) != (int64_t)unit) errs[j]++;
+        if (pinned) eh_mem_unpin(rbuf.data());
         auto t2 = std::chrono::steady_clock::now();
         wsec[j] = std::chrono::duration<double>(t1 - t0).count();
         rsec[j] = std::chrono::duration<double>(t2 - t1).count();

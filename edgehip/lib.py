@@ -72,6 +72,10 @@ class EhLib:
         so.eh_engine_mark_device_failed.restype = ctypes.c_int
         so.eh_metric.argtypes = [ctypes.c_char_p]
         so.eh_metric.restype = ctypes.c_uint64
+        so.eh_mem_pin.argtypes = [ctypes.c_void_p, ctypes.c_size_t]
+        so.eh_mem_pin.restype = ctypes.c_int
+        so.eh_mem_unpin.argtypes = [ctypes.c_void_p]
+        so.eh_mem_unpin.restype = ctypes.c_int
 
     # ---- convenience wrappers ----
     def blake3(self, data: bytes) -> bytes:
@@ -204,6 +208,17 @@ class Engine:
         rc = self.lib._so.eh_engine_mark_device_failed(self.h, dev)
         if rc != 0:
             raise OSError(rc, f"mark_device_failed({dev}) rc={rc}")
+
+    def pin(self, buf, n: int) -> bool:
+        """hipHostRegister a writable buffer (mmap/bytearray) for DMA reads."""
+        mv = memoryview(buf)
+        b = (ctypes.c_ubyte * n).from_buffer(mv.cast("B"))
+        return self.lib._so.eh_mem_pin(b, n) == 0
+
+    def unpin(self, buf) -> None:
+        mv = memoryview(buf)
+        b = (ctypes.c_ubyte * mv.nbytes).from_buffer(mv.cast("B"))
+        self.lib._so.eh_mem_unpin(b)
 
     def object(self):
         return Obj(self)

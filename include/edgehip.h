@@ -54,6 +54,8 @@ int64_t eh_object_copy_range(void* dst_obj, void* src_obj, uint64_t src_off,
 
 /* ---- GPU kernel test hooks (tests/kernels) ---- */
 int eh_gpu_available(void);
+int eh_mem_pin(void* p, size_t len);  /* hipHostRegister caller buffer */
+int eh_mem_unpin(void* p);
 
 /* multi-device topology (striped store) */
 int eh_engine_ndev(void* engine);

@@ -297,3 +297,13 @@ extern "C" int eh_gpu_perm_probe(uint32_t a, uint32_t b, uint32_t* out16) {
         return -1;
     }
 }
+
+// Pin (register) caller-owned host memory so engine D2H/H2D lands as true
+// async DMA instead of runtime-staged copies. Bench/daemon reply buffers
+// use this; no-op failure (rc != 0) on GPU-less boxes.
+extern "C" int eh_mem_pin(void* p, size_t len) {
+    return hipHostRegister(p, len, hipHostRegisterDefault) == hipSuccess ? 0 : -1;
+}
+extern "C" int eh_mem_unpin(void* p) {
+    return hipHostUnregister(p) == hipSuccess ? 0 : -1;
+}

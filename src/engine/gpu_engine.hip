@@ -303,6 +303,8 @@ class GpuEngine : public Engine {
                     dp == probe;
                 (void)hipHostFree(probe);
             }
+            const char* zc = getenv("EDGEHIP_ZC");  // debug toggle: 0 = off
+            if (zc && *zc == '0') host_zero_copy_ = false;
             LOGI("GpuEngine: pinned zero-copy gather %s",
                  host_zero_copy_ ? "enabled" : "unavailable (per-chunk DMA)");
         }
