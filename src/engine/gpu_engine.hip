@@ -148,6 +148,7 @@ struct DevCtx {
         uint8_t* d_rdstage = nullptr;
         DevChunk* d_rchunks = nullptr;
         uint32_t* d_rstatus = nullptr;
+        PtrChunk* d_pdesc = nullptr;  // k_scatter_host run descriptors
         uint8_t* pin_rmeta = nullptr;
         std::mutex mu;
     };
@@ -269,7 +270,9 @@ class GpuEngine : public Engine {
                 HIP_CHECK(hipMalloc(&R.d_rchunks,
                                     sizeof(DevChunk) * (kMaxBatch + kMaxGather)));
                 HIP_CHECK(hipMalloc(&R.d_rstatus, sizeof(uint32_t) * kMaxBatch));
-                HIP_CHECK(hipHostMalloc(&R.pin_rmeta, 64 * KiB, hipHostMallocDefault));
+                HIP_CHECK(hipMalloc(&R.d_pdesc,
+                                    sizeof(PtrChunk) * (kMaxBatch + kMaxGather)));
+                HIP_CHECK(hipHostMalloc(&R.pin_rmeta, 128 * KiB, hipHostMallocDefault));
             }
             HIP_CHECK(hipMalloc(&D.d_rs, (uint64_t)(kRsMaxK + kRsMaxM) * cs));
             HIP_CHECK(hipMalloc(&D.d_rspar, (uint64_t)kMaxBatch * cs));
@@ -335,6 +338,7 @@ class GpuEngine : public Engine {
                 (void)hipFree(R.d_rdstage);
                 (void)hipFree(R.d_rchunks);
                 (void)hipFree(R.d_rstatus);
+                (void)hipFree(R.d_pdesc);
                 (void)hipHostFree(R.pin_rmeta);
                 (void)hipStreamDestroy(R.s);
             }
@@ -1134,7 +1138,7 @@ St GpuEngine::get_chunks(const GetChunk* get, int n) {
             }
         }
         auto* h_dc = (DevChunk*)(R.pin_rmeta);
-        auto* h_status = (uint32_t*)(R.pin_rmeta + 32 * KiB);
+        auto* h_status = (uint32_t*)(R.pin_rmeta + 112 * KiB);
         std::vector<DevReq*> mine;
         for (auto* qp : mine0)
             if (qp->dev == (int)d) mine.push_back(qp);
@@ -1207,15 +1211,34 @@ St GpuEngine::get_chunks(const GetChunk* get, int n) {
             mi = 0;
             // merge arena/blob-contiguous requests into one DMA: per-copy
             // SDMA setup is ~40 us/MiB (profiles/engine_bench_r01) and the
-            // bump allocator makes sequentially-written chunks adjacent
+            // bump allocator makes sequentially-written chunks adjacent.
+            // Direct runs whose user buffer is device-mapped (registered
+            // pinned) go through ONE k_scatter_host launch instead — with
+            // interleaved writers the arena runs fragment to ~1 chunk each
+            // and the per-copy setup became the read wall (r2 profile).
+            auto* h_pd = (PtrChunk*)(R.pin_rmeta + 48 * KiB);
+            int npd = 0;
+            uint32_t pd_maxlen = 0;
+            auto dst_mapped = [&](uint8_t* p) {
+                if (!host_zero_copy_) return false;
+                void* dp = nullptr;
+                return hipHostGetDevicePointer(&dp, p, 0) == hipSuccess && dp == p;
+            };
             const uint8_t* run_src = nullptr;
             uint8_t* run_dst = nullptr;  // direct mode: user buffer
             uint64_t run_out = 0, run_len = 0;
             auto flush_run = [&]() {
                 if (!run_len) return;
-                HIP_CHECK(hipMemcpyAsync(direct ? run_dst : st.blob.p + run_out,
-                                         run_src, run_len, hipMemcpyDeviceToHost,
-                                         R.s));
+                if (direct && dst_mapped(run_dst)) {
+                    h_pd[npd] = PtrChunk{run_dst, (uint64_t)(uintptr_t)run_src,
+                                         (uint32_t)run_len, 0};
+                    pd_maxlen = std::max(pd_maxlen, (uint32_t)run_len);
+                    npd++;
+                } else {
+                    HIP_CHECK(hipMemcpyAsync(direct ? run_dst : st.blob.p + run_out,
+                                             run_src, run_len, hipMemcpyDeviceToHost,
+                                             R.s));
+                }
                 run_len = 0;
             };
             uint64_t gpack = gather_base;
@@ -1248,6 +1271,11 @@ St GpuEngine::get_chunks(const GetChunk* get, int n) {
                 if (!direct && pout != run_out + run_len) flush_run();  // padding gap
             }
             flush_run();
+            if (npd) {
+                HIP_CHECK(hipMemcpyAsync(R.d_pdesc, h_pd, sizeof(PtrChunk) * npd,
+                                         hipMemcpyHostToDevice, R.s));
+                launch_scatter_host(R.d_pdesc, npd, pd_maxlen, R.s);
+            }
             HIP_CHECK(hipStreamSynchronize(R.s));
             metrics().d2h_bytes += pout;
             for (int x = 0; x < mi; x++)

@@ -30,6 +30,10 @@ void launch_batch_copy(const uint8_t* src, uint8_t* dst, const DevChunk* descs, 
 // (replaces per-chunk hipMemcpyAsync SDMA setup on the ingest path)
 void launch_gather_host(const PtrChunk* descs, uint8_t* dst, int ndescs,
                         uint32_t max_len, hipStream_t stream);
+// reads: device bytes -> scattered pinned host destinations, one launch
+// (PtrChunk.src = host dst pointer, PtrChunk.dst_off = abs device src addr)
+void launch_scatter_host(const PtrChunk* descs, int ndescs, uint32_t max_len,
+                         hipStream_t stream);
 
 void launch_cdc_candidates(const uint8_t* data, uint64_t len, uint64_t avg,
                            uint32_t* bitmap, hipStream_t stream, const uint64_t* gear_host);

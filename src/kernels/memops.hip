@@ -75,4 +75,38 @@ void launch_gather_host(const PtrChunk* descs, uint8_t* dst, int ndescs,
                        descs, dst, ndescs);
 }
 
+// Mirror of k_gather_host for reads: push device bytes into scattered PINNED
+// host destinations with one launch (posted PCIe writes run at link rate).
+// Descriptor reuse: PtrChunk.src = host DESTINATION pointer,
+// PtrChunk.dst_off = ABSOLUTE device source address (arena or staging).
+__global__ __launch_bounds__(256) void k_scatter_host(
+    const PtrChunk* __restrict__ descs, int ndescs) {
+    int d = blockIdx.y;
+    if (d >= ndescs) return;
+    const PtrChunk c = descs[d];
+    const uint8_t* s = (const uint8_t*)(uintptr_t)c.dst_off;
+    uint8_t* o = (uint8_t*)c.src;
+    uint32_t len = c.len;
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    uint64_t t = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if ((((uintptr_t)s | (uintptr_t)o) & 15) == 0) {
+        uint32_t vec = len / 16;
+        for (uint64_t v = t; v < vec; v += stride)
+            ((uint4*)o)[v] = ((const uint4*)s)[v];
+        for (uint64_t i = (uint64_t)vec * 16 + t; i < len; i += stride) o[i] = s[i];
+    } else {
+        for (uint64_t i = t; i < len; i += stride) o[i] = s[i];
+    }
+}
+
+void launch_scatter_host(const PtrChunk* descs, int ndescs, uint32_t max_len,
+                         hipStream_t stream) {
+    if (ndescs <= 0) return;
+    uint32_t bx = (max_len + 256 * 512 - 1) / (256 * 512);  // 128 KiB per WG
+    if (bx < 1) bx = 1;
+    if (bx > 16) bx = 16;
+    hipLaunchKernelGGL(k_scatter_host, dim3(bx, ndescs), dim3(256), 0, stream,
+                       descs, ndescs);
+}
+
 }  // namespace edgehip
