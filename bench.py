@@ -81,32 +81,38 @@ class ObjectIo:
     def close(self):
         self.eng.close()
 
-    def run_step(self, step_idx, gib, rand_buf, read_buf, jobs=4, rank=0):
+    def run_step(self, step_idx, gib, rand_buf, read_buf, jobs=4, rank=0,
+                 single_file=False):
         # fio-numjobs analog: `jobs` threads, each streaming its own object
         # (ctypes releases the GIL, so host memcpys and engine batches overlap
         # across jobs exactly like concurrent FUSE worker threads would).
         # Each job owns a DISJOINT ring region and stamps every 1 MiB block
         # unique before writing it — no dedup hits, content incompressible.
+        # single_file: ONE shared object, jobs write offset-partitioned
+        # slices (fio numjobs + offset_increment on one file — the exact
+        # config 3/4 shape: one 64 GiB / 1 TiB object).
         import threading
         unit = 8 * MIB
         per_job = max(1, gib * 1024 // 8 // jobs)
         src = memoryview(rand_buf)
         ring_units = len(rand_buf) // unit // jobs  # units per job region
         errs = []
+        shared = self.eng.object() if single_file else None
 
         phase = [[0.0, 0.0] for _ in range(jobs)]
 
         def job(j):
             try:
                 region = j * ring_units * unit  # this job's exclusive slice
-                o = self.eng.object()
+                o = shared if single_file else self.eng.object()
+                obase = j * per_job * unit if single_file else 0
                 rbuf = None
                 try:
                     t0 = time.perf_counter()
                     for i in range(per_job):
                         off = region + (i % ring_units) * unit
                         stamp_blocks(src, off, unit, rank)
-                        o.write(i * unit, src[off:off + unit])
+                        o.write(obase + i * unit, src[off:off + unit])
                     o.flush()
                     t1 = time.perf_counter()
                     # pinned read buffer: direct-DMA D2H at link rate (into
@@ -116,7 +122,7 @@ class ObjectIo:
                     self.eng.pin(rbuf, unit)
                     total = 0
                     for i in range(per_job):
-                        total += o.read_into(i * unit, unit, rbuf)
+                        total += o.read_into(obase + i * unit, unit, rbuf)
                     if total != per_job * unit:
                         raise RuntimeError("short read")
                     t2 = time.perf_counter()
@@ -125,8 +131,9 @@ class ObjectIo:
                 finally:
                     if rbuf is not None:
                         self.eng.unpin(rbuf)
-                    o.truncate(0)
-                    o.close()
+                    if not single_file:
+                        o.truncate(0)
+                        o.close()
             except Exception as e:  # pragma: no cover
                 errs.append(e)
 
@@ -135,6 +142,9 @@ class ObjectIo:
             t.start()
         for t in ts:
             t.join()
+        if shared is not None:
+            shared.truncate(0)
+            shared.close()
         if errs:
             raise errs[0]
         self.write_secs += max(p[0] for p in phase)
@@ -236,6 +246,10 @@ def main():
     ap.add_argument("--striped", action="store_true",
                     help="single process, ngpus=--gpus, ONE striped store "
                          "(config 4/5 topology) instead of rank-per-GPU dp")
+    ap.add_argument("--single-file", action="store_true",
+                    help="jobs write offset-partitioned slices of ONE object "
+                         "(config 3/4's single 64 GiB / 1 TiB file shape); "
+                         "object-api transport only")
     ap.add_argument("--mnt-base", default="/tmp/edgehip_bench")
     args = ap.parse_args()
 
@@ -329,14 +343,16 @@ def main():
         try:
             for w in range(args.warmup):
                 io.run_step(1000 + w, max(1, args.gib_per_step // 2), rng_buf, read_buf,
-                            jobs=args.jobs, rank=rank)
+                            jobs=args.jobs, rank=rank,
+                            single_file=args.single_file)
             barrier()
             dh0 = metric("dedup_hits")
             t0 = time.perf_counter()
             moved = 0
             for s in range(args.steps):
                 moved += io.run_step(s, args.gib_per_step, rng_buf, read_buf,
-                                     jobs=args.jobs, rank=rank)
+                                     jobs=args.jobs, rank=rank,
+                                     single_file=args.single_file)
             barrier()
             t1 = time.perf_counter()
             dh1 = metric("dedup_hits")

@@ -1,9 +1,16 @@
 #!/usr/bin/env python3
-"""Judged config 5: dedup ingest — N overlapping files (same content with a
-distinct prefix each) through the engine; reports ingest GiB/s and the dedup
-ratio from the fingerprint index."""
+"""Judged config 5: dedup ingest — N overlapping files (identical body,
+distinct 1 MiB head each) through the engine; reports ingest GiB/s and the
+dedup ratio from the fingerprint index.
+
+Honesty (r2): the body is UNIQUE across the whole file length — content at
+offset `o` is a ring block stamped with o's block index, so it is identical
+across files (cross-file dedup, what config 5 measures) but never repeats
+within a file (r1's 192 MiB ring repeated intra-file and inflated the ratio
+to ~200x; the true ratio for 10 overlapping files is ~10x)."""
 
 import argparse
+import struct
 import sys
 import time
 from pathlib import Path
@@ -13,39 +20,50 @@ sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
 from edgehip.lib import Engine  # noqa: E402
 from tests.conftest import rand_bytes  # noqa: E402
 
+MB = 1 << 20
+
 
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--files", type=int, default=10)
-    ap.add_argument("--gib-per-file", type=int, default=2)
+    ap.add_argument("--gib-per-file", type=int, default=16)
     ap.add_argument("--ngpus", type=int, default=1)
+    ap.add_argument("--arena", default="24g")
     args = ap.parse_args()
 
-    mb = 1 << 20
     per = args.gib_per_file << 30
-    base = memoryview(bytearray(rand_bytes(256 * mb, seed=7)))  # shared ring
-    # (memoryview slices pass to ctypes zero-copy; bytes slicing would copy
-    # under the GIL and serialize the concurrent writers)
+    ring = 1 << 30  # 1 GiB base ring; every MiB gets an offset-derived stamp
+    base = bytearray(rand_bytes(ring, seed=7))
 
     import threading
 
     with Engine(f"engine=gpu,ngpus={args.ngpus},fp=blake3,compress=lz4,dedup=1,"
-                "arena=16g,spill=0") as eng:
+                f"arena={args.arena},spill=0") as eng:
         t0 = time.perf_counter()
         objs = [eng.object() for _ in range(args.files)]
         errs = []
 
         def ingest(f):
             # fio-style: all files ingest concurrently (ctypes drops the GIL,
-            # so writer memcpys and engine batches overlap across files)
+            # so writer memcpys and engine batches overlap across files).
+            # Body block at offset o == stamped ring block (o//MiB) — the
+            # SAME bytes for every file, unique within a file. Each writer
+            # stamps into its own 64 MiB scratch (the ring itself is shared).
             try:
                 o = objs[f]
-                # distinct 1 MiB head, identical body => ~100% dedup vs file 0
-                o.write(0, rand_bytes(mb, seed=1000 + f))
-                for off in range(mb, per, 64 * mb):
-                    n = min(64 * mb, per - off)
-                    src_off = (off - mb) % (192 * mb)
-                    o.write(off, base[src_off:src_off + n])
+                o.write(0, rand_bytes(MB, seed=1000 + f))  # distinct head
+                scratch = bytearray(64 * MB)
+                sv = memoryview(scratch)
+                for off in range(MB, per, 64 * MB):
+                    n = min(64 * MB, per - off)
+                    roff = (off - MB) % ring
+                    take = min(n, ring - roff)
+                    sv[0:take] = base[roff:roff + take]
+                    if take < n:
+                        sv[take:n] = base[0:n - take]
+                    for b in range(0, n, MB):
+                        sv[b:b + 8] = struct.pack("<Q", (off + b) // MB)
+                    o.write(off, sv[0:n])
                 o.flush()
             except Exception as e:  # pragma: no cover
                 errs.append(e)
@@ -54,7 +72,9 @@ def main():
         # concurrently and dedup against COMMITTED entries (concurrent
         # identical batches would all publish pending entries and convoy on
         # each other's pipeline syncs — measured 3x slower than seeding)
+        t_seed0 = time.perf_counter()
         ingest(0)
+        t_seed = time.perf_counter() - t_seed0
         ts = [threading.Thread(target=ingest, args=(f,))
               for f in range(1, args.files)]
         for t in ts:
@@ -67,10 +87,12 @@ def main():
         logical = args.files * per
         stored = eng.stored_bytes
         print(f"dedup ingest: {logical / (1 << 30) / dt:.2f} GiB/s logical "
-              f"({dt:.1f}s, {args.files} x {args.gib_per_file} GiB)")
+              f"({dt:.1f}s incl. {t_seed:.1f}s unique seed, "
+              f"{args.files} x {args.gib_per_file} GiB)")
         print(f"stored {stored / (1 << 30):.2f} GiB, dedup ratio "
               f"{logical / max(1, stored):.1f}x, chunks {eng.stored_chunks}")
         for o in objs:
+            o.truncate(0)
             o.close()
 
 
