@@ -50,17 +50,23 @@ def main():
             # SAME bytes for every file, unique within a file. Each writer
             # stamps into its own 64 MiB scratch (the ring itself is shared).
             try:
+                import ctypes
                 o = objs[f]
                 o.write(0, rand_bytes(MB, seed=1000 + f))  # distinct head
                 scratch = bytearray(64 * MB)
                 sv = memoryview(scratch)
+                sb = (ctypes.c_char * len(scratch)).from_buffer(scratch)
+                bb = (ctypes.c_char * ring).from_buffer(base)
                 for off in range(MB, per, 64 * MB):
                     n = min(64 * MB, per - off)
                     roff = (off - MB) % ring
                     take = min(n, ring - roff)
-                    sv[0:take] = base[roff:roff + take]
+                    # ctypes.memmove releases the GIL: 9 writers copy in
+                    # parallel (bytearray slice assignment serialized them
+                    # and made the HARNESS the bottleneck)
+                    ctypes.memmove(sb, ctypes.byref(bb, roff), take)
                     if take < n:
-                        sv[take:n] = base[0:n - take]
+                        ctypes.memmove(ctypes.byref(sb, take), bb, n - take)
                     for b in range(0, n, MB):
                         sv[b:b + 8] = struct.pack("<Q", (off + b) // MB)
                     o.write(off, sv[0:n])
