@@ -62,7 +62,7 @@ Object::~Object() {
 }
 
 void Object::wait_inflight_locked(std::unique_lock<std::mutex>& lk) {
-    flush_cv_.wait(lk, [&] { return inflight_ == 0; });
+    flush_cv_.wait(lk, [&] { return inflight_ == 0 && rs_busy_ == 0; });
 }
 
 // Hand a set of FULL dirty chunks to the background flushers. The task OWNS
@@ -107,6 +107,9 @@ void Object::submit_async_flush_locked(std::vector<uint64_t> idxs) {
                     mark_stripe_dirty(idx);
                     pending_publish_.push_back(refs[i]);
                 }
+                // freshly committed chunks may have completed stripes:
+                // encode parity NOW, while the shards are HBM-resident
+                maybe_async_rs_locked();
             }
             for (auto& it : *items) inflight_view_.erase(it.idx);
             inflight_--;
@@ -142,6 +145,114 @@ int16_t Object::shard_place_(uint64_t stripe, int shard) const {
 int16_t Object::data_place_(uint64_t chunk_idx) const {
     if (conf_->rs_k <= 0) return -1;
     return shard_place_(chunk_idx / conf_->rs_k, (int)(chunk_idx % conf_->rs_k));
+}
+
+// shard_map: the ACTUAL device per shard (manifest.h: 8 x u8, data then
+// parity; 0xFE = host/spill tier, 0xFF = hole/unknown), queried from the
+// store — dedup can pin a shard wherever its first copy landed.
+uint64_t Object::stripe_shard_map(const ChunkRef* data, const ChunkRef* par) const {
+    const int k = conf_->rs_k, m = conf_->rs_m;
+    uint64_t smap = 0;
+    for (int j = 0; j < k + m && j < 8; j++) {
+        const Fp& fp = j < k ? data[j].fp : par[j - k].fp;
+        uint8_t b = 0xFF;
+        if (!fp.is_zero()) {
+            int d = eng_->chunk_device(fp);
+            b = d >= 0 ? (uint8_t)d : d == Engine::kDevHost ? 0xFE : 0xFF;
+        }
+        smap |= (uint64_t)b << (8 * j);
+    }
+    return smap;
+}
+
+// Dispatch stable fully-committed dirty stripes to a background encode (mu_
+// held). Stability: every data chunk committed, none dirty/in-flight, and
+// the stripe not already encoding. Tail / holey stripes wait for flush-time
+// rs_update_locked. Data chunks are addref'd for the encode's lifetime so
+// concurrent overwrite/truncate cannot free them; if the stripe re-dirties
+// while encoding, the fresh parity is discarded (flush re-encodes).
+void Object::maybe_async_rs_locked() {
+    const int k = conf_->rs_k, m = conf_->rs_m;
+    if (k <= 0 || m <= 0 || conf_->cdc) return;
+    constexpr int kMinStripes = 4, kMaxStripes = 32;
+    std::vector<uint64_t> ready;
+    for (auto it = rs_dirty_.begin();
+         it != rs_dirty_.end() && (int)ready.size() < kMaxStripes; ++it) {
+        uint64_t s = *it;
+        if (rs_inflight_.count(s)) continue;
+        bool stable = true;
+        for (int j = 0; j < k && stable; j++) {
+            uint64_t idx = s * k + j;
+            if (idx >= mf_.chunks.size() || mf_.chunks[idx].fp.is_zero() ||
+                dirty_.count(idx) || inflight_view_.count(idx))
+                stable = false;
+        }
+        if (stable) ready.push_back(s);
+    }
+    if ((int)ready.size() < kMinStripes) return;
+    auto batch = std::make_shared<std::vector<uint64_t>>(std::move(ready));
+    auto data = std::make_shared<std::vector<ChunkRef>>(batch->size() * (size_t)k);
+    for (size_t x = 0; x < batch->size(); x++) {
+        uint64_t s = (*batch)[x];
+        rs_dirty_.erase(s);
+        rs_inflight_.insert(s);
+        for (int j = 0; j < k; j++) {
+            ChunkRef& r = (*data)[x * k + j] = mf_.chunks[s * k + j];
+            r.flags |= ChunkRef::kFlagRs;
+            eng_->addref_chunk(r);  // pin across the unlocked encode
+        }
+    }
+    rs_busy_++;
+    FlushPool::inst().submit([this, batch, data] {
+        const int k2 = conf_->rs_k, m2 = conf_->rs_m;
+        const uint32_t cs = (uint32_t)conf_->chunk_size;
+        std::vector<ChunkRef> par(batch->size() * (size_t)m2);
+        std::vector<int16_t> pplace(batch->size() * (size_t)m2);
+        for (size_t x = 0; x < batch->size(); x++)
+            for (int i = 0; i < m2; i++)
+                pplace[x * m2 + i] = shard_place_((*batch)[x], k2 + i);
+        St st = eng_->rs_encode_stripes(data->data(), k2, m2, cs,
+                                        (int)batch->size(), par.data(),
+                                        pplace.data());
+        std::lock_guard<std::mutex> g(mu_);
+        for (size_t x = 0; x < batch->size(); x++) {
+            uint64_t s = (*batch)[x];
+            rs_inflight_.erase(s);
+            bool stale = st != St::ok || rs_dirty_.count(s);
+            for (int j = 0; j < k2 && !stale; j++) {
+                uint64_t idx = s * k2 + j;
+                // overwrite/truncate mid-encode: data no longer matches
+                stale = idx >= mf_.chunks.size() ||
+                        !(mf_.chunks[idx].fp == (*data)[x * k2 + j].fp);
+            }
+            if (stale) {
+                rs_dirty_.insert(s);
+                if (st == St::ok)
+                    for (int i = 0; i < m2; i++) drop_ref(par[x * m2 + i]);
+                continue;
+            }
+            if (mf_.parity.size() < (s + 1) * (uint64_t)m2)
+                mf_.parity.resize((s + 1) * m2);
+            uint64_t smap = stripe_shard_map(data->data() + x * k2,
+                                             par.data() + x * m2);
+            for (int i = 0; i < m2; i++) {
+                drop_ref(mf_.parity[s * m2 + i]);
+                par[x * m2 + i].shard_map = smap;
+                mf_.parity[s * m2 + i] = par[x * m2 + i];
+            }
+            for (int j = 0; j < k2; j++) {
+                uint64_t idx = s * k2 + j;
+                if (idx < mf_.chunks.size() && !mf_.chunks[idx].fp.is_zero()) {
+                    mf_.chunks[idx].flags |= ChunkRef::kFlagRs;
+                    mf_.chunks[idx].shard_map = smap;
+                }
+            }
+        }
+        for (auto& r : *data)
+            if (!r.fp.is_zero()) eng_->release_chunk(r);
+        rs_busy_--;
+        flush_cv_.notify_all();
+    });
 }
 
 void Object::mark_stripe_dirty(uint64_t chunk_idx) {
@@ -327,20 +438,7 @@ St Object::rs_update_locked() {
     for (size_t x = 0; x < ready.size(); x++) {
         uint64_t s = ready[x];
         if (mf_.parity.size() < (s + 1) * m) mf_.parity.resize((s + 1) * m);
-        // shard_map: the ACTUAL device per shard (manifest.h: 8 x u8, data
-        // then parity; 0xFE = host/spill tier, 0xFF = hole/unknown) — the
-        // documented semantics, queried from the store, not the hints
-        // (dedup can pin a shard wherever its first copy landed).
-        uint64_t smap = 0;
-        for (int j = 0; j < k + m && j < 8; j++) {
-            const Fp& fp = j < k ? data[x * k + j].fp : par[x * m + (j - k)].fp;
-            uint8_t b = 0xFF;
-            if (!fp.is_zero()) {
-                int d = eng_->chunk_device(fp);
-                b = d >= 0 ? (uint8_t)d : d == Engine::kDevHost ? 0xFE : 0xFF;
-            }
-            smap |= (uint64_t)b << (8 * j);
-        }
+        uint64_t smap = stripe_shard_map(data.data() + x * k, par.data() + x * m);
         for (int i = 0; i < m; i++) {
             drop_ref(mf_.parity[s * m + i]);
             par[x * m + i].shard_map = smap;

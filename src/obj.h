@@ -72,6 +72,11 @@ class Object {
     void mark_stripe_dirty(uint64_t chunk_idx);
     St rs_update_locked();
     St rs_reconstruct_chunk_locked(uint64_t idx, uint8_t* chunk_buf);
+    // async parity pipeline: encode stable dirty stripes in the background
+    // while their chunks are still HBM-resident (flush-time encode after
+    // spill pulled every shard back from host RAM — the config-4 wall)
+    void maybe_async_rs_locked();
+    uint64_t stripe_shard_map(const ChunkRef* data, const ChunkRef* par) const;
     // stripe-aware placement: device slot for shard `shard` of `stripe`
     // (data shards 0..k-1, parity k..k+m-1); -1 = no hint (fp placement)
     int16_t shard_place_(uint64_t stripe, int shard) const;
@@ -83,6 +88,10 @@ class Object {
     Manifest mf_;                    // committed chunk refs (index = chunk idx)
     std::map<uint64_t, Dirty> dirty_;
     std::set<uint64_t> rs_dirty_;    // stripe indices with stale parity
+    std::set<uint64_t> rs_inflight_; // stripes encoding in the background
+    int rs_busy_ = 0;                // background rs-encode tasks outstanding
+                                     // (separate from inflight_ so writer
+                                     // backpressure ignores parity work)
     int inflight_ = 0;               // background flush batches outstanding
     std::condition_variable flush_cv_;
     St async_err_ = St::ok;
