@@ -18,6 +18,7 @@
 #include <algorithm>
 #include <condition_variable>
 #include <list>
+#include <map>
 #include <shared_mutex>
 #include <thread>
 #include <mutex>
@@ -1403,15 +1404,60 @@ St GpuEngine::rs_encode_stripes(const ChunkRef* data, int k, int m,
         int ns = std::min(slab, nstripes - s0);
         std::lock_guard<std::mutex> rg(C.rs_mu);  // held across the put: it
         HIP_CHECK(hipSetDevice(C.hip_dev));       // reads d_rspar via D2D
+        // Fast path: when every shard of a stripe is arena-resident,
+        // uncompressed and full-length, the GF kernel reads the shards IN
+        // PLACE — local HBM or peer arenas over xGMI (peer access enabled
+        // at init; 7 direct links/GPU so k shards stream from k devices in
+        // parallel). No staging copies, no per-chunk syncs: the gather+sync
+        // per chunk (~0.5 ms/stripe serial) was the async-parity ceiling.
+        // Shared evict locks on every source device pin the arena slots for
+        // the slab; the evictor's exclusive lock waits the few ms.
+        std::map<int, std::shared_lock<std::shared_mutex>> evict_locks;
+        auto pin_dev = [&](int d2) {
+            if (!evict_locks.count(d2))
+                evict_locks.emplace(d2, std::shared_lock<std::shared_mutex>(
+                                            devs_[d2].evict_rw));
+        };
+        bool slow_used = false;
         for (int s = 0; s < ns; s++) {
             const ChunkRef* d0 = data + (size_t)(s0 + s) * k;
-            for (int j = 0; j < k; j++) {
-                St st = gather_chunk_to_dev(d0[j], cdev, C.d_rs + (uint64_t)j * cs);
-                if (st != St::ok) return st;
-            }
-            HIP_CHECK(hipStreamSynchronize(C.s_read));  // gathers may cross streams
             RsDesc d{};
-            for (int j = 0; j < k; j++) d.in[j] = C.d_rs + (uint64_t)j * cs;
+            bool fast = true;
+            int sdev[kRsMaxK];
+            // classify under map_mu_, THEN pin (evict_rw) outside it — lock
+            // order is evict_rw before map_mu_ (DevCtx doc) — then
+            // revalidate: a slot may have spilled between the two.
+            auto classify = [&]() {
+                std::lock_guard<std::mutex> lk(map_mu_);
+                for (int j = 0; j < k; j++) {
+                    if (d0[j].fp.is_zero()) return false;
+                    auto it = map_.find(d0[j].fp);
+                    if (it == map_.end() || it->second.spilled ||
+                        it->second.compressed || it->second.pending ||
+                        it->second.len != cs || device_failed(it->second.dev))
+                        return false;
+                    d.in[j] = devs_[it->second.dev].arena + it->second.off;
+                    sdev[j] = (int)it->second.dev;
+                }
+                return true;
+            };
+            fast = classify();
+            if (fast) {
+                for (int j = 0; j < k; j++) pin_dev(sdev[j]);
+                fast = classify();  // revalidate under the pins
+            }
+            if (!fast) {
+                // drain in-flight fast-path kernels before staging into d_rs
+                if (slow_used == false && s > 0)
+                    HIP_CHECK(hipStreamSynchronize(C.s_read));
+                slow_used = true;
+                for (int j = 0; j < k; j++) {
+                    St st = gather_chunk_to_dev(d0[j], cdev, C.d_rs + (uint64_t)j * cs);
+                    if (st != St::ok) return st;
+                }
+                HIP_CHECK(hipStreamSynchronize(C.s_read));  // cross-stream gathers
+                for (int j = 0; j < k; j++) d.in[j] = C.d_rs + (uint64_t)j * cs;
+            }
             for (int i = 0; i < m; i++)
                 d.out[i] = C.d_rspar + ((uint64_t)s * m + i) * cs;
             for (int i = 0; i < m; i++)
@@ -1420,6 +1466,8 @@ St GpuEngine::rs_encode_stripes(const ChunkRef* data, int k, int m,
             d.m = m;
             d.len = cs;
             launch_gf_matmul(d, C.s_read);
+            // slow path reuses d_rs per stripe: wait out this kernel now
+            if (!fast) HIP_CHECK(hipStreamSynchronize(C.s_read));
         }
         HIP_CHECK(hipStreamSynchronize(C.s_read));
         // parity never leaves the GPU: device-source put (D2D staging);
