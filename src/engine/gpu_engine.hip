@@ -1418,7 +1418,6 @@ St GpuEngine::rs_encode_stripes(const ChunkRef* data, int k, int m,
                 evict_locks.emplace(d2, std::shared_lock<std::shared_mutex>(
                                             devs_[d2].evict_rw));
         };
-        bool slow_used = false;
         for (int s = 0; s < ns; s++) {
             const ChunkRef* d0 = data + (size_t)(s0 + s) * k;
             RsDesc d{};
@@ -1447,10 +1446,13 @@ St GpuEngine::rs_encode_stripes(const ChunkRef* data, int k, int m,
                 fast = classify();  // revalidate under the pins
             }
             if (!fast) {
-                // drain in-flight fast-path kernels before staging into d_rs
-                if (slow_used == false && s > 0)
-                    HIP_CHECK(hipStreamSynchronize(C.s_read));
-                slow_used = true;
+                // Drain in-flight fast kernels and DROP the slab pins first:
+                // gather takes its own shared evict locks (same-thread
+                // shared-lock stacking deadlocks when an exclusive waiter
+                // sits between) and spilled gathers can re-put via the
+                // journal path (eviction needs the exclusive lock).
+                HIP_CHECK(hipStreamSynchronize(C.s_read));
+                evict_locks.clear();
                 for (int j = 0; j < k; j++) {
                     St st = gather_chunk_to_dev(d0[j], cdev, C.d_rs + (uint64_t)j * cs);
                     if (st != St::ok) return st;
@@ -1470,6 +1472,10 @@ St GpuEngine::rs_encode_stripes(const ChunkRef* data, int k, int m,
             if (!fast) HIP_CHECK(hipStreamSynchronize(C.s_read));
         }
         HIP_CHECK(hipStreamSynchronize(C.s_read));
+        // DROP the arena pins before the parity put: the put can trigger
+        // eviction, whose exclusive evict_rw would deadlock against our own
+        // shared locks (hung a 128 GiB spill run for its whole timeout)
+        evict_locks.clear();
         // parity never leaves the GPU: device-source put (D2D staging);
         // stripe-aware hints place each parity shard on its own device
         std::vector<PutChunk> puts((size_t)ns * m);
