@@ -159,6 +159,9 @@ struct DevCtx {
 
     uint8_t* d_rs = nullptr;
     uint8_t* d_rspar = nullptr;  // batched-parity staging (rs_encode_stripes)
+    PtrChunk* d_evdesc = nullptr;   // eviction scatter descriptors
+    PtrChunk* pin_evdesc = nullptr; // pinned staging for them
+    static constexpr int kMaxEvict = 4096;
     unsigned long long* d_index = nullptr;  // G7 fingerprint index (HBM)
     uint64_t index_mask = 0;
     uint64_t stage_sz = 0;
@@ -277,6 +280,10 @@ class GpuEngine : public Engine {
             }
             HIP_CHECK(hipMalloc(&D.d_rs, (uint64_t)(kRsMaxK + kRsMaxM) * cs));
             HIP_CHECK(hipMalloc(&D.d_rspar, (uint64_t)kMaxBatch * cs));
+            HIP_CHECK(hipMalloc(&D.d_evdesc, sizeof(PtrChunk) * DevCtx::kMaxEvict));
+            HIP_CHECK(hipHostMalloc(&D.pin_evdesc,
+                                    sizeof(PtrChunk) * DevCtx::kMaxEvict,
+                                    hipHostMallocDefault));
             // G7: per-device fingerprint index shard in HBM (tags of chunks
             // this device owns); 1/1024 of the arena, power-of-two slots
             {
@@ -344,8 +351,9 @@ class GpuEngine : public Engine {
                 (void)hipStreamDestroy(R.s);
             }
             for (void* p : {(void*)D.arena, (void*)D.d_rs, (void*)D.d_rspar,
-                            (void*)D.d_index})
+                            (void*)D.d_index, (void*)D.d_evdesc})
                 (void)hipFree(p);
+            (void)hipHostFree(D.pin_evdesc);
             (void)hipStreamDestroy(D.s_read);
         }
     }
@@ -965,14 +973,36 @@ St GpuEngine::evict_for(int dev, uint32_t need_bytes) {
                 return St::exists;  // sentinel: transient, retry
         return St::no_space;
     }
-    for (auto& v : victims) {
-        // Pool (pinned) blobs, recycled: fresh-malloc spill paid first-touch
-        // page faults on every eviction (~3 GB/s wall). Pool blocks are
-        // faulted+locked once and reused forever — steady-state spill D2H
-        // runs at PCIe DMA rate. Locked-RAM use is bounded by the spill cap.
-        v.blob = BufPool::inst().get(v.len);
-        HIP_CHECK(hipMemcpyAsync(v.blob.p, O.arena + v.off, v.len, hipMemcpyDeviceToHost,
-                                 O.s_read));
+    // Pool (pinned) blobs, recycled: fresh-malloc spill paid first-touch
+    // page faults on every eviction (~3 GB/s wall). Pool blocks are
+    // faulted+locked once and reused forever — steady-state spill D2H
+    // runs at PCIe DMA rate. Locked-RAM use is bounded by the spill cap.
+    // One k_scatter_host launch pushes the whole pass (per-victim
+    // hipMemcpyAsync SDMA setup was the spill-write wall at 128 GiB scale);
+    // per-victim DMA is the fallback when pinned memory isn't device-mapped.
+    for (auto& v : victims) v.blob = BufPool::inst().get(v.len);
+    if (host_zero_copy_ && BufPool::inst().pinned()) {
+        size_t i = 0;
+        while (i < victims.size()) {
+            int nb = (int)std::min<size_t>(victims.size() - i, DevCtx::kMaxEvict);
+            uint32_t maxlen = 0;
+            for (int x = 0; x < nb; x++) {
+                Victim& v = victims[i + x];
+                O.pin_evdesc[x] = PtrChunk{
+                    v.blob.p, (uint64_t)(uintptr_t)(O.arena + v.off), v.len, 0};
+                maxlen = std::max(maxlen, v.len);
+            }
+            HIP_CHECK(hipMemcpyAsync(O.d_evdesc, O.pin_evdesc,
+                                     sizeof(PtrChunk) * nb, hipMemcpyHostToDevice,
+                                     O.s_read));
+            launch_scatter_host(O.d_evdesc, nb, maxlen, O.s_read);
+            HIP_CHECK(hipStreamSynchronize(O.s_read));
+            i += nb;
+        }
+    } else {
+        for (auto& v : victims)
+            HIP_CHECK(hipMemcpyAsync(v.blob.p, O.arena + v.off, v.len,
+                                     hipMemcpyDeviceToHost, O.s_read));
     }
     HIP_CHECK(hipStreamSynchronize(O.s_read));
     {

@@ -65,14 +65,22 @@ __global__ __launch_bounds__(256) void k_gather_host(
     }
 }
 
+// PCIe copy kernels are latency-bound, not CU-bound: ~256 outstanding WGs
+// saturate the link, and MORE just crowd the compute kernels off the CUs
+// (profiled: 512-WG gathers slowed concurrent blake3/crc/lz4 4x). Budget
+// the x-dim so grid ~= 256 WGs total.
+static uint32_t copy_bx(int ndescs, uint32_t max_len) {
+    uint32_t by_budget = ndescs > 0 ? 256u / (uint32_t)ndescs : 1;
+    uint32_t by_len = (max_len + 256 * 512 - 1) / (256 * 512);  // 128 KiB/WG
+    uint32_t bx = by_budget < by_len ? by_budget : by_len;
+    return bx < 1 ? 1 : bx > 16 ? 16 : bx;
+}
+
 void launch_gather_host(const PtrChunk* descs, uint8_t* dst, int ndescs,
                         uint32_t max_len, hipStream_t stream) {
     if (ndescs <= 0) return;
-    uint32_t bx = (max_len + 256 * 512 - 1) / (256 * 512);  // 128 KiB per WG
-    if (bx < 1) bx = 1;
-    if (bx > 16) bx = 16;
-    hipLaunchKernelGGL(k_gather_host, dim3(bx, ndescs), dim3(256), 0, stream,
-                       descs, dst, ndescs);
+    hipLaunchKernelGGL(k_gather_host, dim3(copy_bx(ndescs, max_len), ndescs),
+                       dim3(256), 0, stream, descs, dst, ndescs);
 }
 
 // Mirror of k_gather_host for reads: push device bytes into scattered PINNED
@@ -102,11 +110,8 @@ __global__ __launch_bounds__(256) void k_scatter_host(
 void launch_scatter_host(const PtrChunk* descs, int ndescs, uint32_t max_len,
                          hipStream_t stream) {
     if (ndescs <= 0) return;
-    uint32_t bx = (max_len + 256 * 512 - 1) / (256 * 512);  // 128 KiB per WG
-    if (bx < 1) bx = 1;
-    if (bx > 16) bx = 16;
-    hipLaunchKernelGGL(k_scatter_host, dim3(bx, ndescs), dim3(256), 0, stream,
-                       descs, ndescs);
+    hipLaunchKernelGGL(k_scatter_host, dim3(copy_bx(ndescs, max_len), ndescs),
+                       dim3(256), 0, stream, descs, ndescs);
 }
 
 }  // namespace edgehip
