@@ -30,6 +30,20 @@ def build_native():
     (VERDICT r1 weak #7); an up-to-date tree is a <1s no-op."""
     subprocess.run(["make", "-j8"], cwd=REPO, check=True,
                    stdout=subprocess.DEVNULL)
+    # GPU warmup with retry: the first HIP init on a freshly-provisioned box
+    # occasionally hiccups (observed ~1-in-5 on the pool); absorb it here so
+    # a cold-start blip can't fail the first collected test under -x
+    import ctypes
+    import time
+    try:
+        so = ctypes.CDLL(str(REPO / "lib" / "libedgehip.so"))
+        so.eh_gpu_available.restype = ctypes.c_int
+        for _ in range(3):
+            if so.eh_gpu_available() > 0:
+                break
+            time.sleep(2)
+    except Exception:
+        pass
     return REPO / "lib" / "libedgehip.so"
 
 
