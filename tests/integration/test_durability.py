@@ -187,3 +187,56 @@ def test_crash_torture_random_kill(tmp_mount, tmp_path):
         with Mount(tmp_mount, opts(jdir)):
             for name, data in synced.items():
                 assert (tmp_mount / name).read_bytes() == data, (it, name)
+
+
+def test_replay_resyncs_past_midlog_corruption(tmp_mount, tmp_path):
+    """Mid-log damage must not silently discard every later fsync-acknowledged
+    record: replay scans for the next record magic and
+    continues; only the damaged record's effect is lost."""
+    jdir = tmp_path / "jrnl"
+    with Mount(tmp_mount, opts(jdir)):
+        for n in ("early.txt", "victim.txt", "late.txt"):
+            (tmp_mount / n).write_text(f"content of {n}\n")
+            fd = os.open(tmp_mount / n, os.O_RDONLY)
+            os.fsync(fd)
+            os.close(fd)
+    log = jdir / "journal.log"
+    raw = bytearray(log.read_bytes())
+    # corrupt one byte inside the record that mentions victim.txt
+    at = raw.find(b"victim.txt")
+    assert at > 0
+    raw[at + 3] ^= 0xFF
+    log.write_bytes(raw)
+    with Mount(tmp_mount, opts(jdir)):
+        names = set(os.listdir(tmp_mount))
+        assert "early.txt" in names
+        assert "late.txt" in names, "records after the corruption were dropped"
+        assert (tmp_mount / "late.txt").read_text() == "content of late.txt\n"
+        stats = (tmp_mount / ".edgehip-stats").read_text()
+        resyncs = [l for l in stats.splitlines() if l.startswith("journal_resyncs")]
+        assert resyncs and int(resyncs[0].split()[1]) >= 1
+
+
+def test_real_file_named_reserved_tmp_survives_exchange(tmp_mount, tmp_path):
+    """r1 journaled RENAME_EXCHANGE via a literal '.eh_xchg_tmp' temp name, so
+    a real file with that name was clobbered on replay.
+    Dedicated EXCHANGE records must leave it untouched."""
+    import ctypes
+    jdir = tmp_path / "jrnl"
+    libc = ctypes.CDLL("libc.so.6", use_errno=True)
+    with Mount(tmp_mount, opts(jdir)):
+        (tmp_mount / ".eh_xchg_tmp").write_text("I am a real file\n")
+        (tmp_mount / "x1").write_text("one\n")
+        (tmp_mount / "x2").write_text("two\n")
+        # renameat2(AT_FDCWD, x1, AT_FDCWD, x2, RENAME_EXCHANGE=2)
+        rc = libc.renameat2(-100, str(tmp_mount / "x1").encode(), -100,
+                            str(tmp_mount / "x2").encode(), 2)
+        assert rc == 0, os.strerror(ctypes.get_errno())
+        for n in ("x1", "x2", ".eh_xchg_tmp"):
+            fd = os.open(tmp_mount / n, os.O_RDONLY)
+            os.fsync(fd)
+            os.close(fd)
+    with Mount(tmp_mount, opts(jdir)):
+        assert (tmp_mount / "x1").read_text() == "two\n"
+        assert (tmp_mount / "x2").read_text() == "one\n"
+        assert (tmp_mount / ".eh_xchg_tmp").read_text() == "I am a real file\n"

@@ -102,3 +102,44 @@ def test_cdc_with_rs():
         o.write(0, data)
         o.commit()
         assert o.read(0, len(data)) == data
+
+
+def test_stash_overlap_drained_not_wedged():
+    """an out-of-order (stashed) piece whose offset an
+    in-order write's length passes must contribute its suffix and drain —
+    the exact-match-only drain left it stuck and wedged every flush."""
+    from edgehip.lib import Engine
+    with Engine(CDC) as eng, eng.object() as o:
+        a = bytes(bytearray(range(256))) * 1024      # 256 KiB
+        b = b"\x5a" * (256 << 10)
+        c = b"\xa5" * (256 << 10)
+        o.write(0, a)                 # in-order
+        o.write(len(a) + (128 << 10), c)   # ahead of stream -> stashed
+        # in-order write OVERLAPS the stash head: covers [len(a), len(a)+256K)
+        o.write(len(a), b)
+        o.flush()                     # must not EINVAL
+        total = len(a) + len(b) + (128 << 10)
+        assert o.size == total
+        assert o.read(0, len(a)) == a
+        assert o.read(len(a), len(b)) == b
+        # the stashed piece's non-overlapped suffix (last 128 KiB of c)
+        assert o.read(len(a) + len(b), 128 << 10) == c[-(128 << 10):]
+        o.truncate(0)
+
+
+def test_stash_gap_recovers_after_commit_error():
+    """A permanently unfilled gap surfaces EINVAL once at flush, then the
+    object recovers (stash dropped) instead of failing forever."""
+    from edgehip.lib import Engine
+    with Engine(CDC) as eng, eng.object() as o:
+        o.write(0, b"x" * (256 << 10))
+        o.write(1 << 20, b"y" * (256 << 10))  # gap at [256K, 1M)
+        try:
+            o.commit()
+            raise AssertionError("commit should fail on the gap")
+        except OSError:
+            pass
+        o.write(256 << 10, b"z" * (256 << 10))  # append continues in-order
+        o.flush()                                # recovered
+        assert o.read(256 << 10, 4) == b"zzzz"
+        o.truncate(0)
