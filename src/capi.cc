@@ -160,6 +160,9 @@ extern "C" int eh_engine_chunk_device(void* e, const uint8_t* fp32) {
 extern "C" int eh_engine_mark_device_failed(void* e, int dev) {
     return (int)((EhEngine*)e)->eng->mark_device_failed(dev);
 }
+extern "C" uint64_t eh_engine_manifest_records(void* e, int dev_slot, int total) {
+    return ((EhEngine*)e)->eng->manifest_records(dev_slot, total != 0);
+}
 
 // global metrics counter by name (bench honesty: exports dedup_hits etc.)
 extern "C" uint64_t eh_metric(const char* name) {

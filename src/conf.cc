@@ -87,6 +87,8 @@ St Conf::apply_kv(const std::string& k, const std::string& v) {
     if (k == "arena") return parse_size(v, &arena_bytes) ? St::ok : bad();
     if (k == "pinned") return parse_size(v, &pinned_bytes) ? St::ok : bad();
     if (k == "spill") return parse_size(v, &spill_bytes) ? St::ok : bad();
+    if (k == "dist_table_slots")
+        return parse_size(v, &dist_table_slots) ? St::ok : bad();
     if (k == "spill_dir") { spill_dir = v; return St::ok; }
     if (k == "threads") return parse_int(v, &fuse_threads) && fuse_threads >= 0 && fuse_threads <= 128 ? St::ok : bad();
     if (k == "max_write") {

@@ -38,6 +38,8 @@ struct Conf {
     uint64_t arena_bytes = 16 * GiB;       // HBM arena per GPU (chunk store+cache)
     uint64_t pinned_bytes = 512 * MiB;     // pinned staging ring per GPU
     uint64_t spill_bytes = 64 * GiB;       // host spill tier cap (0 = off)
+    uint64_t dist_table_slots = 0;         // C4 manifest-ring slots (0 = 1M
+                                           // default; tests shrink to wrap)
     std::string spill_dir;                 // backing dir for spill (empty = RAM)
 
     // fuse frontend

@@ -127,6 +127,13 @@ class Engine {
     // placement skips it. Returns inval on engines without devices.
     virtual St mark_device_failed(int dev_slot) { (void)dev_slot; return St::inval; }
     virtual bool device_failed(int dev_slot) const { (void)dev_slot; return false; }
+    // C4 manifest-ring introspection (multigpu tests): live records
+    // (saturates at capacity) or the monotonic appended total.
+    virtual uint64_t manifest_records(int dev_slot, bool total) const {
+        (void)dev_slot;
+        (void)total;
+        return 0;
+    }
 
     virtual uint64_t stored_bytes() const = 0;
     virtual uint64_t stored_chunks() const = 0;

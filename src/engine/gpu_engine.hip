@@ -297,7 +297,7 @@ class GpuEngine : public Engine {
 
         }
         if (ndev > 1) {
-            if (dist_.init_local(ndev, conf.gpu_first) != St::ok)
+            if (dist_.init_local(ndev, conf.gpu_first, conf.dist_table_slots) != St::ok)
                 throw std::runtime_error("GpuEngine: RCCL communicator init failed");
         }
         // Probe zero-copy: can kernels read hipHostMalloc'd memory at the
@@ -403,6 +403,12 @@ class GpuEngine : public Engine {
     }
     bool device_failed(int dev_slot) const override {
         return dev_slot >= 0 && (failed_mask_.load() >> dev_slot) & 1;
+    }
+
+    uint64_t manifest_records(int dev_slot, bool total) const override {
+        if (!dist_.active()) return 0;
+        return total ? dist_.manifest_total(dev_slot)
+                     : dist_.manifest_count(dev_slot);
     }
 
     St cdc_scan(const uint8_t* data, uint64_t len, uint64_t avg,

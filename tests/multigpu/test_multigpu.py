@@ -202,3 +202,28 @@ def test_striped_spill_roundtrip():
         for off in range(0, len(data), 256 * MIB):
             assert o.read(off, 8 * MIB) == data[off:off + 8 * MIB]
         o.truncate(0)
+
+
+@pytest.mark.skipif("ndev() < 2")
+def test_manifest_ring_wrap_saturates_not_zeroes():
+    """DistCtx manifest table is a RING: with a tiny
+    injected capacity, publishing past it must keep the monotonic total
+    growing while the live count saturates at capacity (the r1 modulo
+    zeroed the count at exactly-full)."""
+    with Engine("engine=gpu,ngpus=2,fp=blake3,compress=lz4,dedup=1,arena=1g,"
+                "spill=0,dist_table_slots=256") as eng:
+        so = eng.lib._so
+        so.eh_engine_manifest_records.argtypes = [ctypes.c_void_p, ctypes.c_int,
+                                                  ctypes.c_int]
+        so.eh_engine_manifest_records.restype = ctypes.c_uint64
+        # each commit publishes in 64-record padded batches over 2 ranks
+        # (128 table records per publish); 4 commits x 1 batch wraps 256
+        for r in range(6):
+            with eng.object() as o:
+                o.write(0, rand_bytes(4 * MIB, seed=100 + r))
+                o.commit()
+                o.truncate(0)
+        total = so.eh_engine_manifest_records(eng.h, 0, 1)
+        live = so.eh_engine_manifest_records(eng.h, 0, 0)
+        assert total > 256, f"expected wrap past capacity, total={total}"
+        assert live == 256, f"live count must saturate at capacity, got {live}"
