@@ -237,6 +237,9 @@ def main():
     ap.add_argument("--spill", default="0", help="host spill tier cap (0=off)")
     ap.add_argument("--rs", default="0", help="erasure coding, e.g. 6+2 (config 4)")
     ap.add_argument("--journal", default="", help="journal dir (durable mode)")
+    ap.add_argument("--prealloc", default="",
+                    help="pre-fault this much pinned pool at init (spill "
+                         "steady state, e.g. 140g) — untimed")
     ap.add_argument("--cdc", type=int, default=0, choices=[0, 1],
                     help="content-defined chunking mode (append-only streams)")
     ap.add_argument("--dedup", type=int, default=1, choices=[0, 1],
@@ -281,6 +284,7 @@ def main():
     opts = (f"engine=gpu,{gpu_part},fp=blake3,compress=lz4,dedup={args.dedup},"
             f"arena={args.arena},spill={args.spill},rs={args.rs},writeback=1"
             + (f",journal={args.journal}_r{local_rank}" if args.journal else "")
+            + (f",prealloc={args.prealloc}" if args.prealloc else "")
             + (",cdc=1,chunk_size=256k" if args.cdc else "")
             if use_gpu else
             "engine=host,fp=blake3,compress=lz4,dedup=1,writeback=1")

@@ -319,6 +319,12 @@ class GpuEngine : public Engine {
             LOGI("GpuEngine: pinned zero-copy gather %s",
                  host_zero_copy_ ? "enabled" : "unavailable (per-chunk DMA)");
         }
+        // pre-fault pinned pool blocks (spill-tier steady state): growing the
+        // pool DURING a spill-heavy run pays hipHostMalloc + first-touch per
+        // 1 MiB block inline with ingest — at 128 GiB scale that was most of
+        // the write wall
+        if (conf.prealloc)
+            BufPool::inst().prealloc(conf.prealloc, (uint32_t)max_chunk_);
         LOGI("GpuEngine: %d dev (first %d), arena %.1f GiB/GPU, spill cap %.1f GiB",
              ndev, conf.gpu_first, conf.arena_bytes / (double)GiB,
              conf.spill_bytes / (double)GiB);
