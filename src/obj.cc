@@ -62,7 +62,11 @@ Object::~Object() {
 }
 
 void Object::wait_inflight_locked(std::unique_lock<std::mutex>& lk) {
-    flush_cv_.wait(lk, [&] { return inflight_ == 0 && rs_busy_ == 0; });
+    // drains background flushes, background parity encodes AND writers in
+    // an unlocked payload copy — callers may then mutate/erase dirty_
+    flush_cv_.wait(lk, [&] {
+        return inflight_ == 0 && rs_busy_ == 0 && write_busy_ == 0;
+    });
 }
 
 // Hand a set of FULL dirty chunks to the background flushers. The task OWNS
@@ -550,6 +554,9 @@ St Object::write(uint64_t off, const uint8_t* data, uint64_t len) {
             // sparse write into a chunk that has committed content underneath
             bool committed = idx < mf_.chunks.size() && !mf_.chunks[idx].fp.is_zero();
             if (committed) {
+                // another writer may be mid-copy into this entry with mu_
+                // dropped; the merge below replaces the blob — wait it out
+                flush_cv_.wait(g, [&] { return it->second.busy == 0; });
                 // merge committed bytes under the already-written ones:
                 Dirty merged;
                 St s = load_chunk_for_rmw(idx, &merged);
@@ -574,7 +581,22 @@ St Object::write(uint64_t off, const uint8_t* data, uint64_t len) {
             }
         }
         Dirty& d = it->second;
-        std::memcpy(d.data.p + o, p, n);
+        // Copy the payload with mu_ DROPPED: concurrent writers to ONE
+        // object (fio numjobs on a single file — config 3/4's shape) were
+        // serialized on this memcpy and capped the single-file write phase
+        // at one core's copy bandwidth. busy pins the entry: no flush /
+        // merge / erase path touches it until the copy lands (all of them
+        // run under mu_ and either check busy or wait write_busy_ == 0).
+        // Map-node references are stable; only erase invalidates them.
+        d.busy++;
+        write_busy_++;
+        uint8_t* dst = d.data.p + o;
+        g.unlock();
+        std::memcpy(dst, p, n);
+        g.lock();
+        d.busy--;
+        write_busy_--;
+        if (d.busy == 0 || write_busy_ == 0) flush_cv_.notify_all();
         d.len = std::max<uint32_t>(d.len, o + n);
         pos += n;
         p += n;
@@ -592,7 +614,8 @@ St Object::write(uint64_t off, const uint8_t* data, uint64_t len) {
     // background flushers — the writer's memcpy stream overlaps the engine
     int full = 0;
     for (auto& kv : dirty_)
-        if (kv.second.len == cs && kv.first != last_idx) full++;
+        if (kv.second.len == cs && kv.second.busy == 0 && kv.first != last_idx)
+            full++;
     if (full >= kPutBatch || (full > 0 && inflight_ == 0 && full >= kPutBatch / 4)) {
         // backpressure: ≤2 batches in flight per object bounds pinned-pool
         // growth (each in-flight chunk pins a pool blob until its put lands)
@@ -604,7 +627,8 @@ St Object::write(uint64_t off, const uint8_t* data, uint64_t len) {
         }
         std::vector<uint64_t> idxs;  // recompute: dirty_ may change in wait
         for (auto& kv : dirty_)
-            if (kv.second.len == cs && kv.first != last_idx) idxs.push_back(kv.first);
+            if (kv.second.len == cs && kv.second.busy == 0 && kv.first != last_idx)
+                idxs.push_back(kv.first);
         if (!idxs.empty()) submit_async_flush_locked(std::move(idxs));
     }
     return St::ok;

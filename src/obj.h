@@ -60,6 +60,9 @@ class Object {
         Blob data;            // pool buffer, chunk_size capacity
         uint32_t len = 0;     // valid prefix length
         bool loaded = false;  // true once committed content was merged
+        int busy = 0;         // writers copying payload with mu_ DROPPED:
+                              // the entry (and its blob) must not be erased,
+                              // flushed, merged or snapshotted while busy>0
     };
 
     St flush_locked(std::unique_lock<std::mutex>& lk);
@@ -92,6 +95,8 @@ class Object {
     int rs_busy_ = 0;                // background rs-encode tasks outstanding
                                      // (separate from inflight_ so writer
                                      // backpressure ignores parity work)
+    int write_busy_ = 0;             // writers in an unlocked payload copy
+                                     // (flush/truncate/punch wait for 0)
     int inflight_ = 0;               // background flush batches outstanding
     std::condition_variable flush_cv_;
     St async_err_ = St::ok;
