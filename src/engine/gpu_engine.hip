@@ -153,7 +153,9 @@ struct DevCtx {
         uint8_t* pin_rmeta = nullptr;
         std::mutex mu;
     };
-    static constexpr int kReadSlots = 8;  // r2: 6+ fio jobs contended on 4
+    static constexpr int kReadSlots = 4;  // more slots measured neutral for
+                                          // bulk reads and dilute the small-
+                                          // read batches (fewer reqs/launch)
     ReadSlot rslots[kReadSlots];
     std::atomic<uint32_t> rrr{0};
 
