@@ -64,6 +64,7 @@ St Conf::apply_kv(const std::string& k, const std::string& v) {
     }
     if (k == "dedup") return parse_bool(v, &dedup) ? St::ok : bad();
     if (k == "verify_read") return parse_bool(v, &verify_read) ? St::ok : bad();
+    if (k == "clone") return parse_bool(v, &fuse_clone) ? St::ok : bad();
     if (k == "crc") return parse_bool(v, &crc) ? St::ok : bad();
     if (k == "prealloc") return parse_size(v, &prealloc) ? St::ok : bad();
     if (k == "rs") {  // "6+2" or "0"

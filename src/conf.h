@@ -53,6 +53,7 @@ struct Conf {
     bool read_direct = true;           // engine reads DMA straight into the caller
                                        // buffer (skip pinned-blob staging + memcpy)
     bool direct_io = false;            // force FOPEN_DIRECT_IO on opens
+    bool fuse_clone = true;            // per-reader FUSE_DEV_IOC_CLONE channels
     bool kernel_cache = true;          // keep_cache on open
 
     // durability: journal dir for namespace + manifests +

@@ -5,6 +5,7 @@
 #include <fcntl.h>
 #include <linux/fuse.h>
 #include <poll.h>
+#include <sys/ioctl.h>
 #include <sys/stat.h>
 #include <sys/mount.h>
 #include <sys/sysmacros.h>
@@ -204,14 +205,32 @@ void FuseMount::wait() {
     mounted_ = false;
 }
 
+// Per-thread device channel: FUSE_DEV_IOC_CLONE gives each reader its own
+// fd sharing the connection — each clone has its own processing queue and
+// lock, so threads stop contending on one fd's queue (SURVEY H1's
+// multi-channel plan). Falls back to the shared fd if the ioctl refuses.
+int FuseMount::clone_channel() {
+    int nfd = ::open("/dev/fuse", O_RDWR | O_CLOEXEC);
+    if (nfd < 0) return -1;
+    uint32_t src = (uint32_t)fd_;
+    if (::ioctl(nfd, FUSE_DEV_IOC_CLONE, &src)) {
+        ::close(nfd);
+        return -1;
+    }
+    return nfd;
+}
+
 void FuseMount::reader_loop(int tid) {
-    (void)tid;
     size_t bufsz = (size_t)conf_->max_write + 128 * KiB;
     std::vector<uint8_t> buf(bufsz);
     std::vector<uint8_t> reply;
     reply.reserve((size_t)conf_->max_write + 4 * KiB);
+    int own = conf_->fuse_clone ? clone_channel() : -1;
+    int ch = own >= 0 ? own : fd_;
+    if (tid == 0)
+        LOGI("fuse: reader channels: %s", own >= 0 ? "per-thread clones" : "shared fd");
     while (!stop_.load(std::memory_order_relaxed)) {
-        ssize_t n = ::read(fd_, buf.data(), bufsz);
+        ssize_t n = ::read(ch, buf.data(), bufsz);
         if (n < 0) {
             if (errno == EINTR || errno == EAGAIN || errno == ENOENT) continue;
             if (errno == ENODEV) break;  // unmounted
@@ -219,12 +238,14 @@ void FuseMount::reader_loop(int tid) {
             break;
         }
         if (n < (ssize_t)sizeof(fuse_in_header)) continue;
-        handle(buf.data(), (size_t)n, &reply);
+        handle(buf.data(), (size_t)n, &reply, ch);
     }
+    if (own >= 0) ::close(own);
     mounted_ = false;
 }
 
-void FuseMount::handle(const uint8_t* req, size_t len, std::vector<uint8_t>* reply_buf) {
+void FuseMount::handle(const uint8_t* req, size_t len, std::vector<uint8_t>* reply_buf,
+                       int ch_fd) {
     auto* h = (const fuse_in_header*)req;
     const uint8_t* arg = req + sizeof(fuse_in_header);
     metrics().fuse_reqs++;
@@ -236,7 +257,7 @@ void FuseMount::handle(const uint8_t* req, size_t len, std::vector<uint8_t>* rep
         oh.len = sizeof(oh);
         oh.error = neg_errno;
         oh.unique = h->unique;
-        ssize_t w = ::write(fd_, &oh, sizeof(oh));
+        ssize_t w = ::write(ch_fd, &oh, sizeof(oh));
         (void)w;
     };
     auto reply_body = [&](const void* body, size_t blen) {
@@ -244,7 +265,7 @@ void FuseMount::handle(const uint8_t* req, size_t len, std::vector<uint8_t>* rep
         oh.len = (uint32_t)(sizeof(oh) + blen);
         oh.unique = h->unique;
         iovec iov[2] = {{&oh, sizeof(oh)}, {(void*)body, blen}};
-        ssize_t w = ::writev(fd_, iov, blen ? 2 : 1);
+        ssize_t w = ::writev(ch_fd, iov, blen ? 2 : 1);
         (void)w;
     };
     auto reply_st = [&](St s) { reply_err(st_errno(s)); };

@@ -37,7 +37,11 @@ class FuseMount {
 
   private:
     void reader_loop(int tid);
-    void handle(const uint8_t* req, size_t len, std::vector<uint8_t>* reply_buf);
+    // ch_fd: the /dev/fuse channel the request arrived on (per-thread clone
+    // via FUSE_DEV_IOC_CLONE when available) — replies go back on it
+    void handle(const uint8_t* req, size_t len, std::vector<uint8_t>* reply_buf,
+                int ch_fd);
+    int clone_channel();  // -1 if the kernel/ioctl refuses
 
     Vfs* vfs_;
     const Conf* conf_;
