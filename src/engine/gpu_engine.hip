@@ -799,7 +799,7 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n,
                 while (!devs_[odev].arena_alloc(stored, &aoff)) {
                     g.unlock();  // eviction D2H must not block other writers
                     St s = evict_for(odev, stored);
-                    if (s == St::exists && ++waits < 20000) {  // transient: another
+                    if (s == St::exists && ++waits < 40000) {  // transient: another
                         std::this_thread::sleep_for(std::chrono::microseconds(500));
                         g.lock();                       // evictor / pending batch
                         continue;
