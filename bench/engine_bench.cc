@@ -62,14 +62,31 @@ int main(int argc, char** argv) {
             }
             if (eh_object_write(o, off, buf.data() + src, unit) != 0) errs[j]++;
         }
+        // keep a copy of the final unit's content for read-back verification
+        std::vector<uint8_t> expect;
+        uint64_t expect_off = (uint64_t)-1;
+        if (per_job >= unit) {
+            uint64_t last = (per_job - unit) / unit;
+            uint64_t src = ((uint64_t)j * region_units + last % region_units) * unit;
+            expect.assign(buf.data() + src, buf.data() + src + unit);
+            expect_off = last * unit;
+        }
         if (eh_object_flush(o) != 0) errs[j]++;
         auto t1 = std::chrono::steady_clock::now();
         std::vector<uint8_t> rbuf(unit);
         // pinned destination: direct-DMA reads land at link rate (D2H into
         // pageable memory degrades to runtime-staged copies)
         bool pinned = eh_mem_pin(rbuf.data(), unit) == 0;
-        for (uint64_t off = 0; off < per_job; off += unit)
+        for (uint64_t off = 0; off < per_job; off += unit) {
             if (eh_object_read(o, off, rbuf.data(), unit) != (int64_t)unit) errs[j]++;
+            // verify bytes (not just sizes): a silently-faulted kernel must
+            // show up as corruption, not as a clean-looking rate
+            if (off == expect_off && std::memcmp(rbuf.data(), expect.data(), unit)) {
+                fprintf(stderr, "job %d: READBACK MISMATCH at off %llu\n", j,
+                        (unsigned long long)off);
+                errs[j]++;
+            }
+        }
         if (pinned) eh_mem_unpin(rbuf.data());
         auto t2 = std::chrono::steady_clock::now();
         wsec[j] = std::chrono::duration<double>(t1 - t0).count();
