@@ -17,7 +17,7 @@ namespace {
 
 class HostEngine : public Engine {
   public:
-    explicit HostEngine(const Conf& conf) : conf_(conf), store_(0) {
+    explicit HostEngine(const Conf& conf) : conf_(conf), store_(0, conf.spill_dir) {
         if (conf.prealloc) BufPool::inst().prealloc(conf.prealloc, (uint32_t)conf.chunk_size);
     }
 

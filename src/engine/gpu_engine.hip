@@ -213,7 +213,8 @@ struct DevCtx {
 
 class GpuEngine : public Engine {
   public:
-    explicit GpuEngine(const Conf& conf) : conf_(conf), spill_(conf.spill_bytes) {
+    explicit GpuEngine(const Conf& conf)
+        : conf_(conf), spill_(conf.spill_bytes, conf.spill_dir) {
         // CDC chunks reach 4x the average (cdc.cc max bound); staging strides
         // are sized for the largest chunk the object layer can produce.
         max_chunk_ = conf.cdc ? 4 * conf.chunk_size : conf.chunk_size;

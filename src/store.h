@@ -17,7 +17,11 @@ namespace edgehip {
 
 class HostStore {
   public:
-    explicit HostStore(uint64_t cap_bytes = 0) : cap_(cap_bytes) {}
+    // cap_bytes: RAM-resident budget (0 = unlimited). dir: optional disk
+    // overflow tier — once RAM is at cap, chunks land as files under
+    // dir/<fp-hex> (config 4's 1 TiB object > HBM + host RAM). Disk-tier
+    // reads pread(2) on demand.
+    explicit HostStore(uint64_t cap_bytes = 0, const std::string& dir = "");
 
     // Stores (or bumps the refcount of) fp, CONSUMING the blob (zero-copy
     // ingest: pool buffers cycle object layer -> store -> pool). Returns
@@ -31,14 +35,19 @@ class HostStore {
     St addref(const Fp& fp);
     St decref(const Fp& fp);  // frees at 0
     St remove(const Fp& fp);  // unconditional erase (fault injection)
-    uint64_t bytes() const { return bytes_.load(); }
+    uint64_t bytes() const { return bytes_.load() + disk_bytes_.load(); }
+    uint64_t disk_bytes() const { return disk_bytes_.load(); }
     uint64_t count() const;
 
   private:
     struct Chunk {
-        Blob data;
+        Blob data;           // RAM tier (data.p != nullptr)
+        uint32_t disk_len = 0;  // disk tier: stored file length
         uint32_t refs = 1;
+        bool on_disk() const { return data.p == nullptr; }
     };
+    std::string chunk_path(const Fp& fp) const;
+    St write_file(const Fp& fp, const uint8_t* p, uint32_t len);
     static constexpr int kShards = 64;
     struct Shard {
         mutable std::mutex mu;
@@ -46,8 +55,10 @@ class HostStore {
     };
     Shard& shard(const Fp& fp) const { return shards_[fp.b[0] & (kShards - 1)]; }
     mutable Shard shards_[kShards];
-    std::atomic<uint64_t> bytes_{0};
+    std::atomic<uint64_t> bytes_{0};       // RAM-resident bytes
+    std::atomic<uint64_t> disk_bytes_{0};  // disk-tier bytes
     uint64_t cap_;
+    std::string dir_;
 };
 
 }  // namespace edgehip
