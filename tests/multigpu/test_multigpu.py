@@ -119,10 +119,13 @@ def test_stripe_placement_invariant_and_shard_map():
 
 
 @pytest.mark.skipif("ndev() < 2")
-def test_degraded_read_one_device_dropped():
-    """Drop every chunk stored on device 1 (simulated shard loss) — reads
-    must reconstruct each lost chunk from the surviving k shards (C2 gather
-    + G6 decode) and repair it in place."""
+def test_degraded_read_device_shards_dropped():
+    """Drop device-1 data chunks — AT MOST m=2 per stripe (RS 6+2 tolerates
+    m losses per stripe; with 2 devices a whole device holds 4 shards of
+    each stripe, beyond tolerance — that case needs >= 4 devices and is
+    covered by test_device_failure...). Reads must reconstruct each lost
+    chunk from the surviving shards (C2 gather + G6 decode) and repair it."""
+    k, m = 6, 2
     with Engine("engine=gpu,ngpus=2,fp=blake3,compress=lz4,dedup=1,rs=6+2,"
                 "arena=2g,spill=0") as eng, eng.object() as o:
         data = rand_bytes(24 << 20, seed=4)
@@ -130,20 +133,27 @@ def test_degraded_read_one_device_dropped():
         mf = o.commit()
         drefs, _ = decode_manifest(mf)
         dropped = 0
-        for r in drefs:
-            if eng.chunk_device(r["fp"]) == 1:
+        per_stripe = {}
+        for i, r in enumerate(drefs):
+            s = i // k
+            if eng.chunk_device(r["fp"]) == 1 and per_stripe.get(s, 0) < m:
                 assert drop_chunk(eng, r["fp"]) == 0
+                per_stripe[s] = per_stripe.get(s, 0) + 1
                 dropped += 1
         assert dropped > 0, "nothing placed on device 1?"
         assert o.read(0, len(data)) == data  # reconstructs via RS
         o.truncate(0)
 
 
-@pytest.mark.skipif("ndev() < 2")
+@pytest.mark.skipif("ndev() < 4")
 def test_device_failure_reads_degrade_writes_restripe():
-    """SURVEY §6.3 device-level failure: mark device 1 failed. Reads of its
-    chunks degrade to RS reconstruct; new writes place only on survivors."""
-    with Engine("engine=gpu,ngpus=2,fp=blake3,compress=lz4,dedup=1,rs=6+2,"
+    """SURVEY §6.3 device-level failure: mark device 1 failed. Whole-device
+    loss is reconstructable only when no device holds more than m=2 shards
+    of a stripe — 6+2 round-robin over 4 devices puts exactly 2 per device,
+    so this needs >= 4 GPUs (over 2 it would hold 4 > m). Reads of the
+    failed device's chunks degrade to RS reconstruct + in-place repair;
+    new writes place only on survivors."""
+    with Engine("engine=gpu,ngpus=4,fp=blake3,compress=lz4,dedup=1,rs=6+2,"
                 "arena=2g,spill=0") as eng:
         with eng.object() as o:
             data = rand_bytes(24 << 20, seed=5)
@@ -156,14 +166,14 @@ def test_device_failure_reads_degrade_writes_restripe():
             # reads reconstruct the failed device's chunks from survivors
             assert o.read(0, len(data)) == data
             o.truncate(0)
-        # writes after the failure land only on surviving devices
+        # writes after the failure avoid the failed device
         with eng.object() as o2:
             data2 = rand_bytes(12 << 20, seed=6)
             o2.write(0, data2)
             mf2 = o2.commit()
             drefs2, prefs2 = decode_manifest(mf2)
             for r in drefs2 + prefs2:
-                assert eng.chunk_device(r["fp"]) == 0
+                assert eng.chunk_device(r["fp"]) != 1
             assert o2.read(0, len(data2)) == data2
             o2.truncate(0)
 
