@@ -1,6 +1,7 @@
 #include "store.h"
 
 #include <errno.h>
+#include <dirent.h>
 #include <fcntl.h>
 #include <sys/stat.h>
 #include <unistd.h>
@@ -13,6 +14,18 @@ HostStore::HostStore(uint64_t cap_bytes, const std::string& dir)
     : cap_(cap_bytes), dir_(dir) {
     if (!dir_.empty()) {
         ::mkdir(dir_.c_str(), 0755);
+        // the disk tier is a CACHE (durability is the journal): stale files
+        // from a previous process are unreachable — start clean
+        if (DIR* d = ::opendir(dir_.c_str())) {
+            size_t purged = 0;
+            while (dirent* e = ::readdir(d)) {
+                if (e->d_name[0] == '.') continue;
+                if (::unlink((dir_ + "/" + e->d_name).c_str()) == 0) purged++;
+            }
+            ::closedir(d);
+            if (purged)
+                LOGI("store: purged %zu stale disk-tier chunks", purged);
+        }
         LOGI("store: disk overflow tier at %s (RAM budget %.1f GiB)",
              dir_.c_str(), cap_ / (double)GiB);
     }
