@@ -91,6 +91,40 @@ int main(int argc, char** argv) {
     bench("batch_copy", 2.0 * total,
           [&] { launch_batch_copy(d_src, d_dst, d_chunks, nchunks, nullptr); });
 
+    // zero-copy PCIe kernels: pull from / push to scattered pinned host
+    // buffers (the ingest / direct-read / eviction transfer path)
+    {
+        void* pin = nullptr;
+        if (hipHostMalloc(&pin, total, hipHostMallocDefault) == hipSuccess) {
+            void* dp = nullptr;
+            bool mapped = hipHostGetDevicePointer(&dp, pin, 0) == hipSuccess &&
+                          dp == pin;
+            if (mapped) {
+                std::vector<PtrChunk> h_pd(nchunks);
+                for (uint64_t i = 0; i < nchunks; i++)
+                    h_pd[i] = PtrChunk{(uint8_t*)pin + i * (1 << 20),
+                                       i * (1ull << 20), 1 << 20, 0};
+                PtrChunk* d_pd = nullptr;
+                (void)hipMalloc(&d_pd, sizeof(PtrChunk) * nchunks);
+                (void)hipMemcpy(d_pd, h_pd.data(), sizeof(PtrChunk) * nchunks,
+                                hipMemcpyHostToDevice);
+                bench("gather_host(PCIe pull)", (double)total, [&] {
+                    launch_gather_host(d_pd, d_dst, (int)nchunks, 1 << 20, nullptr);
+                });
+                // scatter: src = device addr in dst_off, dst = host ptr
+                for (uint64_t i = 0; i < nchunks; i++)
+                    h_pd[i].dst_off = (uint64_t)(uintptr_t)(d_src + i * (1 << 20));
+                (void)hipMemcpy(d_pd, h_pd.data(), sizeof(PtrChunk) * nchunks,
+                                hipMemcpyHostToDevice);
+                bench("scatter_host(PCIe push)", (double)total, [&] {
+                    launch_scatter_host(d_pd, (int)nchunks, 1 << 20, nullptr);
+                });
+                (void)hipFree(d_pd);
+            }
+            (void)hipHostFree(pin);
+        }
+    }
+
     // RS 6+2 over the whole buffer (k shards of total/6)
     {
         uint64_t shard = total / 8 & ~15ull;
