@@ -109,6 +109,20 @@ class Engine {
         return St::io;
     }
 
+    // G2 one-pass GPU path: upload the slab ONCE, scan, select boundaries
+    // (host, byte-identical to cdc_boundaries) and ingest the cut chunks
+    // straight from device staging — the two-phase flow crossed PCIe twice
+    // per byte and capped CDC mode at ~22 GiB/s. ends[i] = slab offsets the
+    // chunks end at (refs aligned 1:1). Default: St::io => caller runs the
+    // scan+put host flow.
+    virtual St cdc_ingest(const uint8_t* data, uint64_t len, uint64_t avg,
+                          bool final_tail, std::vector<size_t>* ends,
+                          std::vector<ChunkRef>* refs) {
+        (void)data; (void)len; (void)avg; (void)final_tail;
+        (void)ends; (void)refs;
+        return St::io;
+    }
+
     // Test hook: drop a chunk's stored bytes, simulating shard loss
     // (tests/faults; the read path must then reconstruct via RS).
     virtual St debug_drop_chunk(const Fp& fp) = 0;

@@ -144,6 +144,12 @@ struct DevCtx {
         adm_cv.notify_all();
     }
 
+    // CDC one-pass ingest: bound concurrent slab holders (each pins a pipe
+    // slot across a nested put that acquires another slot)
+    std::mutex cdc_hmu;
+    std::condition_variable cdc_hcv;
+    int cdc_holders = 0;
+
     struct ReadSlot {
         hipStream_t s{};
         uint8_t* d_rdstage = nullptr;
@@ -451,6 +457,79 @@ class GpuEngine : public Engine {
         return result;
     }
 
+    St cdc_ingest(const uint8_t* data, uint64_t len, uint64_t avg,
+                  bool final_tail, std::vector<size_t>* ends,
+                  std::vector<ChunkRef>* refs) override {
+        if (conf_.journal_ptr) return St::io;  // chunk mirror needs host bytes
+        if (len == 0 || len > devs_[0].stage_sz) return St::io;
+        DevCtx& D = devs_[alive_slot((int)(rr_.fetch_add(1) % devs_.size()))];
+        {
+            std::unique_lock<std::mutex> lk(D.cdc_hmu);
+            D.cdc_hcv.wait(lk, [&] { return D.cdc_holders < kPipeSlots / 2; });
+            D.cdc_holders++;
+        }
+        struct HolderGuard {
+            DevCtx& d;
+            ~HolderGuard() {
+                {
+                    std::lock_guard<std::mutex> g(d.cdc_hmu);
+                    d.cdc_holders--;
+                }
+                d.cdc_hcv.notify_one();
+            }
+        } hg{D};
+        PipeSlot* SLP = D.acquire_slot();
+        PipeSlot& S = *SLP;
+        struct SlotGuard {
+            DevCtx& d;
+            PipeSlot* s;
+            ~SlotGuard() {
+                (void)hipStreamSynchronize(s->s);
+                d.release_slot(s);
+            }
+        } sg{D, SLP};
+        std::vector<uint32_t> bm;
+        {
+            if (hipSetDevice(D.hip_dev) != hipSuccess) return St::gpu;
+            std::memcpy(S.pin_in, data, len);
+            if (hipMemcpyAsync(S.d_stage, S.pin_in, len, hipMemcpyHostToDevice,
+                               S.s) != hipSuccess)
+                return St::gpu;
+            uint32_t* d_bm = (uint32_t*)S.d_segscratch;  // 71 MB scratch
+            launch_cdc_candidates(S.d_stage, len, avg, d_bm, S.s, gear_table());
+            uint64_t words = (len + 31) / 32;
+            bm.resize(words);
+            if (hipMemcpyAsync(bm.data(), d_bm, words * 4, hipMemcpyDeviceToHost,
+                               S.s) != hipSuccess)
+                return St::gpu;
+            if (hipStreamSynchronize(S.s) != hipSuccess) return St::gpu;
+        }
+        ends->clear();
+        cdc_select(bm.data(), len, avg, final_tail, ends);
+        refs->assign(ends->size(), ChunkRef{});
+        if (ends->empty()) return St::ok;
+        // ingest the cut chunks IN PLACE from device staging (the slab stays
+        // pinned by our slot; the nested batches D2D from it on their own
+        // slots, then hash/compress/store as usual)
+        const size_t n = ends->size();
+        std::vector<PutChunk> puts(n);
+        size_t start = 0;
+        for (size_t i = 0; i < n; i++) {
+            puts[i] = PutChunk{S.d_stage + start,
+                               (uint32_t)((*ends)[i] - start), nullptr};
+            start = (*ends)[i];
+        }
+        const int ingest_dev = (int)(&D - devs_.data());
+        for (size_t i = 0; i < n; i += kMaxBatch) {
+            St s = flush_batch(puts.data() + i, refs->data() + i,
+                               (int)std::min<size_t>(kMaxBatch, n - i),
+                               /*device_src=*/true, /*force_dev=*/ingest_dev,
+                               /*device_lz4=*/true);
+            if (s != St::ok) return s;
+        }
+        return St::ok;
+    }
+
     St publish_manifest(const ChunkRef* recs, int n) override {
         if (!dist_.active()) return St::ok;
         // contribute from slot 0 (single writer process); C5 pre-warm: also
@@ -516,8 +595,12 @@ class GpuEngine : public Engine {
     }
 
     St evict_for(int dev, uint32_t need_bytes);  // map_mu_ held
+    // device_src: put[i].data are device pointers on force_dev's device.
+    // device_lz4: compress device-source chunks too (CDC ingest) — parity
+    // callers leave it off (GF-mixed bytes never compress).
     St flush_batch(const PutChunk* put, ChunkRef* refs, int n,
-                   bool device_src = false, int force_dev = -1);
+                   bool device_src = false, int force_dev = -1,
+                   bool device_lz4 = false);
     St gather_chunk_to_dev(const ChunkRef& ref, int ddev, uint8_t* dst);
 
     Conf conf_;
@@ -538,7 +621,7 @@ class GpuEngine : public Engine {
 };
 
 St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n,
-                          bool device_src, int force_dev) {
+                          bool device_src, int force_dev, bool device_lz4) {
     RoctxRange rr_ingest("edgehip::ingest_batch");
     const uint32_t cs = (uint32_t)max_chunk_;
     // proactive eviction: keep ~8% arena headroom so writers rarely hit a
@@ -555,9 +638,11 @@ St GpuEngine::flush_batch(const PutChunk* put, ChunkRef* refs, int n,
             }
         }
     }
-    // device-src batches are RS parity: GF-mixed bytes never compress —
-    // skip the attempt (it was ~30% of batch kernel time for nothing)
-    const bool do_lz4 = conf_.compress == Compress::lz4 && !device_src;
+    // device-src parity batches skip LZ4 (GF-mixed bytes never compress —
+    // the attempt was ~30% of batch kernel time for nothing); CDC device
+    // ingest opts back in via device_lz4
+    const bool do_lz4 =
+        conf_.compress == Compress::lz4 && (!device_src || device_lz4);
     const bool do_fp = conf_.fp == FpAlgo::blake3;
     const int ingest_dev =
         force_dev >= 0 ? force_dev

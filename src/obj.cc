@@ -1030,34 +1030,46 @@ St Object::cdc_cut_locked(bool final_tail) {
         mf_.logical_size = size_;
         return final_tail ? rs_update_locked() : St::ok;
     }
+    // One-pass GPU path: upload the slab once, scan + select + ingest the
+    // chunks from device staging (boundaries byte-identical to the host
+    // flow — same bitmap kernel, same cdc_select). Falls back to the
+    // two-phase host flow on St::io (no GPU / journal mode / oversized).
     std::vector<size_t> ends;
-    {
-        std::vector<uint32_t> bm;
-        if (eng_->cdc_scan(tail_.data(), tail_.size(), conf_->chunk_size, &bm) == St::ok)
-            cdc_select(bm.data(), tail_.size(), conf_->chunk_size, final_tail, &ends);
-        else
-            cdc_boundaries(tail_.data(), tail_.size(), conf_->chunk_size, final_tail, &ends);
+    std::vector<ChunkRef> refs;
+    St es = eng_->cdc_ingest(tail_.data(), tail_.size(), conf_->chunk_size,
+                             final_tail, &ends, &refs);
+    if (es != St::ok && es != St::io) return es;
+    if (es == St::io) {
+        {
+            std::vector<uint32_t> bm;
+            if (eng_->cdc_scan(tail_.data(), tail_.size(), conf_->chunk_size, &bm) == St::ok)
+                cdc_select(bm.data(), tail_.size(), conf_->chunk_size, final_tail, &ends);
+            else
+                cdc_boundaries(tail_.data(), tail_.size(), conf_->chunk_size, final_tail, &ends);
+        }
+        // keep the last (possibly still-moving) piece buffered unless finalizing
+        size_t cut_upto = 0;
+        std::vector<PutChunk> puts;
+        uint64_t next_idx = mf_.chunks.size();
+        for (size_t e : ends) {
+            puts.push_back({tail_.data() + cut_upto, (uint32_t)(e - cut_upto), nullptr,
+                            false, data_place_(next_idx++)});
+            cut_upto = e;
+        }
+        refs.assign(puts.size(), ChunkRef{});
+        if (!puts.empty()) {
+            St s = eng_->put_chunks(puts.data(), refs.data(), (int)puts.size());
+            if (s != St::ok) return s;
+        }
     }
-    // keep the last (possibly still-moving) piece buffered unless finalizing
-    size_t cut_upto = 0;
-    std::vector<PutChunk> puts;
-    uint64_t next_idx = mf_.chunks.size();
-    for (size_t e : ends) {
-        puts.push_back({tail_.data() + cut_upto, (uint32_t)(e - cut_upto), nullptr,
-                        false, data_place_(next_idx++)});
-        cut_upto = e;
-    }
-    if (!puts.empty()) {
-        std::vector<ChunkRef> refs(puts.size());
-        St s = eng_->put_chunks(puts.data(), refs.data(), (int)puts.size());
-        if (s != St::ok) return s;
+    if (!refs.empty()) {
         for (auto& r : refs) {
             mf_.chunks.push_back(r);
             cum_.push_back((cum_.empty() ? 0 : cum_.back()) + r.logical_len);
             mark_stripe_dirty(mf_.chunks.size() - 1);
             pending_publish_.push_back(r);
         }
-        tail_.erase(tail_.begin(), tail_.begin() + cut_upto);
+        tail_.erase(tail_.begin(), tail_.begin() + (ends.empty() ? 0 : ends.back()));
     }
     mf_.logical_size = size_;
     if (final_tail) return rs_update_locked();
