@@ -2,6 +2,7 @@
 
 #include <hip/hip_runtime.h>
 
+#include <map>
 #include <thread>
 
 #include "obs.h"
@@ -46,6 +47,28 @@ void BufPool::prealloc(uint64_t bytes, uint32_t block_sz, int nthreads) {
     }
     for (auto& th : ts) th.join();
     // destructors return everything to the free lists
+}
+
+namespace {
+std::mutex g_pinreg_mu;
+std::map<uintptr_t, size_t> g_pinreg;  // start -> len
+}  // namespace
+
+void pinreg_add(const void* p, size_t n) {
+    std::lock_guard<std::mutex> g(g_pinreg_mu);
+    g_pinreg[(uintptr_t)p] = n;
+}
+void pinreg_del(const void* p) {
+    std::lock_guard<std::mutex> g(g_pinreg_mu);
+    g_pinreg.erase((uintptr_t)p);
+}
+bool pinreg_contains(const void* p, size_t n) {
+    std::lock_guard<std::mutex> g(g_pinreg_mu);
+    if (g_pinreg.empty()) return false;
+    auto it = g_pinreg.upper_bound((uintptr_t)p);
+    if (it == g_pinreg.begin()) return false;
+    --it;
+    return (uintptr_t)p >= it->first && (uintptr_t)p + n <= it->first + it->second;
 }
 
 }  // namespace edgehip

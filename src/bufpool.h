@@ -119,4 +119,14 @@ inline void Blob::release() {
 }
 inline Blob::~Blob() { release(); }
 
+// Registry of CALLER-registered (hipHostRegister) host ranges that device
+// kernels may write directly. hipHostGetDevicePointer alone is NOT a valid
+// probe: on XNACK-capable boxes it succeeds for arbitrary pageable memory,
+// and kernel writes to such pages are silently lost on others (read-path
+// zeros corruption, caught by tests/kernels/test_concurrency.py on a cold
+// box). Only ranges explicitly pinned through eh_mem_pin are eligible.
+void pinreg_add(const void* p, size_t n);
+void pinreg_del(const void* p);
+bool pinreg_contains(const void* p, size_t n);
+
 }  // namespace edgehip

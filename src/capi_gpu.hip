@@ -3,6 +3,7 @@
 // host so pytest can compare against the CPU goldens bit-for-bit.
 #include <vector>
 
+#include "bufpool.h"
 #include "cdc.h"
 #include "kernels/launchers.h"
 #include "lz4.h"
@@ -302,8 +303,11 @@ extern "C" int eh_gpu_perm_probe(uint32_t a, uint32_t b, uint32_t* out16) {
 // async DMA instead of runtime-staged copies. Bench/daemon reply buffers
 // use this; no-op failure (rc != 0) on GPU-less boxes.
 extern "C" int eh_mem_pin(void* p, size_t len) {
-    return hipHostRegister(p, len, hipHostRegisterDefault) == hipSuccess ? 0 : -1;
+    if (hipHostRegister(p, len, hipHostRegisterDefault) != hipSuccess) return -1;
+    edgehip::pinreg_add(p, len);  // eligible for direct kernel writes
+    return 0;
 }
 extern "C" int eh_mem_unpin(void* p) {
+    edgehip::pinreg_del(p);
     return hipHostUnregister(p) == hipSuccess ? 0 : -1;
 }

@@ -1350,17 +1350,20 @@ St GpuEngine::get_chunks(const GetChunk* get, int n) {
             auto* h_pd = (PtrChunk*)(R.pin_rmeta + 48 * KiB);
             int npd = 0;
             uint32_t pd_maxlen = 0;
-            auto dst_mapped = [&](uint8_t* p) {
-                if (!host_zero_copy_) return false;
-                void* dp = nullptr;
-                return hipHostGetDevicePointer(&dp, p, 0) == hipSuccess && dp == p;
+            // Eligible destinations = ranges the CALLER registered via
+            // eh_mem_pin. Probing hipHostGetDevicePointer was UNSOUND: on
+            // XNACK boxes it succeeds for arbitrary pageable memory and the
+            // kernel's writes are silently dropped elsewhere — reads came
+            // back as untouched zeros (caught by the concurrency suite).
+            auto dst_mapped = [&](uint8_t* p, uint64_t n) {
+                return host_zero_copy_ && pinreg_contains(p, n);
             };
             const uint8_t* run_src = nullptr;
             uint8_t* run_dst = nullptr;  // direct mode: user buffer
             uint64_t run_out = 0, run_len = 0;
             auto flush_run = [&]() {
                 if (!run_len) return;
-                if (direct && dst_mapped(run_dst)) {
+                if (direct && dst_mapped(run_dst, run_len)) {
                     h_pd[npd] = PtrChunk{run_dst, (uint64_t)(uintptr_t)run_src,
                                          (uint32_t)run_len, 0};
                     pd_maxlen = std::max(pd_maxlen, (uint32_t)run_len);
