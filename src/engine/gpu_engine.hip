@@ -525,7 +525,13 @@ class GpuEngine : public Engine {
                                (int)std::min<size_t>(kMaxBatch, n - i),
                                /*device_src=*/true, /*force_dev=*/ingest_dev,
                                /*device_lz4=*/true);
-            if (s != St::ok) return s;
+            if (s != St::ok) {
+                // drop the refs of batches already stored: the caller sees
+                // one error for the whole slab (no half-adopted chunks)
+                for (size_t j = 0; j < i; j++)
+                    (void)release_chunk((*refs)[j]);
+                return s;
+            }
         }
         return St::ok;
     }
