@@ -555,8 +555,17 @@ St Object::write(uint64_t off, const uint8_t* data, uint64_t len) {
             bool committed = idx < mf_.chunks.size() && !mf_.chunks[idx].fp.is_zero();
             if (committed) {
                 // another writer may be mid-copy into this entry with mu_
-                // dropped; the merge below replaces the blob — wait it out
-                flush_cv_.wait(g, [&] { return it->second.busy == 0; });
+                // dropped; the merge below replaces the blob — wait it out.
+                // The entry can be ERASED during the wait (a third writer's
+                // async-flush submit picks it once busy drops), so the
+                // predicate must re-find instead of holding `it` (dangling
+                // iterator otherwise); if it vanished, redo this chunk.
+                flush_cv_.wait(g, [&] {
+                    auto i2 = dirty_.find(idx);
+                    return i2 == dirty_.end() || i2->second.busy == 0;
+                });
+                it = dirty_.find(idx);
+                if (it == dirty_.end()) continue;  // re-run for this pos
                 // merge committed bytes under the already-written ones:
                 Dirty merged;
                 St s = load_chunk_for_rmw(idx, &merged);
