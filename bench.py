@@ -125,6 +125,12 @@ class ObjectIo:
                         total += o.read_into(obase + i * unit, unit, rbuf)
                     if total != per_job * unit:
                         raise RuntimeError("short read")
+                    # verify the final unit against the ring (its slot still
+                    # holds the last-stamped content): silent corruption must
+                    # fail the bench, not inflate it
+                    voff = region + ((per_job - 1) % ring_units) * unit
+                    if bytes(rbuf[:unit]) != bytes(src[voff:voff + unit]):
+                        raise RuntimeError(f"job {j}: readback verify mismatch")
                     t2 = time.perf_counter()
                     phase[j][0] = t1 - t0
                     phase[j][1] = t2 - t1
@@ -209,6 +215,11 @@ def run_step(mnt, step_idx, gib, rand_buf, read_buf, jobs=4, rank=0):
                     total += n
             finally:
                 os.close(fd)
+            # the file's last MiB is in rbuf; its ring slot still holds the
+            # last-stamped content — verify (corruption must fail loudly)
+            voff = (region + (per_job - 1) % ring_mb) * MIB
+            if bytes(rbuf) != bytes(src[voff:voff + MIB]):
+                raise RuntimeError(f"job {j}: readback verify mismatch")
             os.unlink(path)
         except Exception as e:  # pragma: no cover
             errs.append(e)
