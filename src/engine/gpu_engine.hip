@@ -390,6 +390,9 @@ class GpuEngine : public Engine {
     St rs_encode_stripes(const ChunkRef* data, int k, int m, uint32_t shard_len,
                          int nstripes, ChunkRef* parity_out,
                          const int16_t* parity_place = nullptr) override;
+    St rs_reconstruct(const ChunkRef* refs, const bool* present, int k, int m,
+                      uint32_t shard_len, int target, uint8_t* out,
+                      uint32_t out_len) override;
 
     int ndev() const override { return (int)devs_.size(); }
 
@@ -1630,6 +1633,105 @@ St GpuEngine::rs_encode_stripes(const ChunkRef* data, int k, int m,
                 ChunkRef::kFlagParity | ChunkRef::kFlagRs;
         metrics().rs_encoded_stripes += ns;
     }
+    return St::ok;
+}
+
+// Degraded-read decode ON DEVICE: read k resident survivor shards in place
+// (local HBM or peer xGMI) and run one GF kernel with the recovery row —
+// the base path pulls every shard to host and decodes at CPU speed, which
+// made device-loss recovery (thousands of chunks) host-bound. Falls back to
+// the base flow (which handles spilled/compressed/short/hole shards) when
+// any condition fails.
+St GpuEngine::rs_reconstruct(const ChunkRef* refs, const bool* present_in, int k,
+                             int m, uint32_t shard_len, int target, uint8_t* out,
+                             uint32_t out_len) {
+    if (shard_len != max_chunk_ || k > kRsMaxK || m > kRsMaxM || k + m > 32 ||
+        target < 0 || target >= k + m || out_len > shard_len ||
+        refs[target].fp.is_zero())
+        return Engine::rs_reconstruct(refs, present_in, k, m, shard_len, target,
+                                      out, out_len);
+    const uint32_t cs = shard_len;
+    const int cdev = alive_slot((int)(rr_.fetch_add(1) % devs_.size()));
+    DevCtx& C = devs_[cdev];
+    std::lock_guard<std::mutex> rg(C.rs_mu);  // lock order: rs_mu, then pins
+    std::map<int, std::shared_lock<std::shared_mutex>> pins;
+    auto pin_dev = [&](int d2) {
+        if (!pins.count(d2))
+            pins.emplace(d2, std::shared_lock<std::shared_mutex>(
+                                 devs_[d2].evict_rw));
+    };
+    int chosen[kRsMaxK + kRsMaxM];
+    const uint8_t* in[kRsMaxK];
+    int sdev[kRsMaxK];
+    int nsel = 0;
+    auto classify = [&]() {
+        nsel = 0;
+        std::lock_guard<std::mutex> lk(map_mu_);
+        for (int i = 0; i < k + m && nsel < k; i++) {
+            if (i == target || refs[i].fp.is_zero()) continue;
+            if (present_in && !present_in[i]) continue;
+            auto it = map_.find(refs[i].fp);
+            if (it == map_.end() || it->second.spilled || it->second.compressed ||
+                it->second.pending || it->second.len != cs ||
+                device_failed(it->second.dev))
+                continue;
+            chosen[nsel] = i;
+            sdev[nsel] = (int)it->second.dev;
+            in[nsel] = devs_[it->second.dev].arena + it->second.off;
+            nsel++;
+        }
+        return nsel == k;
+    };
+    if (!classify())
+        return Engine::rs_reconstruct(refs, present_in, k, m, shard_len, target,
+                                      out, out_len);
+    for (int j = 0; j < k; j++) pin_dev(sdev[j]);
+    if (!classify()) {  // revalidate under the pins (a slot may have spilled)
+        pins.clear();
+        return Engine::rs_reconstruct(refs, present_in, k, m, shard_len, target,
+                                      out, out_len);
+    }
+    // Recovery row: survivors = G_S * data where row e<k is unit(e) and row
+    // e>=k is the Cauchy parity row. target<k  -> row target of G_S^{-1};
+    // target>=k -> C[target-k] * G_S^{-1}.
+    auto M = rs_parity_matrix(k, m);
+    const GF256& gf = gf256();
+    std::vector<uint8_t> A((size_t)k * k, 0);
+    for (int r = 0; r < k; r++) {
+        int e = chosen[r];
+        if (e < k)
+            A[(size_t)r * k + e] = 1;
+        else
+            for (int j = 0; j < k; j++) A[(size_t)r * k + j] = M[(size_t)(e - k) * k + j];
+    }
+    if (!gf_matrix_invert(A, k))
+        return Engine::rs_reconstruct(refs, present_in, k, m, shard_len, target,
+                                      out, out_len);
+    uint8_t row[kRsMaxK];
+    if (target < k) {
+        // d = A * s  ->  d_target = A[target][*] . s
+        for (int j = 0; j < k; j++) row[j] = A[(size_t)target * k + j];
+    } else {
+        for (int j = 0; j < k; j++) {
+            uint8_t acc = 0;
+            for (int t = 0; t < k; t++)
+                acc ^= gf.mul(M[(size_t)(target - k) * k + t], A[(size_t)t * k + j]);
+            row[j] = acc;
+        }
+    }
+    HIP_CHECK(hipSetDevice(C.hip_dev));
+    RsDesc d{};
+    for (int j = 0; j < k; j++) d.in[j] = in[j];
+    d.out[0] = C.d_rs;
+    for (int j = 0; j < k; j++) d.coef[0][j] = row[j];
+    d.k = k;
+    d.m = 1;
+    d.len = cs;
+    launch_gf_matmul(d, C.s_read);
+    HIP_CHECK(hipStreamSynchronize(C.s_read));
+    pins.clear();  // kernel done reading the arenas
+    HIP_CHECK(hipMemcpy(out, C.d_rs, out_len, hipMemcpyDeviceToHost));
+    metrics().rs_reconstructs++;
     return St::ok;
 }
 
