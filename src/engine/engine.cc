@@ -17,7 +17,12 @@ namespace {
 
 class HostEngine : public Engine {
   public:
-    explicit HostEngine(const Conf& conf) : conf_(conf), store_(0, conf.spill_dir) {
+    // arena acts as the host-RAM store budget (the host engine's "HBM");
+    // with spill_dir set, chunks past it overflow to the disk tier — the
+    // same three-tier shape as the GPU engine, CPU-testable.
+    explicit HostEngine(const Conf& conf)
+        : conf_(conf), store_(conf.spill_dir.empty() ? 0 : conf.arena_bytes,
+                              conf.spill_dir) {
         if (conf.prealloc) BufPool::inst().prealloc(conf.prealloc, (uint32_t)conf.chunk_size);
     }
 

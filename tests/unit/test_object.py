@@ -189,3 +189,22 @@ def test_punch_hole_subchunk_at_zero():
         o.write(0, data)
         o.punch_hole(0, 1000)
         assert o.read(0, 2000) == b"\x00" * 1000 + data[1000:2000]
+
+
+def test_host_disk_overflow_tier(tmp_path):
+    """spill_dir on the host engine: chunks past the RAM budget (arena) land
+    as files and reads pread them back — CPU-testable mirror of the GPU
+    engine's three-tier shape."""
+    from edgehip.lib import Engine
+    from tests.conftest import rand_bytes
+    d = tmp_path / "overflow"
+    with Engine(f"engine=host,fp=blake3,compress=none,dedup=0,arena=16m,"
+                f"spill_dir={d}") as eng, eng.object() as o:
+        data = rand_bytes(64 << 20, seed=88)
+        o.write(0, data)
+        o.flush()
+        assert list(d.iterdir()), "nothing overflowed to disk"
+        assert o.read(0, len(data)) == data
+        assert o.read(40 << 20, 1 << 20) == data[40 << 20:41 << 20]
+        o.truncate(0)
+        assert not list(d.iterdir()), "disk chunks must unlink on release"
